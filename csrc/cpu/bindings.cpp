@@ -1,0 +1,228 @@
+// pybind11 bindings for the host-side engine (_pnr_cpu).
+// Unity build: includes the implementation TUs directly.
+#include <pybind11/pybind11.h>
+#include <pybind11/numpy.h>
+#include <pybind11/stl.h>
+#include <memory>
+
+#include "pnr.h"
+#include "rr_build.cpp"
+#include "route_serial.cpp"
+#include "place_serial.cpp"
+#include "sta_serial.cpp"
+
+namespace py = pybind11;
+using namespace pnr;
+
+// zero-copy numpy view over a vector owned by a shared_ptr-held object
+template <typename T, typename Owner>
+static py::array_t<T> vec_view(const std::vector<T>& v, const std::shared_ptr<Owner>& owner) {
+  auto capsule = py::capsule(new std::shared_ptr<Owner>(owner), [](void* p) {
+    delete static_cast<std::shared_ptr<Owner>*>(p);
+  });
+  return py::array_t<T>({(py::ssize_t)v.size()}, {(py::ssize_t)sizeof(T)},
+                        v.data(), capsule);
+}
+
+template <typename T, int Flags>
+static std::vector<T> to_vec(py::array_t<T, Flags> arr) {
+  py::array_t<T, py::array::c_style | py::array::forcecast> a(arr);
+  auto r = a.template unchecked<1>();
+  std::vector<T> v((size_t)r.shape(0));
+  for (py::ssize_t i = 0; i < r.shape(0); ++i) v[i] = r(i);
+  return v;
+}
+
+PYBIND11_MODULE(_pnr_cpu, m) {
+  m.doc() = "parallel_eda_amd host-side engine (rr-graph builder, CPU oracles)";
+
+  py::class_<RRGraph, std::shared_ptr<RRGraph>>(m, "RRGraph")
+      .def_readonly("nx", &RRGraph::nx)
+      .def_readonly("ny", &RRGraph::ny)
+      .def_readonly("W", &RRGraph::W)
+      .def_readonly("L", &RRGraph::L)
+      .def_readonly("num_nodes", &RRGraph::num_nodes)
+      .def_readonly("num_edges", &RRGraph::num_edges)
+      .def_readonly("degree_max", &RRGraph::degree_max)
+      .def_property_readonly("type", [](std::shared_ptr<RRGraph> g) { return vec_view(g->type, g); })
+      .def_property_readonly("xlow", [](std::shared_ptr<RRGraph> g) { return vec_view(g->xlow, g); })
+      .def_property_readonly("ylow", [](std::shared_ptr<RRGraph> g) { return vec_view(g->ylow, g); })
+      .def_property_readonly("xhigh", [](std::shared_ptr<RRGraph> g) { return vec_view(g->xhigh, g); })
+      .def_property_readonly("yhigh", [](std::shared_ptr<RRGraph> g) { return vec_view(g->yhigh, g); })
+      .def_property_readonly("ptc", [](std::shared_ptr<RRGraph> g) { return vec_view(g->ptc, g); })
+      .def_property_readonly("capacity", [](std::shared_ptr<RRGraph> g) { return vec_view(g->capacity, g); })
+      .def_property_readonly("node_R", [](std::shared_ptr<RRGraph> g) { return vec_view(g->R, g); })
+      .def_property_readonly("node_C", [](std::shared_ptr<RRGraph> g) { return vec_view(g->C, g); })
+      .def_property_readonly("row_ptr", [](std::shared_ptr<RRGraph> g) { return vec_view(g->row_ptr, g); })
+      .def_property_readonly("edge_dst", [](std::shared_ptr<RRGraph> g) { return vec_view(g->edge_dst, g); })
+      .def_property_readonly("edge_sw", [](std::shared_ptr<RRGraph> g) { return vec_view(g->edge_sw, g); })
+      .def_property_readonly("tile_source", [](std::shared_ptr<RRGraph> g) { return vec_view(g->tile_source, g); })
+      .def_property_readonly("tile_sink", [](std::shared_ptr<RRGraph> g) { return vec_view(g->tile_sink, g); })
+      .def_property_readonly("sw_R", [](std::shared_ptr<RRGraph> g) {
+        return py::array_t<float>(NUM_SWITCHES, g->sw_R); })
+      .def_property_readonly("sw_Cin", [](std::shared_ptr<RRGraph> g) {
+        return py::array_t<float>(NUM_SWITCHES, g->sw_Cin); })
+      .def_property_readonly("sw_Tdel", [](std::shared_ptr<RRGraph> g) {
+        return py::array_t<float>(NUM_SWITCHES, g->sw_Tdel); })
+      .def_property_readonly("base_cost", [](std::shared_ptr<RRGraph> g) {
+        return py::array_t<float>(6, g->base_cost); })
+      .def("tile_id", &RRGraph::tile_id);
+
+  m.def("build_rr_graph", [](py::dict a) {
+    ArchParams ap{};
+    ap.nx = a["nx"].cast<int>(); ap.ny = a["ny"].cast<int>();
+    ap.W = a["W"].cast<int>(); ap.L = a["L"].cast<int>();
+    ap.fc_in = a["fc_in"].cast<int>(); ap.fc_out = a["fc_out"].cast<int>();
+    ap.clb_in = a["clb_in"].cast<int>(); ap.clb_out = a["clb_out"].cast<int>();
+    ap.io_cap = a["io_cap"].cast<int>();
+    ap.R_wire = a["R_wire"].cast<float>(); ap.C_wire = a["C_wire"].cast<float>();
+    ap.R_sw = a["R_sw"].cast<float>(); ap.C_sw_in = a["C_sw_in"].cast<float>();
+    ap.T_sw = a["T_sw"].cast<float>(); ap.T_opin = a["T_opin"].cast<float>();
+    ap.T_ipin = a["T_ipin"].cast<float>();
+    auto bc = a["base_cost"].cast<std::vector<float>>();
+    for (int i = 0; i < 6; ++i) ap.base_cost[i] = bc[i];
+    std::shared_ptr<RRGraph> g;
+    {
+      py::gil_scoped_release rel;
+      g = std::make_shared<RRGraph>(build_rr_graph(ap));
+    }
+    return g;
+  });
+
+  py::class_<Netlist, std::shared_ptr<Netlist>>(m, "Netlist")
+      .def(py::init([](py::array_t<int8_t> btype, py::array_t<uint8_t> bseq,
+                       py::array_t<int32_t> driver, py::array_t<int64_t> sptr,
+                       py::array_t<int32_t> sinks) {
+        auto nl = std::make_shared<Netlist>();
+        nl->block_type = to_vec(btype);
+        nl->block_is_seq = to_vec(bseq);
+        nl->net_driver = to_vec(driver);
+        nl->net_sink_ptr = to_vec(sptr);
+        nl->net_sinks = to_vec(sinks);
+        nl->num_blocks = (int)nl->block_type.size();
+        nl->num_nets = (int)nl->net_driver.size();
+        if (nl->net_sink_ptr.size() != (size_t)nl->num_nets + 1)
+          throw std::runtime_error("bad net_sink_ptr length");
+        return nl;
+      }))
+      .def_readonly("num_blocks", &Netlist::num_blocks)
+      .def_readonly("num_nets", &Netlist::num_nets)
+      .def_property_readonly("block_type", [](std::shared_ptr<Netlist> n) { return vec_view(n->block_type, n); })
+      .def_property_readonly("block_is_seq", [](std::shared_ptr<Netlist> n) { return vec_view(n->block_is_seq, n); })
+      .def_property_readonly("net_driver", [](std::shared_ptr<Netlist> n) { return vec_view(n->net_driver, n); })
+      .def_property_readonly("net_sink_ptr", [](std::shared_ptr<Netlist> n) { return vec_view(n->net_sink_ptr, n); })
+      .def_property_readonly("net_sinks", [](std::shared_ptr<Netlist> n) { return vec_view(n->net_sinks, n); });
+
+  py::class_<RouterOpts>(m, "RouterOpts")
+      .def(py::init<>())
+      .def_readwrite("pres_fac_init", &RouterOpts::pres_fac_init)
+      .def_readwrite("pres_fac_mult", &RouterOpts::pres_fac_mult)
+      .def_readwrite("acc_fac", &RouterOpts::acc_fac)
+      .def_readwrite("astar_fac", &RouterOpts::astar_fac)
+      .def_readwrite("max_iters", &RouterOpts::max_iters);
+
+  py::class_<SerialRouter>(m, "SerialRouter")
+      .def(py::init([](std::shared_ptr<RRGraph> g, py::array_t<int32_t> src,
+                       py::array_t<int64_t> sptr, py::array_t<int32_t> sinks,
+                       RouterOpts opts) {
+        // keep graph alive via a holder trick: SerialRouter stores raw ptr,
+        // so stash the shared_ptr in a wrapper
+        auto r = new SerialRouter(g.get(), to_vec(src), to_vec(sptr),
+                                  to_vec(sinks), opts);
+        r->graph_holder_ = g;
+        return r;
+      }))
+      .def("route_iteration", [](SerialRouter& r, py::array_t<float, py::array::c_style | py::array::forcecast> crit) {
+        const float* c = crit.size() ? crit.data() : nullptr;
+        int64_t over;
+        { py::gil_scoped_release rel; over = r.route_iteration(c); }
+        return over;
+      })
+      .def("update_costs", &SerialRouter::update_costs)
+      .def("set_pres_fac", &SerialRouter::set_pres_fac)
+      .def("count_overused", &SerialRouter::count_overused)
+      .def("feasible", &SerialRouter::feasible)
+      .def("total_wirelength", &SerialRouter::total_wirelength)
+      .def("heap_pushes", &SerialRouter::heap_pushes)
+      .def("heap_pops", &SerialRouter::heap_pops)
+      .def("sink_delays", [](SerialRouter& r) {
+        py::array_t<float> out((py::ssize_t)r.num_sinks_total());
+        r.sink_delays(out.mutable_data());
+        return out;
+      })
+      .def("occ", [](SerialRouter& r) {
+        return py::array_t<int32_t>((py::ssize_t)r.occ().size(), r.occ().data());
+      })
+      .def("check_routed", [](SerialRouter& r) {
+        std::string err;
+        bool ok = r.check_routed(&err);
+        return py::make_tuple(ok, err);
+      })
+      .def("tree", [](SerialRouter& r, int inet) {
+        const RouteTree& t = r.tree(inet);
+        return py::make_tuple(
+            py::array_t<int32_t>((py::ssize_t)t.nodes.size(), t.nodes.data()),
+            py::array_t<int32_t>((py::ssize_t)t.parent.size(), t.parent.data()),
+            py::array_t<int8_t>((py::ssize_t)t.sw.size(), t.sw.data()),
+            py::array_t<float>((py::ssize_t)t.delay.size(), t.delay.data()));
+      });
+
+  py::class_<SerialPlacer>(m, "SerialPlacer")
+      .def(py::init([](std::shared_ptr<Netlist> nl, int nx, int ny, int io_cap,
+                       py::array_t<float, py::array::c_style | py::array::forcecast> delay_mat,
+                       uint64_t seed) {
+        auto p = new SerialPlacer(nl.get(), nx, ny, io_cap, to_vec(delay_mat), seed);
+        p->netlist_holder_ = nl;
+        return p;
+      }))
+      .def("bb_cost", &SerialPlacer::bb_cost)
+      .def("td_cost", &SerialPlacer::td_cost)
+      .def("recompute_bb_cost", &SerialPlacer::recompute_bb_cost_from_scratch)
+      .def("run_moves", [](SerialPlacer& p, double T, double rlim, int64_t n,
+                           double tt, double bbn, double tdn) {
+        py::gil_scoped_release rel;
+        return p.run_moves(T, rlim, n, tt, bbn, tdn);
+      })
+      .def("last_delta_std", &SerialPlacer::last_delta_std)
+      .def("set_crit", [](SerialPlacer& p, py::array_t<float, py::array::c_style | py::array::forcecast> c) {
+        p.set_crit(c.data(), c.size());
+      })
+      .def("conn_delays", [](SerialPlacer& p) {
+        py::array_t<float> out((py::ssize_t)p.nl_->net_sinks.size());
+        p.get_conn_delays(out.mutable_data());
+        return out;
+      })
+      .def("placement", [](SerialPlacer& p) {
+        int nb = p.nl_->num_blocks;
+        py::array_t<int32_t> x(nb), y(nb), s(nb);
+        p.get_placement(x.mutable_data(), y.mutable_data(), s.mutable_data());
+        return py::make_tuple(x, y, s);
+      })
+      .def("set_placement", [](SerialPlacer& p, py::array_t<int32_t, py::array::c_style | py::array::forcecast> x,
+                               py::array_t<int32_t, py::array::c_style | py::array::forcecast> y,
+                               py::array_t<int32_t, py::array::c_style | py::array::forcecast> s) {
+        p.set_placement(x.data(), y.data(), s.data());
+      })
+      .def("check_place", [](SerialPlacer& p) {
+        std::string err;
+        bool ok = p.check_place(&err);
+        return py::make_tuple(ok, err);
+      });
+
+  py::class_<TimingGraph>(m, "TimingGraph")
+      .def(py::init([](std::shared_ptr<Netlist> nl, float t_clb, float t_out, float t_in) {
+        auto t = new TimingGraph(nl.get(), t_clb, t_out, t_in);
+        t->netlist_holder_ = nl;
+        return t;
+      }))
+      .def("num_levels", &TimingGraph::num_levels)
+      .def("analyze", [](TimingGraph& t, py::array_t<float, py::array::c_style | py::array::forcecast> conn_delay) {
+        py::ssize_t n = conn_delay.size();
+        py::array_t<float> slack(n), crit(n);
+        float cpd = t.analyze(conn_delay.data(), slack.mutable_data(), crit.mutable_data());
+        return py::make_tuple(cpd, slack, crit);
+      })
+      .def("level_of", [](TimingGraph& t) {
+        return py::array_t<int32_t>((py::ssize_t)t.level_of().size(), t.level_of().data());
+      });
+}

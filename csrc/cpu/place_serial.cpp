@@ -1,0 +1,483 @@
+// Serial simulated-annealing placer — the CPU oracle.
+//
+// Re-implements the semantics of the reference placer
+// (vpr/SRC/place/place.c:310 try_place; try_swap:1252; update_bb:2292;
+//  get_net_cost:2204 = HPWL x crossing-count factor; adaptive schedule
+//  update_t:983): range-limited swaps, incremental bounding-box cost with
+//  boundary counts, Metropolis acceptance, success-rate-driven cooling.
+// The outer temperature loop lives in Python (place/placer.py) so the GPU
+// engine and this oracle share schedule + STA refresh logic; this class
+// runs one temperature's worth of moves.
+#include "pnr.h"
+#include <random>
+#include <algorithm>
+
+namespace pnr {
+
+// crossing-count factor q(fanout) — standard Cheng (1994) wirelength
+// correction used by VPR (place.c:197 cross_count); piecewise-linear fit.
+static inline float cross_count(int n) {
+  static const float q3 = 1.0f, q50 = 2.79f;
+  if (n <= 3) return q3;
+  if (n >= 50) return q50 + 0.02616f * (n - 50);
+  return q3 + (q50 - q3) * (n - 3) / 47.0f;
+}
+
+struct Bb { int16_t xmin, xmax, ymin, ymax; int16_t nxmin, nxmax, nymin, nymax; };
+
+class SerialPlacer {
+ public:
+  // delay_mat: (nx+2)*(ny+2) delay lookup by |dx|,|dy| (flattened dy-major:
+  // delay_mat[dx*(ny+2)+dy]); may be empty for bb-only placement.
+  SerialPlacer(const Netlist* nl, int nx, int ny, int io_cap,
+               std::vector<float> delay_mat, uint64_t seed)
+      : nl_(nl), nx_(nx), ny_(ny), io_cap_(io_cap),
+        delay_mat_(std::move(delay_mat)), rng_(seed) {
+    int nb = nl_->num_blocks;
+    bx_.assign(nb, -1); by_.assign(nb, -1); bslot_.assign(nb, 0);
+    // grid occupancy: per location, list of block ids (size cap)
+    gx_ = nx + 2; gy_ = ny + 2;
+    grid_.assign((size_t)gx_ * gy_ * std::max(1, io_cap), -1);
+    grid_cnt_.assign((size_t)gx_ * gy_, 0);
+    build_net_arrays();
+    initial_placement();
+    recompute_bb_all();
+    crit_.assign(nl_->net_sinks.size(), 0.0f);
+    conn_delay_.assign(nl_->net_sinks.size(), 0.0f);
+    recompute_td_all();
+  }
+
+  int cap_at(int x, int y) const {
+    bool io = (x == 0 || x == gx_ - 1 || y == 0 || y == gy_ - 1);
+    if (x >= 1 && x <= nx_ && y >= 1 && y <= ny_) return 1;
+    if (io && ((x >= 1 && x <= nx_) || (y >= 1 && y <= ny_))) return io_cap_;
+    return 0;
+  }
+  bool is_io_loc(int x, int y) const {
+    return (x == 0 || x == gx_ - 1 || y == 0 || y == gy_ - 1);
+  }
+
+  // ---- costs ----
+  double bb_cost() const { return bb_cost_; }
+  double td_cost() const { return td_cost_; }
+
+  double recompute_bb_cost_from_scratch() {
+    // drift check (reference: place.c recompute_bb_cost / check_place:2950)
+    double c = 0;
+    for (int n = 0; n < nl_->num_nets; ++n) c += net_cost_from_scratch(n);
+    return c;
+  }
+
+  // One temperature step: nmoves moves at temperature T, range limit rlim.
+  // timing_tradeoff: 0 => pure bb. crit must be set via set_crit.
+  // Returns success rate.
+  double run_moves(double T, double rlim, int64_t nmoves, double timing_tradeoff,
+                   double bb_norm, double td_norm) {
+    int64_t acc = 0;
+    delta_sum_ = delta_sq_sum_ = 0; delta_n_ = 0;
+    for (int64_t m = 0; m < nmoves; ++m)
+      acc += try_swap(T, rlim, timing_tradeoff, bb_norm, td_norm);
+    return (double)acc / std::max<int64_t>(1, nmoves);
+  }
+
+  // std of move deltas over the last run_moves call — used for the
+  // reference-style starting temperature (place.c:1045 starting_t).
+  double last_delta_std() const {
+    if (delta_n_ < 2) return 0.0;
+    double mean = delta_sum_ / delta_n_;
+    double var = delta_sq_sum_ / delta_n_ - mean * mean;
+    return var > 0 ? std::sqrt(var) : 0.0;
+  }
+
+  void set_crit(const float* crit, int64_t n) {
+    for (int64_t i = 0; i < n; ++i) crit_[i] = crit[i];
+    recompute_td_all();
+  }
+
+  // connection delays by current placement (for STA)
+  void get_conn_delays(float* out) const {
+    for (size_t i = 0; i < conn_delay_.size(); ++i) out[i] = conn_delay_[i];
+  }
+
+  void get_placement(int32_t* x, int32_t* y, int32_t* slot) const {
+    for (int b = 0; b < nl_->num_blocks; ++b) { x[b] = bx_[b]; y[b] = by_[b]; slot[b] = bslot_[b]; }
+  }
+  void set_placement(const int32_t* x, const int32_t* y, const int32_t* slot) {
+    std::fill(grid_.begin(), grid_.end(), -1);
+    std::fill(grid_cnt_.begin(), grid_cnt_.end(), 0);
+    for (int b = 0; b < nl_->num_blocks; ++b) {
+      bx_[b] = x[b]; by_[b] = y[b]; bslot_[b] = slot[b];
+      grid_at(x[b], y[b], slot[b]) = b;
+      grid_cnt_[(size_t)x[b] * gy_ + y[b]]++;
+    }
+    recompute_bb_all();
+    recompute_td_all();
+  }
+
+  bool check_place(std::string* err) const {
+    // every block in a legal location; grid consistent; cost drift small
+    std::vector<int> cnt((size_t)gx_ * gy_, 0);
+    for (int b = 0; b < nl_->num_blocks; ++b) {
+      int x = bx_[b], y = by_[b];
+      if (cap_at(x, y) <= bslot_[b]) { *err = "block in illegal slot"; return false; }
+      bool io = nl_->block_type[b] == 0;
+      if (io != is_io_loc(x, y)) { *err = "type/location mismatch"; return false; }
+      if (grid_at(x, y, bslot_[b]) != b) { *err = "grid inconsistent"; return false; }
+      cnt[(size_t)x * gy_ + y]++;
+    }
+    double fresh = const_cast<SerialPlacer*>(this)->net_cost_sum_check();
+    if (std::abs(fresh - bb_cost_) > 1e-3 * std::max(1.0, fresh)) {
+      *err = "bb cost drift: " + std::to_string(fresh) + " vs " + std::to_string(bb_cost_);
+      return false;
+    }
+    return true;
+  }
+
+  uint64_t rand_u64() { return rng_(); }
+
+ public:
+  std::shared_ptr<Netlist> netlist_holder_;  // lifetime pin for Python bindings
+  const Netlist* nl_;
+  int nx_, ny_, io_cap_, gx_, gy_;
+  std::vector<float> delay_mat_;
+  std::mt19937_64 rng_;
+  std::vector<int32_t> bx_, by_, bslot_;
+
+ private:
+  std::vector<int32_t> grid_;      // (x*gy+y)*cap + slot -> block
+  std::vector<int32_t> grid_cnt_;
+  std::vector<Bb> bbs_;
+  std::vector<float> net_cost_;
+  double bb_cost_ = 0, td_cost_ = 0;
+  std::vector<float> crit_, conn_delay_;
+  // per-block nets (CSR): nets touching each block
+  std::vector<int64_t> blk_net_ptr_;
+  std::vector<int32_t> blk_nets_;
+  std::vector<int64_t> net_nblocks_;  // fanout+1 per net
+
+  int32_t& grid_at(int x, int y, int slot) {
+    return grid_[((size_t)x * gy_ + y) * std::max(1, io_cap_) + slot];
+  }
+  int32_t grid_at(int x, int y, int slot) const {
+    return grid_[((size_t)x * gy_ + y) * std::max(1, io_cap_) + slot];
+  }
+
+  void build_net_arrays() {
+    int nb = nl_->num_blocks, nn = nl_->num_nets;
+    std::vector<int64_t> cnt(nb, 0);
+    auto each_pin = [&](auto&& f) {
+      for (int n = 0; n < nn; ++n) {
+        f(n, nl_->net_driver[n]);
+        for (int64_t s = nl_->net_sink_ptr[n]; s < nl_->net_sink_ptr[n + 1]; ++s)
+          f(n, nl_->net_sinks[s]);
+      }
+    };
+    each_pin([&](int n, int b) { (void)n; cnt[b]++; });
+    blk_net_ptr_.assign(nb + 1, 0);
+    for (int b = 0; b < nb; ++b) blk_net_ptr_[b + 1] = blk_net_ptr_[b] + cnt[b];
+    blk_nets_.assign(blk_net_ptr_[nb], -1);
+    std::vector<int64_t> cur(blk_net_ptr_.begin(), blk_net_ptr_.end() - 1);
+    each_pin([&](int n, int b) { blk_nets_[cur[b]++] = n; });
+    net_nblocks_.assign(nn, 0);
+    for (int n = 0; n < nn; ++n)
+      net_nblocks_[n] = 1 + (nl_->net_sink_ptr[n + 1] - nl_->net_sink_ptr[n]);
+    bbs_.resize(nn);
+    net_cost_.assign(nn, 0.f);
+  }
+
+  void initial_placement() {
+    // random legal placement: CLBs into CLB tiles, IOs into IO slots
+    std::vector<std::pair<int, int>> clb_locs, io_locs;
+    for (int x = 1; x <= nx_; ++x) for (int y = 1; y <= ny_; ++y) clb_locs.push_back({x, y});
+    for (int y = 1; y <= ny_; ++y) { io_locs.push_back({0, y}); io_locs.push_back({gx_ - 1, y}); }
+    for (int x = 1; x <= nx_; ++x) { io_locs.push_back({x, 0}); io_locs.push_back({x, gy_ - 1}); }
+    std::shuffle(clb_locs.begin(), clb_locs.end(), rng_);
+    std::shuffle(io_locs.begin(), io_locs.end(), rng_);
+    size_t ci = 0, ii = 0; int io_slot = 0;
+    for (int b = 0; b < nl_->num_blocks; ++b) {
+      if (nl_->block_type[b] == 1) {
+        if (ci >= clb_locs.size()) throw std::runtime_error("too many CLBs for grid");
+        auto [x, y] = clb_locs[ci++];
+        bx_[b] = x; by_[b] = y; bslot_[b] = 0;
+        grid_at(x, y, 0) = b; grid_cnt_[(size_t)x * gy_ + y]++;
+      } else {
+        if (ii >= io_locs.size()) throw std::runtime_error("too many IOs for grid");
+        auto [x, y] = io_locs[ii];
+        bx_[b] = x; by_[b] = y; bslot_[b] = io_slot;
+        grid_at(x, y, io_slot) = b; grid_cnt_[(size_t)x * gy_ + y]++;
+        if (++io_slot >= io_cap_) { io_slot = 0; ++ii; }
+      }
+    }
+  }
+
+  float net_cost_from_scratch(int n) {
+    Bb& b = bbs_[n];
+    int16_t xmin = 32767, xmax = 0, ymin = 32767, ymax = 0;
+    auto upd = [&](int blk) {
+      int x = bx_[blk], y = by_[blk];
+      if (x < xmin) xmin = x; if (x > xmax) xmax = x;
+      if (y < ymin) ymin = y; if (y > ymax) ymax = y;
+    };
+    upd(nl_->net_driver[n]);
+    for (int64_t s = nl_->net_sink_ptr[n]; s < nl_->net_sink_ptr[n + 1]; ++s)
+      upd(nl_->net_sinks[s]);
+    b.xmin = xmin; b.xmax = xmax; b.ymin = ymin; b.ymax = ymax;
+    // boundary counts
+    b.nxmin = b.nxmax = b.nymin = b.nymax = 0;
+    auto cntb = [&](int blk) {
+      if (bx_[blk] == xmin) b.nxmin++;
+      if (bx_[blk] == xmax) b.nxmax++;
+      if (by_[blk] == ymin) b.nymin++;
+      if (by_[blk] == ymax) b.nymax++;
+    };
+    cntb(nl_->net_driver[n]);
+    for (int64_t s = nl_->net_sink_ptr[n]; s < nl_->net_sink_ptr[n + 1]; ++s)
+      cntb(nl_->net_sinks[s]);
+    float c = cross_count((int)net_nblocks_[n]) *
+              ((b.xmax - b.xmin + 1) + (b.ymax - b.ymin + 1));
+    net_cost_[n] = c;
+    return c;
+  }
+
+  void recompute_bb_all() {
+    bb_cost_ = 0;
+    for (int n = 0; n < nl_->num_nets; ++n) bb_cost_ += net_cost_from_scratch(n);
+  }
+  double net_cost_sum_check() {
+    double c = 0;
+    for (int n = 0; n < nl_->num_nets; ++n) {
+      float saved = net_cost_[n];
+      Bb sb = bbs_[n];
+      c += net_cost_from_scratch(n);
+      bbs_[n] = sb; net_cost_[n] = saved;
+    }
+    return c;
+  }
+
+  float conn_delay(int bsrc, int bsnk) const {
+    if (delay_mat_.empty()) return 0.f;
+    int dx = std::abs(bx_[bsrc] - bx_[bsnk]);
+    int dy = std::abs(by_[bsrc] - by_[bsnk]);
+    return delay_mat_[(size_t)dx * gy_ + dy];
+  }
+
+  void recompute_td_all() {
+    td_cost_ = 0;
+    for (int n = 0; n < nl_->num_nets; ++n) {
+      int drv = nl_->net_driver[n];
+      for (int64_t s = nl_->net_sink_ptr[n]; s < nl_->net_sink_ptr[n + 1]; ++s) {
+        float d = conn_delay(drv, nl_->net_sinks[s]);
+        conn_delay_[s] = d;
+        td_cost_ += (double)crit_[s] * d;
+      }
+    }
+  }
+
+  // incremental bb update for moving block blk from (x0,y0) to (x1,y1).
+  // Mirrors reference update_bb (place.c:2292): O(1) via boundary counts,
+  // full recompute only when a boundary with count 1 moves inward.
+  void update_net_for_move(int n, int x0, int y0, int x1, int y1) {
+    Bb& b = bbs_[n];
+    bool recompute = false;
+    // x dimension
+    if (x1 < x0) {  // moving left
+      if (x0 == b.xmax) { if (b.nxmax == 1) recompute = true; }
+    } else if (x1 > x0) {
+      if (x0 == b.xmin) { if (b.nxmin == 1) recompute = true; }
+    }
+    if (y1 < y0) {
+      if (y0 == b.ymax) { if (b.nymax == 1) recompute = true; }
+    } else if (y1 > y0) {
+      if (y0 == b.ymin) { if (b.nymin == 1) recompute = true; }
+    }
+    if (recompute) { net_cost_from_scratch(n); return; }
+    // grow / boundary-count updates
+    auto leave_x = [&](int x) {
+      if (x == b.xmin) b.nxmin--;
+      if (x == b.xmax) b.nxmax--;
+    };
+    auto leave_y = [&](int y) {
+      if (y == b.ymin) b.nymin--;
+      if (y == b.ymax) b.nymax--;
+    };
+    leave_x(x0); leave_y(y0);
+    if (x1 < b.xmin) { b.xmin = x1; b.nxmin = 1; }
+    else if (x1 == b.xmin) b.nxmin++;
+    if (x1 > b.xmax) { b.xmax = x1; b.nxmax = 1; }
+    else if (x1 == b.xmax) b.nxmax++;
+    if (y1 < b.ymin) { b.ymin = y1; b.nymin = 1; }
+    else if (y1 == b.ymin) b.nymin++;
+    if (y1 > b.ymax) { b.ymax = y1; b.nymax = 1; }
+    else if (y1 == b.ymax) b.nymax++;
+    net_cost_[n] = cross_count((int)net_nblocks_[n]) *
+                   ((b.xmax - b.xmin + 1) + (b.ymax - b.ymin + 1));
+  }
+
+  int try_swap(double T, double rlim, double timing_tradeoff,
+               double bb_norm, double td_norm) {
+    int nb = nl_->num_blocks;
+    int blk = (int)(rng_() % nb);
+    int x0 = bx_[blk], y0 = by_[blk];
+    bool io = nl_->block_type[blk] == 0;
+    // find_to: range-limited destination of matching type (place.c:1520)
+    int irlim = std::max(1, (int)rlim);
+    int x1 = -1, y1 = -1, slot1 = 0;
+    for (int attempt = 0; attempt < 12; ++attempt) {
+      int dx = (int)(rng_() % (2 * irlim + 1)) - irlim;
+      int dy = (int)(rng_() % (2 * irlim + 1)) - irlim;
+      int tx = x0 + dx, ty = y0 + dy;
+      if (tx < 0 || tx >= gx_ || ty < 0 || ty >= gy_) continue;
+      if (is_io_loc(tx, ty) != io) continue;
+      if (cap_at(tx, ty) <= 0) continue;
+      if (tx == x0 && ty == y0) continue;
+      x1 = tx; y1 = ty;
+      slot1 = (int)(rng_() % cap_at(tx, ty));
+      break;
+    }
+    if (x1 < 0) return 0;
+    int other = grid_at(x1, y1, slot1);
+    if (other == blk) return 0;
+
+    // save + compute delta over affected nets
+    saved_bbs_.clear(); saved_costs_.clear(); saved_ids_.clear();
+    double d_bb = 0, d_td = 0;
+    auto affect = [&](int b, int ox, int oy, int nx2, int ny2) {
+      for (int64_t k = blk_net_ptr_[b]; k < blk_net_ptr_[b + 1]; ++k) {
+        int n = blk_nets_[k];
+        if (!mark_net(n)) continue;
+        saved_ids_.push_back(n);
+        saved_bbs_.push_back(bbs_[n]);
+        saved_costs_.push_back(net_cost_[n]);
+        (void)ox; (void)oy; (void)nx2; (void)ny2;
+      }
+    };
+    affect(blk, x0, y0, x1, y1);
+    if (other >= 0) affect(other, x1, y1, x0, y0);
+    double before = 0;
+    for (int n : saved_ids_) before += net_cost_[n];
+    double td_before = 0, td_after = 0;
+    if (timing_tradeoff > 0) td_before = td_of_blocks(blk, other);
+
+    // tentatively move
+    move_block(blk, x1, y1, slot1, other, x0, y0);
+    for (size_t i = 0; i < saved_ids_.size(); ++i) {
+      int n = saved_ids_[i];
+      // use scratch recompute for any net touched by a swap of two blocks;
+      // single-block moves use the incremental path.
+      if (other >= 0) net_cost_from_scratch(n);
+      else {
+        // determine which endpoint moved
+        update_net_for_move_of(n, blk, x0, y0, x1, y1);
+      }
+    }
+    double after = 0;
+    for (int n : saved_ids_) after += net_cost_[n];
+    d_bb = after - before;
+    if (timing_tradeoff > 0) { td_after = td_of_blocks(blk, other); d_td = td_after - td_before; }
+
+    double delta = (1.0 - timing_tradeoff) * d_bb / bb_norm +
+                   timing_tradeoff * d_td / td_norm;
+    delta_sum_ += delta; delta_sq_sum_ += delta * delta; ++delta_n_;
+    bool accept;
+    if (delta <= 0) accept = true;
+    else if (T <= 0) accept = false;
+    else {
+      double u = (double)(rng_() % (1ull << 53)) / (double)(1ull << 53);
+      accept = u < std::exp(-delta / T);
+    }
+    if (accept) {
+      bb_cost_ += d_bb; td_cost_ += d_td;
+      for (int n : saved_ids_) unmark_net(n);
+      update_conn_delays(blk, other);
+      return 1;
+    }
+    // revert
+    move_block(blk, x0, y0, bslot_saved_, other, x1, y1, /*revert_other_slot=*/slot1);
+    for (size_t i = 0; i < saved_ids_.size(); ++i) {
+      bbs_[saved_ids_[i]] = saved_bbs_[i];
+      net_cost_[saved_ids_[i]] = saved_costs_[i];
+      unmark_net(saved_ids_[i]);
+    }
+    return 0;
+  }
+
+  void update_net_for_move_of(int n, int blk, int x0, int y0, int x1, int y1) {
+    (void)blk;
+    update_net_for_move(n, x0, y0, x1, y1);
+  }
+
+  double td_of_blocks(int b1, int b2) {
+    double t = 0;
+    auto acc = [&](int b) {
+      if (b < 0) return;
+      for (int64_t k = blk_net_ptr_[b]; k < blk_net_ptr_[b + 1]; ++k) {
+        int n = blk_nets_[k];
+        if (!mark_net2(n)) continue;
+        int drv = nl_->net_driver[n];
+        for (int64_t s = nl_->net_sink_ptr[n]; s < nl_->net_sink_ptr[n + 1]; ++s)
+          t += (double)crit_[s] * conn_delay(drv, nl_->net_sinks[s]);
+      }
+    };
+    acc(b1); acc(b2);
+    for (int n : marked2_) net_mark2_[n] = 0;
+    marked2_.clear();
+    return t;
+  }
+
+  void update_conn_delays(int b1, int b2) {
+    if (delay_mat_.empty()) return;
+    auto acc = [&](int b) {
+      if (b < 0) return;
+      for (int64_t k = blk_net_ptr_[b]; k < blk_net_ptr_[b + 1]; ++k) {
+        int n = blk_nets_[k];
+        if (!mark_net2(n)) continue;
+        int drv = nl_->net_driver[n];
+        for (int64_t s = nl_->net_sink_ptr[n]; s < nl_->net_sink_ptr[n + 1]; ++s)
+          conn_delay_[s] = conn_delay(drv, nl_->net_sinks[s]);
+      }
+    };
+    acc(b1); acc(b2);
+    for (int n : marked2_) net_mark2_[n] = 0;
+    marked2_.clear();
+  }
+
+  int bslot_saved_ = 0;
+  void move_block(int blk, int x1, int y1, int slot1, int other,
+                  int x0, int y0, int revert_other_slot = -1) {
+    bslot_saved_ = bslot_[blk];
+    grid_at(bx_[blk], by_[blk], bslot_[blk]) = -1;
+    if (other >= 0) grid_at(bx_[other], by_[other], bslot_[other]) = -1;
+    int oslot = (revert_other_slot >= 0) ? revert_other_slot : bslot_saved_;
+    bx_[blk] = x1; by_[blk] = y1; bslot_[blk] = slot1;
+    grid_at(x1, y1, slot1) = blk;
+    if (other >= 0) {
+      bx_[other] = x0; by_[other] = y0; bslot_[other] = oslot;
+      grid_at(x0, y0, oslot) = other;
+    }
+  }
+
+  // net marking for dedup within a move
+  std::vector<uint8_t> net_mark_, net_mark2_;
+  std::vector<int> marked2_;
+  bool mark_net(int n) {
+    if (net_mark_.empty()) net_mark_.assign(nl_->num_nets, 0);
+    if (net_mark_[n]) return false;
+    net_mark_[n] = 1;
+    return true;
+  }
+  void unmark_net(int n) { net_mark_[n] = 0; }
+  bool mark_net2(int n) {
+    if (net_mark2_.empty()) net_mark2_.assign(nl_->num_nets, 0);
+    if (net_mark2_[n]) return false;
+    net_mark2_[n] = 1; marked2_.push_back(n);
+    return true;
+  }
+
+  std::vector<Bb> saved_bbs_;
+  std::vector<float> saved_costs_;
+  std::vector<int> saved_ids_;
+  double delta_sum_ = 0, delta_sq_sum_ = 0;
+  int64_t delta_n_ = 0;
+};
+
+}  // namespace pnr

@@ -1,0 +1,165 @@
+// Static timing analysis — levelized forward/backward sweeps.
+//
+// Re-implements the semantics of the reference STA
+// (vpr/SRC/timing/path_delay.c:1994 do_timing_analysis_new, levelization
+//  path_delay2.c:81): block-granularity timing graph, T_arr max-plus forward
+//  sweep per level, T_req min-minus backward, per-connection slack and
+//  criticality = 1 - slack/cpd. Single clock domain (the reference's
+//  multi-domain loop collapses to one domain on our netlists).
+//
+// The level arrays built here are uploaded to HBM for the GPU STA kernels
+// (csrc/hip/sta_kernel.hip), which run the same sweeps level-synchronously.
+#include "pnr.h"
+#include <algorithm>
+
+namespace pnr {
+
+class TimingGraph {
+ public:
+  TimingGraph(const Netlist* nl, float T_clb, float T_seq_out, float T_seq_in)
+      : nl_(nl), T_clb_(T_clb), T_seq_out_(T_seq_out), T_seq_in_(T_seq_in) {
+    levelize();
+  }
+
+  // conn_delay: per (net,sink) connection delay, aligned with nl_->net_sinks.
+  // Outputs: slack (same alignment), crit = 1 - slack/cpd, returns cpd.
+  float analyze(const float* conn_delay, float* slack, float* crit) {
+    int nb = nl_->num_blocks;
+    t_arr_.assign(nb, 0.0f);
+    t_req_.assign(nb, 3.0e38f);
+    // forward: blocks in topo order. T_arr[b] = output arrival time.
+    for (int b : topo_) {
+      bool seq = nl_->block_is_seq[b];
+      if (seq) { t_arr_[b] = T_seq_out_; continue; }
+      float a = in_arrival(b, conn_delay);
+      t_arr_[b] = a + T_clb_;
+    }
+    // cpd = max arrival at any sequential/output endpoint input
+    float cpd = 0.0f;
+    for (int b = 0; b < nb; ++b) {
+      if (!nl_->block_is_seq[b]) continue;
+      float a = in_arrival(b, conn_delay);
+      if (a + T_seq_in_ > cpd) cpd = a + T_seq_in_;
+    }
+    if (cpd <= 0) cpd = 1e-12f;
+    // backward: T_req at block OUTPUT. For seq endpoint b, required at its
+    // input is cpd - T_seq_in.
+    for (auto it = topo_.rbegin(); it != topo_.rend(); ++it) {
+      int b = *it;
+      float r = 3.0e38f;
+      for (int64_t k = out_ptr_[b]; k < out_ptr_[b + 1]; ++k) {
+        int64_t conn = out_conn_[k];
+        int snk = nl_->net_sinks[conn];
+        float req_in = nl_->block_is_seq[snk] ? (cpd - T_seq_in_)
+                                              : (t_req_[snk] - T_clb_);
+        float rr = req_in - conn_delay[conn];
+        if (rr < r) r = rr;
+      }
+      t_req_[b] = r;  // inf for blocks driving nothing
+    }
+    // per-connection slack + criticality
+    int64_t nconn = (int64_t)nl_->net_sinks.size();
+    for (int64_t c = 0; c < nconn; ++c) {
+      int drv = conn_driver_[c];
+      int snk = nl_->net_sinks[c];
+      float req_in = nl_->block_is_seq[snk] ? (cpd - T_seq_in_)
+                                            : (t_req_[snk] - T_clb_);
+      float s = req_in - (t_arr_[drv] + conn_delay[c]);
+      slack[c] = s;
+      float cr = 1.0f - s / cpd;
+      crit[c] = cr < 0 ? 0.0f : (cr > 1 ? 1.0f : cr);
+    }
+    return cpd;
+  }
+
+  int num_levels() const { return num_levels_; }
+  const std::vector<int32_t>& topo() const { return topo_; }
+  const std::vector<int32_t>& level_of() const { return level_; }
+
+ public:
+  std::shared_ptr<Netlist> netlist_holder_;  // lifetime pin for Python bindings
+  const Netlist* nl_;
+  float T_clb_, T_seq_out_, T_seq_in_;
+
+ private:
+  std::vector<int32_t> topo_, level_;
+  int num_levels_ = 0;
+  std::vector<float> t_arr_, t_req_;
+  // per-block incoming connections (conn index into net_sinks) and outgoing
+  std::vector<int64_t> in_ptr_, out_ptr_;
+  std::vector<int64_t> in_conn_, out_conn_;
+  std::vector<int32_t> conn_driver_;
+
+  float in_arrival(int b, const float* conn_delay) const {
+    float a = 0.0f;
+    for (int64_t k = in_ptr_[b]; k < in_ptr_[b + 1]; ++k) {
+      int64_t c = in_conn_[k];
+      float v = t_arr_[conn_driver_[c]] + conn_delay[c];
+      if (v > a) a = v;
+    }
+    return a;
+  }
+
+  void levelize() {
+    int nb = nl_->num_blocks, nn = nl_->num_nets;
+    int64_t nconn = (int64_t)nl_->net_sinks.size();
+    conn_driver_.assign(nconn, -1);
+    std::vector<int64_t> in_cnt(nb, 0), out_cnt(nb, 0);
+    for (int n = 0; n < nn; ++n) {
+      int drv = nl_->net_driver[n];
+      for (int64_t s = nl_->net_sink_ptr[n]; s < nl_->net_sink_ptr[n + 1]; ++s) {
+        conn_driver_[s] = drv;
+        out_cnt[drv]++;
+        in_cnt[nl_->net_sinks[s]]++;
+      }
+    }
+    in_ptr_.assign(nb + 1, 0); out_ptr_.assign(nb + 1, 0);
+    for (int b = 0; b < nb; ++b) {
+      in_ptr_[b + 1] = in_ptr_[b] + in_cnt[b];
+      out_ptr_[b + 1] = out_ptr_[b] + out_cnt[b];
+    }
+    in_conn_.assign(in_ptr_[nb], 0); out_conn_.assign(out_ptr_[nb], 0);
+    std::vector<int64_t> ic(in_ptr_.begin(), in_ptr_.end() - 1);
+    std::vector<int64_t> oc(out_ptr_.begin(), out_ptr_.end() - 1);
+    for (int n = 0; n < nn; ++n) {
+      for (int64_t s = nl_->net_sink_ptr[n]; s < nl_->net_sink_ptr[n + 1]; ++s) {
+        out_conn_[oc[nl_->net_driver[n]]++] = s;
+        in_conn_[ic[nl_->net_sinks[s]]++] = s;
+      }
+    }
+    // topo order over combinational edges (seq blocks are sources: their
+    // arrival doesn't depend on inputs)
+    std::vector<int32_t> pend(nb, 0);
+    std::vector<int32_t> stack;
+    level_.assign(nb, 0);
+    for (int b = 0; b < nb; ++b) {
+      if (nl_->block_is_seq[b]) { stack.push_back(b); continue; }
+      int cnt = 0;
+      for (int64_t k = in_ptr_[b]; k < in_ptr_[b + 1]; ++k) cnt++;
+      pend[b] = cnt;
+      if (cnt == 0) stack.push_back(b);
+    }
+    topo_.clear(); topo_.reserve(nb);
+    size_t head = 0;
+    std::vector<int32_t> q(std::move(stack));
+    while (head < q.size()) {
+      int b = q[head++];
+      topo_.push_back(b);
+      for (int64_t k = out_ptr_[b]; k < out_ptr_[b + 1]; ++k) {
+        int snk = nl_->net_sinks[out_conn_[k]];
+        if (nl_->block_is_seq[snk]) continue;  // edges into seq don't gate
+        int lv = level_[b] + 1;
+        if (lv > level_[snk]) level_[snk] = lv;
+        if (--pend[snk] == 0) q.push_back(snk);
+      }
+    }
+    if ((int)topo_.size() != nb)
+      throw std::runtime_error("combinational cycle in netlist (" +
+                               std::to_string(topo_.size()) + "/" +
+                               std::to_string(nb) + " levelized)");
+    num_levels_ = 0;
+    for (int b = 0; b < nb; ++b) num_levels_ = std::max(num_levels_, level_[b] + 1);
+  }
+};
+
+}  // namespace pnr

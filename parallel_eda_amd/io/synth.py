@@ -1,0 +1,199 @@
+"""Synthetic netlist generation.
+
+There is no network access for MCNC/VTR/Titan benchmark files, so the named
+BASELINE configs run on synthetic netlists of the named size: random
+Rent's-rule-flavored netlists with locality, acyclic combinational structure
+(sequential blocks break cycles), generated deterministically from a seed.
+
+The generated object is the post-packing netlist the reference reads back
+with read_netlist.c — block-level: IO and CLB blocks, nets with one driver
+and >=1 sinks.
+"""
+from dataclasses import dataclass
+
+import numpy as np
+
+from ..arch.archdef import ArchDef, BLK_IO, BLK_CLB
+from .. import ops
+
+
+@dataclass
+class SynthSpec:
+    n_clb: int
+    n_in: int            # input pads
+    n_out: int           # output pads
+    avg_fanout: float = 3.0
+    seq_frac: float = 0.35   # fraction of CLBs that are sequential
+    max_fanin: int = 16      # block input-pin budget (arch.clb_in)
+    seed: int = 1
+
+
+def spec_for_arch(arch: ArchDef, fill: float = 0.85, seed: int = 1) -> SynthSpec:
+    n_clb = int(arch.nx * arch.ny * fill)
+    n_io = max(4, int(0.12 * n_clb))
+    n_io = min(n_io, arch.num_io_slots() // 2 - 2)
+    return SynthSpec(n_clb=n_clb, n_in=max(2, n_io // 2),
+                     n_out=max(2, n_io // 2), max_fanin=arch.clb_in,
+                     seed=seed)
+
+
+class NetlistPy:
+    """Python-side netlist holding numpy arrays + the C++ handle."""
+
+    def __init__(self, block_type, block_is_seq, net_driver, net_sink_ptr,
+                 net_sinks, names=None):
+        self.block_type = np.asarray(block_type, dtype=np.int8)
+        self.block_is_seq = np.asarray(block_is_seq, dtype=np.uint8)
+        self.net_driver = np.asarray(net_driver, dtype=np.int32)
+        self.net_sink_ptr = np.asarray(net_sink_ptr, dtype=np.int64)
+        self.net_sinks = np.asarray(net_sinks, dtype=np.int32)
+        self.names = names
+        self._cpp = None
+
+    @property
+    def num_blocks(self):
+        return len(self.block_type)
+
+    @property
+    def num_nets(self):
+        return len(self.net_driver)
+
+    @property
+    def num_conns(self):
+        return len(self.net_sinks)
+
+    def cpp(self):
+        if self._cpp is None:
+            self._cpp = ops.cpu().Netlist(
+                self.block_type, self.block_is_seq, self.net_driver,
+                self.net_sink_ptr, self.net_sinks)
+        return self._cpp
+
+    def fanout(self, n):
+        return self.net_sink_ptr[n + 1] - self.net_sink_ptr[n]
+
+
+def synth_netlist(spec: SynthSpec) -> NetlistPy:
+    """Generate a deterministic synthetic netlist.
+
+    Block ids: [0, n_in) input pads; [n_in, n_in+n_out) output pads;
+    [n_in+n_out, ...) CLBs. Combinational edges always go from
+    lower "rank" to higher rank among comb CLBs => acyclic.
+    """
+    rng = np.random.default_rng(spec.seed)
+    n_io = spec.n_in + spec.n_out
+    nb = n_io + spec.n_clb
+    block_type = np.full(nb, BLK_CLB, dtype=np.int8)
+    block_type[:n_io] = BLK_IO
+    block_is_seq = np.zeros(nb, dtype=np.uint8)
+    block_is_seq[:n_io] = 1  # pads are timing endpoints
+    clb0 = n_io
+    seq_mask = rng.random(spec.n_clb) < spec.seq_frac
+    block_is_seq[clb0:] = seq_mask.astype(np.uint8)
+
+    # Drivers: every input pad and every CLB drives exactly one net.
+    drivers = np.concatenate([
+        np.arange(spec.n_in, dtype=np.int32),                      # input pads
+        np.arange(clb0, nb, dtype=np.int32),                       # CLBs
+    ])
+    n_nets = len(drivers)
+
+    # Comb rank: random permutation of CLBs; an edge into a COMB clb must
+    # come from a driver with lower rank (or from a seq block / pad).
+    rank = np.empty(spec.n_clb, dtype=np.int64)
+    rank[:] = rng.permutation(spec.n_clb)
+
+    # Sink candidate pools with block fan-in budgets: a CLB accepts at most
+    # max_fanin incoming connections (it has that many input pins); an
+    # output pad accepts exactly one driver.
+    fanin = np.zeros(nb, dtype=np.int64)
+    max_fanin = np.full(nb, spec.max_fanin, dtype=np.int64)
+    max_fanin[:n_io] = 1  # pads: single pin
+    sink_lists = []
+    out_pad_driven = np.zeros(spec.n_out, dtype=bool)
+    for i, drv in enumerate(drivers):
+        fanout = max(1, int(rng.poisson(spec.avg_fanout)))
+        sinks = set()
+        drv_is_clb = drv >= clb0
+        drv_rank = rank[drv - clb0] if (drv_is_clb and not block_is_seq[drv]) else -1
+        for _ in range(fanout * 4):
+            if len(sinks) >= fanout:
+                break
+            if rng.random() < 0.04 and spec.n_out > 0:
+                # drive an output pad (one driver each)
+                o = int(rng.integers(spec.n_out))
+                blk = spec.n_in + o
+                if out_pad_driven[o] or blk in sinks:
+                    continue
+                sinks.add(blk)
+                out_pad_driven[o] = True
+                fanin[blk] += 1
+                continue
+            c = int(rng.integers(spec.n_clb))
+            blk = clb0 + c
+            if blk == drv or blk in sinks:
+                continue
+            if fanin[blk] >= max_fanin[blk]:
+                continue
+            if not block_is_seq[blk]:
+                # comb sink: need drv_rank < rank[c] for acyclicity
+                if drv_rank >= 0 and rank[c] <= drv_rank:
+                    continue
+            sinks.add(blk)
+            fanin[blk] += 1
+        if not sinks:
+            # fall back: an undriven output pad, else any seq CLB with room
+            placed = False
+            for o in range(spec.n_out):
+                if not out_pad_driven[o]:
+                    sinks.add(spec.n_in + o)
+                    out_pad_driven[o] = True
+                    fanin[spec.n_in + o] += 1
+                    placed = True
+                    break
+            if not placed:
+                for c in rng.permutation(spec.n_clb):
+                    blk = clb0 + int(c)
+                    if blk != drv and block_is_seq[blk] and fanin[blk] < max_fanin[blk]:
+                        sinks.add(blk)
+                        fanin[blk] += 1
+                        placed = True
+                        break
+            if not placed:
+                raise RuntimeError("synth: no legal sink available")
+        sink_lists.append(sorted(sinks))
+
+    # ensure every output pad is driven by exactly one net
+    for o in range(spec.n_out):
+        if not out_pad_driven[o]:
+            n = int(rng.integers(n_nets))
+            sink_lists[n].append(spec.n_in + o)
+            out_pad_driven[o] = True
+            fanin[spec.n_in + o] += 1
+
+    sink_ptr = np.zeros(n_nets + 1, dtype=np.int64)
+    for i, s in enumerate(sink_lists):
+        sink_ptr[i + 1] = sink_ptr[i] + len(s)
+    net_sinks = np.concatenate([np.asarray(s, dtype=np.int32) for s in sink_lists])
+
+    # Every comb CLB must have at least one input (else it's dangling logic);
+    # this holds with high probability — patch the stragglers.
+    has_in = np.zeros(nb, dtype=bool)
+    has_in[net_sinks] = True
+    stragglers = [b for b in range(clb0, nb) if not has_in[b]]
+    if stragglers:
+        extra = []
+        for b in stragglers:
+            # drive from a random input pad's net
+            n = int(rng.integers(spec.n_in)) if spec.n_in else 0
+            extra.append((n, b))
+        # rebuild CSR with extras appended
+        per_net = [list(s) for s in sink_lists]
+        for n, b in extra:
+            per_net[n].append(b)
+        sink_ptr = np.zeros(n_nets + 1, dtype=np.int64)
+        for i, s in enumerate(per_net):
+            sink_ptr[i + 1] = sink_ptr[i] + len(s)
+        net_sinks = np.concatenate([np.asarray(s, dtype=np.int32) for s in per_net])
+
+    return NetlistPy(block_type, block_is_seq, drivers, sink_ptr, net_sinks)

@@ -1,0 +1,31 @@
+"""Static timing analysis wrapper.
+
+Reference semantics: vpr/SRC/timing/path_delay.c:1994 do_timing_analysis_new
+(T_arr forward max-plus per level, T_req backward, slack, criticality) and
+router glue router.cxx:27-40 analyze_timing. CPU engine in
+csrc/cpu/sta_serial.cpp; GPU engine runs the same levelized sweeps as HIP
+kernels (csrc/hip/sta_kernel.hip).
+"""
+import numpy as np
+
+from ..arch.archdef import ArchDef
+from .. import ops
+
+
+class STA:
+    def __init__(self, netlist, arch: ArchDef):
+        cpu = ops.cpu()
+        self.netlist = netlist
+        self.arch = arch
+        self.tg = cpu.TimingGraph(netlist.cpp(), arch.T_clb, arch.T_seq_out,
+                                  arch.T_seq_in)
+
+    @property
+    def num_levels(self):
+        return self.tg.num_levels()
+
+    def analyze(self, conn_delay):
+        """Returns (critical_path_delay, slack[], crit[]) per connection."""
+        conn_delay = np.ascontiguousarray(conn_delay, dtype=np.float32)
+        cpd, slack, crit = self.tg.analyze(conn_delay)
+        return float(cpd), np.asarray(slack), np.asarray(crit)

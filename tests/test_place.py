@@ -1,0 +1,59 @@
+import numpy as np
+import pytest
+
+from parallel_eda_amd.arch.archdef import get_arch
+from parallel_eda_amd.io.synth import synth_netlist, spec_for_arch
+from parallel_eda_amd.place.placer import anneal_place, analytic_delay_matrix
+from parallel_eda_amd.timing.sta import STA
+from parallel_eda_amd import ops
+
+
+@pytest.fixture(scope="module")
+def tiny_setup():
+    arch = get_arch("tiny")
+    nl = synth_netlist(spec_for_arch(arch, fill=0.8, seed=11))
+    return arch, nl
+
+
+def test_initial_placement_legal(tiny_setup):
+    arch, nl = tiny_setup
+    cpu = ops.cpu()
+    p = cpu.SerialPlacer(nl.cpp(), arch.nx, arch.ny, arch.io_cap,
+                         np.zeros(0, dtype=np.float32), 42)
+    ok, err = p.check_place()
+    assert ok, err
+    assert p.bb_cost() > 0
+
+
+def test_anneal_improves_cost(tiny_setup):
+    arch, nl = tiny_setup
+    cpu = ops.cpu()
+    p0 = cpu.SerialPlacer(nl.cpp(), arch.nx, arch.ny, arch.io_cap,
+                          np.zeros(0, dtype=np.float32), 7)
+    init_cost = p0.bb_cost()
+    pl = anneal_place(nl, arch, seed=7, timing_tradeoff=0.0)
+    assert pl.bb_cost < init_cost * 0.85  # anneal must improve HPWL
+    # placement arrays legal
+    assert len(pl.x) == nl.num_blocks
+
+
+def test_incremental_cost_matches_scratch(tiny_setup):
+    arch, nl = tiny_setup
+    cpu = ops.cpu()
+    p = cpu.SerialPlacer(nl.cpp(), arch.nx, arch.ny, arch.io_cap,
+                         np.zeros(0, dtype=np.float32), 3)
+    p.run_moves(1e30, max(arch.nx, arch.ny), 5000, 0.0, 1.0, 1.0)
+    fresh = p.recompute_bb_cost()
+    assert fresh == pytest.approx(p.bb_cost(), rel=1e-4)
+    ok, err = p.check_place()
+    assert ok, err
+
+
+def test_timing_driven_anneal(tiny_setup):
+    arch, nl = tiny_setup
+    sta = STA(nl, arch)
+    pl = anneal_place(nl, arch, seed=7, timing_tradeoff=0.5, sta=sta)
+    assert pl.td_cost > 0
+    dm = analytic_delay_matrix(arch)
+    assert dm.shape == (arch.nx + 2, arch.ny + 2)
+    assert dm[0, 0] > 0 and dm[3, 3] > dm[1, 1]

@@ -1,0 +1,75 @@
+import numpy as np
+import pytest
+
+from parallel_eda_amd.arch.archdef import get_arch, RR_CHANX, RR_CHANY, RR_SOURCE, RR_SINK
+from parallel_eda_amd import rrgraph
+
+
+@pytest.fixture(scope="module")
+def tiny_graph():
+    arch = get_arch("tiny")
+    g = rrgraph.build_rr_graph(arch)
+    return arch, g
+
+
+def test_build_and_validate_tiny(tiny_graph):
+    arch, g = tiny_graph
+    assert g.num_nodes > 0 and g.num_edges > 0
+    assert rrgraph.check_rr_graph(g, arch)
+
+
+def test_build_and_validate_tseng():
+    arch = get_arch("tseng")
+    g = rrgraph.build_rr_graph(arch)
+    assert rrgraph.check_rr_graph(g, arch)
+    s = rrgraph.graph_summary(g)
+    # every CLB + IO tile has SOURCE/SINK
+    n_tiles = arch.nx * arch.ny + 2 * (arch.nx + arch.ny)
+    assert s["SOURCE"] == n_tiles
+    assert s["SINK"] == n_tiles
+
+
+def test_wire_counts(tiny_graph):
+    arch, g = tiny_graph
+    ty = np.asarray(g.type)
+    # W tracks per channel; (ny+1) CHANX channels spanning nx tiles.
+    # wire count per channel track = number of spans
+    n_chanx = int((ty == RR_CHANX).sum())
+    n_chany = int((ty == RR_CHANY).sum())
+    assert n_chanx > 0 and n_chany > 0
+    # each span covers <= L tiles, so at least nx/L wires per track
+    assert n_chanx >= (arch.ny + 1) * arch.W * (arch.nx // arch.L)
+
+
+def test_graph_is_routable_fabric(tiny_graph):
+    """BFS from one CLB SOURCE must reach every SINK (connected fabric)."""
+    arch, g = tiny_graph
+    row_ptr = np.asarray(g.row_ptr)
+    dst = np.asarray(g.edge_dst)
+    ty = np.asarray(g.type)
+    ts = np.asarray(g.tile_source)
+    gy = arch.ny + 2
+    start = ts[1 * gy + 1]
+    assert start >= 0
+    seen = np.zeros(g.num_nodes, dtype=bool)
+    stack = [int(start)]
+    seen[start] = True
+    while stack:
+        v = stack.pop()
+        for e in range(row_ptr[v], row_ptr[v + 1]):
+            w = dst[e]
+            if not seen[w]:
+                seen[w] = True
+                stack.append(int(w))
+    sinks = np.nonzero(ty == RR_SINK)[0]
+    reached = seen[sinks].mean()
+    assert reached == 1.0, f"only {reached:.1%} of sinks reachable"
+
+
+def test_deterministic_build():
+    arch = get_arch("tiny")
+    g1 = rrgraph.build_rr_graph(arch)
+    g2 = rrgraph.build_rr_graph(arch)
+    assert g1.num_nodes == g2.num_nodes
+    assert g1.num_edges == g2.num_edges
+    assert np.array_equal(np.asarray(g1.edge_dst), np.asarray(g2.edge_dst))

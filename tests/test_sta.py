@@ -1,0 +1,54 @@
+import numpy as np
+import pytest
+
+from parallel_eda_amd.arch.archdef import get_arch
+from parallel_eda_amd.io.synth import synth_netlist, SynthSpec, NetlistPy
+from parallel_eda_amd.timing.sta import STA
+
+
+def test_synth_netlist_basic():
+    nl = synth_netlist(SynthSpec(n_clb=50, n_in=5, n_out=5, seed=3))
+    assert nl.num_blocks == 60
+    assert nl.num_nets == 55  # every input pad + every CLB drives one net
+    # all sinks valid block ids, no self-loop driver==sink
+    for n in range(nl.num_nets):
+        s = nl.net_sinks[nl.net_sink_ptr[n]:nl.net_sink_ptr[n + 1]]
+        assert len(s) >= 1
+        assert (s != nl.net_driver[n]).all()
+
+
+def test_sta_levelizes_and_analyzes():
+    arch = get_arch("tiny")
+    nl = synth_netlist(SynthSpec(n_clb=40, n_in=4, n_out=4, seed=5))
+    sta = STA(nl, arch)
+    assert sta.num_levels >= 1
+    d = np.full(nl.num_conns, 1e-9, dtype=np.float32)
+    cpd, slack, crit = sta.analyze(d)
+    assert cpd > 0
+    assert crit.min() >= 0.0 and crit.max() <= 1.0
+    # some connection must be fully critical
+    assert crit.max() > 0.99
+    # zero delays => smaller cpd
+    cpd0, _, _ = sta.analyze(np.zeros_like(d))
+    assert cpd0 < cpd
+
+
+def test_sta_hand_case():
+    # in_pad(0) -> comb A(2) -> comb B(3) -> out_pad(1); delays 1ns each conn
+    block_type = [0, 0, 1, 1]
+    block_is_seq = [1, 1, 0, 0]
+    # nets: 0: pad0 -> A ; 1: A -> B ; 2: B -> pad1
+    driver = [0, 2, 3]
+    sptr = [0, 1, 2, 3]
+    sinks = [2, 3, 1]
+    nl = NetlistPy(block_type, block_is_seq, driver, sptr, sinks)
+    arch = get_arch("tiny")
+    sta = STA(nl, arch)
+    d = np.full(3, 1e-9, dtype=np.float32)
+    cpd, slack, crit = sta.analyze(d)
+    # cpd = T_seq_out + 3 conn delays + 2 comb delays + T_seq_in
+    expect = arch.T_seq_out + 3e-9 + 2 * arch.T_clb + arch.T_seq_in
+    assert cpd == pytest.approx(expect, rel=1e-5)
+    # single path: every connection critical, slack ~ 0
+    assert np.allclose(crit, 1.0, atol=1e-5)
+    assert np.allclose(slack, 0.0, atol=1e-12)
