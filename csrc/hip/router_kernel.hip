@@ -1,0 +1,547 @@
+// CDNA4 wavefront router — the hot engine.
+//
+// Re-designs the reference's per-sink A*/delta-stepping sink router
+// (vpr/SRC/parallel_route/dijkstra.h:16, delta_stepping.h:45,
+//  SinkRouter partitioning_multi_sink_delta_stepping_route.cxx:360-780)
+// for the MI355X execution model: one 256-thread workgroup (4 wave64) per
+// net, delta-stepping bucket expansion with ping-pong frontier buffers in
+// HBM, search state as a packed (back_cost, prev) u64 updated with a single
+// 64-bit atomicMin (deterministic tie-break on prev id), congestion shared
+// across concurrently-routed nets via device-scope atomics on occ
+// (the reference's net-level-parallel family, locking_route.cxx semantics,
+// with pres_cost computed on the fly from occ — no lock needed).
+//
+// Small nets use bb-dense local state (bb tiles x nodes-per-tile); large
+// nets use whole-graph state in a few dedicated slots. Touched-list reset
+// keeps per-sink state reset O(visited), like the reference's sparse reset
+// (partitioning_multi_sink...cxx:769-779).
+#include "pnr_hip.h"
+
+namespace pnrh {
+
+#define WG_THREADS 256
+#define PATH_CAP 4096      // max path nodes per sink backtrack (LDS)
+#define FAIL_FRONTIER 1
+#define FAIL_ROUNDS 2
+#define FAIL_NO_PATH 3
+#define FAIL_TREE_CAP 4
+#define FAIL_PATH_CAP 5
+#define FAIL_TOUCHED 6
+
+#define INF_STATE 0xffffffffffffffffull
+
+__device__ __forceinline__ uint32_t f32_bits(float f) {
+  // order-preserving bits for non-negative floats
+  return __float_as_uint(f);
+}
+__device__ __forceinline__ float bits_f32(uint32_t u) { return __uint_as_float(u); }
+
+__device__ __forceinline__ uint64_t pack_state(float back, int32_t prev) {
+  return ((uint64_t)f32_bits(back) << 32) | (uint32_t)prev;
+}
+
+struct SinkCtx {
+  int32_t sink_node;
+  int16_t sx, sy;
+  float crit;
+  float astar_fac;
+};
+
+__device__ __forceinline__ float expected_cost(const RRDev& g, const RouteParams& P,
+                                               int32_t v, const SinkCtx& S) {
+  int8_t ty = g.type[v];
+  if (ty == 1 /*SINK*/) return 0.0f;
+  int tx = S.sx, ty2 = S.sy;
+  int dx = 0, dy = 0;
+  int xl = g.xlow[v], xh = g.xhigh[v], yl = g.ylow[v], yh = g.yhigh[v];
+  if (xl > tx) dx = xl - tx; else if (xh < tx) dx = tx - xh;
+  if (yl > ty2) dy = yl - ty2; else if (yh < ty2) dy = ty2 - yh;
+  int dist = dx + dy;
+  int nseg = (dist + g.L - 1) / g.L;
+  float cong = nseg * P.seg_base + P.ipin_base;
+  float del = nseg * P.seg_delay + P.ipin_delay;
+  return S.crit * del + (1.0f - S.crit) * cong;
+}
+
+// congestion cost of entering node v (pres computed from occ on the fly;
+// semantics of congestion.cxx:296 update_one_cost_internal's pres formula)
+__device__ __forceinline__ float cong_cost(const RRDev& g, const RouteParams& P,
+                                           const int32_t* occ, const float* acc,
+                                           int32_t v) {
+  int over = occ[v] + 1 - g.capacity[v];
+  float pres = over > 0 ? 1.0f + over * P.pres_fac : 1.0f;
+  return g.base_cost[g.type[v]] * acc[v] * pres;
+}
+
+__device__ __forceinline__ float hop_delay(const RRDev& g, int8_t sw, int32_t v) {
+  return g.sw_Tdel[sw] + g.C[v] * (g.sw_R[sw] + 0.5f * g.R[v]);
+}
+
+struct LocalIdx {
+  // small class: dense bb-local index; large class: global node id
+  int bx0, by0, bw, bh, npt;
+  bool dense;
+  __device__ __forceinline__ int64_t operator()(const RRDev& g, int32_t v) const {
+    if (!dense) return v;
+    int tx = g.xlow[v] - bx0;
+    int ty = g.ylow[v] - by0;
+    if (tx < 0 || ty < 0 || tx >= bw || ty >= bh) return -1;
+    return ((int64_t)tx * bh + ty) * npt + g.idx_in_tile[v];
+  }
+  __device__ __forceinline__ bool in_bb(const RRDev& g, int32_t v) const {
+    int tx = g.xlow[v] - bx0;
+    int ty = g.ylow[v] - by0;
+    return tx >= 0 && ty >= 0 && tx < bw && ty < bh;
+  }
+};
+
+struct WgShared {
+  int fcnt[2];
+  unsigned fmin_next;
+  unsigned best_sink_back;
+  int touched_cnt;
+  int attach_idx;
+  int fail;
+  int n_cur;
+  int path_len;
+  int32_t path[PATH_CAP];
+};
+
+__launch_bounds__(WG_THREADS, 1)
+__global__ void route_nets_kernel(
+    RRDev g, NetsDev nets, TreesDev trees, RouteParams P,
+    const int32_t* __restrict__ queue_small, int32_t n_queue_small,
+    const int32_t* __restrict__ queue_large, int32_t n_queue_large,
+    int32_t* q_cursors,                // [2]
+    int32_t* occ, const float* __restrict__ acc,
+    uint64_t* state_base, int64_t small_cap, int64_t large_cap,
+    int32_t n_small_slots,
+    float4* frontier_base, int64_t f_cap_small, int64_t f_cap_large,
+    int32_t* touched_base, int64_t t_cap_small, int64_t t_cap_large,
+    int32_t* fail_flags) {
+  const int tid = threadIdx.x;
+  const bool is_small = (int)blockIdx.x < n_small_slots;
+  const int slot = blockIdx.x;
+
+  __shared__ WgShared sh;
+
+  uint64_t* state;
+  float4* fr[2];
+  int32_t* touched;
+  int64_t f_cap, t_cap;
+  if (is_small) {
+    state = state_base + (int64_t)slot * small_cap;
+    float4* fb = frontier_base + (int64_t)slot * 2 * f_cap_small;
+    fr[0] = fb; fr[1] = fb + f_cap_small;
+    touched = touched_base + (int64_t)slot * t_cap_small;
+    f_cap = f_cap_small; t_cap = t_cap_small;
+  } else {
+    int ls = slot - n_small_slots;
+    uint64_t* lbase = state_base + (int64_t)n_small_slots * small_cap;
+    state = lbase + (int64_t)ls * large_cap;
+    float4* fb = frontier_base + (int64_t)n_small_slots * 2 * f_cap_small
+               + (int64_t)ls * 2 * f_cap_large;
+    fr[0] = fb; fr[1] = fb + f_cap_large;
+    touched = touched_base + (int64_t)n_small_slots * t_cap_small
+            + (int64_t)ls * t_cap_large;
+    f_cap = f_cap_large; t_cap = t_cap_large;
+  }
+
+  const int32_t* queue = is_small ? queue_small : queue_large;
+  const int32_t n_queue = is_small ? n_queue_small : n_queue_large;
+  int32_t* cursor = q_cursors + (is_small ? 0 : 1);
+
+  for (;;) {
+    // pop a net
+    if (tid == 0) sh.fcnt[0] = atomicAdd(cursor, 1);
+    __syncthreads();
+    int qi = sh.fcnt[0];
+    __syncthreads();
+    if (qi >= n_queue) return;
+    const int32_t inet = queue[qi];
+
+    const int32_t src = nets.src[inet];
+    const int32_t s0 = nets.sink_ptr[inet], s1 = nets.sink_ptr[inet + 1];
+    const int64_t toff = trees.off[inet];
+    const int32_t tcap = (int32_t)(trees.off[inet + 1] - toff);
+    int32_t* t_node = trees.node + toff;
+    int32_t* t_parent = trees.parent + toff;
+    int8_t* t_sw = trees.sw + toff;
+    float* t_delay = trees.delay + toff;
+
+    LocalIdx L;
+    L.dense = is_small;
+    L.bx0 = nets.bb[4 * inet + 0];
+    L.by0 = nets.bb[4 * inet + 1];
+    L.bw = nets.bb[4 * inet + 2] - L.bx0 + 1;
+    L.bh = nets.bb[4 * inet + 3] - L.by0 + 1;
+    L.npt = g.npt;
+    if (!is_small) { L.bx0 = 0; L.by0 = 0; L.bw = g.nx + 2; L.bh = g.ny + 2; }
+
+    // ---- rip-up previous tree (reference: route_tree rip-up, occ -1) ----
+    int32_t old_len = trees.len[inet];
+    for (int k = tid; k < old_len; k += WG_THREADS)
+      atomicSub(&occ[t_node[k]], 1);
+    __syncthreads();
+    // ---- new tree root ----
+    if (tid == 0) {
+      t_node[0] = src; t_parent[0] = -1; t_sw[0] = -1; t_delay[0] = 0.0f;
+      trees.len[inet] = 1;
+      atomicAdd(&occ[src], 1);
+      sh.fail = 0;
+    }
+    __syncthreads();
+    int tree_len = 1;
+
+    // ---- route each sink (pre-ordered by criticality on host) ----
+    for (int32_t si = s0; si < s1; ++si) {
+      SinkCtx S;
+      S.sink_node = nets.sink_rr[si];
+      S.sx = g.xlow[S.sink_node];
+      S.sy = g.ylow[S.sink_node];
+      S.crit = nets.crit[si];
+      S.astar_fac = P.astar_fac;
+      // per-sink bucket width: half an edge-step in this sink's cost units
+      const float delta = 0.5f * (S.crit * P.seg_delay +
+                                  (1.0f - S.crit) * P.seg_base);
+
+      if (tid == 0) {
+        sh.fcnt[0] = 0; sh.fcnt[1] = 0;
+        sh.fmin_next = 0xffffffffu;
+        sh.best_sink_back = 0xffffffffu;
+        sh.touched_cnt = 0;
+        sh.path_len = -1;
+      }
+      __syncthreads();
+
+      // seed from the current route tree (reference: SinkRouter seeds
+      // from tree nodes inside bb, partitioning_multi_sink...:707-745)
+      for (int k = tid; k < tree_len; k += WG_THREADS) {
+        int32_t v = t_node[k];
+        if (g.type[v] == 1 /*SINK*/) continue;
+        if (!L.in_bb(g, v)) continue;
+        float back = S.crit * t_delay[k];
+        float tot = back + S.astar_fac * expected_cost(g, P, v, S);
+        int64_t li = L(g, v);
+        state[li] = pack_state(back, v);  // prev==self marks a tree seed
+        int ti = atomicAdd(&sh.touched_cnt, 1);
+        if (ti < t_cap) touched[ti] = (int32_t)li;
+        int fi = atomicAdd(&sh.fcnt[0], 1);
+        if (fi < f_cap)
+          fr[0][fi] = make_float4(tot, back, __int_as_float(v), __int_as_float(v));
+        atomicMin(&sh.fmin_next, f32_bits(tot));
+      }
+      __syncthreads();
+      if (sh.fcnt[0] > f_cap) { if (tid == 0) sh.fail = FAIL_FRONTIER; }
+      __syncthreads();
+
+      int cur = 0;
+      int n_cur = sh.fcnt[0];
+      unsigned fmin = sh.fmin_next;
+      int rounds = 0;
+
+      while (!sh.fail) {
+        if (n_cur == 0) { if (tid == 0 && sh.best_sink_back == 0xffffffffu) sh.fail = FAIL_NO_PATH; break; }
+        // terminate when the sink's settled cost is <= min frontier f
+        if (sh.best_sink_back != 0xffffffffu && sh.best_sink_back <= fmin) break;
+        if (++rounds > P.max_rounds) { if (tid == 0) sh.fail = FAIL_ROUNDS; break; }
+        const float thr = bits_f32(fmin) + delta;
+        const int nxt = cur ^ 1;
+        if (tid == 0) { sh.fcnt[nxt] = 0; sh.fmin_next = 0xffffffffu; }
+        __syncthreads();
+
+        for (int i = tid; i < n_cur; i += WG_THREADS) {
+          float4 e = fr[cur][i];
+          float tot = e.x, back = e.y;
+          int32_t v = __float_as_int(e.z);
+          int32_t prev = __float_as_int(e.w);
+          int64_t li = L(g, v);
+          if (state[li] != pack_state(back, prev)) continue;  // stale entry
+          if (tot > thr) {
+            // keep for a later bucket
+            int fi = atomicAdd(&sh.fcnt[nxt], 1);
+            if (fi < f_cap) fr[nxt][fi] = e;
+            atomicMin(&sh.fmin_next, f32_bits(tot));
+            continue;
+          }
+          if (v == S.sink_node) continue;  // settled sink; no expansion
+          // expand
+          int32_t e0 = g.row_ptr[v], e1 = g.row_ptr[v + 1];
+          for (int32_t ei = e0; ei < e1; ++ei) {
+            int32_t w = g.edge_dst[ei];
+            int8_t ty = g.type[w];
+            if (ty == 1 && w != S.sink_node) continue;       // other SINK
+            if (ty == 3 /*IPIN*/ && (g.xlow[w] != S.sx || g.ylow[w] != S.sy))
+              continue;                                      // wrong-tile IPIN
+            if (!L.in_bb(g, w)) continue;                    // bb prune
+            int8_t sw = g.edge_sw[ei];
+            float back_new = back + S.crit * hop_delay(g, sw, w) +
+                             (1.0f - S.crit) * cong_cost(g, P, occ, acc, w);
+            float tot_new = back_new + S.astar_fac * expected_cost(g, P, w, S);
+            int64_t lw = L(g, w);
+            uint64_t pk = pack_state(back_new, v);
+            uint64_t old = atomicMin((unsigned long long*)&state[lw],
+                                     (unsigned long long)pk);
+            if (pk < old) {
+              if (old == INF_STATE) {
+                int ti = atomicAdd(&sh.touched_cnt, 1);
+                if (ti < t_cap) touched[ti] = (int32_t)lw;
+              }
+              int fi = atomicAdd(&sh.fcnt[nxt], 1);
+              if (fi < f_cap)
+                fr[nxt][fi] = make_float4(tot_new, back_new,
+                                          __int_as_float(w), __int_as_float(v));
+              atomicMin(&sh.fmin_next, f32_bits(tot_new));
+              if (w == S.sink_node) atomicMin(&sh.best_sink_back, f32_bits(back_new));
+            }
+          }
+        }
+        __syncthreads();
+        if (tid == 0) {
+          if (sh.fcnt[nxt] > f_cap) sh.fail = FAIL_FRONTIER;
+          if (sh.touched_cnt > t_cap) sh.fail = FAIL_TOUCHED;
+        }
+        __syncthreads();
+        n_cur = min((int64_t)sh.fcnt[nxt], f_cap);
+        fmin = sh.fmin_next;
+        cur = nxt;
+      }
+      __syncthreads();
+
+      // ---- backtrack + commit (reference: backtrack
+      //      partitioning_multi_sink...:613-680 + route_tree add) ----
+      if (!sh.fail && tid == 0) {
+        int n = 0;
+        int32_t v = S.sink_node;
+        for (;;) {
+          uint64_t st = state[(size_t)L(g, v)];
+          int32_t prev = (int32_t)(st & 0xffffffffu);
+          if (prev == v) break;  // reached a tree seed
+          if (n >= PATH_CAP) { sh.fail = FAIL_PATH_CAP; break; }
+          sh.path[n++] = v;
+          v = prev;
+        }
+        sh.path_len = n;
+        sh.attach_idx = -1;
+        // stash attach node in fmin_next slot for the parallel scan
+        sh.fmin_next = (unsigned)v;
+      }
+      __syncthreads();
+      if (sh.fail) break;
+      // parallel scan for the attach node's tree index
+      {
+        int32_t attach_node = (int32_t)sh.fmin_next;
+        for (int k = tid; k < tree_len; k += WG_THREADS)
+          if (t_node[k] == attach_node) sh.attach_idx = k;
+      }
+      __syncthreads();
+      if (tid == 0) {
+        int ai = sh.attach_idx;
+        if (ai < 0) { sh.fail = FAIL_NO_PATH; }
+        else {
+          float dacc = t_delay[ai];
+          int parent = ai;
+          int len = tree_len;
+          for (int k = sh.path_len - 1; k >= 0; --k) {
+            int32_t u = sh.path[k];
+            int32_t pu = t_node[parent];
+            // find the switch of edge pu->u
+            int8_t sw = 0;
+            for (int32_t ei = g.row_ptr[pu]; ei < g.row_ptr[pu + 1]; ++ei)
+              if (g.edge_dst[ei] == u) { sw = g.edge_sw[ei]; break; }
+            dacc += hop_delay(g, sw, u);
+            if (len >= tcap) { sh.fail = FAIL_TREE_CAP; break; }
+            t_node[len] = u; t_parent[len] = parent; t_sw[len] = sw;
+            t_delay[len] = dacc;
+            parent = len;
+            ++len;
+            atomicAdd(&occ[u], 1);
+          }
+          tree_len = len;
+          trees.len[inet] = len;  // always accurate, even on FAIL_TREE_CAP,
+                                  // so the retry's rip-up stays balanced
+          if (!sh.fail) trees.sink_delay[si] = dacc;
+        }
+      }
+      __syncthreads();
+      if (sh.fail) break;
+      // broadcast updated tree_len to all threads
+      if (tid == 0) sh.fcnt[0] = tree_len;
+      __syncthreads();
+      tree_len = sh.fcnt[0];
+
+      // ---- sparse state reset (touched list) ----
+      int nt = min((int64_t)sh.touched_cnt, t_cap);
+      for (int k = tid; k < nt; k += WG_THREADS)
+        state[(size_t)(uint32_t)touched[k]] = INF_STATE;
+      __syncthreads();
+    }
+
+    if (sh.fail && tid == 0) fail_flags[inet] = sh.fail;
+    __syncthreads();
+    // if this net failed mid-sink, its touched entries were reset above only
+    // on success; do a full reset of touched here for safety
+    if (sh.fail) {
+      int nt = min((int64_t)sh.touched_cnt, t_cap);
+      for (int k = tid; k < nt; k += WG_THREADS)
+        state[(size_t)(uint32_t)touched[k]] = INF_STATE;
+      // if the touched list overflowed, fall back to a full clear
+      if (sh.touched_cnt > t_cap) {
+        int64_t cap = is_small ? small_cap : large_cap;
+        for (int64_t k = tid; k < cap; k += WG_THREADS) state[k] = INF_STATE;
+      }
+      __syncthreads();
+    }
+  }
+}
+
+// ---------------- congestion sweeps ----------------
+// reference: congestion.h:176-193 update_costs — acc_cost += overuse*acc_fac
+__global__ void update_acc_kernel(const int32_t* __restrict__ occ,
+                                  const int16_t* __restrict__ cap,
+                                  float* acc, float acc_fac, int32_t n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n; i += gridDim.x * blockDim.x) {
+    int over = occ[i] - cap[i];
+    if (over > 0) acc[i] += over * acc_fac;
+  }
+}
+
+// overuse census (reference: feasible_routing router.h:103 + census)
+__global__ void overuse_count_kernel(const int32_t* __restrict__ occ,
+                                     const int16_t* __restrict__ cap,
+                                     const int8_t* __restrict__ type,
+                                     int32_t* out /*[8]*/, int32_t n) {
+  __shared__ int32_t cnt[8];
+  if (threadIdx.x < 8) cnt[threadIdx.x] = 0;
+  __syncthreads();
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n; i += gridDim.x * blockDim.x) {
+    if (occ[i] > cap[i]) atomicAdd(&cnt[type[i]], 1);
+  }
+  __syncthreads();
+  if (threadIdx.x < 8 && cnt[threadIdx.x])
+    atomicAdd(&out[threadIdx.x], cnt[threadIdx.x]);
+}
+
+// occupancy recount from route trees (debug cross-check; reference:
+// recalculate_occ partitioning_multi_sink...:6194-6216)
+__global__ void recount_occ_kernel(TreesDev trees, const int32_t* __restrict__ net_ids,
+                                   int32_t n_nets, int32_t* recount) {
+  int inet_i = blockIdx.x;
+  if (inet_i >= n_nets) return;
+  int32_t inet = net_ids[inet_i];
+  int64_t off = trees.off[inet];
+  int32_t len = trees.len[inet];
+  for (int k = threadIdx.x; k < len; k += blockDim.x)
+    atomicAdd(&recount[trees.node[off + k]], 1);
+}
+
+}  // namespace pnrh
+
+// ---------------- C ABI launchers ----------------
+using namespace pnrh;
+
+extern "C" {
+
+struct RouteLaunchArgs {
+  // RRDev
+  const int8_t* type; const int16_t* xlow; const int16_t* ylow;
+  const int16_t* xhigh; const int16_t* yhigh; const int16_t* capacity;
+  const float* R; const float* C;
+  const int32_t* row_ptr; const int32_t* edge_dst; const int8_t* edge_sw;
+  const float* sw_R; const float* sw_Tdel; const float* base_cost;
+  const int32_t* idx_in_tile;
+  int32_t num_nodes, nx, ny, L, npt;
+  // NetsDev
+  const int32_t* net_src; const int32_t* sink_ptr; const int32_t* sink_rr;
+  const float* crit; const int16_t* bb; int32_t num_nets;
+  // TreesDev
+  const int64_t* tree_off; int32_t* tree_node; int32_t* tree_parent;
+  int8_t* tree_sw; float* tree_delay; int32_t* tree_len; float* sink_delay;
+  // params
+  float astar_fac, pres_fac, seg_delay, ipin_delay, seg_base, ipin_base;
+  int32_t max_rounds;
+  // queues
+  const int32_t* queue_small; int32_t n_queue_small;
+  const int32_t* queue_large; int32_t n_queue_large;
+  int32_t* q_cursors;
+  int32_t* occ; const float* acc;
+  uint64_t* state_base; int64_t small_cap; int64_t large_cap;
+  int32_t n_small_slots; int32_t n_large_slots;
+  float4* frontier_base; int64_t f_cap_small; int64_t f_cap_large;
+  int32_t* touched_base; int64_t t_cap_small; int64_t t_cap_large;
+  int32_t* fail_flags;
+};
+
+int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
+  RRDev g{a->type, a->xlow, a->ylow, a->xhigh, a->yhigh, a->capacity,
+          a->R, a->C, a->row_ptr, a->edge_dst, a->edge_sw,
+          a->sw_R, a->sw_Tdel, a->base_cost, a->idx_in_tile,
+          a->num_nodes, a->nx, a->ny, a->L, a->npt};
+  NetsDev nets{a->net_src, a->sink_ptr, a->sink_rr, a->crit, a->bb, a->num_nets};
+  TreesDev trees{a->tree_off, a->tree_node, a->tree_parent, a->tree_sw,
+                 a->tree_delay, a->tree_len, a->sink_delay};
+  RouteParams P{};
+  P.astar_fac = a->astar_fac; P.pres_fac = a->pres_fac;
+  P.seg_delay = a->seg_delay; P.ipin_delay = a->ipin_delay;
+  P.seg_base = a->seg_base; P.ipin_base = a->ipin_base;
+  P.max_rounds = a->max_rounds;
+  int grid = a->n_small_slots + a->n_large_slots;
+  hipLaunchKernelGGL(route_nets_kernel, dim3(grid), dim3(WG_THREADS), 0,
+                     (hipStream_t)stream,
+                     g, nets, trees, P,
+                     a->queue_small, a->n_queue_small,
+                     a->queue_large, a->n_queue_large,
+                     a->q_cursors, a->occ, a->acc,
+                     a->state_base, a->small_cap, a->large_cap,
+                     a->n_small_slots,
+                     a->frontier_base, a->f_cap_small, a->f_cap_large,
+                     a->touched_base, a->t_cap_small, a->t_cap_large,
+                     a->fail_flags);
+  return (int)hipGetLastError();
+}
+
+int pnr_update_acc(const int32_t* occ, const int16_t* cap, float* acc,
+                   float acc_fac, int32_t n, void* stream) {
+  int grid = min((n + 255) / 256, 2048);
+  hipLaunchKernelGGL(update_acc_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream, occ, cap, acc, acc_fac, n);
+  return (int)hipGetLastError();
+}
+
+int pnr_overuse_count(const int32_t* occ, const int16_t* cap, const int8_t* type,
+                      int32_t* out, int32_t n, void* stream) {
+  int grid = min((n + 255) / 256, 2048);
+  hipLaunchKernelGGL(overuse_count_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream, occ, cap, type, out, n);
+  return (int)hipGetLastError();
+}
+
+int pnr_recount_occ(const int64_t* tree_off, int32_t* tree_node, int32_t* tree_len,
+                    const int32_t* net_ids, int32_t n_nets, int32_t* recount,
+                    void* stream) {
+  TreesDev t{};
+  t.off = tree_off; t.node = tree_node; t.len = tree_len;
+  hipLaunchKernelGGL(recount_occ_kernel, dim3(n_nets), dim3(256), 0,
+                     (hipStream_t)stream, t, net_ids, n_nets, recount);
+  return (int)hipGetLastError();
+}
+
+}  // extern "C"
+
+namespace pnrh {
+__global__ void fill_u64_kernel(uint64_t* p, uint64_t v, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n; i += (int64_t)gridDim.x * blockDim.x) p[i] = v;
+}
+}
+
+extern "C" int pnr_fill_u64_launch(uint64_t* p, uint64_t v, int64_t n, void* stream) {
+  int64_t g64 = (n + 255) / 256;
+  int grid = (int)(g64 < 4096 ? g64 : 4096);
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(pnrh::fill_u64_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream, p, v, n);
+  return (int)hipGetLastError();
+}

@@ -1,0 +1,349 @@
+"""GPU PathFinder engine: HBM-resident rr graph + wavefront router kernel.
+
+Host side of csrc/hip/router_kernel.hip: uploads the SoA graph, sizes the
+per-slot search scratch, runs the rip-up-and-reroute outer loop with
+criticality-ordered sinks, and recovers failed nets by growing their
+bounding boxes (reference: bb_factor growth; route failures re-tried).
+"""
+import numpy as np
+
+from ..arch.archdef import ArchDef, RR_SINK
+from .. import ops
+from ..ops import hip_api
+from .router import net_rr_terminals, RouteResult
+
+
+def _torch():
+    import torch
+    return torch
+
+
+class GpuRouter:
+    def __init__(self, g, arch: ArchDef, src_rr, sink_ptr, sink_rr,
+                 device="cuda:0", astar_fac=1.2, n_small_slots=512,
+                 n_large_slots=8, bb_margin=3, max_rounds=200000,
+                 occ=None):
+        torch = _torch()
+        self.torch = torch
+        self.device = device
+        self.arch = arch
+        self.g = g
+        self.num_nets = len(src_rr)
+        self.astar_fac = astar_fac
+        self.bb_margin = bb_margin
+        self.max_rounds = max_rounds
+
+        if g.num_edges >= 2**31:
+            raise ValueError("edge count exceeds int32 CSR limit")
+
+        # ---- graph upload (SoA, untransformed) ----
+        def up(a, dtype=None):
+            t = torch.from_numpy(np.ascontiguousarray(a))
+            if dtype is not None:
+                t = t.to(dtype)
+            return t.to(device)
+
+        self.t_type = up(np.asarray(g.type))
+        xlow = np.asarray(g.xlow); ylow = np.asarray(g.ylow)
+        self.t_xlow = up(xlow); self.t_ylow = up(ylow)
+        self.t_xhigh = up(np.asarray(g.xhigh)); self.t_yhigh = up(np.asarray(g.yhigh))
+        self.t_cap = up(np.asarray(g.capacity))
+        self.t_R = up(np.asarray(g.node_R)); self.t_C = up(np.asarray(g.node_C))
+        self.t_row_ptr = up(np.asarray(g.row_ptr).astype(np.int32))
+        self.t_edge_dst = up(np.asarray(g.edge_dst))
+        self.t_edge_sw = up(np.asarray(g.edge_sw))
+        self.t_sw_R = up(np.asarray(g.sw_R)); self.t_sw_Tdel = up(np.asarray(g.sw_Tdel))
+        self.t_base_cost = up(np.asarray(g.base_cost))
+
+        # dense per-tile node index (anchor = (xlow, ylow))
+        gy = arch.ny + 2
+        tile = xlow.astype(np.int64) * gy + ylow.astype(np.int64)
+        order = np.argsort(tile, kind="stable")
+        sorted_tile = tile[order]
+        starts = np.r_[0, np.nonzero(np.diff(sorted_tile))[0] + 1]
+        counts = np.diff(np.r_[starts, len(tile)])
+        idx_in_tile = np.empty(len(tile), dtype=np.int32)
+        pos = np.arange(len(tile)) - np.repeat(starts, counts)
+        idx_in_tile[order] = pos.astype(np.int32)
+        self.npt = int(counts.max())
+        self.t_idx_in_tile = up(idx_in_tile)
+
+        # ---- nets ----
+        self.src_rr = np.asarray(src_rr, dtype=np.int32)
+        self.sink_ptr = np.asarray(sink_ptr, dtype=np.int32)
+        self.sink_rr = np.asarray(sink_rr, dtype=np.int32)
+        self.n_sinks = len(self.sink_rr)
+        self.t_net_src = up(self.src_rr)
+        self.t_sink_ptr = up(self.sink_ptr)
+
+        # bounding boxes (terminal bb + margin)
+        self.bb_margin_per_net = np.full(self.num_nets, bb_margin, dtype=np.int32)
+        self.bb = self._compute_bbs()
+        self.t_bb = up(self.bb)
+
+        # slot classes
+        areas = self._bb_areas(self.bb)
+        self.bb_max_small_area = int(min(max(int(areas.max()), 64), 4096))
+        self.n_small_slots = n_small_slots
+        self.n_large_slots = n_large_slots
+        self.small_cap = self.bb_max_small_area * self.npt
+        self.large_cap = ((arch.nx + 2) * (arch.ny + 2)) * self.npt
+        # if the whole chip fits in a small slot, drop the large class size
+        if self.large_cap <= self.small_cap:
+            self.large_cap = self.small_cap
+
+        self.f_cap_small = 1 << 16
+        self.f_cap_large = 1 << 21
+        self.t_cap_small = self.small_cap
+        self.t_cap_large = self.large_cap
+
+        nb = torch.int64
+        self.t_state = torch.empty(
+            n_small_slots * self.small_cap + n_large_slots * self.large_cap,
+            dtype=nb, device=device)
+        self.t_frontier = torch.empty(
+            (n_small_slots * 2 * self.f_cap_small +
+             n_large_slots * 2 * self.f_cap_large) * 4,
+            dtype=torch.float32, device=device)
+        self.t_touched = torch.empty(
+            n_small_slots * self.t_cap_small + n_large_slots * self.t_cap_large,
+            dtype=torch.int32, device=device)
+
+        # trees: per-net capacity
+        nsinks = np.diff(self.sink_ptr)
+        bw = self.bb[:, 2] - self.bb[:, 0] + 1
+        bh = self.bb[:, 3] - self.bb[:, 1] + 1
+        cap = 32 + nsinks * (2 * (bw.astype(np.int64) + bh.astype(np.int64)) + 64)
+        self.tree_off = np.r_[0, np.cumsum(cap)].astype(np.int64)
+        total_tree = int(self.tree_off[-1])
+        self.t_tree_off = up(self.tree_off)
+        self.t_tree_node = torch.zeros(total_tree, dtype=torch.int32, device=device)
+        self.t_tree_parent = torch.zeros(total_tree, dtype=torch.int32, device=device)
+        self.t_tree_sw = torch.zeros(total_tree, dtype=torch.int8, device=device)
+        self.t_tree_delay = torch.zeros(total_tree, dtype=torch.float32, device=device)
+        self.t_tree_len = torch.zeros(self.num_nets, dtype=torch.int32, device=device)
+        self.t_sink_delay = torch.zeros(self.n_sinks, dtype=torch.float32, device=device)
+
+        # congestion
+        if occ is None:
+            self.t_occ = torch.zeros(g.num_nodes, dtype=torch.int32, device=device)
+        else:
+            self.t_occ = occ
+        self.t_acc = torch.ones(g.num_nodes, dtype=torch.float32, device=device)
+        self.t_fail = torch.zeros(self.num_nets, dtype=torch.int32, device=device)
+        self.t_cursors = torch.zeros(2, dtype=torch.int32, device=device)
+        self.t_overuse = torch.zeros(8, dtype=torch.int32, device=device)
+
+        # lookahead constants (same as serial oracle)
+        self.seg_delay = float(arch.T_sw + arch.C_wire * arch.L *
+                               (arch.R_sw + 0.5 * arch.R_wire * arch.L))
+        self.ipin_delay = float(arch.T_ipin)
+        self.seg_base = float(arch.base_costs()[4])
+        self.ipin_base = float(arch.base_costs()[3])
+
+        # init state to INF once
+        self._fill_state()
+        self.lib = hip_api.lib()
+
+    # ---- helpers ----
+    def _compute_bbs(self):
+        xlow = np.asarray(self.g.xlow); ylow = np.asarray(self.g.ylow)
+        nx, ny = self.arch.nx, self.arch.ny
+        bb = np.zeros((self.num_nets, 4), dtype=np.int16)
+        for n in range(self.num_nets):
+            terms = np.r_[self.src_rr[n],
+                          self.sink_rr[self.sink_ptr[n]:self.sink_ptr[n + 1]]]
+            xs = xlow[terms]; ys = ylow[terms]
+            m = self.bb_margin_per_net[n]
+            bb[n] = (max(0, xs.min() - m), max(0, ys.min() - m),
+                     min(nx + 1, xs.max() + m), min(ny + 1, ys.max() + m))
+        return bb
+
+    def _bb_areas(self, bb):
+        return ((bb[:, 2] - bb[:, 0] + 1).astype(np.int64) *
+                (bb[:, 3] - bb[:, 1] + 1).astype(np.int64))
+
+    def _fill_state(self):
+        rc = hip_api.lib().pnr_fill_u64_launch(
+            ct_ptr(self.t_state), 0xFFFFFFFFFFFFFFFF, self.t_state.numel(),
+            self._stream())
+        hip_api.check(rc, "fill_state")
+
+    def _stream(self):
+        return self.torch.cuda.current_stream().cuda_stream
+
+    # ---- one PathFinder iteration ----
+    def route_iteration(self, crit, pres_fac):
+        """crit: per-sink criticality aligned with sink_rr (original order).
+        Returns (overused_count, sink_delays aligned with original order)."""
+        torch = self.torch
+        # order sinks by decreasing criticality within each net
+        net_of_sink = np.repeat(np.arange(self.num_nets), np.diff(self.sink_ptr))
+        perm = np.lexsort((-crit, net_of_sink))
+        t_sink_rr = torch.from_numpy(self.sink_rr[perm]).to(self.device)
+        t_crit = torch.from_numpy(np.ascontiguousarray(crit[perm], dtype=np.float32)).to(self.device)
+
+        areas = self._bb_areas(self.bb)
+        small = np.nonzero(areas <= self.bb_max_small_area)[0].astype(np.int32)
+        large = np.nonzero(areas > self.bb_max_small_area)[0].astype(np.int32)
+        # route hardest (most sinks) first within each class for load balance
+        nsinks = np.diff(self.sink_ptr)
+        small = small[np.argsort(-nsinks[small], kind="stable")]
+        large = large[np.argsort(-nsinks[large], kind="stable")]
+
+        attempts = 0
+        while True:
+            q_small = torch.from_numpy(small).to(self.device)
+            q_large = torch.from_numpy(large).to(self.device)
+            self.t_cursors.zero_()
+            self.t_fail.zero_()
+            args = self._make_args(t_sink_rr, t_crit, q_small, q_large, pres_fac)
+            rc = self.lib.pnr_route_nets(hip_api.ct.byref(args), self._stream())
+            hip_api.check(rc, "route_nets")
+            torch.cuda.synchronize(self.device)
+            fail = self.t_fail.cpu().numpy()
+            failed = np.nonzero(fail)[0]
+            if len(failed) == 0:
+                break
+            attempts += 1
+            if attempts > 6:
+                raise RuntimeError(
+                    f"router: {len(failed)} nets failed after retries "
+                    f"(codes {np.unique(fail[failed])})")
+            # grow bb of failed nets and retry just those
+            self.bb_margin_per_net[failed] = np.minimum(
+                self.bb_margin_per_net[failed] * 2 + 4,
+                max(self.arch.nx, self.arch.ny) + 2)
+            self.bb = self._compute_bbs()
+            self.t_bb.copy_(torch.from_numpy(self.bb).to(self.device))
+            # state may be dirty for failed slots; refill (cheap)
+            self._fill_state()
+            areas = self._bb_areas(self.bb)
+            fsmall = failed[areas[failed] <= self.bb_max_small_area].astype(np.int32)
+            flarge = failed[areas[failed] > self.bb_max_small_area].astype(np.int32)
+            small, large = fsmall, flarge
+
+        overused = int((self.t_occ > self.t_cap.to(torch.int32)).sum().item())
+        sd = self.t_sink_delay.cpu().numpy()
+        sink_delays = np.empty_like(sd)
+        sink_delays[perm] = sd
+        return overused, sink_delays
+
+    def update_acc(self, acc_fac):
+        rc = self.lib.pnr_update_acc(
+            ct_ptr(self.t_occ), ct_ptr(self.t_cap), ct_ptr(self.t_acc),
+            hip_api.ct.c_float(acc_fac), self.g.num_nodes, self._stream())
+        hip_api.check(rc, "update_acc")
+
+    def wirelength(self):
+        torch = self.torch
+        chan = (self.t_type == 4) | (self.t_type == 5)
+        length = (self.t_xhigh - self.t_xlow + self.t_yhigh - self.t_ylow + 1).to(torch.int64)
+        return int((self.t_occ.to(torch.int64) * length * chan.to(torch.int64)).sum().item())
+
+    def check_occ_recount(self):
+        """Debug cross-check: occ == recount over route trees."""
+        torch = self.torch
+        recount = torch.zeros_like(self.t_occ)
+        ids = torch.arange(self.num_nets, dtype=torch.int32, device=self.device)
+        rc = self.lib.pnr_recount_occ(
+            ct_ptr(self.t_tree_off), ct_ptr(self.t_tree_node),
+            ct_ptr(self.t_tree_len), ct_ptr(ids), self.num_nets,
+            ct_ptr(recount), self._stream())
+        hip_api.check(rc, "recount_occ")
+        torch.cuda.synchronize(self.device)
+        return bool((recount == self.t_occ).all().item())
+
+    def get_tree(self, inet):
+        o0, o1 = int(self.tree_off[inet]), int(self.tree_off[inet + 1])
+        ln = int(self.t_tree_len[inet].item())
+        return (self.t_tree_node[o0:o0 + ln].cpu().numpy(),
+                self.t_tree_parent[o0:o0 + ln].cpu().numpy(),
+                self.t_tree_sw[o0:o0 + ln].cpu().numpy(),
+                self.t_tree_delay[o0:o0 + ln].cpu().numpy())
+
+    def _make_args(self, t_sink_rr, t_crit, q_small, q_large, pres_fac):
+        a = hip_api.RouteLaunchArgs()
+        a.type = ct_ptr(self.t_type); a.xlow = ct_ptr(self.t_xlow)
+        a.ylow = ct_ptr(self.t_ylow); a.xhigh = ct_ptr(self.t_xhigh)
+        a.yhigh = ct_ptr(self.t_yhigh); a.capacity = ct_ptr(self.t_cap)
+        a.R = ct_ptr(self.t_R); a.C = ct_ptr(self.t_C)
+        a.row_ptr = ct_ptr(self.t_row_ptr); a.edge_dst = ct_ptr(self.t_edge_dst)
+        a.edge_sw = ct_ptr(self.t_edge_sw); a.sw_R = ct_ptr(self.t_sw_R)
+        a.sw_Tdel = ct_ptr(self.t_sw_Tdel); a.base_cost = ct_ptr(self.t_base_cost)
+        a.idx_in_tile = ct_ptr(self.t_idx_in_tile)
+        a.num_nodes = self.g.num_nodes; a.nx = self.arch.nx; a.ny = self.arch.ny
+        a.L = self.arch.L; a.npt = self.npt
+        a.net_src = ct_ptr(self.t_net_src); a.sink_ptr = ct_ptr(self.t_sink_ptr)
+        a.sink_rr = ct_ptr(t_sink_rr); a.crit = ct_ptr(t_crit)
+        a.bb = ct_ptr(self.t_bb); a.num_nets = self.num_nets
+        a.tree_off = ct_ptr(self.t_tree_off); a.tree_node = ct_ptr(self.t_tree_node)
+        a.tree_parent = ct_ptr(self.t_tree_parent); a.tree_sw = ct_ptr(self.t_tree_sw)
+        a.tree_delay = ct_ptr(self.t_tree_delay); a.tree_len = ct_ptr(self.t_tree_len)
+        a.sink_delay = ct_ptr(self.t_sink_delay)
+        a.astar_fac = self.astar_fac; a.pres_fac = pres_fac
+        a.seg_delay = self.seg_delay; a.ipin_delay = self.ipin_delay
+        a.seg_base = self.seg_base; a.ipin_base = self.ipin_base
+        a.max_rounds = self.max_rounds
+        a.queue_small = ct_ptr(q_small); a.n_queue_small = len(q_small)
+        a.queue_large = ct_ptr(q_large); a.n_queue_large = len(q_large)
+        a.q_cursors = ct_ptr(self.t_cursors)
+        a.occ = ct_ptr(self.t_occ); a.acc = ct_ptr(self.t_acc)
+        a.state_base = ct_ptr(self.t_state)
+        a.small_cap = self.small_cap; a.large_cap = self.large_cap
+        a.n_small_slots = self.n_small_slots; a.n_large_slots = self.n_large_slots
+        a.frontier_base = ct_ptr(self.t_frontier)
+        a.f_cap_small = self.f_cap_small; a.f_cap_large = self.f_cap_large
+        a.touched_base = ct_ptr(self.t_touched)
+        a.t_cap_small = self.t_cap_small; a.t_cap_large = self.t_cap_large
+        a.fail_flags = ct_ptr(self.t_fail)
+        self._args_keepalive = (t_sink_rr, t_crit, q_small, q_large)
+        return a
+
+
+def ct_ptr(t):
+    return hip_api.ct.c_void_p(t.data_ptr())
+
+
+def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
+                         pres_fac_init=0.5, pres_fac_mult=1.3, acc_fac=1.0,
+                         astar_fac=1.2, verbose=False, device="cuda:0"):
+    """GPU PathFinder outer loop — mirrors route.router.pathfinder_route."""
+    net_ids, src_rr, sink_ptr, sink_rr, conn_index = net_rr_terminals(
+        netlist, placement, g, arch)
+    router = GpuRouter(g, arch, src_rr, sink_ptr.astype(np.int32), sink_rr,
+                       device=device, astar_fac=astar_fac)
+    n_rsinks = len(sink_rr)
+    crit = np.zeros(n_rsinks, dtype=np.float32)
+    conn_delay = np.zeros(netlist.num_conns, dtype=np.float32)
+    pres_fac = 0.0   # first iteration: congestion-blind (VPR style)
+    cpd = 0.0
+    history = []
+    it = 0
+    overused = -1
+    for it in range(1, max_iters + 1):
+        overused, sink_delays = router.route_iteration(crit, pres_fac)
+        history.append(dict(iter=it, overused=int(overused), cpd=cpd))
+        if verbose:
+            print(f"[gpu] iter {it}: overused={overused} cpd={cpd*1e9:.2f}ns")
+        if sta is not None:
+            for k, (idx, inv) in enumerate(conn_index):
+                conn_delay[idx] = sink_delays[sink_ptr[k]:sink_ptr[k + 1]][inv]
+            cpd, slack, c = sta.analyze(conn_delay)
+            for k, (idx, inv) in enumerate(conn_index):
+                seg = np.zeros(sink_ptr[k + 1] - sink_ptr[k], dtype=np.float32)
+                np.maximum.at(seg, inv, c[idx])
+                crit[sink_ptr[k]:sink_ptr[k + 1]] = seg
+        if overused == 0:
+            break
+        pres_fac = pres_fac_init if pres_fac == 0.0 else pres_fac * pres_fac_mult
+        router.update_acc(acc_fac)
+
+    ok = overused == 0
+    if ok and not router.check_occ_recount():
+        raise RuntimeError("GPU route: occ recount mismatch")
+    return RouteResult(success=ok, iterations=it, overused=int(overused),
+                       wirelength=router.wirelength(), crit_path_delay=cpd,
+                       stats={"history": history,
+                              "num_routed_nets": len(net_ids)},
+                       router=router)
