@@ -1,0 +1,165 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: PathFinder routing throughput on the bitcoin_miner-scale
+synthetic config (BASELINE.json config 5), 1-8 MI355X.
+
+A step = one full PathFinder iteration: rip-up & reroute of EVERY net
+(criticality-ordered sinks, GPU wavefront kernel), the RCCL occ all-reduce
+(N>1), replicated STA, and the acc-cost sweep — i.e. the complete per-
+iteration work of the place+route flow's routing stage, which dominates the
+reference's wall-clock metric. Setup (synthetic netlist+placement, rr-graph
+build, upload) is untimed.
+
+Launch (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  # N>1: python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+  #        --master-addr 127.0.0.1 bench.py --gpus N ...
+"""
+import argparse
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent))
+
+import numpy as np
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=4)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--config", type=str, default="bitcoin_miner",
+                    help="named arch scale (BASELINE configs)")
+    ap.add_argument("--fill", type=float, default=0.6)
+    ap.add_argument("--seed", type=int, default=1)
+    ap.add_argument("--verbose", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+    from parallel_eda_amd.arch.archdef import get_arch
+    from parallel_eda_amd.io.synth import synth_placed_netlist
+    from parallel_eda_amd.route.router import net_rr_terminals, ConnMap
+    from parallel_eda_amd.route.gpu_router import GpuRouter
+    from parallel_eda_amd.timing.sta import STA
+    from parallel_eda_amd.parallel.dist import (init_dist, DistRouteLoop,
+                                                GpuEngine)
+    from parallel_eda_amd import rrgraph
+
+    rank, ws, local = init_dist()
+    assert ws == args.gpus or ws == 1, f"WORLD_SIZE {ws} != --gpus {args.gpus}"
+    device = f"cuda:{local}" if torch.cuda.is_available() else "cpu"
+    if device == "cpu":
+        print(json.dumps({"error": "no GPU available; bench requires MI355X"}))
+        return 1
+    torch.cuda.set_device(device)
+
+    def log(*a):
+        if rank == 0 and args.verbose:
+            print(*a, file=sys.stderr, flush=True)
+
+    t_setup = time.perf_counter()
+    arch = get_arch(args.config)
+    nl, pl = synth_placed_netlist(arch, fill=args.fill, seed=args.seed)
+    log(f"netlist: {nl.num_blocks} blocks, {nl.num_nets} nets, "
+        f"{nl.num_conns} conns")
+    g = rrgraph.build_rr_graph(arch)
+    log(f"rr graph: {g.num_nodes} nodes, {g.num_edges} edges")
+    sta = STA(nl, arch)
+    net_ids, src_rr, sink_ptr, sink_rr, conn_index = net_rr_terminals(
+        nl, pl, g, arch)
+    n_rsinks = len(sink_rr)
+    cmap = ConnMap(conn_index, sink_ptr, nl.num_conns, n_rsinks)
+    router = GpuRouter(g, arch, src_rr, sink_ptr.astype(np.int32), sink_rr,
+                       device=device)
+    engine = GpuEngine(router)
+    loop = DistRouteLoop(engine, len(net_ids), router.bb, n_rsinks,
+                         sink_ptr, rank=rank, world_size=ws)
+    log(f"setup {time.perf_counter()-t_setup:.1f}s; routed nets "
+        f"{len(net_ids)}, sinks {n_rsinks}, my nets {len(loop.my_nets)}")
+
+    crit = np.zeros(n_rsinks, dtype=np.float32)
+    conn_delay = np.zeros(nl.num_conns, dtype=np.float32)
+    pres_fac = 0.0
+    cpd = 0.0
+
+    dist = None
+    if ws > 1:
+        import torch.distributed as dist
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        torch.cuda.synchronize(device)
+
+    def step():
+        nonlocal pres_fac, cpd, crit
+        over, sd = loop.iteration(crit, pres_fac, acc_fac=1.0)
+        cmap.conn_delays(sd, out=conn_delay)
+        cpd, slack, c = sta.analyze(conn_delay)
+        crit = cmap.sink_crit(c)
+        pres_fac = 0.5 if pres_fac == 0.0 else min(pres_fac * 1.3, 1000.0)
+        return over
+
+    for w in range(args.warmup):
+        t0 = time.perf_counter()
+        over = step()
+        log(f"warmup {w}: overused={over} cpd={cpd*1e9:.2f}ns "
+            f"{time.perf_counter()-t0:.1f}s")
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for k in range(args.steps):
+        over = step()
+        log(f"step {k}: overused={over} cpd={cpd*1e9:.2f}ns")
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+    if dist is not None:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    # whole-job aggregate: sink connections (re)routed per second across
+    # all GPUs — each step reroutes every net once, split across ranks.
+    value = n_rsinks * args.steps / elapsed
+    if rank == 0:
+        out = {
+            "metric": "routed_sink_connections_per_s "
+                      "(place+route flow, routing-dominated; "
+                      "bitcoin_miner-scale synthetic on stratixiv-like fabric)",
+            "value": round(value, 2),
+            "unit": "sinks/s",
+            "n_gpus": args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic (no network: synthetic netlist+placement of "
+                    "the named scale, random seed %d)" % args.seed,
+            "config": {
+                "model": args.config,
+                "arch": arch.name,
+                "grid": f"{arch.nx}x{arch.ny}", "W": arch.W,
+                "rr_nodes": g.num_nodes, "rr_edges": int(g.num_edges),
+                "nets": int(nl.num_nets), "sinks": int(n_rsinks),
+                "global_batch": int(n_rsinks), "seq_len": 0,
+                "parallelism": f"spatial-partition dp{args.gpus} + "
+                               f"RCCL occ all-reduce per iteration",
+                "final_overused": int(over),
+                "crit_path_ns": round(cpd * 1e9, 3),
+            },
+        }
+        print(json.dumps(out))
+    if dist is not None:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

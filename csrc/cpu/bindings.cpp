@@ -138,6 +138,20 @@ PYBIND11_MODULE(_pnr_cpu, m) {
         { py::gil_scoped_release rel; over = r.route_iteration(c); }
         return over;
       })
+      .def("route_subset", [](SerialRouter& r,
+                              py::array_t<float, py::array::c_style | py::array::forcecast> crit,
+                              py::array_t<int32_t, py::array::c_style | py::array::forcecast> ids) {
+        const float* c = crit.size() ? crit.data() : nullptr;
+        int64_t over;
+        { py::gil_scoped_release rel; over = r.route_subset(c, ids.data(), ids.size()); }
+        return over;
+      })
+      .def("set_occ", [](SerialRouter& r,
+                         py::array_t<int32_t, py::array::c_style | py::array::forcecast> occ) {
+        if ((int64_t)occ.size() != (int64_t)r.g_->num_nodes)
+          throw std::runtime_error("bad occ length");
+        r.set_occ(occ.data());
+      })
       .def("update_costs", &SerialRouter::update_costs)
       .def("set_pres_fac", &SerialRouter::set_pres_fac)
       .def("count_overused", &SerialRouter::count_overused)

@@ -76,6 +76,23 @@ class SerialRouter {
     return count_overused();
   }
 
+  // Route only the given nets (distributed partitioning / selective reroute;
+  // reference: mpi router routes partition_nets[rank], mpi_route...cxx:936).
+  int64_t route_subset(const float* crit, const int32_t* ids, int64_t n) {
+    for (int64_t i = 0; i < n; ++i) route_net(ids[i], crit);
+    return count_overused();
+  }
+
+  // Replace occupancy wholesale (after a distributed occ all-reduce) and
+  // refresh pres_cost from it.
+  void set_occ(const int32_t* occ) {
+    for (int v = 0; v < g_->num_nodes; ++v) {
+      occ_[v] = occ[v];
+      int over = occ_[v] + 1 - g_->capacity[v];
+      pres_[v] = (over > 0) ? 1.0f + over * pres_fac_ : 1.0f;
+    }
+  }
+
   // PathFinder cost-schedule update between iterations
   // (reference: congestion.h:176-193 update_costs).
   void update_costs(float pres_fac, float acc_fac) {
@@ -303,6 +320,10 @@ class SerialRouter {
 
   void push(std::priority_queue<HeapEnt>& heap, int32_t v, float tot,
             float back, float rup, int32_t prev, int32_t sw) {
+    // a node seeded from the route tree is final for this sink's search:
+    // re-entering the tree through real wires would duplicate tree nodes
+    // on backtrack (and costs more congestion anyway)
+    if (prev_node_[v] <= -2) return;
     if (path_cost_[v] >= 1e29f) touched_.push_back(v);
     if (tot < path_cost_[v]) {
       path_cost_[v] = tot;

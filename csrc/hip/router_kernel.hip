@@ -223,7 +223,13 @@ __global__ void route_nets_kernel(
         float back = S.crit * t_delay[k];
         float tot = back + S.astar_fac * expected_cost(g, P, v, S);
         int64_t li = L(g, v);
-        state[li] = pack_state(back, v);  // prev==self marks a tree seed
+        // prev==self marks a tree seed; stored back 0.0 makes the seed
+        // UNBEATABLE by any ordinary entry (all edge costs > 0), so the
+        // backtrack's prev==self stop condition is stable and a path can
+        // never re-enter the tree (which would duplicate tree nodes and
+        // leave phantom occupancy). The frontier entry carries the REAL
+        // back cost for expansion.
+        state[li] = pack_state(0.0f, v);
         int ti = atomicAdd(&sh.touched_cnt, 1);
         if (ti < t_cap) touched[ti] = (int32_t)li;
         int fi = atomicAdd(&sh.fcnt[0], 1);
@@ -256,7 +262,11 @@ __global__ void route_nets_kernel(
           int32_t v = __float_as_int(e.z);
           int32_t prev = __float_as_int(e.w);
           int64_t li = L(g, v);
-          if (state[li] != pack_state(back, prev)) continue;  // stale entry
+          // stale check: seeds (prev==self) store back 0.0 in state while
+          // their entry carries the real back cost
+          const uint64_t expect = (prev == v) ? pack_state(0.0f, v)
+                                              : pack_state(back, prev);
+          if (state[li] != expect) continue;  // stale entry
           if (tot > thr) {
             // keep for a later bucket
             int fi = atomicAdd(&sh.fcnt[nxt], 1);

@@ -73,6 +73,149 @@ class NetlistPy:
         return self.net_sink_ptr[n + 1] - self.net_sink_ptr[n]
 
 
+def synth_placed_netlist(arch: ArchDef, fill: float = 0.85,
+                         avg_fanout: float = 3.0, locality: float = 6.0,
+                         seq_frac: float = 0.35, seed: int = 1):
+    """Generate a netlist WITH a placement, nets geometrically local.
+
+    For large configs (LU32PEEng/bgm/bitcoin_miner scale) a full SA anneal
+    as benchmark *setup* would dominate; real routing benchmarks start from
+    a quality placement, so we synthesize netlist+placement jointly: blocks
+    sit on the grid and each net connects a driver to sinks drawn from a
+    local neighborhood (geometric decay, Rent-flavored long tail).
+    Returns (NetlistPy, Placement-like object with x/y/slot arrays).
+    """
+    from ..place.placer import Placement
+    rng = np.random.default_rng(seed)
+    nx, ny, io_cap = arch.nx, arch.ny, arch.io_cap
+    n_clb = int(nx * ny * fill)
+    n_io_pairs = max(2, min(int(0.06 * n_clb), (nx + ny) * io_cap // 2 - 2))
+    n_in = n_out = n_io_pairs
+    n_io = n_in + n_out
+    nb = n_io + n_clb
+
+    block_type = np.full(nb, BLK_CLB, dtype=np.int8)
+    block_type[:n_io] = BLK_IO
+    block_is_seq = np.zeros(nb, dtype=np.uint8)
+    block_is_seq[:n_io] = 1
+    clb0 = n_io
+    block_is_seq[clb0:] = (rng.random(n_clb) < seq_frac).astype(np.uint8)
+
+    # placement: CLBs into random distinct tiles; IOs into perimeter slots
+    clb_tiles = rng.choice(nx * ny, size=n_clb, replace=False)
+    cx = (clb_tiles // ny + 1).astype(np.int32)
+    cy = (clb_tiles % ny + 1).astype(np.int32)
+    io_locs = ([(0, y) for y in range(1, ny + 1)] +
+               [(nx + 1, y) for y in range(1, ny + 1)] +
+               [(x, 0) for x in range(1, nx + 1)] +
+               [(x, ny + 1) for x in range(1, nx + 1)])
+    io_slots = [(x, y, s) for (x, y) in io_locs for s in range(io_cap)]
+    sel = rng.choice(len(io_slots), size=n_io, replace=False)
+    bx = np.zeros(nb, dtype=np.int32)
+    by = np.zeros(nb, dtype=np.int32)
+    bslot = np.zeros(nb, dtype=np.int32)
+    for i, k in enumerate(sel):
+        x, y, s = io_slots[k]
+        bx[i], by[i], bslot[i] = x, y, s
+    bx[clb0:], by[clb0:] = cx, cy
+
+    # spatial index of CLBs for locality draws
+    order = np.lexsort((cy, cx))
+    # nets: every input pad + every CLB drives one net
+    drivers = np.concatenate([
+        np.arange(n_in, dtype=np.int32),
+        np.arange(clb0, nb, dtype=np.int32)])
+    n_nets = len(drivers)
+    rank = rng.permutation(n_clb)
+
+    fanin = np.zeros(nb, dtype=np.int64)
+    max_fanin = np.full(nb, arch.clb_in, dtype=np.int64)
+    max_fanin[:n_io] = 1
+    # KD-free locality: bucket CLBs per tile-cell grid
+    cell = max(2, int(locality))
+    ncx = (nx + cell - 1) // cell
+    ncy = (ny + cell - 1) // cell
+    buckets = [[] for _ in range(ncx * ncy)]
+    for i in range(n_clb):
+        bxi = min((cx[i] - 1) // cell, ncx - 1)
+        byi = min((cy[i] - 1) // cell, ncy - 1)
+        buckets[bxi * ncy + byi].append(i)
+
+    sink_lists = []
+    out_pad_driven = np.zeros(n_out, dtype=bool)
+    for drv in drivers:
+        fanout = max(1, int(rng.poisson(avg_fanout)))
+        if drv < n_in:
+            dxx, dyy = 1 + rng.integers(nx), 1 + rng.integers(ny)
+        else:
+            dxx, dyy = bx[drv], by[drv]
+        drv_is_comb = drv >= clb0 and not block_is_seq[drv]
+        drv_rank = rank[drv - clb0] if drv_is_comb else -1
+        sinks = set()
+        for _ in range(fanout * 6):
+            if len(sinks) >= fanout:
+                break
+            if rng.random() < 0.03 and n_out > 0:
+                o = int(rng.integers(n_out))
+                blk = n_in + o
+                if out_pad_driven[o] or blk in sinks:
+                    continue
+                sinks.add(blk)
+                out_pad_driven[o] = True
+                fanin[blk] += 1
+                continue
+            # locality: pick a cell near the driver (geometric radius),
+            # occasionally a uniform long-range sink
+            if rng.random() < 0.08:
+                c = int(rng.integers(n_clb))
+            else:
+                r = rng.geometric(1.0 / max(1.0, locality / cell))
+                ang = rng.random() * 2 * np.pi
+                tx = int(dxx + np.cos(ang) * r * cell)
+                ty = int(dyy + np.sin(ang) * r * cell)
+                bxi = min(max((tx - 1) // cell, 0), ncx - 1)
+                byi = min(max((ty - 1) // cell, 0), ncy - 1)
+                bl = buckets[bxi * ncy + byi]
+                if not bl:
+                    continue
+                c = bl[int(rng.integers(len(bl)))]
+            blk = clb0 + c
+            if blk == drv or blk in sinks:
+                continue
+            if fanin[blk] >= max_fanin[blk]:
+                continue
+            if not block_is_seq[blk] and drv_rank >= 0 and rank[c] <= drv_rank:
+                continue
+            sinks.add(blk)
+            fanin[blk] += 1
+        if not sinks:
+            for c in rng.permutation(n_clb)[:64]:
+                blk = clb0 + int(c)
+                if blk != drv and (block_is_seq[blk] or drv_rank < 0 or
+                                   rank[int(c)] > drv_rank) \
+                        and fanin[blk] < max_fanin[blk]:
+                    sinks.add(blk)
+                    fanin[blk] += 1
+                    break
+        if not sinks:
+            o = int(np.argmin(out_pad_driven))
+            sinks.add(n_in + o)
+            out_pad_driven[o] = True
+        sink_lists.append(sorted(sinks))
+    for o in range(n_out):
+        if not out_pad_driven[o]:
+            n = int(rng.integers(n_nets))
+            sink_lists[n].append(n_in + o)
+
+    sink_ptr = np.zeros(n_nets + 1, dtype=np.int64)
+    for i, s in enumerate(sink_lists):
+        sink_ptr[i + 1] = sink_ptr[i] + len(s)
+    net_sinks = np.concatenate([np.asarray(s, dtype=np.int32) for s in sink_lists])
+    nl = NetlistPy(block_type, block_is_seq, drivers, sink_ptr, net_sinks)
+    pl = Placement(bx, by, bslot)
+    return nl, pl
+
+
 def synth_netlist(spec: SynthSpec) -> NetlistPy:
     """Generate a deterministic synthetic netlist.
 

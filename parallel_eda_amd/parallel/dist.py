@@ -1,0 +1,172 @@
+"""Multi-GPU (and multi-process CPU) distributed routing.
+
+Replaces the reference's MPI net-partitioned PathFinder
+(mpi_route_load_balanced_nonblocking_send_recv_encoded.cxx:402) with the
+MI355X-native model from SURVEY.md section 5.8: one process per GPU over
+RCCL/xGMI ("nccl" backend; "gloo" for CPU tests), every rank holding the
+FULL rr graph + congestion, nets partitioned SPATIALLY (region cuts = GPU
+boundaries), and one integer occ all-reduce per PathFinder iteration in
+place of the reference's per-net broadcast stream. STA is replicated per
+rank (removes the reference's root-Scatterv serialization, A:1473-1496).
+
+Integer occ deltas + fixed wave schedules make the whole distributed
+iteration DETERMINISTIC (integer sums are order-independent).
+"""
+import os
+
+import numpy as np
+
+
+def _dist():
+    import torch.distributed as dist
+    return dist
+
+
+def init_dist():
+    """Initialize torch.distributed from torchrun env; no-op if WORLD_SIZE<=1.
+
+    Returns (rank, world_size, local_rank)."""
+    import torch
+    dist = _dist()
+    ws = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local = int(os.environ.get("LOCAL_RANK", "0"))
+    if ws > 1 and not dist.is_initialized():
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29517")
+        dist.init_process_group(backend=backend, rank=rank, world_size=ws)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local)
+    return rank, ws, local
+
+
+def spatial_partition(bb, world_size):
+    """Assign nets to ranks by bb-center strips along x (region-tree top
+    cuts; reference analogue: fpga_bipartition / build_net_tree
+    partitioning_multi_sink...cxx:3064,3295). Balanced by net count.
+    Returns rank_of_net (int array)."""
+    n = len(bb)
+    cx = (bb[:, 0].astype(np.int64) + bb[:, 2].astype(np.int64))
+    order = np.argsort(cx, kind="stable")
+    rank_of = np.zeros(n, dtype=np.int32)
+    chunks = np.array_split(order, world_size)
+    for r, ch in enumerate(chunks):
+        rank_of[ch] = r
+    return rank_of
+
+
+class CpuEngine:
+    """SerialRouter adapter for DistRouteLoop (gloo CPU testing path)."""
+
+    def __init__(self, router, num_nodes):
+        self.r = router
+        self.num_nodes = num_nodes
+
+    def route_subset(self, crit, pres_fac, net_ids):
+        import numpy as _np
+        self.r.set_pres_fac(pres_fac)
+        # refresh pres array from occ under the new pres_fac
+        self.r.set_occ(_np.asarray(self.r.occ()))
+        self.r.route_subset(np.ascontiguousarray(crit, dtype=np.float32),
+                            np.asarray(net_ids, dtype=np.int32))
+
+    def occ_tensor(self):
+        import torch
+        return torch.from_numpy(np.asarray(self.r.occ()).copy())
+
+    def set_occ(self, t):
+        self.r.set_occ(t.cpu().numpy())
+
+    def sink_delays_local(self, net_ids):
+        return np.asarray(self.r.sink_delays())
+
+    def update_acc(self, acc_fac):
+        # acc only; pres is refreshed at the next route_subset
+        self.r.update_costs(self.r_pres_fac if hasattr(self, "r_pres_fac") else 0.0,
+                            acc_fac)
+
+    def num_overused(self):
+        return int(self.r.count_overused())
+
+
+class GpuEngine:
+    """GpuRouter adapter for DistRouteLoop."""
+
+    def __init__(self, router):
+        self.g = router
+        self._last_sd = np.zeros(router.n_sinks, dtype=np.float32)
+
+    def route_subset(self, crit, pres_fac, net_ids):
+        _, sd = self.g.route_iteration(crit, pres_fac, net_subset=net_ids)
+        self._last_sd = sd
+
+    def occ_tensor(self):
+        return self.g.t_occ
+
+    def set_occ(self, t):
+        self.g.t_occ.copy_(t)
+
+    def sink_delays_local(self, net_ids):
+        return self._last_sd
+
+    def update_acc(self, acc_fac):
+        self.g.update_acc(acc_fac)
+
+    def num_overused(self):
+        import torch
+        return int((self.g.t_occ > self.g.t_cap.to(torch.int32)).sum().item())
+
+
+class DistRouteLoop:
+    """Distributed PathFinder outer loop over an engine.
+
+    The engine abstracts CPU-oracle vs GPU-kernel routing:
+      engine.route_subset(crit, pres_fac, net_ids) -> None
+      engine.occ_tensor() -> torch int32 tensor (device or cpu)
+      engine.set_occ(tensor) -> None
+      engine.sink_delays_local(net_ids) -> np.float32 aligned with sinks
+      engine.update_acc(acc_fac)
+      engine.num_overused() -> int   (from its occ)
+    """
+
+    def __init__(self, engine, num_nets, bb, n_rsinks, sink_ptr,
+                 rank=0, world_size=1):
+        self.engine = engine
+        self.rank = rank
+        self.ws = world_size
+        self.rank_of = spatial_partition(bb, world_size) if world_size > 1 \
+            else np.zeros(num_nets, dtype=np.int32)
+        self.my_nets = np.nonzero(self.rank_of == rank)[0]
+        # sinks owned by my nets
+        mask = np.zeros(n_rsinks, dtype=bool)
+        for n in self.my_nets:
+            mask[sink_ptr[n]:sink_ptr[n + 1]] = True
+        self.my_sink_mask = mask
+
+    def iteration(self, crit, pres_fac, acc_fac):
+        """One distributed PathFinder iteration. Returns
+        (overused_global, sink_delays_global)."""
+        import torch
+        dist = _dist() if self.ws > 1 else None
+        eng = self.engine
+        if self.ws > 1:
+            occ_before = eng.occ_tensor().clone()
+        eng.route_subset(crit, pres_fac, self.my_nets)
+        sd = eng.sink_delays_local(self.my_nets)
+        if self.ws > 1:
+            occ = eng.occ_tensor()
+            delta = occ - occ_before
+            dist.all_reduce(delta, op=dist.ReduceOp.SUM)
+            occ_new = occ_before + delta
+            eng.set_occ(occ_new)
+            # sink delays: mask to my sinks, sum across ranks
+            sd_t = torch.from_numpy(np.where(self.my_sink_mask, sd, 0.0)
+                                    .astype(np.float32))
+            if occ.is_cuda:
+                sd_t = sd_t.to(occ.device)
+            dist.all_reduce(sd_t, op=dist.ReduceOp.SUM)
+            sd = sd_t.cpu().numpy()
+        over = eng.num_overused()
+        eng.update_acc(acc_fac)
+        return over, sd

@@ -80,6 +80,8 @@ class GpuRouter:
         self.bb_margin_per_net = np.full(self.num_nets, bb_margin, dtype=np.int32)
         self.bb = self._compute_bbs()
         self.t_bb = up(self.bb)
+        self._bb_version = 0
+        self._waves_cache = None
 
         # slot classes
         areas = self._bb_areas(self.bb)
@@ -172,10 +174,50 @@ class GpuRouter:
     def _stream(self):
         return self.torch.cuda.current_stream().cuda_stream
 
+    # ---- ParaDRo-style wave schedule (reference:
+    #      partitioning_multi_sink...cxx:3563-4450 overlap graph + coloring).
+    # Nets whose bbs share a coarse grid cell go in different waves; one
+    # kernel launch per wave. Within a wave bbs are disjoint, so concurrent
+    # nets never read or write each other's congestion => the whole
+    # iteration is DETERMINISTIC (fixed schedule, like the reference's
+    # partitioning router).
+    def _schedule_waves(self, net_ids):
+        # cache the full-set schedule; it only changes when bbs grow
+        full = len(net_ids) == self.num_nets
+        if full and getattr(self, "_waves_cache", None) is not None \
+                and self._waves_cache[0] == self._bb_version:
+            return self._waves_cache[1]
+        waves = self._schedule_waves_impl(net_ids)
+        if full:
+            self._waves_cache = (self._bb_version, waves)
+        return waves
+
+    def _schedule_waves_impl(self, net_ids):
+        cell = 8
+        ncx = (self.arch.nx + 2 + cell - 1) // cell
+        ncy = (self.arch.ny + 2 + cell - 1) // cell
+        next_free = np.zeros(ncx * ncy, dtype=np.int32)
+        areas = self._bb_areas(self.bb)
+        order = net_ids[np.argsort(-areas[net_ids], kind="stable")]
+        wave_of = np.zeros(len(order), dtype=np.int32)
+        bb = self.bb
+        for i, n in enumerate(order):
+            cx0 = bb[n, 0] // cell; cy0 = bb[n, 1] // cell
+            cx1 = bb[n, 2] // cell; cy1 = bb[n, 3] // cell
+            blockv = next_free[:].reshape(ncx, ncy)[cx0:cx1 + 1, cy0:cy1 + 1]
+            w = int(blockv.max())
+            blockv[:] = np.maximum(blockv, w + 1)
+            wave_of[i] = w
+        n_waves = int(wave_of.max()) + 1 if len(order) else 0
+        waves = [order[wave_of == w] for w in range(n_waves)]
+        return waves
+
     # ---- one PathFinder iteration ----
-    def route_iteration(self, crit, pres_fac):
+    def route_iteration(self, crit, pres_fac, net_subset=None):
         """crit: per-sink criticality aligned with sink_rr (original order).
-        Returns (overused_count, sink_delays aligned with original order)."""
+        Returns (overused_count, sink_delays aligned with original order).
+        net_subset: optional array of net ids to (re)route; others keep
+        their route trees (multi-GPU partitioning / selective reroute)."""
         torch = self.torch
         # order sinks by decreasing criticality within each net
         net_of_sink = np.repeat(np.arange(self.num_nets), np.diff(self.sink_ptr))
@@ -183,23 +225,23 @@ class GpuRouter:
         t_sink_rr = torch.from_numpy(self.sink_rr[perm]).to(self.device)
         t_crit = torch.from_numpy(np.ascontiguousarray(crit[perm], dtype=np.float32)).to(self.device)
 
-        areas = self._bb_areas(self.bb)
-        small = np.nonzero(areas <= self.bb_max_small_area)[0].astype(np.int32)
-        large = np.nonzero(areas > self.bb_max_small_area)[0].astype(np.int32)
-        # route hardest (most sinks) first within each class for load balance
-        nsinks = np.diff(self.sink_ptr)
-        small = small[np.argsort(-nsinks[small], kind="stable")]
-        large = large[np.argsort(-nsinks[large], kind="stable")]
-
+        todo = (np.arange(self.num_nets, dtype=np.int64) if net_subset is None
+                else np.asarray(net_subset, dtype=np.int64))
         attempts = 0
         while True:
-            q_small = torch.from_numpy(small).to(self.device)
-            q_large = torch.from_numpy(large).to(self.device)
-            self.t_cursors.zero_()
             self.t_fail.zero_()
-            args = self._make_args(t_sink_rr, t_crit, q_small, q_large, pres_fac)
-            rc = self.lib.pnr_route_nets(hip_api.ct.byref(args), self._stream())
-            hip_api.check(rc, "route_nets")
+            waves = self._schedule_waves(todo)
+            areas = self._bb_areas(self.bb)
+            for wave in waves:
+                small = wave[areas[wave] <= self.bb_max_small_area].astype(np.int32)
+                large = wave[areas[wave] > self.bb_max_small_area].astype(np.int32)
+                q_small = torch.from_numpy(small).to(self.device)
+                q_large = torch.from_numpy(large).to(self.device)
+                self.t_cursors.zero_()
+                args = self._make_args(t_sink_rr, t_crit, q_small, q_large, pres_fac)
+                rc = self.lib.pnr_route_nets(hip_api.ct.byref(args), self._stream())
+                hip_api.check(rc, "route_nets")
+                # no host sync between waves: stream order serializes them
             torch.cuda.synchronize(self.device)
             fail = self.t_fail.cpu().numpy()
             failed = np.nonzero(fail)[0]
@@ -216,12 +258,10 @@ class GpuRouter:
                 max(self.arch.nx, self.arch.ny) + 2)
             self.bb = self._compute_bbs()
             self.t_bb.copy_(torch.from_numpy(self.bb).to(self.device))
+            self._bb_version += 1
             # state may be dirty for failed slots; refill (cheap)
             self._fill_state()
-            areas = self._bb_areas(self.bb)
-            fsmall = failed[areas[failed] <= self.bb_max_small_area].astype(np.int32)
-            flarge = failed[areas[failed] > self.bb_max_small_area].astype(np.int32)
-            small, large = fsmall, flarge
+            todo = failed
 
         overused = int((self.t_occ > self.t_cap.to(torch.int32)).sum().item())
         sd = self.t_sink_delay.cpu().numpy()
@@ -311,9 +351,11 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
     """GPU PathFinder outer loop — mirrors route.router.pathfinder_route."""
     net_ids, src_rr, sink_ptr, sink_rr, conn_index = net_rr_terminals(
         netlist, placement, g, arch)
+    from .router import ConnMap
     router = GpuRouter(g, arch, src_rr, sink_ptr.astype(np.int32), sink_rr,
                        device=device, astar_fac=astar_fac)
     n_rsinks = len(sink_rr)
+    cmap = ConnMap(conn_index, sink_ptr, netlist.num_conns, n_rsinks)
     crit = np.zeros(n_rsinks, dtype=np.float32)
     conn_delay = np.zeros(netlist.num_conns, dtype=np.float32)
     pres_fac = 0.0   # first iteration: congestion-blind (VPR style)
@@ -327,13 +369,9 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
         if verbose:
             print(f"[gpu] iter {it}: overused={overused} cpd={cpd*1e9:.2f}ns")
         if sta is not None:
-            for k, (idx, inv) in enumerate(conn_index):
-                conn_delay[idx] = sink_delays[sink_ptr[k]:sink_ptr[k + 1]][inv]
+            cmap.conn_delays(sink_delays, out=conn_delay)
             cpd, slack, c = sta.analyze(conn_delay)
-            for k, (idx, inv) in enumerate(conn_index):
-                seg = np.zeros(sink_ptr[k + 1] - sink_ptr[k], dtype=np.float32)
-                np.maximum.at(seg, inv, c[idx])
-                crit[sink_ptr[k]:sink_ptr[k + 1]] = seg
+            crit = cmap.sink_crit(c)
         if overused == 0:
             break
         pres_fac = pres_fac_init if pres_fac == 0.0 else pres_fac * pres_fac_mult

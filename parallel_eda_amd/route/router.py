@@ -56,6 +56,53 @@ def net_rr_terminals(netlist, placement, g, arch: ArchDef):
             conn_index)
 
 
+class ConnMap:
+    """Vectorized mapping between netlist connections and routed sinks.
+
+    Routed sinks are deduped per (net, sink-tile); each netlist connection
+    (net, sink-block) maps onto one routed sink. Used to scatter routed
+    sink delays to connection delays for STA, and gather connection
+    criticalities back (max per routed sink).
+    """
+
+    def __init__(self, conn_index, sink_ptr, num_conns, n_rsinks):
+        conn_all, rsink_all = [], []
+        for k, (idx, inv) in enumerate(conn_index):
+            conn_all.append(idx)
+            rsink_all.append(sink_ptr[k] + inv)
+        if conn_all:
+            self.conn = np.concatenate(conn_all).astype(np.int64)
+            self.rsink = np.concatenate(rsink_all).astype(np.int64)
+        else:
+            self.conn = np.zeros(0, dtype=np.int64)
+            self.rsink = np.zeros(0, dtype=np.int64)
+        self.num_conns = num_conns
+        self.n_rsinks = n_rsinks
+
+    def conn_delays(self, sink_delays, out=None):
+        if out is None:
+            out = np.zeros(self.num_conns, dtype=np.float32)
+        out[self.conn] = sink_delays[self.rsink]
+        return out
+
+    def sink_crit(self, conn_crit, out=None, max_crit=0.99, crit_exp=1.0):
+        """Max connection criticality per routed sink, clamped to max_crit.
+
+        The clamp (reference: VPR max_criticality, router opts) is
+        essential: at crit == 1.0 the congestion term (1-crit)*cong
+        vanishes and fully-critical nets ignore pres_cost forever.
+        """
+        if out is None:
+            out = np.zeros(self.n_rsinks, dtype=np.float32)
+        else:
+            out[:] = 0.0
+        np.maximum.at(out, self.rsink, conn_crit[self.conn])
+        if crit_exp != 1.0:
+            np.power(out, crit_exp, out=out)
+        np.minimum(out, max_crit, out=out)
+        return out
+
+
 @dataclass
 class RouteResult:
     success: bool
@@ -87,6 +134,7 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
     router = cpu.SerialRouter(g, src_rr, sink_ptr, sink_rr, opts)
 
     n_rsinks = len(sink_rr)
+    cmap = ConnMap(conn_index, sink_ptr, netlist.num_conns, n_rsinks)
     crit = np.zeros(n_rsinks, dtype=np.float32)
     conn_delay = np.zeros(netlist.num_conns, dtype=np.float32)
     pres_fac = pres_fac_init
@@ -104,16 +152,9 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
                   f"cpd={cpd*1e9:.2f}ns")
         if sta is not None:
             # net delays -> connection delays -> STA -> criticality
-            sd = router.sink_delays()
-            for (idx, inv), k in zip(conn_index, range(len(net_ids))):
-                conn_delay[idx] = sd[sink_ptr[k]:sink_ptr[k + 1]][inv]
+            cmap.conn_delays(router.sink_delays(), out=conn_delay)
             cpd, slack, c = sta.analyze(conn_delay)
-            # map connection crits back to routed-sink crits (max over conns
-            # sharing a sink tile)
-            for (idx, inv), k in zip(conn_index, range(len(net_ids))):
-                seg = np.zeros(sink_ptr[k + 1] - sink_ptr[k], dtype=np.float32)
-                np.maximum.at(seg, inv, c[idx])
-                crit[sink_ptr[k]:sink_ptr[k + 1]] = seg
+            crit = cmap.sink_crit(c)
         if overused == 0:
             break
         pres_fac = pres_fac_init if it == 1 else pres_fac * pres_fac_mult
