@@ -238,5 +238,22 @@ PYBIND11_MODULE(_pnr_cpu, m) {
       })
       .def("level_of", [](TimingGraph& t) {
         return py::array_t<int32_t>((py::ssize_t)t.level_of().size(), t.level_of().data());
+      })
+      .def("level_arrays", [](TimingGraph& t) {
+        std::vector<int32_t> blocks, start;
+        t.level_arrays(blocks, start);
+        return py::make_tuple(
+            py::array_t<int32_t>((py::ssize_t)blocks.size(), blocks.data()),
+            py::array_t<int32_t>((py::ssize_t)start.size(), start.data()));
+      })
+      .def("csr_arrays", [](TimingGraph& t) {
+        auto arr64 = [](const std::vector<int64_t>& v) {
+          return py::array_t<int64_t>((py::ssize_t)v.size(), v.data());
+        };
+        return py::make_tuple(
+            arr64(t.in_ptr()), arr64(t.in_conn()),
+            arr64(t.out_ptr()), arr64(t.out_conn()),
+            py::array_t<int32_t>((py::ssize_t)t.conn_driver().size(),
+                                 t.conn_driver().data()));
       });
 }

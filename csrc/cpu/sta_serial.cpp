@@ -76,6 +76,23 @@ class TimingGraph {
   const std::vector<int32_t>& topo() const { return topo_; }
   const std::vector<int32_t>& level_of() const { return level_; }
 
+  // GPU upload accessors: blocks sorted by level + CSRs
+  void level_arrays(std::vector<int32_t>& blocks,
+                    std::vector<int32_t>& start) const {
+    int nb = nl_->num_blocks;
+    start.assign(num_levels_ + 1, 0);
+    for (int b = 0; b < nb; ++b) start[level_[b] + 1]++;
+    for (int l = 0; l < num_levels_; ++l) start[l + 1] += start[l];
+    blocks.assign(nb, -1);
+    std::vector<int32_t> cur(start.begin(), start.end() - 1);
+    for (int b = 0; b < nb; ++b) blocks[cur[level_[b]]++] = b;
+  }
+  const std::vector<int64_t>& in_ptr() const { return in_ptr_; }
+  const std::vector<int64_t>& in_conn() const { return in_conn_; }
+  const std::vector<int64_t>& out_ptr() const { return out_ptr_; }
+  const std::vector<int64_t>& out_conn() const { return out_conn_; }
+  const std::vector<int32_t>& conn_driver() const { return conn_driver_; }
+
  public:
   std::shared_ptr<Netlist> netlist_holder_;  // lifetime pin for Python bindings
   const Netlist* nl_;

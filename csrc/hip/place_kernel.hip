@@ -1,0 +1,354 @@
+// CDNA4 batched simulated-annealing placer.
+//
+// Re-designs the reference's serial try_swap loop (vpr/SRC/place/place.c:1252
+// try_swap, update_bb:2292, get_net_cost:2204, comp_delta_td_cost) as
+// batched parallel moves: each thread proposes one range-limited move/swap,
+// evaluates the EXACT bb+timing delta against the frozen pre-batch state,
+// runs Metropolis, then claims its touched nets + grid locations with
+// atomicMin (move-index priority => deterministic winner set); a second
+// kernel applies the conflict-free winners, whose deltas are exact and
+// additive (disjoint nets/locations => sequential-equivalent batch).
+#include "pnr_hip.h"
+
+namespace pnrh {
+
+struct PlaceDev {
+  // netlist
+  const int32_t* net_blk_ptr;   // [num_nets+1] CSR: net -> member blocks
+  const int32_t* net_blks;      //   (driver first, then sinks; may repeat)
+  const int32_t* blk_net_ptr;   // [num_blocks+1] CSR: block -> nets (deduped)
+  const int32_t* blk_nets;
+  const int8_t* blk_type;       // 0=IO 1=CLB
+  const float* net_q;           // crossing factor per net
+  // timing
+  const int32_t* net_sink_ptr;  // [num_nets+1] conn ranges
+  const float* conn_crit;       // per conn
+  const float* delay_mat;       // [gx*gy] |dx|*gy+|dy|
+  // placement state
+  int32_t* bx;
+  int32_t* by;
+  int32_t* bslot;
+  int32_t* grid;                // [(x*gy+y)*cap + slot] -> block or -1
+  float* net_cost;              // current bb cost per net
+  float* net_tcost;             // current timing cost per net
+  int32_t num_blocks, num_nets, gx, gy, cap, nx, ny, io_cap;
+};
+
+struct MovesDev {
+  int32_t* mv_blk;      // proposer's block
+  int32_t* mv_to;       // encoded dest (x*gy+y)*cap+slot
+  int32_t* mv_other;    // block swapped with (or -1)
+  float* mv_dbb;        // exact bb delta
+  float* mv_dtd;        // exact td delta
+  uint8_t* mv_flags;    // 1 = Metropolis-accepted candidate; 2 = winner
+  int32_t* net_claim;   // [num_nets] atomicMin move index
+  int32_t* loc_claim;   // [gx*gy] atomicMin move index (tile granularity)
+  int32_t* counters;    // [4]: attempts, accepted, winners, conflicts
+  int32_t n_moves;
+};
+
+__device__ __forceinline__ uint32_t rng_hash(uint32_t a, uint32_t b, uint32_t c) {
+  uint32_t h = a * 0x9E3779B9u ^ b * 0x85EBCA6Bu ^ c * 0xC2B2AE35u;
+  h ^= h >> 16; h *= 0x7FEB352Du; h ^= h >> 15; h *= 0x846CA68Bu; h ^= h >> 16;
+  return h;
+}
+
+__device__ __forceinline__ float cross_count_dev(int n) {
+  const float q3 = 1.0f, q50 = 2.79f;
+  if (n <= 3) return q3;
+  if (n >= 50) return q50 + 0.02616f * (n - 50);
+  return q3 + (q50 - q3) * (n - 3) / 47.0f;
+}
+
+__device__ __forceinline__ bool is_io_loc(const PlaceDev& p, int x, int y) {
+  return x == 0 || x == p.gx - 1 || y == 0 || y == p.gy - 1;
+}
+__device__ __forceinline__ int cap_at(const PlaceDev& p, int x, int y) {
+  bool io = is_io_loc(p, x, y);
+  if (!io && x >= 1 && x <= p.nx && y >= 1 && y <= p.ny) return 1;
+  if (io && ((x >= 1 && x <= p.nx) != (y >= 1 && y <= p.ny))) return p.io_cap;
+  return 0;
+}
+
+// bb cost of net n with up to 2 positional overrides
+__device__ float net_bb_cost(const PlaceDev& p, int n, int32_t b1, int x1,
+                             int y1, int32_t b2, int x2, int y2) {
+  int xmin = 1 << 28, xmax = -1, ymin = 1 << 28, ymax = -1;
+  int32_t e0 = p.net_blk_ptr[n], e1 = p.net_blk_ptr[n + 1];
+  for (int32_t e = e0; e < e1; ++e) {
+    int32_t b = p.net_blks[e];
+    int x = p.bx[b], y = p.by[b];
+    if (b == b1) { x = x1; y = y1; }
+    else if (b == b2) { x = x2; y = y2; }
+    xmin = min(xmin, x); xmax = max(xmax, x);
+    ymin = min(ymin, y); ymax = max(ymax, y);
+  }
+  return p.net_q[n] * ((xmax - xmin + 1) + (ymax - ymin + 1));
+}
+
+// timing cost of net n (sum over conns of crit * delay) with overrides
+__device__ float net_td_cost(const PlaceDev& p, int n, int32_t b1, int x1,
+                             int y1, int32_t b2, int x2, int y2) {
+  if (p.delay_mat == nullptr) return 0.0f;
+  int32_t e0 = p.net_blk_ptr[n], e1 = p.net_blk_ptr[n + 1];
+  int32_t drv = p.net_blks[e0];
+  int dx0 = p.bx[drv], dy0 = p.by[drv];
+  if (drv == b1) { dx0 = x1; dy0 = y1; }
+  else if (drv == b2) { dx0 = x2; dy0 = y2; }
+  float t = 0.0f;
+  int32_t c0 = p.net_sink_ptr[n];
+  for (int32_t e = e0 + 1; e < e1; ++e) {
+    int32_t b = p.net_blks[e];
+    int x = p.bx[b], y = p.by[b];
+    if (b == b1) { x = x1; y = y1; }
+    else if (b == b2) { x = x2; y = y2; }
+    int dx = abs(x - dx0), dy = abs(y - dy0);
+    t += p.conn_crit[c0 + (e - e0 - 1)] * p.delay_mat[dx * p.gy + dy];
+  }
+  return t;
+}
+
+#define MAX_MOVE_NETS 160
+
+__global__ void place_propose_kernel(PlaceDev p, MovesDev m, float T,
+                                     int rlim, float timing_tradeoff,
+                                     float inv_bb_norm, float inv_td_norm,
+                                     uint32_t seed, uint32_t batch) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= m.n_moves) return;
+  m.mv_flags[i] = 0;
+  // pick block + destination
+  uint32_t r0 = rng_hash(seed, batch, i * 4 + 0);
+  int32_t blk = r0 % p.num_blocks;
+  bool io = p.blk_type[blk] == 0;
+  int x0 = p.bx[blk], y0 = p.by[blk];
+  int x1 = -1, y1 = -1, slot1 = 0;
+  for (int att = 0; att < 8; ++att) {
+    uint32_t r1 = rng_hash(seed, batch, i * 131 + 7 * att + 1);
+    uint32_t r2 = rng_hash(seed, batch, i * 131 + 7 * att + 2);
+    int tx = x0 + (int)(r1 % (2 * rlim + 1)) - rlim;
+    int ty = y0 + (int)(r2 % (2 * rlim + 1)) - rlim;
+    if (tx < 0 || tx >= p.gx || ty < 0 || ty >= p.gy) continue;
+    if (is_io_loc(p, tx, ty) != io) continue;
+    int c = cap_at(p, tx, ty);
+    if (c <= 0) continue;
+    if (tx == x0 && ty == y0) continue;
+    x1 = tx; y1 = ty;
+    slot1 = (int)(rng_hash(seed, batch, i * 131 + 7 * att + 3) % c);
+    break;
+  }
+  atomicAdd(&m.counters[0], 1);
+  if (x1 < 0) return;
+  int32_t other = p.grid[((int64_t)x1 * p.gy + y1) * p.cap + slot1];
+  if (other == blk) return;
+
+  // collect affected nets (dedup)
+  int32_t nets[MAX_MOVE_NETS];
+  int nn = 0;
+  for (int pass = 0; pass < 2; ++pass) {
+    int32_t b = pass == 0 ? blk : other;
+    if (b < 0) continue;
+    for (int32_t k = p.blk_net_ptr[b]; k < p.blk_net_ptr[b + 1]; ++k) {
+      int32_t n = p.blk_nets[k];
+      bool dup = false;
+      for (int j = 0; j < nn; ++j) if (nets[j] == n) { dup = true; break; }
+      if (!dup) {
+        if (nn >= MAX_MOVE_NETS) return;  // very-high-fanin block: skip move
+        nets[nn++] = n;
+      }
+    }
+  }
+  // exact deltas with overrides: blk -> (x1,y1); other -> (x0,y0)
+  float dbb = 0.0f, dtd = 0.0f;
+  int ox = other >= 0 ? x0 : -1000, oy = other >= 0 ? y0 : -1000;
+  for (int j = 0; j < nn; ++j) {
+    int32_t n = nets[j];
+    dbb += net_bb_cost(p, n, blk, x1, y1, other, ox, oy) - p.net_cost[n];
+    if (timing_tradeoff > 0.0f)
+      dtd += net_td_cost(p, n, blk, x1, y1, other, ox, oy) - p.net_tcost[n];
+  }
+  float delta = (1.0f - timing_tradeoff) * dbb * inv_bb_norm +
+                timing_tradeoff * dtd * inv_td_norm;
+  bool accept;
+  if (delta <= 0.0f) accept = true;
+  else if (T <= 0.0f) accept = false;
+  else {
+    float u = (rng_hash(seed, batch, i * 131 + 5) & 0xFFFFFF) * (1.0f / 16777216.0f);
+    accept = u < __expf(-delta / T);
+  }
+  if (!accept) return;
+  atomicAdd(&m.counters[1], 1);
+  m.mv_blk[i] = blk;
+  m.mv_to[i] = ((x1 * p.gy + y1) * p.cap + slot1);
+  m.mv_other[i] = other;
+  m.mv_dbb[i] = dbb;
+  m.mv_dtd[i] = dtd;
+  m.mv_flags[i] = 1;
+  // claim nets + the two tile locations (tile granularity serializes
+  // slot-level races within a tile)
+  for (int j = 0; j < nn; ++j) atomicMin(&m.net_claim[nets[j]], i);
+  atomicMin(&m.loc_claim[x0 * p.gy + y0], i);
+  atomicMin(&m.loc_claim[x1 * p.gy + y1], i);
+}
+
+__global__ void place_resolve_kernel(PlaceDev p, MovesDev m) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= m.n_moves || m.mv_flags[i] != 1) return;
+  int32_t blk = m.mv_blk[i];
+  int32_t other = m.mv_other[i];
+  int32_t to = m.mv_to[i];
+  int x1 = to / p.cap / p.gy, y1 = (to / p.cap) % p.gy;
+  int x0 = p.bx[blk], y0 = p.by[blk];
+  bool win = m.loc_claim[x0 * p.gy + y0] == i &&
+             m.loc_claim[x1 * p.gy + y1] == i;
+  if (win) {
+    for (int pass = 0; pass < 2 && win; ++pass) {
+      int32_t b = pass == 0 ? blk : other;
+      if (b < 0) continue;
+      for (int32_t k = p.blk_net_ptr[b]; k < p.blk_net_ptr[b + 1]; ++k)
+        if (m.net_claim[p.blk_nets[k]] != i) { win = false; break; }
+    }
+  }
+  if (!win) { m.mv_flags[i] = 0; atomicAdd(&m.counters[3], 1); return; }
+  m.mv_flags[i] = 2;
+  atomicAdd(&m.counters[2], 1);
+}
+
+__global__ void place_apply_kernel(PlaceDev p, MovesDev m, float timing_tradeoff,
+                                   double* cost_acc /*[2]: dbb, dtd*/) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= m.n_moves || m.mv_flags[i] != 2) return;
+  int32_t blk = m.mv_blk[i];
+  int32_t other = m.mv_other[i];
+  int32_t to = m.mv_to[i];
+  int slot1 = to % p.cap;
+  int x1 = to / p.cap / p.gy, y1 = (to / p.cap) % p.gy;
+  int x0 = p.bx[blk], y0 = p.by[blk], s0 = p.bslot[blk];
+  // grid + placement updates (winner owns both tiles exclusively)
+  p.grid[((int64_t)x0 * p.gy + y0) * p.cap + s0] = other >= 0 ? other : -1;
+  p.grid[((int64_t)x1 * p.gy + y1) * p.cap + slot1] = blk;
+  p.bx[blk] = x1; p.by[blk] = y1; p.bslot[blk] = slot1;
+  if (other >= 0) { p.bx[other] = x0; p.by[other] = y0; p.bslot[other] = s0; }
+  // refresh cached per-net costs for affected nets (exclusive ownership)
+  for (int pass = 0; pass < 2; ++pass) {
+    int32_t b = pass == 0 ? blk : other;
+    if (b < 0) continue;
+    for (int32_t k = p.blk_net_ptr[b]; k < p.blk_net_ptr[b + 1]; ++k) {
+      int32_t n = p.blk_nets[k];
+      if (pass == 1) {
+        // skip nets already refreshed via blk
+        bool shared = false;
+        for (int32_t k2 = p.blk_net_ptr[blk]; k2 < p.blk_net_ptr[blk + 1]; ++k2)
+          if (p.blk_nets[k2] == n) { shared = true; break; }
+        if (shared) continue;
+      }
+      p.net_cost[n] = net_bb_cost(p, n, -1, 0, 0, -1, 0, 0);
+      if (timing_tradeoff > 0.0f)
+        p.net_tcost[n] = net_td_cost(p, n, -1, 0, 0, -1, 0, 0);
+    }
+  }
+  unsafeAtomicAdd(&cost_acc[0], (double)m.mv_dbb[i]);
+  unsafeAtomicAdd(&cost_acc[1], (double)m.mv_dtd[i]);
+}
+
+__global__ void place_reset_claims_kernel(int32_t* net_claim, int32_t nn,
+                                          int32_t* loc_claim, int32_t nl,
+                                          int32_t* counters) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  int total = max(nn, nl);
+  for (; i < total; i += gridDim.x * blockDim.x) {
+    if (i < nn) net_claim[i] = 0x7FFFFFFF;
+    if (i < nl) loc_claim[i] = 0x7FFFFFFF;
+    if (i < 4) counters[i] = 0;
+  }
+}
+
+// full net-cost refresh (init + periodic drift resync).
+// Host must zero out[0..1] before the launch.
+__global__ void place_refresh_costs_kernel(PlaceDev p, float timing_tradeoff,
+                                           double* out /*[2]*/) {
+  int n = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; n < p.num_nets; n += gridDim.x * blockDim.x) {
+    float c = net_bb_cost(p, n, -1, 0, 0, -1, 0, 0);
+    p.net_cost[n] = c;
+    unsafeAtomicAdd(&out[0], (double)c);
+    float t = (timing_tradeoff > 0.0f)
+                  ? net_td_cost(p, n, -1, 0, 0, -1, 0, 0) : 0.0f;
+    p.net_tcost[n] = t;
+    unsafeAtomicAdd(&out[1], (double)t);
+  }
+}
+
+}  // namespace pnrh
+
+using namespace pnrh;
+
+extern "C" {
+
+struct PlaceLaunchArgs {
+  const int32_t* net_blk_ptr; const int32_t* net_blks;
+  const int32_t* blk_net_ptr; const int32_t* blk_nets;
+  const int8_t* blk_type; const float* net_q;
+  const int32_t* net_sink_ptr; const float* conn_crit; const float* delay_mat;
+  int32_t* bx; int32_t* by; int32_t* bslot; int32_t* grid;
+  float* net_cost; float* net_tcost;
+  int32_t num_blocks, num_nets, gx, gy, cap, nx, ny, io_cap;
+  // moves
+  int32_t* mv_blk; int32_t* mv_to; int32_t* mv_other;
+  float* mv_dbb; float* mv_dtd; uint8_t* mv_flags;
+  int32_t* net_claim; int32_t* loc_claim; int32_t* counters;
+  int32_t n_moves;
+  // schedule
+  float T; int32_t rlim; float timing_tradeoff;
+  float inv_bb_norm, inv_td_norm;
+  uint32_t seed, batch;
+  double* cost_acc;
+};
+
+static void unpack(const PlaceLaunchArgs* a, PlaceDev& p, MovesDev& m) {
+  p.net_blk_ptr = a->net_blk_ptr; p.net_blks = a->net_blks;
+  p.blk_net_ptr = a->blk_net_ptr; p.blk_nets = a->blk_nets;
+  p.blk_type = a->blk_type; p.net_q = a->net_q;
+  p.net_sink_ptr = a->net_sink_ptr; p.conn_crit = a->conn_crit;
+  p.delay_mat = a->delay_mat;
+  p.bx = a->bx; p.by = a->by; p.bslot = a->bslot; p.grid = a->grid;
+  p.net_cost = a->net_cost; p.net_tcost = a->net_tcost;
+  p.num_blocks = a->num_blocks; p.num_nets = a->num_nets;
+  p.gx = a->gx; p.gy = a->gy; p.cap = a->cap;
+  p.nx = a->nx; p.ny = a->ny; p.io_cap = a->io_cap;
+  m.mv_blk = a->mv_blk; m.mv_to = a->mv_to; m.mv_other = a->mv_other;
+  m.mv_dbb = a->mv_dbb; m.mv_dtd = a->mv_dtd; m.mv_flags = a->mv_flags;
+  m.net_claim = a->net_claim; m.loc_claim = a->loc_claim;
+  m.counters = a->counters; m.n_moves = a->n_moves;
+}
+
+int pnr_place_batch(const PlaceLaunchArgs* a, void* stream) {
+  PlaceDev p; MovesDev m;
+  unpack(a, p, m);
+  hipStream_t s = (hipStream_t)stream;
+  int rg = (max(max(a->num_nets, a->gx * a->gy), a->n_moves) + 255) / 256;
+  rg = rg < 2048 ? rg : 2048;
+  hipLaunchKernelGGL(place_reset_claims_kernel, dim3(rg), dim3(256), 0, s,
+                     m.net_claim, p.num_nets, m.loc_claim, p.gx * p.gy,
+                     m.counters);
+  int mg = (m.n_moves + 255) / 256;
+  hipLaunchKernelGGL(place_propose_kernel, dim3(mg), dim3(256), 0, s,
+                     p, m, a->T, a->rlim, a->timing_tradeoff,
+                     a->inv_bb_norm, a->inv_td_norm, a->seed, a->batch);
+  hipLaunchKernelGGL(place_resolve_kernel, dim3(mg), dim3(256), 0, s, p, m);
+  hipLaunchKernelGGL(place_apply_kernel, dim3(mg), dim3(256), 0, s,
+                     p, m, a->timing_tradeoff, a->cost_acc);
+  return (int)hipGetLastError();
+}
+
+int pnr_place_refresh(const PlaceLaunchArgs* a, void* stream) {
+  PlaceDev p; MovesDev m;
+  unpack(a, p, m);
+  hipStream_t s = (hipStream_t)stream;
+  // single-block prologue zeroes out; then grid-stride accumulate
+  hipLaunchKernelGGL(place_refresh_costs_kernel, dim3(1024), dim3(256), 0, s,
+                     p, a->timing_tradeoff, a->cost_acc);
+  return (int)hipGetLastError();
+}
+
+}  // extern "C"

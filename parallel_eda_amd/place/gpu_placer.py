@@ -1,0 +1,337 @@
+"""GPU SA placer driver (host side of csrc/hip/place_kernel.hip).
+
+Shares the adaptive outer schedule with the CPU oracle (placer.py) but runs
+move batches as propose/resolve/apply kernel triples. Batched moves with
+exclusive net+tile claims are sequential-equivalent (disjoint winners,
+exact deltas), so the anneal semantics match the reference's try_swap loop
+at a per-batch granularity.
+"""
+import ctypes as ct
+
+import numpy as np
+
+from ..arch.archdef import ArchDef
+from ..ops import hip_api
+from .placer import Placement, analytic_delay_matrix
+
+
+class PlaceLaunchArgs(ct.Structure):
+    _fields_ = [
+        ("net_blk_ptr", ct.c_void_p), ("net_blks", ct.c_void_p),
+        ("blk_net_ptr", ct.c_void_p), ("blk_nets", ct.c_void_p),
+        ("blk_type", ct.c_void_p), ("net_q", ct.c_void_p),
+        ("net_sink_ptr", ct.c_void_p), ("conn_crit", ct.c_void_p),
+        ("delay_mat", ct.c_void_p),
+        ("bx", ct.c_void_p), ("by", ct.c_void_p), ("bslot", ct.c_void_p),
+        ("grid", ct.c_void_p),
+        ("net_cost", ct.c_void_p), ("net_tcost", ct.c_void_p),
+        ("num_blocks", ct.c_int32), ("num_nets", ct.c_int32),
+        ("gx", ct.c_int32), ("gy", ct.c_int32), ("cap", ct.c_int32),
+        ("nx", ct.c_int32), ("ny", ct.c_int32), ("io_cap", ct.c_int32),
+        ("mv_blk", ct.c_void_p), ("mv_to", ct.c_void_p), ("mv_other", ct.c_void_p),
+        ("mv_dbb", ct.c_void_p), ("mv_dtd", ct.c_void_p), ("mv_flags", ct.c_void_p),
+        ("net_claim", ct.c_void_p), ("loc_claim", ct.c_void_p),
+        ("counters", ct.c_void_p), ("n_moves", ct.c_int32),
+        ("T", ct.c_float), ("rlim", ct.c_int32), ("timing_tradeoff", ct.c_float),
+        ("inv_bb_norm", ct.c_float), ("inv_td_norm", ct.c_float),
+        ("seed", ct.c_uint32), ("batch", ct.c_uint32),
+        ("cost_acc", ct.c_void_p),
+    ]
+
+
+def _lib():
+    lib = hip_api.lib()
+    if not hasattr(lib, "_place_ready"):
+        lib.pnr_place_batch.restype = ct.c_int
+        lib.pnr_place_batch.argtypes = [ct.POINTER(PlaceLaunchArgs), ct.c_void_p]
+        lib.pnr_place_refresh.restype = ct.c_int
+        lib.pnr_place_refresh.argtypes = [ct.POINTER(PlaceLaunchArgs), ct.c_void_p]
+        lib._place_ready = True
+    return lib
+
+
+def ptr(t):
+    return ct.c_void_p(t.data_ptr())
+
+
+class GpuPlacer:
+    def __init__(self, netlist, arch: ArchDef, seed=7, timing=False,
+                 device="cuda:0", n_moves=None):
+        import torch
+        self.torch = torch
+        self.device = device
+        self.arch = arch
+        self.nl = netlist
+        self.seed = seed
+        self.timing = timing
+        nb = netlist.num_blocks
+        nn = netlist.num_nets
+        gx, gy = arch.nx + 2, arch.ny + 2
+        self.gx, self.gy = gx, gy
+        self.cap = max(1, arch.io_cap)
+        self.batch_counter = 0
+
+        # net -> member blocks CSR (driver first)
+        counts = 1 + np.diff(netlist.net_sink_ptr)
+        net_blk_ptr = np.r_[0, np.cumsum(counts)].astype(np.int32)
+        net_blks = np.empty(net_blk_ptr[-1], dtype=np.int32)
+        net_blks[net_blk_ptr[:-1]] = netlist.net_driver
+        sidx = np.concatenate([
+            np.arange(net_blk_ptr[i] + 1, net_blk_ptr[i + 1])
+            for i in range(nn)]) if nn else np.zeros(0, dtype=np.int64)
+        net_blks[sidx] = netlist.net_sinks
+        # block -> nets CSR (deduped per block)
+        blk_lists = [[] for _ in range(nb)]
+        for n in range(nn):
+            for b in set(net_blks[net_blk_ptr[n]:net_blk_ptr[n + 1]].tolist()):
+                blk_lists[b].append(n)
+        blk_net_ptr = np.zeros(nb + 1, dtype=np.int32)
+        for b in range(nb):
+            blk_net_ptr[b + 1] = blk_net_ptr[b] + len(blk_lists[b])
+        blk_nets = np.concatenate([np.asarray(l, dtype=np.int32)
+                                   for l in blk_lists]) if nb else np.zeros(0, np.int32)
+        q = np.array([_cross_count(int(c)) for c in counts], dtype=np.float32)
+
+        def up(a, dtype=None):
+            t = torch.from_numpy(np.ascontiguousarray(a))
+            return t.to(device)
+
+        self.t_net_blk_ptr = up(net_blk_ptr)
+        self.t_net_blks = up(net_blks)
+        self.t_blk_net_ptr = up(blk_net_ptr)
+        self.t_blk_nets = up(blk_nets)
+        self.t_blk_type = up(netlist.block_type)
+        self.t_net_q = up(q)
+        self.t_net_sink_ptr = up(netlist.net_sink_ptr.astype(np.int32))
+        self.t_conn_crit = torch.zeros(netlist.num_conns, dtype=torch.float32,
+                                       device=device)
+        if timing:
+            dm = analytic_delay_matrix(arch)
+            self.t_delay_mat = up(dm.ravel())
+        else:
+            self.t_delay_mat = None
+
+        # initial placement (host, deterministic)
+        bx, by, bslot, grid = self._initial_placement(seed)
+        self.t_bx = up(bx); self.t_by = up(by); self.t_bslot = up(bslot)
+        self.t_grid = up(grid)
+        self.t_net_cost = torch.zeros(nn, dtype=torch.float32, device=device)
+        self.t_net_tcost = torch.zeros(nn, dtype=torch.float32, device=device)
+
+        self.n_moves = n_moves or int(np.clip(nb // 4, 256, 1 << 16))
+        nm = self.n_moves
+        self.t_mv_blk = torch.zeros(nm, dtype=torch.int32, device=device)
+        self.t_mv_to = torch.zeros(nm, dtype=torch.int32, device=device)
+        self.t_mv_other = torch.zeros(nm, dtype=torch.int32, device=device)
+        self.t_mv_dbb = torch.zeros(nm, dtype=torch.float32, device=device)
+        self.t_mv_dtd = torch.zeros(nm, dtype=torch.float32, device=device)
+        self.t_mv_flags = torch.zeros(nm, dtype=torch.uint8, device=device)
+        self.t_net_claim = torch.zeros(nn, dtype=torch.int32, device=device)
+        self.t_loc_claim = torch.zeros(gx * gy, dtype=torch.int32, device=device)
+        self.t_counters = torch.zeros(4, dtype=torch.int32, device=device)
+        self.t_cost_acc = torch.zeros(2, dtype=torch.float64, device=device)
+
+        self.lib = _lib()
+        self.bb_cost, self.td_cost = self.refresh_costs()
+
+    def _initial_placement(self, seed):
+        arch = self.arch
+        nl = self.nl
+        rng = np.random.default_rng(seed)
+        nb = nl.num_blocks
+        bx = np.zeros(nb, dtype=np.int32)
+        by = np.zeros(nb, dtype=np.int32)
+        bslot = np.zeros(nb, dtype=np.int32)
+        grid = np.full((self.gx, self.gy, self.cap), -1, dtype=np.int32)
+        clbs = np.nonzero(nl.block_type == 1)[0]
+        ios = np.nonzero(nl.block_type == 0)[0]
+        tiles = rng.permutation(arch.nx * arch.ny)[:len(clbs)]
+        if len(tiles) < len(clbs):
+            raise ValueError("too many CLBs for grid")
+        bx[clbs] = tiles // arch.ny + 1
+        by[clbs] = tiles % arch.ny + 1
+        grid[bx[clbs], by[clbs], 0] = clbs
+        io_locs = ([(0, y) for y in range(1, arch.ny + 1)] +
+                   [(self.gx - 1, y) for y in range(1, arch.ny + 1)] +
+                   [(x, 0) for x in range(1, arch.nx + 1)] +
+                   [(x, self.gy - 1) for x in range(1, arch.nx + 1)])
+        slots = [(x, y, s) for (x, y) in io_locs for s in range(arch.io_cap)]
+        sel = rng.permutation(len(slots))[:len(ios)]
+        for b, k in zip(ios, sel):
+            x, y, s = slots[k]
+            bx[b], by[b], bslot[b] = x, y, s
+            grid[x, y, s] = b
+        return bx, by, bslot, grid.reshape(-1)
+
+    def _args(self, T=0.0, rlim=1, tt=0.0, inv_bb=1.0, inv_td=1.0):
+        a = PlaceLaunchArgs()
+        a.net_blk_ptr = ptr(self.t_net_blk_ptr); a.net_blks = ptr(self.t_net_blks)
+        a.blk_net_ptr = ptr(self.t_blk_net_ptr); a.blk_nets = ptr(self.t_blk_nets)
+        a.blk_type = ptr(self.t_blk_type); a.net_q = ptr(self.t_net_q)
+        a.net_sink_ptr = ptr(self.t_net_sink_ptr)
+        a.conn_crit = ptr(self.t_conn_crit)
+        a.delay_mat = ptr(self.t_delay_mat) if self.t_delay_mat is not None else None
+        a.bx = ptr(self.t_bx); a.by = ptr(self.t_by); a.bslot = ptr(self.t_bslot)
+        a.grid = ptr(self.t_grid)
+        a.net_cost = ptr(self.t_net_cost); a.net_tcost = ptr(self.t_net_tcost)
+        a.num_blocks = self.nl.num_blocks; a.num_nets = self.nl.num_nets
+        a.gx = self.gx; a.gy = self.gy; a.cap = self.cap
+        a.nx = self.arch.nx; a.ny = self.arch.ny; a.io_cap = self.arch.io_cap
+        a.mv_blk = ptr(self.t_mv_blk); a.mv_to = ptr(self.t_mv_to)
+        a.mv_other = ptr(self.t_mv_other); a.mv_dbb = ptr(self.t_mv_dbb)
+        a.mv_dtd = ptr(self.t_mv_dtd); a.mv_flags = ptr(self.t_mv_flags)
+        a.net_claim = ptr(self.t_net_claim); a.loc_claim = ptr(self.t_loc_claim)
+        a.counters = ptr(self.t_counters); a.n_moves = self.n_moves
+        a.T = T; a.rlim = max(1, int(rlim)); a.timing_tradeoff = tt
+        a.inv_bb_norm = inv_bb; a.inv_td_norm = inv_td
+        a.seed = self.seed & 0xFFFFFFFF; a.batch = self.batch_counter
+        a.cost_acc = ptr(self.t_cost_acc)
+        return a
+
+    def _stream(self):
+        return self.torch.cuda.current_stream().cuda_stream
+
+    def refresh_costs(self):
+        self.t_cost_acc.zero_()
+        a = self._args(tt=1.0 if self.timing else 0.0)
+        rc = self.lib.pnr_place_refresh(ct.byref(a), self._stream())
+        hip_api.check(rc, "place_refresh")
+        self.torch.cuda.synchronize(self.device)
+        bb, td = self.t_cost_acc.cpu().numpy()
+        self.bb_cost, self.td_cost = float(bb), float(td)
+        return self.bb_cost, self.td_cost
+
+    def run_batches(self, T, rlim, n_batches, tt, bb_norm, td_norm):
+        """Run n_batches move batches; returns (success_rate, attempts)."""
+        att = acc = win = 0
+        for _ in range(n_batches):
+            self.batch_counter += 1
+            a = self._args(T=T, rlim=rlim, tt=tt,
+                           inv_bb=1.0 / bb_norm, inv_td=1.0 / td_norm)
+            rc = self.lib.pnr_place_batch(ct.byref(a), self._stream())
+            hip_api.check(rc, "place_batch")
+        self.torch.cuda.synchronize(self.device)
+        # counters reflect only the LAST batch (reset per batch): read them
+        # for rate estimation
+        c = self.t_counters.cpu().numpy()
+        att, acc, win, conf = int(c[0]), int(c[1]), int(c[2]), int(c[3])
+        rej = acc - win - conf  # accepted-but-lost are conflicts, not rejects
+        srate = win / max(1, win + (att - acc))
+        return srate, att
+
+    def set_crit(self, conn_crit):
+        t = self.torch.from_numpy(np.ascontiguousarray(conn_crit,
+                                                       dtype=np.float32))
+        self.t_conn_crit.copy_(t.to(self.device))
+
+    def placement(self):
+        return Placement(self.t_bx.cpu().numpy(), self.t_by.cpu().numpy(),
+                         self.t_bslot.cpu().numpy(),
+                         bb_cost=self.bb_cost, td_cost=self.td_cost)
+
+    def check_place(self):
+        """Host-side validation of GPU placement state (check_place:2950)."""
+        bx = self.t_bx.cpu().numpy(); by = self.t_by.cpu().numpy()
+        bslot = self.t_bslot.cpu().numpy()
+        grid = self.t_grid.cpu().numpy().reshape(self.gx, self.gy, self.cap)
+        nl = self.nl
+        for b in range(nl.num_blocks):
+            io = nl.block_type[b] == 0
+            x, y = bx[b], by[b]
+            on_io = (x == 0 or x == self.gx - 1 or y == 0 or y == self.gy - 1)
+            if io != on_io:
+                return False, f"block {b} type/loc mismatch"
+            if grid[x, y, bslot[b]] != b:
+                return False, f"grid inconsistent at block {b}"
+        occ = (grid >= 0).sum()
+        if occ != nl.num_blocks:
+            return False, f"grid count {occ} != blocks {nl.num_blocks}"
+        return True, ""
+
+
+def _cross_count(n):
+    q3, q50 = 1.0, 2.79
+    if n <= 3:
+        return q3
+    if n >= 50:
+        return q50 + 0.02616 * (n - 50)
+    return q3 + (q50 - q3) * (n - 3) / 47.0
+
+
+def anneal_place_gpu(netlist, arch, seed=7, timing_tradeoff=0.5, inner_num=1.0,
+                     sta=None, crit_exp=1.0, verbose=False, device="cuda:0"):
+    """GPU anneal with the same adaptive schedule as the CPU oracle."""
+    timing = sta is not None and timing_tradeoff > 0
+    placer = GpuPlacer(netlist, arch, seed=seed, timing=timing, device=device)
+    nb = netlist.num_blocks
+    move_lim = max(256, int(inner_num * (nb ** 1.3333)))
+    n_batches = max(1, move_lim // placer.n_moves)
+    rlim = float(max(arch.nx, arch.ny))
+    tt = timing_tradeoff if timing else 0.0
+
+    def refresh_crit():
+        if not timing:
+            return
+        # conn delays from current placement via the delay matrix
+        bx = placer.t_bx.cpu().numpy(); by = placer.t_by.cpu().numpy()
+        drv = netlist.net_driver
+        d_per_conn = np.empty(netlist.num_conns, dtype=np.float32)
+        dm = analytic_delay_matrix(arch)
+        net_of_conn = np.repeat(np.arange(netlist.num_nets),
+                                np.diff(netlist.net_sink_ptr))
+        dx = np.abs(bx[netlist.net_sinks] - bx[drv[net_of_conn]])
+        dy = np.abs(by[netlist.net_sinks] - by[drv[net_of_conn]])
+        d_per_conn = dm[dx, dy]
+        cpd, slack, c = sta.analyze(d_per_conn)
+        placer.set_crit(np.asarray(c) ** crit_exp)
+        placer.refresh_costs()
+
+    refresh_crit()
+    norm_mode = tt > 0
+
+    def norms():
+        if norm_mode:
+            return (max(placer.bb_cost, 1e-12), max(placer.td_cost, 1e-30))
+        return (1.0, 1.0)
+
+    bb_norm, td_norm = norms()
+    # starting T: probe with a hot batch, estimate from delta scale
+    placer.run_batches(1e30, rlim, 2, tt, bb_norm, td_norm)
+    placer.refresh_costs()
+    t = 20.0 * placer.bb_cost / max(1, netlist.num_nets) / bb_norm
+    history = []
+    itemp = 0
+    while True:
+        refresh_crit()
+        bb_norm, td_norm = norms()
+        srate, _ = placer.run_batches(t, rlim, n_batches, tt, bb_norm, td_norm)
+        placer.refresh_costs()  # exact resync every temperature
+        cost = placer.bb_cost
+        history.append((t, cost, srate, rlim))
+        if verbose:
+            print(f"[gpu] T={t:.3e} bb={cost:.1f} acc={srate:.2f} rlim={rlim:.0f}")
+        if srate > 0.96:
+            t *= 0.5
+        elif srate > 0.8:
+            t *= 0.9
+        elif srate > 0.15 and rlim > 1:
+            t *= 0.95
+        else:
+            t *= 0.8
+        rlim = min(max(rlim * (1.0 - 0.44 + srate), 1.0),
+                   float(max(arch.nx, arch.ny)))
+        itemp += 1
+        exit_cost = 1.0 if norm_mode else cost
+        if t < 0.005 * exit_cost / max(1, netlist.num_nets):
+            break
+        if itemp > 500:
+            break
+    placer.run_batches(0.0, 1.0, n_batches, tt, bb_norm, td_norm)
+    placer.refresh_costs()
+    ok, err = placer.check_place()
+    if not ok:
+        raise RuntimeError(f"gpu check_place failed: {err}")
+    pl = placer.placement()
+    pl.stats = {"temps": itemp, "move_lim": move_lim, "history": history,
+                "engine": "gpu"}
+    return pl
