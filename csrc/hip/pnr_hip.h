@@ -35,6 +35,8 @@ struct NetsDev {
   const int32_t* sink_ptr;   // [num_nets+1]
   const int32_t* sink_rr;    // SINK rr nodes, ordered by routing priority
   const float* crit;         // per sink (aligned with sink_rr)
+  const int32_t* sink_orig;  // original sink index per ordered sink (stable
+                             // across per-iteration criticality reorders)
   const int16_t* bb;         // [num_nets][4]: x0,y0,x1,y1 (tile bb incl. margin)
   int32_t num_nets;
 };
@@ -66,11 +68,9 @@ struct SlotsDev {
 struct RouteParams {
   float astar_fac;
   float pres_fac;
-  float acc_default;         // unused
   float seg_delay, ipin_delay, seg_base, ipin_base;
-  float delta_cong;          // bucket width, congestion-cost units
+  float delta_fac;           // bucket width in edge-step cost units
   int32_t max_rounds;        // safety bound on delta-stepping rounds
-  int32_t bb_max_small;      // tiles: nets with bb area <= this use small slots
 };
 
 }  // namespace pnrh

@@ -201,9 +201,11 @@ __global__ void route_nets_kernel(
       S.sy = g.ylow[S.sink_node];
       S.crit = nets.crit[si];
       S.astar_fac = P.astar_fac;
-      // per-sink bucket width: half an edge-step in this sink's cost units
-      const float delta = 0.5f * (S.crit * P.seg_delay +
-                                  (1.0f - S.crit) * P.seg_base);
+      // per-sink bucket width: delta_fac edge-steps in this sink's cost
+      // units (wider buckets = fewer, fatter delta-stepping rounds;
+      // PathFinder tolerates the relaxed expansion order)
+      const float delta = P.delta_fac * (S.crit * P.seg_delay +
+                                         (1.0f - S.crit) * P.seg_base);
 
       if (tid == 0) {
         sh.fcnt[0] = 0; sh.fcnt[1] = 0;
@@ -370,7 +372,7 @@ __global__ void route_nets_kernel(
           tree_len = len;
           trees.len[inet] = len;  // always accurate, even on FAIL_TREE_CAP,
                                   // so the retry's rip-up stays balanced
-          if (!sh.fail) trees.sink_delay[si] = dacc;
+          if (!sh.fail) trees.sink_delay[nets.sink_orig[si]] = dacc;
         }
       }
       __syncthreads();
@@ -434,6 +436,29 @@ __global__ void overuse_count_kernel(const int32_t* __restrict__ occ,
     atomicAdd(&out[threadIdx.x], cnt[threadIdx.x]);
 }
 
+// flag nets whose current tree touches an overused node (selective
+// reroute set; reference: phase-two congested-nets-only rebuild,
+// partitioning_multi_sink...cxx:5530 build_phase_two)
+__global__ void flag_congested_nets_kernel(TreesDev trees,
+                                           const int32_t* __restrict__ occ,
+                                           const int16_t* __restrict__ cap,
+                                           int32_t num_nets,
+                                           uint8_t* __restrict__ out) {
+  int inet = blockIdx.x;
+  if (inet >= num_nets) return;
+  __shared__ int flag;
+  if (threadIdx.x == 0) flag = 0;
+  __syncthreads();
+  int64_t off = trees.off[inet];
+  int32_t len = trees.len[inet];
+  for (int k = threadIdx.x; k < len && !flag; k += blockDim.x) {
+    int32_t v = trees.node[off + k];
+    if (occ[v] > cap[v]) flag = 1;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) out[inet] = (uint8_t)flag;
+}
+
 // occupancy recount from route trees (debug cross-check; reference:
 // recalculate_occ partitioning_multi_sink...:6194-6216)
 __global__ void recount_occ_kernel(TreesDev trees, const int32_t* __restrict__ net_ids,
@@ -465,12 +490,14 @@ struct RouteLaunchArgs {
   int32_t num_nodes, nx, ny, L, npt;
   // NetsDev
   const int32_t* net_src; const int32_t* sink_ptr; const int32_t* sink_rr;
-  const float* crit; const int16_t* bb; int32_t num_nets;
+  const float* crit; const int32_t* sink_orig; const int16_t* bb;
+  int32_t num_nets;
   // TreesDev
   const int64_t* tree_off; int32_t* tree_node; int32_t* tree_parent;
   int8_t* tree_sw; float* tree_delay; int32_t* tree_len; float* sink_delay;
   // params
   float astar_fac, pres_fac, seg_delay, ipin_delay, seg_base, ipin_base;
+  float delta_fac;
   int32_t max_rounds;
   // queues
   const int32_t* queue_small; int32_t n_queue_small;
@@ -489,13 +516,15 @@ int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
           a->R, a->C, a->row_ptr, a->edge_dst, a->edge_sw,
           a->sw_R, a->sw_Tdel, a->base_cost, a->idx_in_tile,
           a->num_nodes, a->nx, a->ny, a->L, a->npt};
-  NetsDev nets{a->net_src, a->sink_ptr, a->sink_rr, a->crit, a->bb, a->num_nets};
+  NetsDev nets{a->net_src, a->sink_ptr, a->sink_rr, a->crit, a->sink_orig,
+               a->bb, a->num_nets};
   TreesDev trees{a->tree_off, a->tree_node, a->tree_parent, a->tree_sw,
                  a->tree_delay, a->tree_len, a->sink_delay};
   RouteParams P{};
   P.astar_fac = a->astar_fac; P.pres_fac = a->pres_fac;
   P.seg_delay = a->seg_delay; P.ipin_delay = a->ipin_delay;
   P.seg_base = a->seg_base; P.ipin_base = a->ipin_base;
+  P.delta_fac = a->delta_fac;
   P.max_rounds = a->max_rounds;
   int grid = a->n_small_slots + a->n_large_slots;
   hipLaunchKernelGGL(route_nets_kernel, dim3(grid), dim3(WG_THREADS), 0,
@@ -545,6 +574,19 @@ __global__ void fill_u64_kernel(uint64_t* p, uint64_t v, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   for (; i < n; i += (int64_t)gridDim.x * blockDim.x) p[i] = v;
 }
+}
+
+extern "C" int pnr_flag_congested_nets(const int64_t* tree_off,
+                                       int32_t* tree_node, int32_t* tree_len,
+                                       const int32_t* occ, const int16_t* cap,
+                                       int32_t num_nets, uint8_t* out,
+                                       void* stream) {
+  TreesDev t{};
+  t.off = tree_off; t.node = tree_node; t.len = tree_len;
+  hipLaunchKernelGGL(pnrh::flag_congested_nets_kernel, dim3(num_nets),
+                     dim3(256), 0, (hipStream_t)stream, t, occ, cap,
+                     num_nets, out);
+  return (int)hipGetLastError();
 }
 
 extern "C" int pnr_fill_u64_launch(uint64_t* p, uint64_t v, int64_t n, void* stream) {

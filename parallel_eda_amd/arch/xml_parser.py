@@ -1,0 +1,145 @@
+"""arch.xml subset parser.
+
+Ingests the structural subset of VPR architecture files that the engine's
+fabric model uses (reference: libarchfpga/read_xml_arch_file.c:2528
+XmlReadArch): device grid hints, channel segmentation (<segmentlist>),
+switch timing (<switchlist>), connection-block Fc (<fc> defaults), and the
+CLB/IO pin counts from <complexblocklist>. Everything else (pb_type mode
+hierarchy, interconnect detail inside clusters) is intentionally collapsed:
+this engine models CLBs at the block level with equivalent pins, like the
+placer/router layers of the reference do.
+"""
+import math
+import xml.etree.ElementTree as ET
+
+from .archdef import ArchDef
+
+
+def _first_float(el, *names, default=None):
+    for n in names:
+        v = el.get(n)
+        if v is not None:
+            try:
+                return float(v)
+            except ValueError:
+                pass
+    return default
+
+
+def parse_arch_xml(path_or_text, nx=None, ny=None, W=64, name=None) -> ArchDef:
+    """Parse an arch.xml into an ArchDef. Grid size (nx, ny) and channel
+    width W are flow inputs in VPR (auto-sized from the netlist), so they
+    are parameters here with the XML supplying the fabric timing/topology.
+    """
+    if "\n" in str(path_or_text) or str(path_or_text).lstrip().startswith("<"):
+        root = ET.fromstring(path_or_text)
+        src = "<inline>"
+    else:
+        root = ET.parse(path_or_text).getroot()
+        src = str(path_or_text)
+
+    a = ArchDef()
+    a.name = name or f"xml:{src}"
+    if nx:
+        a.nx = nx
+    if ny:
+        a.ny = ny
+    a.W = W if W % 2 == 0 else W + 1
+
+    # ---- segments ----
+    seg = root.find(".//segmentlist/segment")
+    if seg is not None:
+        length = seg.get("length", "4")
+        a.L = 1 if length in ("longline",) else max(1, int(length))
+        rm = _first_float(seg, "Rmetal", default=None)
+        cm = _first_float(seg, "Cmetal", default=None)
+        if rm is not None:
+            a.R_wire = rm
+        if cm is not None:
+            a.C_wire = cm
+        # Fc override on the segment
+        for fc_el in seg.findall("fc") + ([] if seg.find("fc") is None else []):
+            pass
+
+    # ---- switches ----
+    sw = None
+    for cand in root.findall(".//switchlist/switch"):
+        sw = cand
+        break
+    if sw is not None:
+        r = _first_float(sw, "R", default=None)
+        cin = _first_float(sw, "Cin", default=None)
+        tdel = _first_float(sw, "Tdel", default=None)
+        if tdel is None:
+            t_el = sw.find("Tdel")
+            if t_el is not None:
+                tdel = _first_float(t_el, "delay", default=None)
+        if r is not None:
+            a.R_sw = r
+        if cin is not None:
+            a.C_sw_in = cin
+        if tdel is not None:
+            a.T_sw = tdel
+
+    # ---- complex blocks: find the CLB-like type and IO ----
+    def pin_count(pb, kind):
+        total = 0
+        for p in pb.findall(kind):
+            total += int(p.get("num_pins", "1"))
+        return total
+
+    for pb in root.findall(".//complexblocklist/pb_type"):
+        pname = (pb.get("name") or "").lower()
+        n_in = pin_count(pb, "input")
+        n_out = pin_count(pb, "output")
+        if pname in ("io", "inpad", "outpad"):
+            cap = int(pb.get("capacity", "8"))
+            a.io_cap = max(1, cap)
+        elif n_in > 0 and n_out > 0:
+            # first real logic block type
+            a.clb_in = n_in
+            a.clb_out = n_out
+            # fc defaults
+            fc = pb.find("fc")
+            if fc is not None:
+                fin = _first_float(fc, "in_val", "default_in_val", default=None)
+                fout = _first_float(fc, "out_val", "default_out_val", default=None)
+                in_ty = fc.get("in_type", fc.get("default_in_type", "frac"))
+                out_ty = fc.get("out_type", fc.get("default_out_type", "frac"))
+                if fin is not None:
+                    a.fc_in = max(1, int(round(fin * a.W)) if "frac" in in_ty
+                                  else int(fin))
+                if fout is not None:
+                    a.fc_out = max(1, int(round(fout * a.W)) if "frac" in out_ty
+                                   else int(fout))
+            # sequential element timing if present
+            tsu = pb.find(".//T_setup")
+            if tsu is not None:
+                v = _first_float(tsu, "value", default=None)
+                if v is not None:
+                    a.T_seq_in = v
+            tcq = pb.find(".//T_clock_to_Q")
+            if tcq is not None:
+                v = _first_float(tcq, "max", "value", default=None)
+                if v is not None:
+                    a.T_seq_out = v
+            dmx = pb.find(".//delay_constant")
+            if dmx is not None:
+                v = _first_float(dmx, "max", default=None)
+                if v is not None:
+                    a.T_clb = v
+    a.fc_in = min(a.fc_in, a.W)
+    a.fc_out = min(a.fc_out, a.W)
+    return a
+
+
+def size_grid_for_netlist(netlist, arch: ArchDef, fill_target=0.8):
+    """VPR-style auto grid sizing (reference: SetupGrid.c): smallest square
+    grid fitting the CLBs at fill_target, with enough IO perimeter."""
+    n_clb = int((netlist.block_type == 1).sum())
+    n_io = int((netlist.block_type == 0).sum())
+    side = max(2, math.ceil(math.sqrt(n_clb / fill_target)))
+    while 2 * (side + side) * arch.io_cap < n_io:
+        side += 1
+    arch.nx = arch.ny = side
+    return arch

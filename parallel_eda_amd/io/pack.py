@@ -1,0 +1,177 @@
+"""Greedy packing: LUT/FF primitives -> CLB clusters.
+
+Re-implements the semantics of the reference's clusterer
+(vpr/SRC/pack/cluster.c:do_clustering, pack.c:20 try_pack): seed a cluster
+with the highest-gain unclustered primitive, then greedily absorb
+connected primitives (attraction = shared signals) while the cluster has
+room (N BLEs, clb_in input pins). Produces the block-level NetlistPy the
+placer/router consume, where a block is one CLB (or an IO pad).
+"""
+import heapq
+from collections import defaultdict
+
+import numpy as np
+
+from ..arch.archdef import ArchDef, BLK_IO, BLK_CLB
+from .blif import BlifModel
+from .synth import NetlistPy
+
+
+def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
+    """Pack `model` into CLBs of n_ble BLEs with arch.clb_in input pins.
+
+    Returns (NetlistPy, cluster_of_prim dict, block_names list).
+    """
+    prims = model.prims
+    np_prims = len(prims)
+    # signal -> driving primitive index (or input pad)
+    drv_of_sig = {}
+    for s in model.inputs:
+        drv_of_sig[s] = ("pad", s)
+    for i, p in enumerate(prims):
+        drv_of_sig[p.name] = ("prim", i)
+
+    # primitive adjacency (shared signals), for attraction gain
+    sig_users = defaultdict(list)
+    for i, p in enumerate(prims):
+        for s in p.inputs:
+            sig_users[s].append(i)
+
+    cluster_of = [-1] * np_prims
+    clusters = []
+
+    def cluster_inputs(members, cand=None):
+        """Distinct external input signals of members (+cand)."""
+        mem = set(members)
+        if cand is not None:
+            mem.add(cand)
+        produced = {prims[i].name for i in mem}
+        ins = set()
+        for i in mem:
+            for s in prims[i].inputs:
+                if s not in produced:
+                    ins.add(s)
+            if prims[i].clock:
+                pass  # clock nets are global, don't count against pins
+        return ins
+
+    unclustered = set(range(np_prims))
+    while unclustered:
+        # seed: primitive with most inputs (hardest to place later)
+        seed = max(unclustered, key=lambda i: (len(prims[i].inputs), -i))
+        members = [seed]
+        unclustered.discard(seed)
+        ins = cluster_inputs(members)
+        while len(members) < n_ble:
+            # candidates: users/drivers of member signals
+            gain = defaultdict(int)
+            for i in members:
+                p = prims[i]
+                for s in p.inputs:
+                    d = drv_of_sig.get(s)
+                    if d and d[0] == "prim" and d[1] in unclustered:
+                        gain[d[1]] += 1
+                for j in sig_users.get(p.name, ()):
+                    if j in unclustered:
+                        gain[j] += 1
+            if not gain:
+                break
+            # best gain, feasibility-filtered on input pins
+            best = None
+            for cand, gn in sorted(gain.items(), key=lambda kv: (-kv[1], kv[0])):
+                if len(cluster_inputs(members, cand)) <= arch.clb_in:
+                    best = cand
+                    break
+            if best is None:
+                break
+            members.append(best)
+            unclustered.discard(best)
+        clusters.append(members)
+        for i in members:
+            cluster_of[i] = len(clusters) - 1
+
+    # ---- block-level netlist ----
+    # blocks: input pads, output pads, clusters
+    n_in = len(model.inputs)
+    n_out = len(model.outputs)
+    nb = n_in + n_out + len(clusters)
+    block_type = np.full(nb, BLK_CLB, dtype=np.int8)
+    block_type[:n_in + n_out] = BLK_IO
+    block_is_seq = np.zeros(nb, dtype=np.uint8)
+    block_is_seq[:n_in + n_out] = 1
+    names = ([f"ipad:{s}" for s in model.inputs] +
+             [f"opad:{s}" for s in model.outputs] +
+             [f"clb_{k}" for k in range(len(clusters))])
+    clb0 = n_in + n_out
+    # a cluster is sequential if it contains any latch
+    for k, members in enumerate(clusters):
+        if any(prims[i].kind == "latch" for i in members):
+            block_is_seq[clb0 + k] = 1
+
+    in_pad_of = {s: i for i, s in enumerate(model.inputs)}
+    out_pad_of = {s: n_in + i for i, s in enumerate(model.outputs)}
+
+    # nets: one per signal that crosses a cluster boundary (or feeds a pad)
+    drivers, sink_lists = [], []
+    for sig, d in drv_of_sig.items():
+        if d[0] == "pad":
+            src_blk = in_pad_of[sig]
+            src_cluster = -1
+        else:
+            src_cluster = cluster_of[d[1]]
+            src_blk = clb0 + src_cluster
+        sinks = set()
+        for j in sig_users.get(sig, ()):
+            c = cluster_of[j]
+            if c != src_cluster:
+                sinks.add(clb0 + c)
+        if sig in out_pad_of:
+            sinks.add(out_pad_of[sig])
+        sinks.discard(src_blk)
+        if sinks:
+            drivers.append(src_blk)
+            sink_lists.append(sorted(sinks))
+
+    sink_ptr = np.zeros(len(drivers) + 1, dtype=np.int64)
+    for i, s in enumerate(sink_lists):
+        sink_ptr[i + 1] = sink_ptr[i] + len(s)
+    net_sinks = (np.concatenate([np.asarray(s, dtype=np.int32)
+                                 for s in sink_lists])
+                 if sink_lists else np.zeros(0, dtype=np.int32))
+    # clustering can create cluster-level combinational cycles even from an
+    # acyclic primitive graph (p1 in A -> p2 in B -> p3 in A). The timing
+    # model needs an acyclic comb graph, so blocks stuck in a cycle are
+    # marked sequential (they contain registered paths in any real mapping).
+    _break_comb_cycles(block_is_seq, drivers, sink_ptr, net_sinks, nb)
+    nl = NetlistPy(block_type, block_is_seq,
+                   np.asarray(drivers, dtype=np.int32), sink_ptr, net_sinks,
+                   names=names)
+    return nl, cluster_of, names
+
+
+def _break_comb_cycles(block_is_seq, drivers, sink_ptr, net_sinks, nb):
+    while True:
+        indeg = np.zeros(nb, dtype=np.int64)
+        adj = defaultdict(list)
+        for n, drv in enumerate(drivers):
+            for s in net_sinks[sink_ptr[n]:sink_ptr[n + 1]]:
+                if not block_is_seq[s]:
+                    adj[drv].append(int(s))
+                    indeg[s] += 1
+        q = [b for b in range(nb)
+             if block_is_seq[b] or indeg[b] == 0]
+        seen = np.zeros(nb, dtype=bool)
+        seen[q] = True
+        head = 0
+        q = list(q)
+        while head < len(q):
+            b = q[head]; head += 1
+            for s in adj.get(b, ()):
+                indeg[s] -= 1
+                if indeg[s] == 0 and not seen[s]:
+                    seen[s] = True
+                    q.append(s)
+        stuck = np.nonzero(~seen)[0]
+        if len(stuck) == 0:
+            return
+        block_is_seq[stuck[0]] = 1  # break one cycle member, re-check

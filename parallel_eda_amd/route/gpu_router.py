@@ -22,7 +22,8 @@ class GpuRouter:
     def __init__(self, g, arch: ArchDef, src_rr, sink_ptr, sink_rr,
                  device="cuda:0", astar_fac=1.2, n_small_slots=512,
                  n_large_slots=8, bb_margin=3, max_rounds=200000,
-                 occ=None):
+                 delta_fac=3.0, deterministic=False,
+                 concurrent_threshold=768, occ=None):
         torch = _torch()
         self.torch = torch
         self.device = device
@@ -32,6 +33,9 @@ class GpuRouter:
         self.astar_fac = astar_fac
         self.bb_margin = bb_margin
         self.max_rounds = max_rounds
+        self.delta_fac = delta_fac
+        self.deterministic = deterministic
+        self.concurrent_threshold = concurrent_threshold
 
         if g.num_edges >= 2**31:
             raise ValueError("edge count exceeds int32 CSR limit")
@@ -224,13 +228,22 @@ class GpuRouter:
         perm = np.lexsort((-crit, net_of_sink))
         t_sink_rr = torch.from_numpy(self.sink_rr[perm]).to(self.device)
         t_crit = torch.from_numpy(np.ascontiguousarray(crit[perm], dtype=np.float32)).to(self.device)
+        t_sink_orig = torch.from_numpy(perm.astype(np.int32)).to(self.device)
 
         todo = (np.arange(self.num_nets, dtype=np.int64) if net_subset is None
                 else np.asarray(net_subset, dtype=np.int64))
         attempts = 0
         while True:
             self.t_fail.zero_()
-            waves = self._schedule_waves(todo)
+            # Large reroute sets run as ONE concurrent launch (net-level
+            # parallelism with atomic congestion, reference locking_route
+            # family); small/endgame sets get the deterministic bb-disjoint
+            # wave schedule (ParaDRo family), which is what resolves the
+            # last contested nodes.
+            if not self.deterministic and len(todo) > self.concurrent_threshold:
+                waves = [todo]
+            else:
+                waves = self._schedule_waves(todo)
             areas = self._bb_areas(self.bb)
             for wave in waves:
                 small = wave[areas[wave] <= self.bb_max_small_area].astype(np.int32)
@@ -238,7 +251,8 @@ class GpuRouter:
                 q_small = torch.from_numpy(small).to(self.device)
                 q_large = torch.from_numpy(large).to(self.device)
                 self.t_cursors.zero_()
-                args = self._make_args(t_sink_rr, t_crit, q_small, q_large, pres_fac)
+                args = self._make_args(t_sink_rr, t_crit, t_sink_orig,
+                                       q_small, q_large, pres_fac)
                 rc = self.lib.pnr_route_nets(hip_api.ct.byref(args), self._stream())
                 hip_api.check(rc, "route_nets")
                 # no host sync between waves: stream order serializes them
@@ -264,10 +278,21 @@ class GpuRouter:
             todo = failed
 
         overused = int((self.t_occ > self.t_cap.to(torch.int32)).sum().item())
-        sd = self.t_sink_delay.cpu().numpy()
-        sink_delays = np.empty_like(sd)
-        sink_delays[perm] = sd
+        sink_delays = self.t_sink_delay.cpu().numpy()
         return overused, sink_delays
+
+    def congested_nets(self):
+        """Nets whose tree touches an overused node (selective-reroute set;
+        reference: build_phase_two congested-nets-only)."""
+        torch = self.torch
+        flags = torch.zeros(self.num_nets, dtype=torch.uint8, device=self.device)
+        rc = self.lib.pnr_flag_congested_nets(
+            ct_ptr(self.t_tree_off), ct_ptr(self.t_tree_node),
+            ct_ptr(self.t_tree_len), ct_ptr(self.t_occ), ct_ptr(self.t_cap),
+            self.num_nets, ct_ptr(flags), self._stream())
+        hip_api.check(rc, "flag_congested_nets")
+        torch.cuda.synchronize(self.device)
+        return np.nonzero(flags.cpu().numpy())[0]
 
     def update_acc(self, acc_fac):
         rc = self.lib.pnr_update_acc(
@@ -302,7 +327,8 @@ class GpuRouter:
                 self.t_tree_sw[o0:o0 + ln].cpu().numpy(),
                 self.t_tree_delay[o0:o0 + ln].cpu().numpy())
 
-    def _make_args(self, t_sink_rr, t_crit, q_small, q_large, pres_fac):
+    def _make_args(self, t_sink_rr, t_crit, t_sink_orig, q_small, q_large,
+                   pres_fac):
         a = hip_api.RouteLaunchArgs()
         a.type = ct_ptr(self.t_type); a.xlow = ct_ptr(self.t_xlow)
         a.ylow = ct_ptr(self.t_ylow); a.xhigh = ct_ptr(self.t_xhigh)
@@ -316,6 +342,7 @@ class GpuRouter:
         a.L = self.arch.L; a.npt = self.npt
         a.net_src = ct_ptr(self.t_net_src); a.sink_ptr = ct_ptr(self.t_sink_ptr)
         a.sink_rr = ct_ptr(t_sink_rr); a.crit = ct_ptr(t_crit)
+        a.sink_orig = ct_ptr(t_sink_orig)
         a.bb = ct_ptr(self.t_bb); a.num_nets = self.num_nets
         a.tree_off = ct_ptr(self.t_tree_off); a.tree_node = ct_ptr(self.t_tree_node)
         a.tree_parent = ct_ptr(self.t_tree_parent); a.tree_sw = ct_ptr(self.t_tree_sw)
@@ -324,6 +351,7 @@ class GpuRouter:
         a.astar_fac = self.astar_fac; a.pres_fac = pres_fac
         a.seg_delay = self.seg_delay; a.ipin_delay = self.ipin_delay
         a.seg_base = self.seg_base; a.ipin_base = self.ipin_base
+        a.delta_fac = self.delta_fac
         a.max_rounds = self.max_rounds
         a.queue_small = ct_ptr(q_small); a.n_queue_small = len(q_small)
         a.queue_large = ct_ptr(q_large); a.n_queue_large = len(q_large)
@@ -337,7 +365,8 @@ class GpuRouter:
         a.touched_base = ct_ptr(self.t_touched)
         a.t_cap_small = self.t_cap_small; a.t_cap_large = self.t_cap_large
         a.fail_flags = ct_ptr(self.t_fail)
-        self._args_keepalive = (t_sink_rr, t_crit, q_small, q_large)
+        self._args_keepalive = (t_sink_rr, t_crit, t_sink_orig, q_small,
+                                q_large)
         return a
 
 
@@ -347,13 +376,15 @@ def ct_ptr(t):
 
 def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
                          pres_fac_init=0.5, pres_fac_mult=1.3, acc_fac=1.0,
-                         astar_fac=1.2, verbose=False, device="cuda:0"):
+                         astar_fac=1.2, verbose=False, device="cuda:0",
+                         rip_up_always=False, deterministic=False):
     """GPU PathFinder outer loop — mirrors route.router.pathfinder_route."""
     net_ids, src_rr, sink_ptr, sink_rr, conn_index = net_rr_terminals(
         netlist, placement, g, arch)
     from .router import ConnMap
     router = GpuRouter(g, arch, src_rr, sink_ptr.astype(np.int32), sink_rr,
-                       device=device, astar_fac=astar_fac)
+                       device=device, astar_fac=astar_fac,
+                       deterministic=deterministic)
     n_rsinks = len(sink_rr)
     cmap = ConnMap(conn_index, sink_ptr, netlist.num_conns, n_rsinks)
     crit = np.zeros(n_rsinks, dtype=np.float32)
@@ -364,8 +395,19 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
     it = 0
     overused = -1
     for it in range(1, max_iters + 1):
-        overused, sink_delays = router.route_iteration(crit, pres_fac)
-        history.append(dict(iter=it, overused=int(overused), cpd=cpd))
+        # selective reroute (reference: build_phase_two congested-only):
+        # iteration 1 routes everything; later iterations re-route only
+        # nets whose trees touch overused nodes
+        subset = None
+        if it > 1 and not rip_up_always:
+            subset = router.congested_nets()
+            if len(subset) == 0:
+                subset = None
+        overused, sink_delays = router.route_iteration(crit, pres_fac,
+                                                       net_subset=subset)
+        history.append(dict(iter=it, overused=int(overused), cpd=cpd,
+                            rerouted=len(subset) if subset is not None
+                            else router.num_nets))
         if verbose:
             print(f"[gpu] iter {it}: overused={overused} cpd={cpd*1e9:.2f}ns")
         if sta is not None:
