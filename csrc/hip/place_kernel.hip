@@ -351,4 +351,6 @@ int pnr_place_refresh(const PlaceLaunchArgs* a, void* stream) {
   return (int)hipGetLastError();
 }
 
+int64_t pnr_place_args_sizeof() { return (int64_t)sizeof(PlaceLaunchArgs); }
+
 }  // extern "C"

@@ -567,6 +567,8 @@ int pnr_recount_occ(const int64_t* tree_off, int32_t* tree_node, int32_t* tree_l
   return (int)hipGetLastError();
 }
 
+int64_t pnr_route_args_sizeof() { return (int64_t)sizeof(RouteLaunchArgs); }
+
 }  // extern "C"
 
 namespace pnrh {

@@ -149,4 +149,6 @@ int pnr_sta_analyze(const StaLaunchArgs* a, void* stream) {
   return (int)hipGetLastError();
 }
 
+int64_t pnr_sta_args_sizeof() { return (int64_t)sizeof(StaLaunchArgs); }
+
 }  // extern "C"

@@ -48,3 +48,48 @@ def run_flow(arch_name: str = "tseng", seed: int = 1, timing_driven: bool = True
     return FlowResult(placement=placement, route=route,
                       cpd=route.crit_path_delay, wirelength=route.wirelength,
                       times=times)
+
+
+def min_channel_width(netlist, placement, arch: ArchDef, w_lo=8, w_hi=None,
+                      max_route_iters=40, engine="cpu", verbose=False):
+    """Binary search for the minimum routable channel width.
+
+    Reference: base/place_and_route.c:432 binary_search_place_and_route —
+    placement fixed, the rr graph is rebuilt and routed at each candidate W.
+    Returns (w_min, result_at_w_min).
+    """
+    import copy
+    w_hi = w_hi or max(arch.W * 2, 32)
+    best = None
+    best_w = None
+
+    def try_w(w):
+        a = copy.copy(arch)
+        a.W = w + (w % 2)
+        g = rrgraph.build_rr_graph(a)
+        res = pathfinder_route(netlist, placement, g, a, sta=None,
+                               max_iters=max_route_iters, engine=engine)
+        if verbose:
+            print(f"W={a.W}: {'routed' if res.success else 'FAILED'} "
+                  f"(overused={res.overused})")
+        return res
+
+    # make sure hi is routable; grow if not
+    while True:
+        res = try_w(w_hi)
+        if res.success:
+            best, best_w = res, w_hi
+            break
+        w_hi *= 2
+        if w_hi > 4096:
+            raise RuntimeError("unroutable even at W=4096")
+    lo, hi = w_lo, w_hi
+    while lo < hi:
+        mid = (lo + hi) // 2
+        res = try_w(mid)
+        if res.success:
+            best, best_w = res, mid
+            hi = mid
+        else:
+            lo = mid + 1
+    return best_w + (best_w % 2), best

@@ -20,7 +20,8 @@ class RouteLaunchArgs(ct.Structure):
         ("L", ct.c_int32), ("npt", ct.c_int32),
         # NetsDev
         ("net_src", ct.c_void_p), ("sink_ptr", ct.c_void_p), ("sink_rr", ct.c_void_p),
-        ("crit", ct.c_void_p), ("bb", ct.c_void_p), ("num_nets", ct.c_int32),
+        ("crit", ct.c_void_p), ("sink_orig", ct.c_void_p), ("bb", ct.c_void_p),
+        ("num_nets", ct.c_int32),
         # TreesDev
         ("tree_off", ct.c_void_p), ("tree_node", ct.c_void_p),
         ("tree_parent", ct.c_void_p), ("tree_sw", ct.c_void_p),
@@ -30,6 +31,7 @@ class RouteLaunchArgs(ct.Structure):
         ("astar_fac", ct.c_float), ("pres_fac", ct.c_float),
         ("seg_delay", ct.c_float), ("ipin_delay", ct.c_float),
         ("seg_base", ct.c_float), ("ipin_base", ct.c_float),
+        ("delta_fac", ct.c_float),
         ("max_rounds", ct.c_int32),
         # queues
         ("queue_small", ct.c_void_p), ("n_queue_small", ct.c_int32),
@@ -67,6 +69,14 @@ def lib():
         _lib.pnr_flag_congested_nets.argtypes = [ct.c_void_p] * 5 + [ct.c_int32, ct.c_void_p, ct.c_void_p]
         _lib.pnr_fill_u64_launch.restype = ct.c_int
         _lib.pnr_fill_u64_launch.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_int64, ct.c_void_p]
+        _lib.pnr_route_args_sizeof.restype = ct.c_int64
+        # ABI guard: the ctypes mirror must match the C struct exactly —
+        # a silent mismatch turns into near-null GPU pointer faults.
+        c_size = _lib.pnr_route_args_sizeof()
+        py_size = ct.sizeof(RouteLaunchArgs)
+        if c_size != py_size:
+            raise RuntimeError(
+                f"RouteLaunchArgs ABI mismatch: C {c_size} vs ctypes {py_size}")
     return _lib
 
 

@@ -46,6 +46,9 @@ def _lib():
         lib.pnr_place_batch.argtypes = [ct.POINTER(PlaceLaunchArgs), ct.c_void_p]
         lib.pnr_place_refresh.restype = ct.c_int
         lib.pnr_place_refresh.argtypes = [ct.POINTER(PlaceLaunchArgs), ct.c_void_p]
+        lib.pnr_place_args_sizeof.restype = ct.c_int64
+        if lib.pnr_place_args_sizeof() != ct.sizeof(PlaceLaunchArgs):
+            raise RuntimeError("PlaceLaunchArgs ABI mismatch")
         lib._place_ready = True
     return lib
 

@@ -36,6 +36,9 @@ def _lib():
     if not hasattr(lib, "_sta_ready"):
         lib.pnr_sta_analyze.restype = ct.c_int
         lib.pnr_sta_analyze.argtypes = [ct.POINTER(StaLaunchArgs), ct.c_void_p]
+        lib.pnr_sta_args_sizeof.restype = ct.c_int64
+        if lib.pnr_sta_args_sizeof() != ct.sizeof(StaLaunchArgs):
+            raise RuntimeError("StaLaunchArgs ABI mismatch")
         lib._sta_ready = True
     return lib
 
