@@ -1,0 +1,156 @@
+"""Command-line driver: the reference's single `Router` binary equivalent.
+
+Mirrors vpr/SRC/main.c:310 main() + ReadOptions.c flag surface (the knobs
+in t_router_opts / s_placer_opts): positional `circuit arch` plus stage and
+engine flags; stages can be skipped by supplying their output files, like
+VPR (--place_file to skip placement, etc.).
+
+Usage examples:
+  python -m parallel_eda_amd circuit.blif arch.xml --route_chan_width 64
+  python -m parallel_eda_amd --synth tseng --engine gpu --stats_dir out/
+"""
+import argparse
+import sys
+import time
+
+
+def build_parser():
+    p = argparse.ArgumentParser(prog="parallel_eda_amd")
+    p.add_argument("circuit", nargs="?", help=".blif netlist")
+    p.add_argument("arch", nargs="?", help="arch.xml architecture")
+    p.add_argument("--synth", type=str, default=None,
+                   help="use a named synthetic config instead of files")
+    p.add_argument("--fill", type=float, default=0.6)
+    p.add_argument("--seed", type=int, default=1)
+    # engine selection (reference: --router_algorithm, OptionTokens.c:76-84)
+    p.add_argument("--engine", choices=["cpu", "gpu"], default="cpu")
+    p.add_argument("--router_algorithm",
+                   choices=["timing_driven", "breadth_first"],
+                   default="timing_driven")
+    # placer opts (reference: s_placer_opts)
+    p.add_argument("--place_file", type=str, default=None,
+                   help="read placement instead of annealing")
+    p.add_argument("--timing_tradeoff", type=float, default=0.5)
+    p.add_argument("--inner_num", type=float, default=1.0)
+    # router opts (reference: s_router_opts vpr_types.h:724-770)
+    p.add_argument("--route_chan_width", type=int, default=None)
+    p.add_argument("--max_router_iterations", type=int, default=60)
+    p.add_argument("--initial_pres_fac", type=float, default=0.5)
+    p.add_argument("--pres_fac_mult", type=float, default=1.3)
+    p.add_argument("--acc_fac", type=float, default=1.0)
+    p.add_argument("--astar_fac", type=float, default=1.2)
+    p.add_argument("--bb_factor", type=int, default=3)
+    p.add_argument("--min_channel_width", action="store_true",
+                   help="binary search the minimum routable W")
+    p.add_argument("--deterministic", action="store_true",
+                   help="force the fixed wave schedule (GPU)")
+    # outputs
+    p.add_argument("--out_place", type=str, default=None)
+    p.add_argument("--out_route", type=str, default=None)
+    p.add_argument("--stats_dir", type=str, default=None)
+    p.add_argument("--verbose", "-v", action="store_true")
+    return p
+
+
+def main(argv=None):
+    args = build_parser().parse_args(argv)
+    from .arch.archdef import get_arch
+    from .arch.xml_parser import parse_arch_xml, size_grid_for_netlist
+    from .io.synth import synth_netlist, spec_for_arch
+    from .io.blif import read_blif
+    from .io.pack import pack_blif
+    from .io.place_file import write_place, read_place
+    from .io.route_file import write_route
+    from .place.placer import anneal_place
+    from .route.router import pathfinder_route, net_rr_terminals
+    from .timing.sta import STA
+    from .utils.stats import StatsWriter, routing_stats
+    from . import rrgraph
+    from .flow import min_channel_width
+
+    t_start = time.perf_counter()
+    if args.synth:
+        arch = get_arch(args.synth)
+        netlist = synth_netlist(spec_for_arch(arch, fill=args.fill,
+                                              seed=args.seed))
+        print(f"synthetic netlist '{args.synth}': {netlist.num_blocks} blocks,"
+              f" {netlist.num_nets} nets")
+    else:
+        if not args.circuit or not args.arch:
+            print("error: need circuit.blif + arch.xml (or --synth NAME)",
+                  file=sys.stderr)
+            return 2
+        model = read_blif(args.circuit)
+        arch = parse_arch_xml(args.arch, W=args.route_chan_width or 64)
+        netlist, _, _ = pack_blif(model, arch)
+        size_grid_for_netlist(netlist, arch)
+        print(f"read {args.circuit}: {len(model.prims)} primitives -> "
+              f"{netlist.num_blocks} blocks on {arch.nx}x{arch.ny} grid")
+    if args.route_chan_width:
+        arch.W = args.route_chan_width + (args.route_chan_width % 2)
+
+    timing = args.router_algorithm == "timing_driven"
+    sta = STA(netlist, arch) if timing else None
+
+    # ---- placement ----
+    t0 = time.perf_counter()
+    if args.place_file:
+        placement = read_place(args.place_file, netlist)
+        print(f"read placement from {args.place_file}")
+    else:
+        placement = anneal_place(
+            netlist, arch, seed=args.seed,
+            timing_tradeoff=args.timing_tradeoff if timing else 0.0,
+            inner_num=args.inner_num, sta=sta, verbose=args.verbose,
+            engine=args.engine)
+        print(f"placement: bb_cost={placement.bb_cost:.1f} "
+              f"({time.perf_counter()-t0:.2f}s)")
+    if args.out_place:
+        write_place(args.out_place, placement, netlist, arch)
+        print(f"wrote {args.out_place}")
+
+    # ---- routing ----
+    if args.min_channel_width:
+        w, res = min_channel_width(netlist, placement, arch,
+                                   engine=args.engine, verbose=args.verbose)
+        print(f"minimum channel width: {w}")
+    else:
+        t0 = time.perf_counter()
+        g = rrgraph.build_rr_graph(arch)
+        print(f"rr graph: {g.num_nodes} nodes, {g.num_edges} edges "
+              f"({time.perf_counter()-t0:.2f}s)")
+        t0 = time.perf_counter()
+        sw = StatsWriter(args.stats_dir) if args.stats_dir else None
+        res = pathfinder_route(
+            netlist, placement, g, arch, sta=sta,
+            max_iters=args.max_router_iterations,
+            pres_fac_init=args.initial_pres_fac,
+            pres_fac_mult=args.pres_fac_mult, acc_fac=args.acc_fac,
+            astar_fac=args.astar_fac, verbose=args.verbose,
+            engine=args.engine)
+        rt = time.perf_counter() - t0
+        if not res.success:
+            print(f"ROUTING FAILED: {res.overused} overused nodes after "
+                  f"{res.iterations} iterations")
+            return 1
+        print(f"routed in {res.iterations} iterations ({rt:.2f}s): "
+              f"wirelength={res.wirelength} "
+              f"crit_path={res.crit_path_delay*1e9:.3f}ns")
+        net_ids, *_ = net_rr_terminals(netlist, placement, g, arch)
+        st = routing_stats(g, arch, net_ids, lambda k: res.router.tree(k))
+        print(f"  segments={st['total_segments']} bends={st['total_bends']} "
+              f"avg_wl/net={st['avg_wirelength_per_net']:.1f}")
+        if sw:
+            for h in res.stats["history"]:
+                sw.iteration(h["iter"], h["overused"], cpd=h.get("cpd", 0.0))
+            sw.final(res.success, res.wirelength, res.crit_path_delay, st)
+        if args.out_route:
+            write_route(args.out_route, g, arch, net_ids,
+                        lambda k: res.router.tree(k), netlist=netlist)
+            print(f"wrote {args.out_route}")
+    print(f"entire flow took {time.perf_counter()-t_start:.2f}s")
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -108,3 +108,37 @@ def test_place_route_files_roundtrip(tmp_path):
     assert "Routing:" in text
     assert text.count("Net ") == len(net_ids)
     assert "SOURCE" in text and "SINK" in text and "CHAN" in text
+
+
+def test_net_file_roundtrip(tmp_path):
+    from parallel_eda_amd.io.net_file import write_net, read_net, check_netlist
+    arch = get_arch("tiny")
+    nl = synth_netlist(spec_for_arch(arch, fill=0.5, seed=5))
+    nl.names = [f"b{i}" for i in range(nl.num_blocks)]
+    errs, dangling = check_netlist(nl)
+    assert not errs
+    p = tmp_path / "t.net"
+    write_net(p, nl)
+    nl2 = read_net(p)
+    assert nl2.num_blocks == nl.num_blocks
+    assert nl2.num_nets == nl.num_nets
+    # same connectivity as multisets of (driver, sinks)
+    def canon(n):
+        out = []
+        for i in range(n.num_nets):
+            s = sorted(n.net_sinks[n.net_sink_ptr[i]:n.net_sink_ptr[i+1]].tolist())
+            out.append((int(n.net_driver[i]), tuple(s)))
+        return sorted(out)
+    assert canon(nl) == canon(nl2)
+
+
+def test_cli_synth_flow(tmp_path):
+    from parallel_eda_amd.__main__ import main
+    rc = main(["--synth", "tiny", "--fill", "0.5", "--seed", "3",
+               "--out_place", str(tmp_path / "o.place"),
+               "--out_route", str(tmp_path / "o.route"),
+               "--stats_dir", str(tmp_path / "stats")])
+    assert rc == 0
+    assert (tmp_path / "o.place").exists()
+    assert (tmp_path / "o.route").exists()
+    assert (tmp_path / "stats" / "final_stats.txt").exists()
