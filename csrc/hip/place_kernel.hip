@@ -251,15 +251,17 @@ __global__ void place_apply_kernel(PlaceDev p, MovesDev m, float timing_tradeoff
   unsafeAtomicAdd(&cost_acc[1], (double)m.mv_dtd[i]);
 }
 
+// claims reset per batch; counters accumulate across batches (host zeroes
+// them per run_batches call) so the schedule sees the TRUE accept rate.
 __global__ void place_reset_claims_kernel(int32_t* net_claim, int32_t nn,
                                           int32_t* loc_claim, int32_t nl,
                                           int32_t* counters) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
   int total = max(nn, nl);
+  (void)counters;
   for (; i < total; i += gridDim.x * blockDim.x) {
     if (i < nn) net_claim[i] = 0x7FFFFFFF;
     if (i < nl) loc_claim[i] = 0x7FFFFFFF;
-    if (i < 4) counters[i] = 0;
   }
 }
 

@@ -204,21 +204,28 @@ class GpuPlacer:
         self.bb_cost, self.td_cost = float(bb), float(td)
         return self.bb_cost, self.td_cost
 
-    def run_batches(self, T, rlim, n_batches, tt, bb_norm, td_norm):
-        """Run n_batches move batches; returns (success_rate, attempts)."""
-        att = acc = win = 0
-        for _ in range(n_batches):
-            self.batch_counter += 1
-            a = self._args(T=T, rlim=rlim, tt=tt,
-                           inv_bb=1.0 / bb_norm, inv_td=1.0 / td_norm)
-            rc = self.lib.pnr_place_batch(ct.byref(a), self._stream())
-            hip_api.check(rc, "place_batch")
-        self.torch.cuda.synchronize(self.device)
-        # counters reflect only the LAST batch (reset per batch): read them
-        # for rate estimation
-        c = self.t_counters.cpu().numpy()
-        att, acc, win, conf = int(c[0]), int(c[1]), int(c[2]), int(c[3])
-        rej = acc - win - conf  # accepted-but-lost are conflicts, not rejects
+    def run_batches(self, T, rlim, target_moves, tt, bb_norm, td_norm,
+                    max_batches=256):
+        """Launch move batches until ~target_moves moves have been DECIDED
+        (applied winners + Metropolis rejects; claim-conflict losers don't
+        count — they were never decided). Returns (success_rate, attempts).
+        Mirrors the serial placer's move_lim semantics under batching."""
+        self.t_counters.zero_()
+        batches = 0
+        win = att = acc = 0
+        while batches < max_batches:
+            for _ in range(4):
+                self.batch_counter += 1
+                a = self._args(T=T, rlim=rlim, tt=tt,
+                               inv_bb=1.0 / bb_norm, inv_td=1.0 / td_norm)
+                rc = self.lib.pnr_place_batch(ct.byref(a), self._stream())
+                hip_api.check(rc, "place_batch")
+                batches += 1
+            self.torch.cuda.synchronize(self.device)
+            c = self.t_counters.cpu().numpy()
+            att, acc, win = int(c[0]), int(c[1]), int(c[2])
+            if win + (att - acc) >= target_moves:
+                break
         srate = win / max(1, win + (att - acc))
         return srate, att
 
@@ -268,7 +275,6 @@ def anneal_place_gpu(netlist, arch, seed=7, timing_tradeoff=0.5, inner_num=1.0,
     placer = GpuPlacer(netlist, arch, seed=seed, timing=timing, device=device)
     nb = netlist.num_blocks
     move_lim = max(256, int(inner_num * (nb ** 1.3333)))
-    n_batches = max(1, move_lim // placer.n_moves)
     rlim = float(max(arch.nx, arch.ny))
     tt = timing_tradeoff if timing else 0.0
 
@@ -299,7 +305,8 @@ def anneal_place_gpu(netlist, arch, seed=7, timing_tradeoff=0.5, inner_num=1.0,
 
     bb_norm, td_norm = norms()
     # starting T: probe with a hot batch, estimate from delta scale
-    placer.run_batches(1e30, rlim, 2, tt, bb_norm, td_norm)
+    placer.run_batches(1e30, rlim, min(move_lim, 2048), tt, bb_norm, td_norm,
+                       max_batches=16)
     placer.refresh_costs()
     t = 20.0 * placer.bb_cost / max(1, netlist.num_nets) / bb_norm
     history = []
@@ -307,7 +314,7 @@ def anneal_place_gpu(netlist, arch, seed=7, timing_tradeoff=0.5, inner_num=1.0,
     while True:
         refresh_crit()
         bb_norm, td_norm = norms()
-        srate, _ = placer.run_batches(t, rlim, n_batches, tt, bb_norm, td_norm)
+        srate, _ = placer.run_batches(t, rlim, move_lim, tt, bb_norm, td_norm)
         placer.refresh_costs()  # exact resync every temperature
         cost = placer.bb_cost
         history.append((t, cost, srate, rlim))
@@ -329,7 +336,7 @@ def anneal_place_gpu(netlist, arch, seed=7, timing_tradeoff=0.5, inner_num=1.0,
             break
         if itemp > 500:
             break
-    placer.run_batches(0.0, 1.0, n_batches, tt, bb_norm, td_norm)
+    placer.run_batches(0.0, 1.0, move_lim, tt, bb_norm, td_norm)
     placer.refresh_costs()
     ok, err = placer.check_place()
     if not ok:

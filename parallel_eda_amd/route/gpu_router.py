@@ -21,7 +21,7 @@ def _torch():
 class GpuRouter:
     def __init__(self, g, arch: ArchDef, src_rr, sink_ptr, sink_rr,
                  device="cuda:0", astar_fac=1.2, n_small_slots=512,
-                 n_large_slots=8, bb_margin=3, max_rounds=200000,
+                 n_large_slots=64, bb_margin=3, max_rounds=200000,
                  delta_fac=3.0, deterministic=False,
                  concurrent_threshold=768, occ=None):
         torch = _torch()
@@ -87,9 +87,12 @@ class GpuRouter:
         self._bb_version = 0
         self._waves_cache = None
 
-        # slot classes
+        # slot classes: the small class covers the 95th-percentile net bb
+        # (state stays cheap, touched-list reset keeps clears O(visited));
+        # the rest use whole-chip state in the large class.
         areas = self._bb_areas(self.bb)
-        self.bb_max_small_area = int(min(max(int(areas.max()), 64), 4096))
+        p95 = int(np.percentile(areas, 95)) if len(areas) else 64
+        self.bb_max_small_area = int(min(max(p95, 64), 16384))
         self.n_small_slots = n_small_slots
         self.n_large_slots = n_large_slots
         self.small_cap = self.bb_max_small_area * self.npt
@@ -99,7 +102,7 @@ class GpuRouter:
             self.large_cap = self.small_cap
 
         self.f_cap_small = 1 << 16
-        self.f_cap_large = 1 << 21
+        self.f_cap_large = 1 << 19
         self.t_cap_small = self.small_cap
         self.t_cap_large = self.large_cap
 
@@ -233,6 +236,7 @@ class GpuRouter:
         todo = (np.arange(self.num_nets, dtype=np.int64) if net_subset is None
                 else np.asarray(net_subset, dtype=np.int64))
         attempts = 0
+        self.last_retries = []
         while True:
             self.t_fail.zero_()
             # Large reroute sets run as ONE concurrent launch (net-level
@@ -241,7 +245,10 @@ class GpuRouter:
             # wave schedule (ParaDRo family), which is what resolves the
             # last contested nodes.
             if not self.deterministic and len(todo) > self.concurrent_threshold:
-                waves = [todo]
+                # one concurrent launch; biggest work first for load balance
+                areas_t = self._bb_areas(self.bb)[todo]
+                nsk = (self.sink_ptr[todo + 1] - self.sink_ptr[todo]).astype(np.int64)
+                waves = [todo[np.argsort(-(areas_t * nsk), kind="stable")]]
             else:
                 waves = self._schedule_waves(todo)
             areas = self._bb_areas(self.bb)
@@ -262,6 +269,8 @@ class GpuRouter:
             if len(failed) == 0:
                 break
             attempts += 1
+            self.last_retries.append(
+                (len(failed), np.unique(fail[failed]).tolist()))
             if attempts > 6:
                 raise RuntimeError(
                     f"router: {len(failed)} nets failed after retries "
