@@ -118,7 +118,7 @@ __global__ void route_nets_kernel(
     int32_t n_small_slots,
     float4* frontier_base, int64_t f_cap_small, int64_t f_cap_large,
     int32_t* touched_base, int64_t t_cap_small, int64_t t_cap_large,
-    int32_t* fail_flags) {
+    int32_t* fail_flags, unsigned long long* stats /*[8] or null*/) {
   const int tid = threadIdx.x;
   const bool is_small = (int)blockIdx.x < n_small_slots;
   const int slot = blockIdx.x;
@@ -176,7 +176,9 @@ __global__ void route_nets_kernel(
     L.bw = nets.bb[4 * inet + 2] - L.bx0 + 1;
     L.bh = nets.bb[4 * inet + 3] - L.by0 + 1;
     L.npt = g.npt;
-    if (!is_small) { L.bx0 = 0; L.by0 = 0; L.bw = g.nx + 2; L.bh = g.ny + 2; }
+    // NOTE: the large class uses GLOBAL indexing (L.dense=false) but keeps
+    // the net's own bb for pruning — searching the whole chip for a
+    // bb-bounded net was the dominant cost of early profiles.
 
     // ---- rip-up previous tree (reference: route_tree rip-up, occ -1) ----
     int32_t old_len = trees.len[inet];
@@ -247,6 +249,7 @@ __global__ void route_nets_kernel(
       int n_cur = sh.fcnt[0];
       unsigned fmin = sh.fmin_next;
       int rounds = 0;
+      int64_t scanned = 0;
 
       while (!sh.fail) {
         if (n_cur == 0) { if (tid == 0 && sh.best_sink_back == 0xffffffffu) sh.fail = FAIL_NO_PATH; break; }
@@ -313,12 +316,19 @@ __global__ void route_nets_kernel(
           if (sh.fcnt[nxt] > f_cap) sh.fail = FAIL_FRONTIER;
           if (sh.touched_cnt > t_cap) sh.fail = FAIL_TOUCHED;
         }
+        scanned += n_cur;
         __syncthreads();
         n_cur = min((int64_t)sh.fcnt[nxt], f_cap);
         fmin = sh.fmin_next;
         cur = nxt;
       }
       __syncthreads();
+      if (tid == 0 && stats) {
+        atomicAdd(&stats[0], (unsigned long long)rounds);
+        atomicAdd(&stats[1], (unsigned long long)scanned);
+        atomicAdd(&stats[2], 1ull);  // sinks attempted
+        atomicAdd(&stats[3], (unsigned long long)sh.touched_cnt);
+      }
 
       // ---- backtrack + commit (reference: backtrack
       //      partitioning_multi_sink...:613-680 + route_tree add) ----
@@ -509,6 +519,7 @@ struct RouteLaunchArgs {
   float4* frontier_base; int64_t f_cap_small; int64_t f_cap_large;
   int32_t* touched_base; int64_t t_cap_small; int64_t t_cap_large;
   int32_t* fail_flags;
+  unsigned long long* stats;   // [8] search counters or null
 };
 
 int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
@@ -537,7 +548,7 @@ int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
                      a->n_small_slots,
                      a->frontier_base, a->f_cap_small, a->f_cap_large,
                      a->touched_base, a->t_cap_small, a->t_cap_large,
-                     a->fail_flags);
+                     a->fail_flags, a->stats);
   return (int)hipGetLastError();
 }
 

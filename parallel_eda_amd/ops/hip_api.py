@@ -46,6 +46,7 @@ class RouteLaunchArgs(ct.Structure):
         ("touched_base", ct.c_void_p), ("t_cap_small", ct.c_int64),
         ("t_cap_large", ct.c_int64),
         ("fail_flags", ct.c_void_p),
+        ("stats", ct.c_void_p),
     ]
 
 

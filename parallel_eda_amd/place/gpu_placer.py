@@ -304,11 +304,23 @@ def anneal_place_gpu(netlist, arch, seed=7, timing_tradeoff=0.5, inner_num=1.0,
         return (1.0, 1.0)
 
     bb_norm, td_norm = norms()
-    # starting T: probe with a hot batch, estimate from delta scale
+    # starting T = 20 * std(move deltas) (reference: place.c:1045
+    # starting_t), measured from a hot probe batch's accepted deltas
     placer.run_batches(1e30, rlim, min(move_lim, 2048), tt, bb_norm, td_norm,
                        max_batches=16)
+    torch = placer.torch
+    flags = placer.t_mv_flags
+    sel = flags >= 1
+    if int(sel.sum().item()) >= 8:
+        d = (1.0 - tt) * placer.t_mv_dbb[sel] / bb_norm
+        if tt > 0:
+            d = d + tt * placer.t_mv_dtd[sel] / td_norm
+        t = float(20.0 * d.std().item())
+    else:
+        t = 20.0 * placer.bb_cost / max(1, netlist.num_nets) / bb_norm
     placer.refresh_costs()
-    t = 20.0 * placer.bb_cost / max(1, netlist.num_nets) / bb_norm
+    if t <= 0:
+        t = 1.0
     history = []
     itemp = 0
     while True:

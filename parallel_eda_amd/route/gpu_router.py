@@ -141,6 +141,7 @@ class GpuRouter:
         self.t_acc = torch.ones(g.num_nodes, dtype=torch.float32, device=device)
         self.t_fail = torch.zeros(self.num_nets, dtype=torch.int32, device=device)
         self.t_cursors = torch.zeros(2, dtype=torch.int32, device=device)
+        self.t_stats = torch.zeros(8, dtype=torch.int64, device=device)
         self.t_overuse = torch.zeros(8, dtype=torch.int32, device=device)
 
         # lookahead constants (same as serial oracle)
@@ -244,7 +245,8 @@ class GpuRouter:
             # family); small/endgame sets get the deterministic bb-disjoint
             # wave schedule (ParaDRo family), which is what resolves the
             # last contested nodes.
-            if not self.deterministic and len(todo) > self.concurrent_threshold:
+            if not self.deterministic and (
+                    len(todo) > self.concurrent_threshold or attempts > 0):
                 # one concurrent launch; biggest work first for load balance
                 areas_t = self._bb_areas(self.bb)[todo]
                 nsk = (self.sink_ptr[todo + 1] - self.sink_ptr[todo]).astype(np.int64)
@@ -289,6 +291,14 @@ class GpuRouter:
         overused = int((self.t_occ > self.t_cap.to(torch.int32)).sum().item())
         sink_delays = self.t_sink_delay.cpu().numpy()
         return overused, sink_delays
+
+    def search_stats(self):
+        s = self.t_stats.cpu().numpy()
+        return dict(rounds=int(s[0]), scanned=int(s[1]), sinks=int(s[2]),
+                    touched=int(s[3]))
+
+    def reset_search_stats(self):
+        self.t_stats.zero_()
 
     def congested_nets(self):
         """Nets whose tree touches an overused node (selective-reroute set;
@@ -374,6 +384,7 @@ class GpuRouter:
         a.touched_base = ct_ptr(self.t_touched)
         a.t_cap_small = self.t_cap_small; a.t_cap_large = self.t_cap_large
         a.fail_flags = ct_ptr(self.t_fail)
+        a.stats = ct_ptr(self.t_stats)
         self._args_keepalive = (t_sink_rr, t_crit, t_sink_orig, q_small,
                                 q_large)
         return a
