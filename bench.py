@@ -105,9 +105,11 @@ def main():
 
     for w in range(args.warmup):
         t0 = time.perf_counter()
+        router.reset_search_stats()
         over = step()
         log(f"warmup {w}: overused={over} cpd={cpd*1e9:.2f}ns "
-            f"{time.perf_counter()-t0:.1f}s")
+            f"{time.perf_counter()-t0:.1f}s stats={router.search_stats()} "
+            f"retries={router.last_retries}")
 
     barrier_sync()
     t0 = time.perf_counter()

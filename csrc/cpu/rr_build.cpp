@@ -155,13 +155,33 @@ RRGraph build_rr_graph(const ArchParams& ap) {
   g.num_nodes = (int)g.type.size();
 
   // ---------------- edge generation (two passes) ----------------
-  // Helper lookups ------------------------------------------------
-  // Wire starting at tile pos (drivable there): INC (t even) with
-  // span_low == pos; DEC (t odd) with span_high == pos.
-  auto wire_starts_at = [&](int t, int pos, int N) -> bool {
-    if ((t & 1) == 0) return span_low(pos, t, L) == pos;
-    return span_high(pos, t, L, N) == pos;
+  // Precomputed per-position track lists (span arithmetic is O(L) per
+  // query; at Titan scale the naive form dominated the build at ~18 Gops).
+  // starts_at[N][pos]: tracks whose wire is DRIVEN at tile pos
+  //   (INC with span_low==pos, DEC with span_high==pos)
+  // inc_end[pos]: INC tracks whose wire ENDS (xhigh) at pos
+  // dec_lowstart[pos]: DEC tracks whose span_low == pos
+  struct TrackLists {
+    std::vector<std::vector<int16_t>> starts_at, inc_end, dec_lowstart;
+    void build(int N, int W, int L) {
+      starts_at.assign(N + 1, {});
+      inc_end.assign(N + 1, {});
+      dec_lowstart.assign(N + 1, {});
+      for (int pos = 1; pos <= N; ++pos)
+        for (int t = 0; t < W; ++t) {
+          if ((t & 1) == 0) {
+            if (span_low(pos, t, L) == pos) starts_at[pos].push_back(t);
+            if (span_high(pos, t, L, N) == pos) inc_end[pos].push_back(t);
+          } else {
+            if (span_high(pos, t, L, N) == pos) starts_at[pos].push_back(t);
+            if (span_low(pos, t, L) == pos) dec_lowstart[pos].push_back(t);
+          }
+        }
+    }
   };
+  TrackLists tlx, tly;
+  tlx.build(nx, W, L);
+  tly.build(ny, W, L);
   // channel access for a tile side: returns (is_x, chan_index, pos)
   struct SideRef { bool is_x; int chan; int pos; bool valid; };
   auto side_ref = [&](int x, int y, int8_t side) -> SideRef {
@@ -182,13 +202,13 @@ RRGraph build_rr_graph(const ArchParams& ap) {
   auto sb_in_list = [&](int i, int j, int side, std::vector<int32_t>& in) {
     in.clear();
     if (side == 0) { if (i < 1) return;
-      for (int t = 0; t < W; t += 2) if (span_high(i, t, L, nx) == i) in.push_back(chanx[j].at(t, i));
+      for (int16_t t : tlx.inc_end[i]) in.push_back(chanx[j].at(t, i));
     } else if (side == 1) { if (i + 1 > nx) return;
-      for (int t = 1; t < W; t += 2) if (span_low(i + 1, t, L) == i + 1) in.push_back(chanx[j].at(t, i + 1));
+      for (int16_t t : tlx.dec_lowstart[i + 1]) in.push_back(chanx[j].at(t, i + 1));
     } else if (side == 2) { if (j < 1) return;
-      for (int t = 0; t < W; t += 2) if (span_high(j, t, L, ny) == j) in.push_back(chany[i].at(t, j));
+      for (int16_t t : tly.inc_end[j]) in.push_back(chany[i].at(t, j));
     } else { if (j + 1 > ny) return;
-      for (int t = 1; t < W; t += 2) if (span_low(j + 1, t, L) == j + 1) in.push_back(chany[i].at(t, j + 1));
+      for (int16_t t : tly.dec_lowstart[j + 1]) in.push_back(chany[i].at(t, j + 1));
     }
   };
 
@@ -205,10 +225,10 @@ RRGraph build_rr_graph(const ArchParams& ap) {
       for (size_t pi = 0; pi < tp.opins.size(); ++pi) {
         SideRef sr = side_ref(x, y, tp.opin_side[pi]);
         if (!sr.valid) continue;
-        int N = sr.is_x ? nx : ny;
         cand.clear();
-        for (int t = 0; t < W; ++t)
-          if (wire_starts_at(t, sr.pos, N)) cand.push_back(chan_node(sr.is_x, sr.chan, t, sr.pos));
+        const auto& starts = (sr.is_x ? tlx : tly).starts_at[sr.pos];
+        for (int16_t t : starts)
+          cand.push_back(chan_node(sr.is_x, sr.chan, t, sr.pos));
         if (cand.empty()) continue;
         int n = (int)cand.size();
         int fc = ap.fc_out < n ? ap.fc_out : n;

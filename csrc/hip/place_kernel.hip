@@ -118,7 +118,7 @@ __global__ void place_propose_kernel(PlaceDev p, MovesDev m, float T,
   if (i >= m.n_moves) return;
   m.mv_flags[i] = 0;
   // pick block + destination
-  uint32_t r0 = rng_hash(seed, batch, i * 4 + 0);
+  uint32_t r0 = rng_hash(seed ^ 0x5BD1E995u, batch, i);
   int32_t blk = r0 % p.num_blocks;
   bool io = p.blk_type[blk] == 0;
   int x0 = p.bx[blk], y0 = p.by[blk];
@@ -137,10 +137,10 @@ __global__ void place_propose_kernel(PlaceDev p, MovesDev m, float T,
     slot1 = (int)(rng_hash(seed, batch, i * 131 + 7 * att + 3) % c);
     break;
   }
-  atomicAdd(&m.counters[0], 1);
   if (x1 < 0) return;
   int32_t other = p.grid[((int64_t)x1 * p.gy + y1) * p.cap + slot1];
   if (other == blk) return;
+  atomicAdd(&m.counters[0], 1);  // valid proposals only
 
   // collect affected nets (dedup)
   int32_t nets[MAX_MOVE_NETS];

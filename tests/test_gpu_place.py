@@ -48,8 +48,11 @@ def test_gpu_anneal_quality_vs_cpu(tseng_case):
     arch, nl = tseng_case
     pl_cpu = anneal_place(nl, arch, seed=7, timing_tradeoff=0.0)
     pl_gpu = anneal_place_gpu(nl, arch, seed=7, timing_tradeoff=0.0)
+    hist = pl_gpu.stats["history"]
     assert pl_gpu.bb_cost <= pl_cpu.bb_cost * 1.15, (
-        f"GPU bb {pl_gpu.bb_cost:.1f} vs CPU {pl_cpu.bb_cost:.1f}")
+        f"GPU bb {pl_gpu.bb_cost:.1f} vs CPU {pl_cpu.bb_cost:.1f}; "
+        f"temps={pl_gpu.stats['temps']} hist[:4]={hist[:4]} "
+        f"hist[-4:]={hist[-4:]}")
 
 
 def test_gpu_timing_anneal(tseng_case):
