@@ -121,7 +121,7 @@ class GpuPlacer:
         self.t_net_cost = torch.zeros(nn, dtype=torch.float32, device=device)
         self.t_net_tcost = torch.zeros(nn, dtype=torch.float32, device=device)
 
-        self.n_moves = n_moves or int(np.clip(nb // 4, 256, 1 << 16))
+        self.n_moves = n_moves or int(np.clip(nb // 8, 64, 1 << 16))
         nm = self.n_moves
         self.t_mv_blk = torch.zeros(nm, dtype=torch.int32, device=device)
         self.t_mv_to = torch.zeros(nm, dtype=torch.int32, device=device)
@@ -226,7 +226,11 @@ class GpuPlacer:
             att, acc, win = int(c[0]), int(c[1]), int(c[2])
             if win + (att - acc) >= target_moves:
                 break
-        srate = win / max(1, win + (att - acc))
+        else:
+            self.torch.cuda.synchronize(self.device)
+        # temperature control wants the Metropolis acceptance fraction;
+        # claim-conflict losers are neither accepts nor rejects
+        srate = acc / max(1, att)
         return srate, att
 
     def set_crit(self, conn_crit):
@@ -269,10 +273,12 @@ def _cross_count(n):
 
 
 def anneal_place_gpu(netlist, arch, seed=7, timing_tradeoff=0.5, inner_num=1.0,
-                     sta=None, crit_exp=1.0, verbose=False, device="cuda:0"):
+                     sta=None, crit_exp=1.0, verbose=False, device="cuda:0",
+                     n_moves=None):
     """GPU anneal with the same adaptive schedule as the CPU oracle."""
     timing = sta is not None and timing_tradeoff > 0
-    placer = GpuPlacer(netlist, arch, seed=seed, timing=timing, device=device)
+    placer = GpuPlacer(netlist, arch, seed=seed, timing=timing, device=device,
+                       n_moves=n_moves)
     nb = netlist.num_blocks
     move_lim = max(256, int(inner_num * (nb ** 1.3333)))
     rlim = float(max(arch.nx, arch.ny))
