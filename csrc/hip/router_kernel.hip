@@ -118,7 +118,8 @@ __global__ void route_nets_kernel(
     int32_t n_small_slots,
     float4* frontier_base, int64_t f_cap_small, int64_t f_cap_large,
     int32_t* touched_base, int64_t t_cap_small, int64_t t_cap_large,
-    int32_t* fail_flags, unsigned long long* stats /*[8] or null*/) {
+    int32_t* fail_flags, unsigned long long* stats /*[8] or null*/,
+    unsigned long long* net_scans /* per-net scan counts or null */) {
   const int tid = threadIdx.x;
   const bool is_small = (int)blockIdx.x < n_small_slots;
   const int slot = blockIdx.x;
@@ -250,6 +251,7 @@ __global__ void route_nets_kernel(
       unsigned fmin = sh.fmin_next;
       int rounds = 0;
       int64_t scanned = 0;
+      (void)net_scans;
 
       while (!sh.fail) {
         if (n_cur == 0) { if (tid == 0 && sh.best_sink_back == 0xffffffffu) sh.fail = FAIL_NO_PATH; break; }
@@ -328,6 +330,7 @@ __global__ void route_nets_kernel(
         atomicAdd(&stats[1], (unsigned long long)scanned);
         atomicAdd(&stats[2], 1ull);  // sinks attempted
         atomicAdd(&stats[3], (unsigned long long)sh.touched_cnt);
+        if (net_scans) atomicAdd(&net_scans[inet], (unsigned long long)scanned);
       }
 
       // ---- backtrack + commit (reference: backtrack
@@ -520,6 +523,7 @@ struct RouteLaunchArgs {
   int32_t* touched_base; int64_t t_cap_small; int64_t t_cap_large;
   int32_t* fail_flags;
   unsigned long long* stats;   // [8] search counters or null
+  unsigned long long* net_scans;  // per-net scan counters or null
 };
 
 int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
@@ -548,7 +552,7 @@ int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
                      a->n_small_slots,
                      a->frontier_base, a->f_cap_small, a->f_cap_large,
                      a->touched_base, a->t_cap_small, a->t_cap_large,
-                     a->fail_flags, a->stats);
+                     a->fail_flags, a->stats, a->net_scans);
   return (int)hipGetLastError();
 }
 

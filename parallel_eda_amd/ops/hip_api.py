@@ -47,6 +47,7 @@ class RouteLaunchArgs(ct.Structure):
         ("t_cap_large", ct.c_int64),
         ("fail_flags", ct.c_void_p),
         ("stats", ct.c_void_p),
+        ("net_scans", ct.c_void_p),
     ]
 
 

@@ -142,6 +142,8 @@ class GpuRouter:
         self.t_fail = torch.zeros(self.num_nets, dtype=torch.int32, device=device)
         self.t_cursors = torch.zeros(2, dtype=torch.int32, device=device)
         self.t_stats = torch.zeros(8, dtype=torch.int64, device=device)
+        self.t_net_scans = torch.zeros(self.num_nets, dtype=torch.int64,
+                                       device=device)
         self.t_overuse = torch.zeros(8, dtype=torch.int32, device=device)
 
         # lookahead constants (same as serial oracle)
@@ -299,6 +301,16 @@ class GpuRouter:
 
     def reset_search_stats(self):
         self.t_stats.zero_()
+        self.t_net_scans.zero_()
+
+    def top_cost_nets(self, k=10):
+        import torch
+        v, idx = torch.topk(self.t_net_scans, min(k, self.num_nets))
+        idx = idx.cpu().numpy(); v = v.cpu().numpy()
+        nsk = self.sink_ptr[idx + 1] - self.sink_ptr[idx]
+        areas = self._bb_areas(self.bb)[idx]
+        return [(int(i), int(s), int(n), int(a))
+                for i, s, n, a in zip(idx, v, nsk, areas)]
 
     def congested_nets(self):
         """Nets whose tree touches an overused node (selective-reroute set;
@@ -385,6 +397,7 @@ class GpuRouter:
         a.t_cap_small = self.t_cap_small; a.t_cap_large = self.t_cap_large
         a.fail_flags = ct_ptr(self.t_fail)
         a.stats = ct_ptr(self.t_stats)
+        a.net_scans = ct_ptr(self.t_net_scans)
         self._args_keepalive = (t_sink_rr, t_crit, t_sink_orig, q_small,
                                 q_large)
         return a
