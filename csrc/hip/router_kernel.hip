@@ -30,6 +30,16 @@ namespace pnrh {
 
 #define INF_STATE 0xffffffffffffffffull
 
+#ifdef PNR_DEBUG_BOUNDS
+#define DBG_LI(li, cap, code) \
+  (((li) < 0 || (li) >= (cap)) ? (atomicExch(&sh.fail, (code)), (int64_t)0) : (li))
+#define DBG_NODE(v, code) \
+  (((v) < 0 || (v) >= g.num_nodes) ? (atomicExch(&sh.fail, (code)), 0) : (v))
+#else
+#define DBG_LI(li, cap, code) (li)
+#define DBG_NODE(v, code) (v)
+#endif
+
 __device__ __forceinline__ uint32_t f32_bits(float f) {
   // order-preserving bits for non-negative floats
   return __float_as_uint(f);
@@ -148,6 +158,8 @@ __global__ void route_nets_kernel(
     f_cap = f_cap_large; t_cap = t_cap_large;
   }
 
+  const int64_t s_cap = is_small ? small_cap : large_cap;
+  (void)s_cap;
   const int32_t* queue = is_small ? queue_small : queue_large;
   const int32_t n_queue = is_small ? n_queue_small : n_queue_large;
   int32_t* cursor = q_cursors + (is_small ? 0 : 1);
@@ -266,9 +278,9 @@ __global__ void route_nets_kernel(
         for (int i = tid; i < n_cur; i += WG_THREADS) {
           float4 e = fr[cur][i];
           float tot = e.x, back = e.y;
-          int32_t v = __float_as_int(e.z);
+          int32_t v = DBG_NODE(__float_as_int(e.z), 95);
           int32_t prev = __float_as_int(e.w);
-          int64_t li = L(g, v);
+          int64_t li = DBG_LI(L(g, v), s_cap, 91);
           // stale check: seeds (prev==self) store back 0.0 in state while
           // their entry carries the real back cost
           const uint64_t expect = (prev == v) ? pack_state(0.0f, v)
@@ -295,7 +307,7 @@ __global__ void route_nets_kernel(
             float back_new = back + S.crit * hop_delay(g, sw, w) +
                              (1.0f - S.crit) * cong_cost(g, P, occ, acc, w);
             float tot_new = back_new + S.astar_fac * expected_cost(g, P, w, S);
-            int64_t lw = L(g, w);
+            int64_t lw = DBG_LI(L(g, w), s_cap, 92);
             uint64_t pk = pack_state(back_new, v);
             uint64_t old = atomicMin((unsigned long long*)&state[lw],
                                      (unsigned long long)pk);
@@ -339,8 +351,12 @@ __global__ void route_nets_kernel(
         int n = 0;
         int32_t v = S.sink_node;
         for (;;) {
-          uint64_t st = state[(size_t)L(g, v)];
+          v = DBG_NODE(v, 94);
+          uint64_t st = state[(size_t)DBG_LI(L(g, v), s_cap, 93)];
           int32_t prev = (int32_t)(st & 0xffffffffu);
+#ifdef PNR_DEBUG_BOUNDS
+          if (sh.fail) break;
+#endif
           if (prev == v) break;  // reached a tree seed
           if (n >= PATH_CAP) { sh.fail = FAIL_PATH_CAP; break; }
           sh.path[n++] = v;

@@ -25,6 +25,8 @@ def cpu():
 
 
 def hip_lib_path() -> Path:
+    if os.environ.get("PNR_HIP_DEBUG"):
+        return _PKG_DIR / "libpnr_hip_dbg.so"
     return _PKG_DIR / "libpnr_hip.so"
 
 

@@ -36,6 +36,15 @@ def main():
     ] + [str(s) for s in srcs]
     print("build_hip:", " ".join(cmd))
     subprocess.check_call(cmd, cwd=ROOT)
+    # debug variant with device-side bounds checks (loaded via
+    # PNR_HIP_DEBUG=1; see ops/__init__.py)
+    dbg = OUT.with_name("libpnr_hip_dbg.so")
+    cmd_dbg = [
+        hipcc, "--offload-arch=gfx950", "-O2", "-std=c++17",
+        "-DPNR_DEBUG_BOUNDS", "-fPIC", "-shared", "-fvisibility=default",
+        "-o", str(dbg),
+    ] + [str(s) for s in srcs]
+    subprocess.check_call(cmd_dbg, cwd=ROOT)
     return 0
 
 
