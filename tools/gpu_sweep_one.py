@@ -30,3 +30,7 @@ st = r.search_stats()
 print(f"astar={astar} delta={delta}: t_last={t_hist[-1]:.2f}s t_all={[round(t,2) for t in t_hist]} "
       f"over={over} wl={r.wirelength()} scan/sink={st['scanned']//max(1,st['sinks'])} "
       f"touched/sink={st['touched']//max(1,st['sinks'])}", flush=True)
+print("top cost nets (net, scans, nsinks, bb_area):", r.top_cost_nets(12), flush=True)
+tot = int(r.t_net_scans.sum().item())
+topsum = sum(t[1] for t in r.top_cost_nets(12))
+print(f"top12 share of last-iter scans: {topsum}/{tot} = {topsum/max(1,tot):.1%}", flush=True)
