@@ -267,8 +267,12 @@ __global__ void route_nets_kernel(
 
       while (!sh.fail) {
         if (n_cur == 0) { if (tid == 0 && sh.best_sink_back == 0xffffffffu) sh.fail = FAIL_NO_PATH; break; }
-        // terminate when the sink's settled cost is <= min frontier f
-        if (sh.best_sink_back != 0xffffffffu && sh.best_sink_back <= fmin) break;
+        // terminate when the sink's settled cost beats the min frontier f
+        // STRICTLY: the == bucket must still be processed so that
+        // equal-cost tie-breaks reach their fixpoint — with an admissible
+        // heuristic this makes the search result order-independent
+        // (deterministic mode relies on it; see docs/PARITY.md §2.3)
+        if (sh.best_sink_back != 0xffffffffu && sh.best_sink_back < fmin) break;
         if (++rounds > P.max_rounds) { if (tid == 0) sh.fail = FAIL_ROUNDS; break; }
         const float thr = bits_f32(fmin) + delta;
         const int nxt = cur ^ 1;

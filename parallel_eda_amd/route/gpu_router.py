@@ -30,7 +30,10 @@ class GpuRouter:
         self.arch = arch
         self.g = g
         self.num_nets = len(src_rr)
-        self.astar_fac = astar_fac
+        # deterministic mode needs an ADMISSIBLE lookahead (astar_fac<=1):
+        # with an inflated heuristic the termination point depends on
+        # which paths the racing relaxations discovered first
+        self.astar_fac = min(astar_fac, 1.0) if deterministic else astar_fac
         self.bb_margin = bb_margin
         self.max_rounds = max_rounds
         self.delta_fac = delta_fac
