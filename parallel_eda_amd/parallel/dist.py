@@ -41,18 +41,30 @@ def init_dist():
     return rank, ws, local
 
 
-def spatial_partition(bb, world_size):
+def spatial_partition(bb, world_size, weight=None):
     """Assign nets to ranks by bb-center strips along x (region-tree top
     cuts; reference analogue: fpga_bipartition / build_net_tree
-    partitioning_multi_sink...cxx:3064,3295). Balanced by net count.
+    partitioning_multi_sink...cxx:3064,3295), balanced by estimated route
+    cost (reference's measured-time load balancing, mpi_route...cxx:249,
+    approximated by nsinks x bb semiperimeter until measured times land).
     Returns rank_of_net (int array)."""
     n = len(bb)
     cx = (bb[:, 0].astype(np.int64) + bb[:, 2].astype(np.int64))
     order = np.argsort(cx, kind="stable")
     rank_of = np.zeros(n, dtype=np.int32)
-    chunks = np.array_split(order, world_size)
-    for r, ch in enumerate(chunks):
-        rank_of[ch] = r
+    if weight is None:
+        chunks = np.array_split(order, world_size)
+        for r, ch in enumerate(chunks):
+            rank_of[ch] = r
+        return rank_of
+    w = np.asarray(weight, dtype=np.float64)[order]
+    cum = np.cumsum(w)
+    total = cum[-1] if len(cum) else 0.0
+    bounds = np.searchsorted(cum, total * np.arange(1, world_size) / world_size)
+    prev = 0
+    for r, b in enumerate(list(bounds) + [n]):
+        rank_of[order[prev:b]] = r
+        prev = b
     return rank_of
 
 
@@ -135,8 +147,14 @@ class DistRouteLoop:
         self.engine = engine
         self.rank = rank
         self.ws = world_size
-        self.rank_of = spatial_partition(bb, world_size) if world_size > 1 \
-            else np.zeros(num_nets, dtype=np.int32)
+        if world_size > 1:
+            nsk = np.diff(np.asarray(sink_ptr))
+            semi = ((bb[:, 2] - bb[:, 0]).astype(np.int64) +
+                    (bb[:, 3] - bb[:, 1]).astype(np.int64) + 2)
+            self.rank_of = spatial_partition(bb, world_size,
+                                             weight=nsk * semi)
+        else:
+            self.rank_of = np.zeros(num_nets, dtype=np.int32)
         self.my_nets = np.nonzero(self.rank_of == rank)[0]
         # sinks owned by my nets
         mask = np.zeros(n_rsinks, dtype=bool)

@@ -89,3 +89,42 @@ def test_gpu_congestion_kernels(tiny_placed):
     over = (occ - cap.to(torch.int32)).clamp(min=0).to(torch.float32)
     expect = 1.0 + 0.5 * over
     assert torch.allclose(acc, expect)
+
+
+def test_gpu_checkpoint_resume(tseng_placed):
+    """Save mid-route, restore into a fresh router, finish identically."""
+    import torch
+    from parallel_eda_amd.route.gpu_router import GpuRouter
+    from parallel_eda_amd.utils.checkpoint import (save_router_state,
+                                                   load_router_state)
+    arch, nl, pl, g = tseng_placed
+    net_ids, src_rr, sink_ptr, sink_rr, _ = net_rr_terminals(nl, pl, g, arch)
+
+    def fresh():
+        return GpuRouter(g, arch, src_rr, sink_ptr.astype(np.int32), sink_rr,
+                         deterministic=True)
+
+    r1 = fresh()
+    crit = np.zeros(len(sink_rr), dtype=np.float32)
+    pres = 0.0
+    for it in range(3):
+        r1.route_iteration(crit, pres)
+        pres = 0.5 if pres == 0.0 else pres * 1.3
+        r1.update_acc(1.0)
+    import tempfile, os
+    path = os.path.join(tempfile.mkdtemp(), "ckpt.npz")
+    save_router_state(path, r1, pres, 3)
+
+    r2 = fresh()
+    pres2, it2 = load_router_state(path, r2)
+    assert it2 == 3 and pres2 == pres
+    # both continue identically (deterministic mode)
+    for it in range(3):
+        o1, _ = r1.route_iteration(crit, pres)
+        o2, _ = r2.route_iteration(crit, pres2)
+        assert o1 == o2
+        pres *= 1.3
+        pres2 = pres
+        r1.update_acc(1.0)
+        r2.update_acc(1.0)
+    assert torch.equal(r1.t_occ, r2.t_occ)
