@@ -10,7 +10,7 @@ import pytest
 from parallel_eda_amd.arch.archdef import get_arch
 from parallel_eda_amd.io.synth import synth_netlist, spec_for_arch
 from parallel_eda_amd.place.placer import anneal_place
-from parallel_eda_amd.route.router import pathfinder_route
+from parallel_eda_amd.route.router import pathfinder_route, net_rr_terminals
 from parallel_eda_amd.timing.sta import STA
 from parallel_eda_amd import rrgraph
 

@@ -22,8 +22,10 @@ t_hist = []
 for it in range(iters):
     t0 = time.perf_counter()
     r.reset_search_stats()
+    print(f"  iter {it} start", flush=True)
     over, sd = r.route_iteration(crit, pres)
     t_hist.append(time.perf_counter() - t0)
+    print(f"  iter {it} done over={over} retries={r.last_retries}", flush=True)
     pres = 0.5 if pres == 0.0 else pres * 1.3
     r.update_acc(1.0)
 st = r.search_stats()
