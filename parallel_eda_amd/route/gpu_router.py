@@ -259,7 +259,9 @@ class GpuRouter:
             else:
                 waves = self._schedule_waves(todo)
             areas = self._bb_areas(self.bb)
-            for wave in waves:
+            import os
+            dbg = os.environ.get("PNR_ROUTE_DEBUG")
+            for wi, wave in enumerate(waves):
                 small = wave[areas[wave] <= self.bb_max_small_area].astype(np.int32)
                 large = wave[areas[wave] > self.bb_max_small_area].astype(np.int32)
                 q_small = torch.from_numpy(small).to(self.device)
@@ -267,8 +269,14 @@ class GpuRouter:
                 self.t_cursors.zero_()
                 args = self._make_args(t_sink_rr, t_crit, t_sink_orig,
                                        q_small, q_large, pres_fac)
+                if dbg:
+                    print(f"    [launch] attempt={attempts} wave={wi}/{len(waves)} "
+                          f"small={len(small)} large={len(large)}", flush=True)
                 rc = self.lib.pnr_route_nets(hip_api.ct.byref(args), self._stream())
                 hip_api.check(rc, "route_nets")
+                if dbg:
+                    torch.cuda.synchronize(self.device)
+                    print(f"    [done]", flush=True)
                 # no host sync between waves: stream order serializes them
             torch.cuda.synchronize(self.device)
             fail = self.t_fail.cpu().numpy()
