@@ -71,6 +71,7 @@ struct RouteParams {
   float seg_delay, ipin_delay, seg_base, ipin_base;
   float delta_fac;           // bucket width in edge-step cost units
   int32_t max_rounds;        // safety bound on delta-stepping rounds
+  int32_t strict_term;       // deterministic mode: process the == bucket
 };
 
 }  // namespace pnrh

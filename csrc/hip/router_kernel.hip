@@ -195,8 +195,23 @@ __global__ void route_nets_kernel(
 
     // ---- rip-up previous tree (reference: route_tree rip-up, occ -1) ----
     int32_t old_len = trees.len[inet];
-    for (int k = tid; k < old_len; k += WG_THREADS)
-      atomicSub(&occ[t_node[k]], 1);
+#ifdef PNR_DEBUG_BOUNDS
+    if (tid == 0 && (old_len < 0 || old_len > tcap)) {
+      fail_flags[inet] = 97;
+      printf("rip-up: net %d len %d cap %d\n", inet, old_len, tcap);
+    }
+#endif
+    for (int k = tid; k < old_len && k < tcap; k += WG_THREADS) {
+      int32_t rv = t_node[k];
+#ifdef PNR_DEBUG_BOUNDS
+      if (rv < 0 || rv >= g.num_nodes) {
+        fail_flags[inet] = 98;
+        printf("rip-up: net %d k %d bad node %d\n", inet, k, rv);
+        continue;
+      }
+#endif
+      atomicSub(&occ[rv], 1);
+    }
     __syncthreads();
     // ---- new tree root ----
     if (tid == 0) {
@@ -267,12 +282,15 @@ __global__ void route_nets_kernel(
 
       while (!sh.fail) {
         if (n_cur == 0) { if (tid == 0 && sh.best_sink_back == 0xffffffffu) sh.fail = FAIL_NO_PATH; break; }
-        // terminate when the sink's settled cost beats the min frontier f
-        // STRICTLY: the == bucket must still be processed so that
-        // equal-cost tie-breaks reach their fixpoint — with an admissible
-        // heuristic this makes the search result order-independent
-        // (deterministic mode relies on it; see docs/PARITY.md §2.3)
-        if (sh.best_sink_back != 0xffffffffu && sh.best_sink_back < fmin) break;
+        // terminate when the sink's settled cost beats the min frontier f.
+        // Deterministic mode uses STRICT <: the == bucket must be fully
+        // processed so equal-cost tie-breaks reach their fixpoint (with an
+        // admissible heuristic the result is then order-independent).
+        // Normal mode uses <= — at pres_fac 0 the fabric is full of exact
+        // cost ties and processing every tie plateau explodes the frontier.
+        if (sh.best_sink_back != 0xffffffffu &&
+            (P.strict_term ? (sh.best_sink_back < fmin)
+                           : (sh.best_sink_back <= fmin))) break;
         if (++rounds > P.max_rounds) { if (tid == 0) sh.fail = FAIL_ROUNDS; break; }
         const float thr = bits_f32(fmin) + delta;
         const int nxt = cur ^ 1;
@@ -390,6 +408,13 @@ __global__ void route_nets_kernel(
           for (int k = sh.path_len - 1; k >= 0; --k) {
             int32_t u = sh.path[k];
             int32_t pu = t_node[parent];
+#ifdef PNR_DEBUG_BOUNDS
+            if (u < 0 || u >= g.num_nodes || pu < 0 || pu >= g.num_nodes) {
+              sh.fail = 99;
+              printf("commit: net %d bad u=%d pu=%d\n", inet, u, pu);
+              break;
+            }
+#endif
             // find the switch of edge pu->u
             int8_t sw = 0;
             for (int32_t ei = g.row_ptr[pu]; ei < g.row_ptr[pu + 1]; ++ei)
@@ -532,6 +557,7 @@ struct RouteLaunchArgs {
   float astar_fac, pres_fac, seg_delay, ipin_delay, seg_base, ipin_base;
   float delta_fac;
   int32_t max_rounds;
+  int32_t strict_term;
   // queues
   const int32_t* queue_small; int32_t n_queue_small;
   const int32_t* queue_large; int32_t n_queue_large;
@@ -561,6 +587,7 @@ int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
   P.seg_base = a->seg_base; P.ipin_base = a->ipin_base;
   P.delta_fac = a->delta_fac;
   P.max_rounds = a->max_rounds;
+  P.strict_term = a->strict_term;
   int grid = a->n_small_slots + a->n_large_slots;
   hipLaunchKernelGGL(route_nets_kernel, dim3(grid), dim3(WG_THREADS), 0,
                      (hipStream_t)stream,

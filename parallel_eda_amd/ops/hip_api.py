@@ -33,6 +33,7 @@ class RouteLaunchArgs(ct.Structure):
         ("seg_base", ct.c_float), ("ipin_base", ct.c_float),
         ("delta_fac", ct.c_float),
         ("max_rounds", ct.c_int32),
+        ("strict_term", ct.c_int32),
         # queues
         ("queue_small", ct.c_void_p), ("n_queue_small", ct.c_int32),
         ("queue_large", ct.c_void_p), ("n_queue_large", ct.c_int32),

@@ -395,6 +395,9 @@ class GpuRouter:
         a.seg_base = self.seg_base; a.ipin_base = self.ipin_base
         a.delta_fac = self.delta_fac
         a.max_rounds = self.max_rounds
+        import os as _os
+        a.strict_term = 1 if (self.deterministic or
+                              _os.environ.get("PNR_FORCE_STRICT")) else 0
         a.queue_small = ct_ptr(q_small); a.n_queue_small = len(q_small)
         a.queue_large = ct_ptr(q_large); a.n_queue_large = len(q_large)
         a.q_cursors = ct_ptr(self.t_cursors)
