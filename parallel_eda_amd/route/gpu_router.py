@@ -250,7 +250,10 @@ class GpuRouter:
             # family); small/endgame sets get the deterministic bb-disjoint
             # wave schedule (ParaDRo family), which is what resolves the
             # last contested nodes.
-            if not self.deterministic and (
+            import os as _os
+            if attempts > 0 and _os.environ.get("PNR_RETRY_SERIAL"):
+                waves = [np.asarray([n]) for n in todo]   # bisection mode
+            elif not self.deterministic and (
                     len(todo) > self.concurrent_threshold or attempts > 0):
                 # one concurrent launch; biggest work first for load balance
                 areas_t = self._bb_areas(self.bb)[todo]
@@ -271,7 +274,9 @@ class GpuRouter:
                                        q_small, q_large, pres_fac)
                 if dbg:
                     print(f"    [launch] attempt={attempts} wave={wi}/{len(waves)} "
-                          f"small={len(small)} large={len(large)}", flush=True)
+                          f"small={len(small)} large={len(large)} "
+                          f"nets={wave[:4].tolist() if len(wave) <= 8 else ''}",
+                          flush=True)
                 rc = self.lib.pnr_route_nets(hip_api.ct.byref(args), self._stream())
                 hip_api.check(rc, "route_nets")
                 if dbg:
