@@ -30,11 +30,15 @@ namespace pnrh {
 
 #define INF_STATE 0xffffffffffffffffull
 
+// Debug bounds checks must be CONTROL-FLOW NEUTRAL: they clamp the index
+// and record a bit in sh.dbg (read only at net end). Writing sh.fail from
+// arbitrary threads mid-round would diverge the round loop's exit and
+// corrupt the workgroup at the next barrier.
 #ifdef PNR_DEBUG_BOUNDS
 #define DBG_LI(li, cap, code) \
-  (((li) < 0 || (li) >= (cap)) ? (atomicExch(&sh.fail, (code)), (int64_t)0) : (li))
+  (((li) < 0 || (li) >= (cap)) ? (atomicOr(&sh.dbg, 1 << ((code) - 90)), (int64_t)0) : (li))
 #define DBG_NODE(v, code) \
-  (((v) < 0 || (v) >= g.num_nodes) ? (atomicExch(&sh.fail, (code)), 0) : (v))
+  (((v) < 0 || (v) >= g.num_nodes) ? (atomicOr(&sh.dbg, 1 << ((code) - 90)), 0) : (v))
 #else
 #define DBG_LI(li, cap, code) (li)
 #define DBG_NODE(v, code) (v)
@@ -106,6 +110,7 @@ struct LocalIdx {
 };
 
 struct WgShared {
+  int dbg;
   int fcnt[2];
   unsigned fmin_next;
   unsigned best_sink_back;
@@ -196,16 +201,13 @@ __global__ void route_nets_kernel(
     // ---- rip-up previous tree (reference: route_tree rip-up, occ -1) ----
     int32_t old_len = trees.len[inet];
 #ifdef PNR_DEBUG_BOUNDS
-    if (tid == 0 && (old_len < 0 || old_len > tcap)) {
-      fail_flags[inet] = 97;
+    if (tid == 0 && (old_len < 0 || old_len > tcap))
       printf("rip-up: net %d len %d cap %d\n", inet, old_len, tcap);
-    }
 #endif
     for (int k = tid; k < old_len && k < tcap; k += WG_THREADS) {
       int32_t rv = t_node[k];
 #ifdef PNR_DEBUG_BOUNDS
       if (rv < 0 || rv >= g.num_nodes) {
-        fail_flags[inet] = 98;
         printf("rip-up: net %d k %d bad node %d\n", inet, k, rv);
         continue;
       }
@@ -219,6 +221,7 @@ __global__ void route_nets_kernel(
       trees.len[inet] = 1;
       atomicAdd(&occ[src], 1);
       sh.fail = 0;
+      sh.dbg = 0;
     }
     __syncthreads();
     int tree_len = 1;
@@ -377,7 +380,7 @@ __global__ void route_nets_kernel(
           uint64_t st = state[(size_t)DBG_LI(L(g, v), s_cap, 93)];
           int32_t prev = (int32_t)(st & 0xffffffffu);
 #ifdef PNR_DEBUG_BOUNDS
-          if (sh.fail) break;
+          if (sh.dbg) { sh.fail = FAIL_NO_PATH; break; }  // tid0-only loop
 #endif
           if (prev == v) break;  // reached a tree seed
           if (n >= PATH_CAP) { sh.fail = FAIL_PATH_CAP; break; }
@@ -410,7 +413,8 @@ __global__ void route_nets_kernel(
             int32_t pu = t_node[parent];
 #ifdef PNR_DEBUG_BOUNDS
             if (u < 0 || u >= g.num_nodes || pu < 0 || pu >= g.num_nodes) {
-              sh.fail = 99;
+              sh.fail = FAIL_NO_PATH;  // tid0-only loop: safe to abort
+              sh.dbg |= 1 << 9;
               printf("commit: net %d bad u=%d pu=%d\n", inet, u, pu);
               break;
             }
@@ -447,7 +451,10 @@ __global__ void route_nets_kernel(
       __syncthreads();
     }
 
-    if (sh.fail && tid == 0) fail_flags[inet] = sh.fail;
+    if (tid == 0) {
+      if (sh.dbg) fail_flags[inet] = 64 + sh.dbg;   // debug-detect bits
+      else if (sh.fail) fail_flags[inet] = sh.fail;
+    }
     __syncthreads();
     // if this net failed mid-sink, its touched entries were reset above only
     // on success; do a full reset of touched here for safety
