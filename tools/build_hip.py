@@ -40,7 +40,7 @@ def main():
     # PNR_HIP_DEBUG=1; see ops/__init__.py)
     dbg = OUT.with_name("libpnr_hip_dbg.so")
     cmd_dbg = [
-        hipcc, "--offload-arch=gfx950", "-O2", "-std=c++17",
+        hipcc, "--offload-arch=gfx950", "-O1", "-g", "-std=c++17",
         "-DPNR_DEBUG_BOUNDS", "-fPIC", "-shared", "-fvisibility=default",
         "-o", str(dbg),
     ] + [str(s) for s in srcs]
