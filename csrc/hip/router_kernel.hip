@@ -253,6 +253,14 @@ __global__ void route_nets_kernel(
       // from tree nodes inside bb, partitioning_multi_sink...:707-745)
       for (int k = tid; k < tree_len; k += WG_THREADS) {
         int32_t v = t_node[k];
+#ifdef PNR_DEBUG_BOUNDS
+        if (v < 0 || v >= g.num_nodes) {
+          atomicOr(&sh.dbg, 1 << 10);
+          printf("SEEDJUNK net=%d k=%d v=0x%08x tree_len=%d si=%d\n",
+                 inet, k, (unsigned)v, tree_len, (int)(si - s0));
+          continue;
+        }
+#endif
         if (g.type[v] == 1 /*SINK*/) continue;
         if (!L.in_bb(g, v)) continue;
         float back = S.crit * t_delay[k];
@@ -323,6 +331,14 @@ __global__ void route_nets_kernel(
           int32_t e0 = g.row_ptr[v], e1 = g.row_ptr[v + 1];
           for (int32_t ei = e0; ei < e1; ++ei) {
             int32_t w = g.edge_dst[ei];
+#ifdef PNR_DEBUG_BOUNDS
+            if (w < 0 || w >= g.num_nodes) {
+              atomicOr(&sh.dbg, 1 << 11);
+              printf("EDGEJUNK net=%d v=%d ei=%d e0=%d e1=%d w=0x%08x\n",
+                     inet, v, ei, e0, e1, (unsigned)w);
+              continue;
+            }
+#endif
             int8_t ty = g.type[w];
             if (ty == 1 && w != S.sink_node) continue;       // other SINK
             if (ty == 3 /*IPIN*/ && (g.xlow[w] != S.sx || g.ylow[w] != S.sy))

@@ -21,7 +21,7 @@ def _torch():
 class GpuRouter:
     def __init__(self, g, arch: ArchDef, src_rr, sink_ptr, sink_rr,
                  device="cuda:0", astar_fac=1.2, n_small_slots=1024,
-                 n_large_slots=64, bb_margin=3, max_rounds=200000,
+                 n_large_slots=64, bb_margin=4, max_rounds=200000,
                  delta_fac=3.0, deterministic=False,
                  concurrent_threshold=768, occ=None):
         torch = _torch()
@@ -104,7 +104,7 @@ class GpuRouter:
         if self.large_cap <= self.small_cap:
             self.large_cap = self.small_cap
 
-        self.f_cap_small = 1 << 16
+        self.f_cap_small = 1 << 17
         self.f_cap_large = 1 << 21
         self.t_cap_small = self.small_cap
         self.t_cap_large = self.large_cap
