@@ -54,6 +54,17 @@ __device__ __forceinline__ uint64_t pack_state(float back, int32_t prev) {
   return ((uint64_t)f32_bits(back) << 32) | (uint32_t)prev;
 }
 
+// state[] is updated with device-scope atomicMin, which executes at L2 and
+// BYPASSES the CU's vector L1. A plain load can hit a stale L1 line (e.g.
+// the INF written by the touched-list reset) and miss the atomic's value —
+// the backtrack then extracts prev = -1 and wild-walks (the intermittent
+// memory faults of profiles/README.md's fault-hunt log). Every read of
+// state[] therefore goes through an agent-scope atomic load (L1-bypassing).
+__device__ __forceinline__ uint64_t load_state(const uint64_t* p) {
+  return __hip_atomic_load((const unsigned long long*)p, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+}
+
 struct SinkCtx {
   int32_t sink_node;
   int16_t sx, sy;
@@ -318,7 +329,7 @@ __global__ void route_nets_kernel(
           // their entry carries the real back cost
           const uint64_t expect = (prev == v) ? pack_state(0.0f, v)
                                               : pack_state(back, prev);
-          if (state[li] != expect) continue;  // stale entry
+          if (load_state(&state[li]) != expect) continue;  // stale entry
           if (tot > thr) {
             // keep for a later bucket
             int fi = atomicAdd(&sh.fcnt[nxt], 1);
@@ -393,7 +404,9 @@ __global__ void route_nets_kernel(
         int32_t v = S.sink_node;
         for (;;) {
           v = DBG_NODE(v, 94);
-          uint64_t st = state[(size_t)DBG_LI(L(g, v), s_cap, 93)];
+          uint64_t st = load_state(&state[(size_t)DBG_LI(L(g, v), s_cap, 93)]);
+          // belt-and-braces: a torn/stale INF would wild-walk; stop instead
+          if (st == INF_STATE) { sh.fail = FAIL_NO_PATH; break; }
           int32_t prev = (int32_t)(st & 0xffffffffu);
 #ifdef PNR_DEBUG_BOUNDS
           if (sh.dbg) { sh.fail = FAIL_NO_PATH; break; }  // tid0-only loop
