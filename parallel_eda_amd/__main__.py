@@ -13,6 +13,8 @@ import argparse
 import sys
 import time
 
+import numpy as np
+
 
 def build_parser():
     p = argparse.ArgumentParser(prog="parallel_eda_amd")
@@ -45,6 +47,10 @@ def build_parser():
     p.add_argument("--deterministic", action="store_true",
                    help="force the fixed wave schedule (GPU)")
     # outputs
+    p.add_argument("--sdc", type=str, default=None,
+                   help="SDC constraints (create_clock -period, subset)")
+    p.add_argument("--timing_report", type=str, default=None,
+                   help="write a critical-path report here after routing")
     p.add_argument("--out_place", type=str, default=None)
     p.add_argument("--out_route", type=str, default=None)
     p.add_argument("--stats_dir", type=str, default=None)
@@ -91,6 +97,12 @@ def main(argv=None):
 
     timing = args.router_algorithm == "timing_driven"
     sta = STA(netlist, arch) if timing else None
+    if args.sdc:
+        from .timing.report import parse_sdc
+        with open(args.sdc) as f:
+            period = parse_sdc(f.read())
+        if period:
+            print(f"SDC: target clock period {period*1e9:.3f} ns")
 
     # ---- placement ----
     t0 = time.perf_counter()
@@ -148,6 +160,18 @@ def main(argv=None):
             write_route(args.out_route, g, arch, net_ids,
                         lambda k: res.router.tree(k), netlist=netlist)
             print(f"wrote {args.out_route}")
+        if args.timing_report and sta is not None:
+            from .route.router import ConnMap
+            from .timing.report import write_timing_report
+            _, src_rr2, sink_ptr2, sink_rr2, ci2 = net_rr_terminals(
+                netlist, placement, g, arch)
+            cmap = ConnMap(ci2, sink_ptr2, netlist.num_conns, len(sink_rr2))
+            sd = np.asarray(res.router.sink_delays()) \
+                if hasattr(res.router, "sink_delays") \
+                else res.router.t_sink_delay.cpu().numpy()
+            conn_delay = cmap.conn_delays(sd)
+            write_timing_report(args.timing_report, netlist, sta, conn_delay)
+            print(f"wrote {args.timing_report}")
     print(f"entire flow took {time.perf_counter()-t_start:.2f}s")
     return 0
 

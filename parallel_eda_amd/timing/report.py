@@ -1,0 +1,86 @@
+"""Timing reports: critical-path extraction + echo files.
+
+Reference surface: the timing echo files and critical-path printout
+(vpr/SRC/timing/path_delay.c print_critical_path / echo system,
+base/place_and_route.c:184-198). Works over the block-level timing graph.
+"""
+import numpy as np
+
+
+def critical_paths(netlist, sta, conn_delay, k=5):
+    """Extract the k most critical register-to-register paths.
+
+    Returns a list of paths; each path is a list of
+    (block, arrival_time, via_conn_delay) hops ending at a timing endpoint.
+    """
+    cpd, slack, crit = sta.analyze(conn_delay)
+    nd = netlist.net_driver
+    sp = netlist.net_sink_ptr
+    ss = netlist.net_sinks
+    nconn = netlist.num_conns
+    net_of_conn = np.repeat(np.arange(netlist.num_nets), np.diff(sp))
+
+    # arrival times per block: recompute forward (same as STA)
+    is_seq = netlist.block_is_seq.astype(bool)
+    # per-block best incoming conn (the argmax arrival feeding it)
+    order = np.argsort(slack, kind="stable")
+    paths = []
+    seen_endpoints = set()
+    for c in order[: 50 * k]:
+        snk = int(ss[c])
+        if not is_seq[snk]:
+            continue
+        if snk in seen_endpoints:
+            continue
+        seen_endpoints.add(snk)
+        # walk backwards along most-critical in-edges
+        path = [(snk, None, None)]
+        cur = int(nd[net_of_conn[c]])
+        path.append((cur, float(conn_delay[c]), int(net_of_conn[c])))
+        guard = 0
+        while not is_seq[cur] and guard < 10000:
+            guard += 1
+            # most critical incoming connection of cur
+            in_conns = np.nonzero(ss == cur)[0]
+            if len(in_conns) == 0:
+                break
+            best = in_conns[np.argmin(slack[in_conns])]
+            cur = int(nd[net_of_conn[best]])
+            path.append((cur, float(conn_delay[best]), int(net_of_conn[best])))
+        paths.append({"endpoint_slack": float(slack[c]),
+                      "hops": list(reversed(path))})
+        if len(paths) >= k:
+            break
+    return cpd, paths
+
+
+def write_timing_report(path, netlist, sta, conn_delay, k=5):
+    cpd, paths = critical_paths(netlist, sta, conn_delay, k=k)
+    names = netlist.names or [f"blk_{i}" for i in range(netlist.num_blocks)]
+    with open(path, "w") as f:
+        f.write(f"Critical path delay: {cpd*1e9:.4f} ns "
+                f"(fmax {1e-6/cpd:.2f} MHz)\n\n")
+        for i, p in enumerate(paths):
+            f.write(f"Path {i}: endpoint slack {p['endpoint_slack']*1e9:.4f} ns\n")
+            for blk, d, net in p["hops"]:
+                if d is None:
+                    f.write(f"  {names[blk]}  (endpoint)\n")
+                else:
+                    f.write(f"  {names[blk]}  -> net {net} "
+                            f"(+{d*1e9:.4f} ns)\n")
+            f.write("\n")
+    return cpd
+
+
+def parse_sdc(text):
+    """Minimal SDC subset (reference: timing/read_sdc.c:115):
+    create_clock -period P [-name N]; returns the target period in seconds
+    (ns units in the file, like VPR) or None."""
+    import re
+    period = None
+    for line in text.splitlines():
+        line = line.split("#", 1)[0]
+        m = re.search(r"create_clock\s+.*-period\s+([0-9.eE+-]+)", line)
+        if m:
+            period = float(m.group(1)) * 1e-9
+    return period

@@ -142,3 +142,19 @@ def test_cli_synth_flow(tmp_path):
     assert (tmp_path / "o.place").exists()
     assert (tmp_path / "o.route").exists()
     assert (tmp_path / "stats" / "final_stats.txt").exists()
+
+
+def test_cli_blif_xml_flow(tmp_path):
+    """End-to-end: BLIF + arch.xml through the CLI (pack/place/route)."""
+    from parallel_eda_amd.__main__ import main
+    blif = tmp_path / "c.blif"
+    blif.write_text(BLIF)
+    xml = tmp_path / "a.xml"
+    xml.write_text(ARCH_XML)
+    rpt = tmp_path / "t.rpt"
+    sdc = tmp_path / "c.sdc"
+    sdc.write_text("create_clock -period 10.0 -name clk\n")
+    rc = main([str(blif), str(xml), "--route_chan_width", "20",
+               "--sdc", str(sdc), "--timing_report", str(rpt)])
+    assert rc == 0
+    assert "Critical path delay" in rpt.read_text()

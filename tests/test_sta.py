@@ -52,3 +52,25 @@ def test_sta_hand_case():
     # single path: every connection critical, slack ~ 0
     assert np.allclose(crit, 1.0, atol=1e-5)
     assert np.allclose(slack, 0.0, atol=1e-12)
+
+
+def test_timing_report_and_sdc(tmp_path):
+    from parallel_eda_amd.timing.report import (write_timing_report,
+                                                critical_paths, parse_sdc)
+    arch = get_arch("tiny")
+    nl = synth_netlist(SynthSpec(n_clb=40, n_in=4, n_out=4, seed=5))
+    nl.names = [f"b{i}" for i in range(nl.num_blocks)]
+    sta = STA(nl, arch)
+    rng = np.random.default_rng(1)
+    d = (rng.random(nl.num_conns) * 1e-9).astype(np.float32)
+    cpd, paths = critical_paths(nl, sta, d, k=3)
+    assert cpd > 0 and len(paths) >= 1
+    # most critical path's endpoint slack ~ 0
+    assert abs(paths[0]["endpoint_slack"]) < cpd * 0.05
+    p = tmp_path / "timing.rpt"
+    write_timing_report(p, nl, sta, d)
+    text = p.read_text()
+    assert "Critical path delay" in text and "Path 0" in text
+    assert parse_sdc("create_clock -period 5.0 -name clk [get_ports clk]\n"
+                     ) == pytest.approx(5e-9)
+    assert parse_sdc("# nothing\n") is None
