@@ -7,6 +7,7 @@ dir creation, :5926-5932 iter stats, :6347-6362 final stats), and the
 wirelength/bends/channel-occupancy report of base/stats.c:27
 routing_stats_new / :355 get_num_bends_and_length.
 """
+import hashlib
 import json
 import os
 import time
@@ -55,6 +56,39 @@ class StatsWriter:
         with open(os.path.join(self.dir, "final_stats.json"), "w") as f:
             json.dump(data, f, indent=1)
         return data
+
+
+def routing_serial_num(net_ids, tree_fn):
+    """Cross-run routing fingerprint (reference: get_serial_num,
+    route_common.c — a hash of every net's traceback, printed so two runs
+    can be diffed without storing .route files)."""
+    h = hashlib.sha256()
+    for k in range(len(net_ids)):
+        nodes, parents, sws, delays = tree_fn(k)
+        h.update(np.ascontiguousarray(nodes, dtype=np.int32).tobytes())
+        h.update(np.ascontiguousarray(parents, dtype=np.int32).tobytes())
+    return h.hexdigest()[:16]
+
+
+def mem_usage_mb():
+    """Host RSS in MiB (reference: main.c:287 get_mem_usage via
+    /proc/self/statm)."""
+    try:
+        with open("/proc/self/statm") as f:
+            pages = int(f.read().split()[1])
+        return pages * os.sysconf("SC_PAGE_SIZE") / (1 << 20)
+    except Exception:
+        return -1.0
+
+
+def gpu_mem_usage_mb(device=None):
+    try:
+        import torch
+        if not torch.cuda.is_available():
+            return -1.0
+        return torch.cuda.memory_allocated(device) / (1 << 20)
+    except Exception:
+        return -1.0
 
 
 def routing_stats(g, arch, net_ids, tree_fn):

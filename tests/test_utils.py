@@ -57,3 +57,24 @@ def test_min_channel_width():
         g2 = rrgraph.build_rr_graph(a2)
         res2 = pathfinder_route(nl, pl, g2, a2, sta=None, max_iters=30)
         assert not res2.success
+
+
+def test_serial_num_and_mem(tiny_routed):
+    from parallel_eda_amd.utils.stats import routing_serial_num, mem_usage_mb
+    arch, nl, pl, g, res, net_ids = tiny_routed
+    s1 = routing_serial_num(net_ids, lambda k: res.router.tree(k))
+    s2 = routing_serial_num(net_ids, lambda k: res.router.tree(k))
+    assert s1 == s2 and len(s1) == 16
+    assert mem_usage_mb() > 10
+
+
+def test_verilog_writer(tmp_path, tiny_routed):
+    from parallel_eda_amd.io.verilog import write_verilog
+    arch, nl, pl, g, res, net_ids = tiny_routed
+    nl.names = [f"b{i}" for i in range(nl.num_blocks)]
+    p = tmp_path / "out.v"
+    write_verilog(p, nl)
+    text = p.read_text()
+    assert text.startswith("module")
+    assert "endmodule" in text
+    assert text.count("wire ") == nl.num_nets
