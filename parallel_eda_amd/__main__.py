@@ -150,8 +150,11 @@ def main(argv=None):
               f"crit_path={res.crit_path_delay*1e9:.3f}ns")
         net_ids, *_ = net_rr_terminals(netlist, placement, g, arch)
         st = routing_stats(g, arch, net_ids, lambda k: res.router.tree(k))
+        from .utils.stats import routing_serial_num, mem_usage_mb
+        serial = routing_serial_num(net_ids, lambda k: res.router.tree(k))
         print(f"  segments={st['total_segments']} bends={st['total_bends']} "
               f"avg_wl/net={st['avg_wirelength_per_net']:.1f}")
+        print(f"  serial_num {serial}  host_mem {mem_usage_mb():.0f} MiB")
         if sw:
             for h in res.stats["history"]:
                 sw.iteration(h["iter"], h["overused"], cpd=h.get("cpd", 0.0))

@@ -455,9 +455,12 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
             subset = router.congested_nets()
             if len(subset) == 0:
                 subset = None
+        router.reset_search_stats()
         overused, sink_delays = router.route_iteration(crit, pres_fac,
                                                        net_subset=subset)
+        st = router.search_stats()
         history.append(dict(iter=it, overused=int(overused), cpd=cpd,
+                            rounds=st["rounds"], scanned=st["scanned"],
                             rerouted=len(subset) if subset is not None
                             else router.num_nets))
         if verbose:
