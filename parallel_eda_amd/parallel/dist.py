@@ -110,7 +110,8 @@ class GpuEngine:
         self._last_sd = np.zeros(router.n_sinks, dtype=np.float32)
 
     def route_subset(self, crit, pres_fac, net_ids):
-        _, sd = self.g.route_iteration(crit, pres_fac, net_subset=net_ids)
+        _, sd = self.g.route_iteration(crit, pres_fac, net_subset=net_ids,
+                                       fail_ok=True)
         self._last_sd = sd
 
     def occ_tensor(self):

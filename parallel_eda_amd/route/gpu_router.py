@@ -226,11 +226,14 @@ class GpuRouter:
         return waves
 
     # ---- one PathFinder iteration ----
-    def route_iteration(self, crit, pres_fac, net_subset=None):
+    def route_iteration(self, crit, pres_fac, net_subset=None, fail_ok=False):
         """crit: per-sink criticality aligned with sink_rr (original order).
         Returns (overused_count, sink_delays aligned with original order).
         net_subset: optional array of net ids to (re)route; others keep
-        their route trees (multi-GPU partitioning / selective reroute)."""
+        their route trees (multi-GPU partitioning / selective reroute).
+        fail_ok: exhausted retries leave the stragglers partially routed
+        instead of raising (multi-rank benches must not kill a rank
+        mid-collective)."""
         torch = self.torch
         # order sinks by decreasing criticality within each net
         net_of_sink = np.repeat(np.arange(self.num_nets), np.diff(self.sink_ptr))
@@ -292,6 +295,12 @@ class GpuRouter:
             self.last_retries.append(
                 (len(failed), np.unique(fail[failed]).tolist()))
             if attempts > 6:
+                if fail_ok:
+                    import sys
+                    print(f"router: giving up on {len(failed)} nets after "
+                          f"retries (codes {np.unique(fail[failed])})",
+                          file=sys.stderr, flush=True)
+                    break
                 raise RuntimeError(
                     f"router: {len(failed)} nets failed after retries "
                     f"(codes {np.unique(fail[failed])})")
