@@ -70,6 +70,9 @@ struct RouteParams {
   float pres_fac;
   float seg_delay, ipin_delay, seg_base, ipin_base;
   float delta_fac;           // bucket width in edge-step cost units
+  float cong_mult;           // lookahead congestion-cost scale (mean used-
+                             // wire cost); keeps A* focused when pres/acc
+                             // inflate real edge costs far above base
   int32_t max_rounds;        // safety bound on delta-stepping rounds
   int32_t strict_term;       // deterministic mode: process the == bucket
 };

@@ -83,7 +83,7 @@ __device__ __forceinline__ float expected_cost(const RRDev& g, const RouteParams
   if (yl > ty2) dy = yl - ty2; else if (yh < ty2) dy = ty2 - yh;
   int dist = dx + dy;
   int nseg = (dist + g.L - 1) / g.L;
-  float cong = nseg * P.seg_base + P.ipin_base;
+  float cong = nseg * P.seg_base * P.cong_mult + P.ipin_base;
   float del = nseg * P.seg_delay + P.ipin_delay;
   return S.crit * del + (1.0f - S.crit) * cong;
 }
@@ -592,6 +592,7 @@ struct RouteLaunchArgs {
   // params
   float astar_fac, pres_fac, seg_delay, ipin_delay, seg_base, ipin_base;
   float delta_fac;
+  float cong_mult;
   int32_t max_rounds;
   int32_t strict_term;
   // queues
@@ -622,6 +623,7 @@ int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
   P.seg_delay = a->seg_delay; P.ipin_delay = a->ipin_delay;
   P.seg_base = a->seg_base; P.ipin_base = a->ipin_base;
   P.delta_fac = a->delta_fac;
+  P.cong_mult = a->cong_mult < 1.0f ? 1.0f : a->cong_mult;
   P.max_rounds = a->max_rounds;
   P.strict_term = a->strict_term;
   int grid = a->n_small_slots + a->n_large_slots;

@@ -32,6 +32,7 @@ class RouteLaunchArgs(ct.Structure):
         ("seg_delay", ct.c_float), ("ipin_delay", ct.c_float),
         ("seg_base", ct.c_float), ("ipin_base", ct.c_float),
         ("delta_fac", ct.c_float),
+        ("cong_mult", ct.c_float),
         ("max_rounds", ct.c_int32),
         ("strict_term", ct.c_int32),
         # queues

@@ -246,6 +246,24 @@ class GpuRouter:
                 else np.asarray(net_subset, dtype=np.int64))
         attempts = 0
         self.last_retries = []
+        # congestion-aware lookahead: scale the heuristic's wire cost by
+        # the mean effective cost of used wires so A* stays focused when
+        # pres/acc inflate edge costs far beyond base (keeps the estimate
+        # admissible-ish w.r.t. congested regions; disable with
+        # cong_aware_lookahead=False)
+        if getattr(self, "cong_aware_lookahead", True) and pres_fac > 0:
+            t = self.torch
+            chan = (self.t_type == 4) | (self.t_type == 5)
+            used = chan & (self.t_occ > 0)
+            if bool(used.any().item()):
+                over = (self.t_occ + 1 - self.t_cap.to(t.int32)).clamp(min=0)
+                pres_t = 1.0 + pres_fac * over.to(t.float32)
+                m = float((self.t_acc * pres_t)[used].mean().item())
+                self._cong_mult = float(np.clip(m, 1.0, 64.0))
+            else:
+                self._cong_mult = 1.0
+        else:
+            self._cong_mult = 1.0
         while True:
             self.t_fail.zero_()
             # Large reroute sets run as ONE concurrent launch (net-level
@@ -408,6 +426,7 @@ class GpuRouter:
         a.seg_delay = self.seg_delay; a.ipin_delay = self.ipin_delay
         a.seg_base = self.seg_base; a.ipin_base = self.ipin_base
         a.delta_fac = self.delta_fac
+        a.cong_mult = getattr(self, "_cong_mult", 1.0)
         a.max_rounds = self.max_rounds
         import os as _os
         a.strict_term = 1 if (self.deterministic or
