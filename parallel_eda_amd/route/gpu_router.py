@@ -251,7 +251,11 @@ class GpuRouter:
         # pres/acc inflate edge costs far beyond base (keeps the estimate
         # admissible-ish w.r.t. congested regions; disable with
         # cong_aware_lookahead=False)
-        if getattr(self, "cong_aware_lookahead", True) and pres_fac > 0:
+        # A/B at LU32 measured 5.5x WORSE with this on (scan/sink 45k ->
+        # 430k): inflating h without scaling the bucket width explodes the
+        # kept-entry rescans. Off by default; revisit with the bucketed
+        # frontier (docs/ROADMAP.md item 1).
+        if getattr(self, "cong_aware_lookahead", False) and pres_fac > 0:
             t = self.torch
             chan = (self.t_type == 4) | (self.t_type == 5)
             used = chan & (self.t_occ > 0)
