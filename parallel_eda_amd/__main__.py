@@ -33,6 +33,10 @@ def build_parser():
     p.add_argument("--place_file", type=str, default=None,
                    help="read placement instead of annealing")
     p.add_argument("--timing_tradeoff", type=float, default=0.5)
+    p.add_argument("--delay_matrix", choices=["analytic", "routed"],
+                   default="analytic",
+                   help="placer delay LUT: analytic Elmore or router-measured"
+                        " (reference: compute_delay_lookup_tables)")
     p.add_argument("--inner_num", type=float, default=1.0)
     # router opts (reference: s_router_opts vpr_types.h:724-770)
     p.add_argument("--route_chan_width", type=int, default=None)
@@ -114,7 +118,7 @@ def main(argv=None):
             netlist, arch, seed=args.seed,
             timing_tradeoff=args.timing_tradeoff if timing else 0.0,
             inner_num=args.inner_num, sta=sta, verbose=args.verbose,
-            engine=args.engine)
+            engine=args.engine, delay_matrix=args.delay_matrix)
         print(f"placement: bb_cost={placement.bb_cost:.1f} "
               f"({time.perf_counter()-t0:.2f}s)")
     if args.out_place:

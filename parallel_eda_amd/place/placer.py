@@ -45,7 +45,8 @@ def analytic_delay_matrix(arch: ArchDef) -> np.ndarray:
 
 def anneal_place(netlist, arch: ArchDef, seed: int = 7, timing_tradeoff: float = 0.5,
                  inner_num: float = 1.0, sta=None, crit_exp: float = 1.0,
-                 verbose: bool = False, engine: str = "cpu") -> Placement:
+                 verbose: bool = False, engine: str = "cpu",
+                 delay_matrix: str = "analytic") -> Placement:
     """Run the full SA schedule; returns final Placement.
 
     sta: optional TimingGraph wrapper (timing.sta.STA) for criticality
@@ -58,8 +59,14 @@ def anneal_place(netlist, arch: ArchDef, seed: int = 7, timing_tradeoff: float =
                                 inner_num=inner_num, sta=sta,
                                 crit_exp=crit_exp, verbose=verbose)
     cpu = ops.cpu()
-    dm = analytic_delay_matrix(arch) if (sta is not None and timing_tradeoff > 0) \
-        else np.zeros(0, dtype=np.float32)
+    if sta is not None and timing_tradeoff > 0:
+        if delay_matrix == "routed":
+            from .delay_matrix import routed_delay_matrix
+            dm = routed_delay_matrix(arch)
+        else:
+            dm = analytic_delay_matrix(arch)
+    else:
+        dm = np.zeros(0, dtype=np.float32)
     placer = cpu.SerialPlacer(netlist.cpp(), arch.nx, arch.ny, arch.io_cap,
                               np.ascontiguousarray(dm.ravel()), seed)
     nb = netlist.num_blocks

@@ -57,3 +57,29 @@ def test_timing_driven_anneal(tiny_setup):
     dm = analytic_delay_matrix(arch)
     assert dm.shape == (arch.nx + 2, arch.ny + 2)
     assert dm[0, 0] > 0 and dm[3, 3] > dm[1, 1]
+
+
+def test_routed_delay_matrix():
+    """Router-measured delay LUT (compute_delay_lookup_tables analogue):
+    monotone in distance and consistent with the analytic Elmore model."""
+    from parallel_eda_amd.place.delay_matrix import routed_delay_matrix
+    arch = get_arch("tiny")
+    dm = routed_delay_matrix(arch)
+    assert dm.shape == (arch.nx + 2, arch.ny + 2)
+    assert (dm > 0).all()
+    # roughly monotone along the diagonal
+    diag = [dm[i, i] for i in range(arch.nx)]
+    assert diag[-1] > diag[0]
+    # within a small factor of the analytic model at mid distance
+    am = analytic_delay_matrix(arch)
+    mid = arch.nx // 2
+    ratio = dm[mid, mid] / am[mid, mid]
+    assert 0.3 < ratio < 3.0, f"routed/analytic ratio {ratio}"
+
+
+def test_timing_anneal_with_routed_matrix(tiny_setup):
+    arch, nl = tiny_setup
+    sta = STA(nl, arch)
+    pl = anneal_place(nl, arch, seed=7, timing_tradeoff=0.5, sta=sta,
+                      delay_matrix="routed")
+    assert pl.td_cost > 0
