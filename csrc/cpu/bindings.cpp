@@ -239,6 +239,17 @@ PYBIND11_MODULE(_pnr_cpu, m) {
       .def("level_of", [](TimingGraph& t) {
         return py::array_t<int32_t>((py::ssize_t)t.level_of().size(), t.level_of().data());
       })
+      .def("analyze_domains", [](TimingGraph& t,
+                                 py::array_t<float, py::array::c_style | py::array::forcecast> conn_delay,
+                                 py::array_t<int32_t, py::array::c_style | py::array::forcecast> block_clock,
+                                 py::array_t<float, py::array::c_style | py::array::forcecast> periods) {
+        py::ssize_t n = conn_delay.size();
+        py::array_t<float> slack(n), crit(n);
+        float wp = t.analyze_domains(conn_delay.data(), block_clock.data(),
+                                     periods.data(), (int)periods.size(),
+                                     slack.mutable_data(), crit.mutable_data());
+        return py::make_tuple(wp, slack, crit);
+      })
       .def("level_arrays", [](TimingGraph& t) {
         std::vector<int32_t> blocks, start;
         t.level_arrays(blocks, start);

@@ -76,6 +76,93 @@ class TimingGraph {
   const std::vector<int32_t>& topo() const { return topo_; }
   const std::vector<int32_t>& level_of() const { return level_; }
 
+  // Multi-clock analysis (reference: do_timing_analysis_new's per
+  // (src_domain, sink_domain) pair loop, path_delay.c:1996-2085).
+  // block_clock: domain id per block (-1 for combinational); periods[K].
+  // Single-cycle setup constraint against the DESTINATION domain's period.
+  // Outputs: per-conn worst slack and criticality (max over pairs, each
+  // normalized by its pair's constraint). Returns the worst cpd/period
+  // ratio times its period (the binding domain's achieved period).
+  float analyze_domains(const float* conn_delay, const int32_t* block_clock,
+                        const float* periods, int K,
+                        float* slack, float* crit) {
+    int nb = nl_->num_blocks;
+    int64_t nconn = (int64_t)nl_->net_sinks.size();
+    const float NEG = -3.0e38f, POS = 3.0e38f;
+    std::vector<std::vector<float>> arr(K), req(K);
+    // forward per source domain
+    for (int ci = 0; ci < K; ++ci) {
+      auto& a = arr[ci];
+      a.assign(nb, NEG);
+      for (int b : topo_) {
+        if (nl_->block_is_seq[b]) {
+          a[b] = (block_clock[b] == ci) ? T_seq_out_ : NEG;
+          continue;
+        }
+        float m = NEG;
+        for (int64_t k = in_ptr_[b]; k < in_ptr_[b + 1]; ++k) {
+          int64_t c = in_conn_[k];
+          float v = a[conn_driver_[c]];
+          if (v > NEG) v += conn_delay[c];
+          if (v > m) m = v;
+        }
+        a[b] = (m > NEG) ? m + T_clb_ : NEG;
+      }
+    }
+    // backward per sink domain (required time at block outputs)
+    for (int cj = 0; cj < K; ++cj) {
+      auto& r = req[cj];
+      r.assign(nb, POS);
+      float req_ep = periods[cj] - T_seq_in_;
+      for (auto it = topo_.rbegin(); it != topo_.rend(); ++it) {
+        int b = *it;
+        float m = POS;
+        for (int64_t k = out_ptr_[b]; k < out_ptr_[b + 1]; ++k) {
+          int64_t c = out_conn_[k];
+          int snk = nl_->net_sinks[c];
+          float ri;
+          if (nl_->block_is_seq[snk])
+            ri = (block_clock[snk] == cj) ? req_ep : POS;
+          else
+            ri = (r[snk] < POS) ? r[snk] - T_clb_ : POS;
+          if (ri < POS) ri -= conn_delay[c];
+          if (ri < m) m = ri;
+        }
+        r[b] = m;
+      }
+    }
+    for (int64_t c = 0; c < nconn; ++c) { slack[c] = POS; crit[c] = 0.0f; }
+    float worst_ratio = 0.0f;  // achieved/required; >1 means violated
+    float worst_period = periods[0];
+    for (int ci = 0; ci < K; ++ci)
+      for (int cj = 0; cj < K; ++cj) {
+        float constraint = periods[cj];
+        for (int64_t c = 0; c < nconn; ++c) {
+          int drv = conn_driver_[c];
+          int snk = nl_->net_sinks[c];
+          if (arr[ci][drv] <= NEG) continue;
+          float ri;
+          if (nl_->block_is_seq[snk])
+            ri = (block_clock[snk] == cj) ? constraint - T_seq_in_ : POS;
+          else
+            ri = (req[cj][snk] < POS) ? req[cj][snk] - T_clb_ : POS;
+          if (ri >= POS) continue;
+          float s = ri - (arr[ci][drv] + conn_delay[c]);
+          if (s < slack[c]) slack[c] = s;
+          float cr = 1.0f - s / constraint;
+          if (cr > crit[c]) crit[c] = cr < 0 ? 0.0f : (cr > 1 ? 1.0f : cr);
+          float achieved = constraint - s;
+          if (achieved / constraint > worst_ratio) {
+            worst_ratio = achieved / constraint;
+            worst_period = achieved;
+          }
+        }
+      }
+    for (int64_t c = 0; c < nconn; ++c)
+      if (slack[c] >= POS) slack[c] = 0.0f;
+    return worst_period;
+  }
+
   // GPU upload accessors: blocks sorted by level + CSRs
   void level_arrays(std::vector<int32_t>& blocks,
                     std::vector<int32_t>& start) const {

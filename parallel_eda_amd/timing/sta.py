@@ -29,3 +29,15 @@ class STA:
         conn_delay = np.ascontiguousarray(conn_delay, dtype=np.float32)
         cpd, slack, crit = self.tg.analyze(conn_delay)
         return float(cpd), np.asarray(slack), np.asarray(crit)
+
+    def analyze_domains(self, conn_delay, block_clock, periods):
+        """Multi-clock analysis (reference: do_timing_analysis_new per
+        (src,sink)-domain pairs). block_clock: per-block domain id (-1
+        comb); periods: seconds per domain. Returns
+        (worst_achieved_period, slack[], crit[]) — slack/crit are the
+        worst/max over domain pairs."""
+        conn_delay = np.ascontiguousarray(conn_delay, dtype=np.float32)
+        bc = np.ascontiguousarray(block_clock, dtype=np.int32)
+        pr = np.ascontiguousarray(periods, dtype=np.float32)
+        wp, slack, crit = self.tg.analyze_domains(conn_delay, bc, pr)
+        return float(wp), np.asarray(slack), np.asarray(crit)

@@ -74,3 +74,33 @@ def test_timing_report_and_sdc(tmp_path):
     assert parse_sdc("create_clock -period 5.0 -name clk [get_ports clk]\n"
                      ) == pytest.approx(5e-9)
     assert parse_sdc("# nothing\n") is None
+
+
+def test_multi_clock_domains():
+    """Two-domain hand case: in0(clkA) -> A(comb) -> ff1(clkA),
+    in1(clkB) -> B(comb) -> ff2(clkB), cross path A -> ff2."""
+    from parallel_eda_amd.io.synth import NetlistPy
+    arch = get_arch("tiny")
+    # blocks: 0=in0(A) 1=in1(B) 2=A comb 3=B comb 4=ff1(A) 5=ff2(B)
+    block_type = [0, 0, 1, 1, 1, 1]
+    block_is_seq = [1, 1, 0, 0, 1, 1]
+    # nets: 0: in0->A; 1: in1->B; 2: A->{ff1, ff2}; 3: B->ff2
+    driver = [0, 1, 2, 3]
+    sptr = [0, 1, 2, 4, 5]
+    sinks = [2, 3, 4, 5, 5]
+    nl = NetlistPy(block_type, block_is_seq, driver, sptr, sinks)
+    sta = STA(nl, arch)
+    block_clock = np.array([0, 1, -1, -1, 0, 1], dtype=np.int32)
+    periods = np.array([5e-9, 2e-9], dtype=np.float32)
+    d = np.full(5, 1e-9, dtype=np.float32)
+    wp, slack, crit = sta.analyze_domains(d, block_clock, periods)
+    # path in0->A->ff1 (A-domain): arrival = Tout + 1n + Tclb + 1n
+    arrA = arch.T_seq_out + 1e-9 + arch.T_clb + 1e-9
+    # conn 2 (A->ff1, clkA): slack = 5n - Tsu - arrA
+    assert slack[2] == pytest.approx(5e-9 - arch.T_seq_in - arrA, rel=1e-5)
+    # conn 3 (A->ff2, clkB): tighter 2n constraint
+    assert slack[3] == pytest.approx(2e-9 - arch.T_seq_in - arrA, rel=1e-5)
+    # crit of the cross-domain conn must exceed the same-domain one
+    assert crit[3] > crit[2]
+    # worst achieved period comes from the tight B constraint paths
+    assert wp == pytest.approx(arrA + arch.T_seq_in, rel=1e-5)
