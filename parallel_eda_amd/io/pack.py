@@ -103,10 +103,28 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
              [f"opad:{s}" for s in model.outputs] +
              [f"clb_{k}" for k in range(len(clusters))])
     clb0 = n_in + n_out
-    # a cluster is sequential if it contains any latch
+    # a cluster is sequential if it contains any latch; its clock domain
+    # is the (majority) latch clock. Clock signals are GLOBAL nets: latch
+    # clock pins are not data sinks, so clock-pad nets end up sinkless and
+    # are dropped from routing (VPR routes clocks on dedicated networks).
+    clock_names = []
+    clock_id = {}
+    block_clock = np.full(nb, -1, dtype=np.int32)
     for k, members in enumerate(clusters):
+        latch_clocks = [prims[i].clock for i in members
+                        if prims[i].kind == "latch" and prims[i].clock]
         if any(prims[i].kind == "latch" for i in members):
             block_is_seq[clb0 + k] = 1
+        if latch_clocks:
+            from collections import Counter
+            cname = Counter(latch_clocks).most_common(1)[0][0]
+            if cname not in clock_id:
+                clock_id[cname] = len(clock_names)
+                clock_names.append(cname)
+            block_clock[clb0 + k] = clock_id[cname]
+    # IO pads: assign to domain 0 if any clock exists
+    if clock_names:
+        block_clock[:n_in + n_out] = 0
 
     in_pad_of = {s: i for i, s in enumerate(model.inputs)}
     out_pad_of = {s: n_in + i for i, s in enumerate(model.outputs)}
@@ -146,6 +164,8 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
     nl = NetlistPy(block_type, block_is_seq,
                    np.asarray(drivers, dtype=np.int32), sink_ptr, net_sinks,
                    names=names)
+    nl.block_clock = block_clock
+    nl.clock_names = clock_names
     return nl, cluster_of, names
 
 

@@ -48,6 +48,8 @@ class NetlistPy:
         self.net_sink_ptr = np.asarray(net_sink_ptr, dtype=np.int64)
         self.net_sinks = np.asarray(net_sinks, dtype=np.int32)
         self.names = names
+        self.block_clock = None   # optional per-block clock domain ids
+        self.clock_names = []     # domain id -> clock name
         self._cpp = None
 
     @property

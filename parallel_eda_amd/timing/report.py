@@ -74,13 +74,29 @@ def write_timing_report(path, netlist, sta, conn_delay, k=5):
 
 def parse_sdc(text):
     """Minimal SDC subset (reference: timing/read_sdc.c:115):
-    create_clock -period P [-name N]; returns the target period in seconds
-    (ns units in the file, like VPR) or None."""
+    create_clock -period P [-name N]; returns the (last) target period in
+    seconds (ns units in the file, like VPR) or None."""
+    clocks = parse_sdc_clocks(text)
+    return list(clocks.values())[-1] if clocks else None
+
+
+def parse_sdc_clocks(text):
+    """All create_clock constraints as {name: period_seconds}; unnamed
+    clocks get the port expression or 'clk<i>'."""
     import re
-    period = None
+    clocks = {}
     for line in text.splitlines():
         line = line.split("#", 1)[0]
-        m = re.search(r"create_clock\s+.*-period\s+([0-9.eE+-]+)", line)
-        if m:
-            period = float(m.group(1)) * 1e-9
-    return period
+        m = re.search(r"create_clock\s+(.*)", line)
+        if not m:
+            continue
+        rest = m.group(1)
+        pm = re.search(r"-period\s+([0-9.eE+-]+)", rest)
+        if not pm:
+            continue
+        nm = re.search(r"-name\s+(\S+)", rest)
+        gp = re.search(r"\[\s*get_ports\s+\{?\s*(\S+?)\s*\}?\s*\]", rest)
+        name = (nm.group(1) if nm else
+                (gp.group(1) if gp else f"clk{len(clocks)}"))
+        clocks[name] = float(pm.group(1)) * 1e-9
+    return clocks

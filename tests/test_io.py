@@ -158,3 +158,40 @@ def test_cli_blif_xml_flow(tmp_path):
                "--sdc", str(sdc), "--timing_report", str(rpt)])
     assert rc == 0
     assert "Critical path delay" in rpt.read_text()
+
+
+BLIF2CLK = """
+.model two
+.inputs a b clkA clkB
+.outputs y z
+.names a n1
+1 1
+.latch n1 q1 re clkA 0
+.names b n2
+1 1
+.latch n2 q2 re clkB 0
+.names q1 y
+1 1
+.names q2 z
+1 1
+.end
+"""
+
+
+def test_two_clock_blif_flow(tmp_path):
+    from parallel_eda_amd.__main__ import main
+    from parallel_eda_amd.io.pack import pack_blif
+    from parallel_eda_amd.io.blif import parse_blif
+    m = parse_blif(BLIF2CLK)
+    arch = get_arch("tiny")
+    nl, _, _ = pack_blif(m, arch, n_ble=2)
+    assert len(nl.clock_names) == 2
+    assert (np.asarray(nl.block_clock) >= 0).sum() >= 2
+    blif = tmp_path / "c.blif"; blif.write_text(BLIF2CLK)
+    xml = tmp_path / "a.xml"; xml.write_text(ARCH_XML)
+    sdc = tmp_path / "c.sdc"
+    sdc.write_text("create_clock -period 8.0 -name clkA\n"
+                   "create_clock -period 4.0 -name clkB\n")
+    rc = main([str(blif), str(xml), "--route_chan_width", "20",
+               "--sdc", str(sdc)])
+    assert rc == 0
