@@ -155,6 +155,7 @@ PYBIND11_MODULE(_pnr_cpu, m) {
       .def("update_costs", &SerialRouter::update_costs)
       .def("set_pres_fac", &SerialRouter::set_pres_fac)
       .def("count_overused", &SerialRouter::count_overused)
+      .def("unrouted_sinks", &SerialRouter::unrouted_sinks)
       .def("feasible", &SerialRouter::feasible)
       .def("total_wirelength", &SerialRouter::total_wirelength)
       .def("heap_pushes", &SerialRouter::heap_pushes)

@@ -72,9 +72,14 @@ class SerialRouter {
   // aligned with sinks_. Returns number of overused rr nodes after the pass.
   int64_t route_iteration(const float* crit) {
     heap_pushes_ = heap_pops_ = 0;
+    unrouted_sinks_ = 0;
     for (int inet = 0; inet < num_nets(); ++inet) route_net(inet, crit);
     return count_overused();
   }
+
+  // sinks the last pass could not reach at all (disconnected fabric or
+  // capacity-saturated SOURCE class) — must be surfaced, never silent
+  int64_t unrouted_sinks() const { return unrouted_sinks_; }
 
   // Route only the given nets (distributed partitioning / selective reroute;
   // reference: mpi router routes partition_nets[rank], mpi_route...cxx:936).
@@ -282,7 +287,8 @@ class SerialRouter {
       }
     }
     if (found_prev == -1) {
-      // unroutable with current costs — shouldn't happen on our fabrics
+      // genuinely unreachable (should not happen on a validated fabric)
+      ++unrouted_sinks_;
       reset_touched();
       return 0.0f;
     }
@@ -357,6 +363,7 @@ class SerialRouter {
   std::vector<float> sink_delays_;
   std::vector<RouteTree> trees_;
   int64_t heap_pushes_ = 0, heap_pops_ = 0;
+  int64_t unrouted_sinks_ = 0;
 
  private:
   std::vector<float> path_cost_, back_cost_, R_up_;

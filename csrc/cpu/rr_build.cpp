@@ -249,11 +249,16 @@ RRGraph build_rr_graph(const ArchParams& ap) {
       }
     }
     // Switch blocks, constructed per OUTGOING wire: each wire's driver mux
-    // at its start SB takes one incoming wire from each of the 3 non-U-turn
-    // arrival sides (Fs=3; guarantees every wire is drivable, like a real
-    // unidir SB mux). out_dir: 0=E 1=W 2=N 3=S; allowed in-sides exclude
-    // the U-turn (out E excludes in-from-E, etc.).
-    auto connect_out_wire = [&](int32_t wnode, int sb_i, int sb_j, int out_dir) {
+    // at its start SB takes (a) the SAME-TRACK straight continuation when
+    // it exists (the standard unidir "straight through" — guarantees long
+    // straight chains to the chip edge), and (b) one incoming wire from
+    // each other non-U-turn side, selected BIJECTIVELY by the out-wire's
+    // position in its side's start list (a pure modular-track rotation can
+    // systematically orphan the fan-out of whole track classes: a fuzzed
+    // 3x7/W28/L3 fabric had sources reaching <10% of sinks).
+    // out_dir: 0=E 1=W 2=N 3=S.
+    auto connect_out_wire = [&](int32_t wnode, int sb_i, int sb_j, int out_dir,
+                                int out_pos) {
       static const int ins[4][3] = {
           {0, 2, 3},   // out E: in W(straight), S, N
           {1, 2, 3},   // out W: in E(straight), S, N
@@ -264,32 +269,43 @@ RRGraph build_rr_graph(const ArchParams& ap) {
         int s = ins[out_dir][k];
         sb_in_list(sb_i, sb_j, s, cand);
         if (cand.empty()) continue;
-        int idx = (t_out / 2 + k + sb_i + sb_j) % (int)cand.size();
+        int idx = -1;
+        if (k == 0) {
+          // straight side: prefer the same track's incoming wire
+          for (size_t c = 0; c < cand.size(); ++c)
+            if (g.ptc[cand[c]] == t_out) { idx = (int)c; break; }
+        }
+        if (idx < 0)
+          idx = (out_pos + k + sb_i + 2 * sb_j) % (int)cand.size();
         emit(cand[idx], wnode, SW_SB);
       }
     };
-    for (int y = 0; y <= ny; ++y)
+    for (int y = 0; y <= ny; ++y) {
+      std::vector<int> pos_inc(nx + 2, 0), pos_dec(nx + 2, 0);
       for (int t = 0; t < W; ++t) {
         int p = 1;
         while (p <= nx) {
           int b = span_high(p, t, L, nx);
           int32_t id = chanx[y].at(t, p);
-          if ((t & 1) == 0) connect_out_wire(id, p - 1, y, 0);  // INC: starts at SB(p-1,y), travels E
-          else connect_out_wire(id, b, y, 1);                   // DEC: starts at SB(b,y), travels W
+          if ((t & 1) == 0) connect_out_wire(id, p - 1, y, 0, pos_inc[p]++);
+          else connect_out_wire(id, b, y, 1, pos_dec[b]++);
           p = b + 1;
         }
       }
-    for (int x = 0; x <= nx; ++x)
+    }
+    for (int x = 0; x <= nx; ++x) {
+      std::vector<int> pos_inc(ny + 2, 0), pos_dec(ny + 2, 0);
       for (int t = 0; t < W; ++t) {
         int p = 1;
         while (p <= ny) {
           int b = span_high(p, t, L, ny);
           int32_t id = chany[x].at(t, p);
-          if ((t & 1) == 0) connect_out_wire(id, x, p - 1, 2);  // INC: starts at SB(x,p-1), travels N
-          else connect_out_wire(id, x, b, 3);                   // DEC: starts at SB(x,b), travels S
+          if ((t & 1) == 0) connect_out_wire(id, x, p - 1, 2, pos_inc[p]++);
+          else connect_out_wire(id, x, b, 3, pos_dec[b]++);
           p = b + 1;
         }
       }
+    }
   };
 
   // pass 1: count

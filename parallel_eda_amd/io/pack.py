@@ -7,7 +7,6 @@ connected primitives (attraction = shared signals) while the cluster has
 room (N BLEs, clb_in input pins). Produces the block-level NetlistPy the
 placer/router consume, where a block is one CLB (or an IO pad).
 """
-import heapq
 from collections import defaultdict
 
 import numpy as np

@@ -7,8 +7,7 @@ bounding boxes (reference: bb_factor growth; route failures re-tried).
 """
 import numpy as np
 
-from ..arch.archdef import ArchDef, RR_SINK
-from .. import ops
+from ..arch.archdef import ArchDef
 from ..ops import hip_api
 from .router import net_rr_terminals, RouteResult
 

@@ -155,12 +155,14 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
             cmap.conn_delays(router.sink_delays(), out=conn_delay)
             cpd, slack, c = sta.analyze(conn_delay)
             crit = cmap.sink_crit(c)
-        if overused == 0:
+        if overused == 0 and router.unrouted_sinks() == 0:
             break
         pres_fac = pres_fac_init if it == 1 else pres_fac * pres_fac_mult
         router.update_costs(pres_fac, acc_fac)
 
-    ok = overused == 0
+    ok = overused == 0 and router.unrouted_sinks() == 0
+    if router.unrouted_sinks() > 0 and verbose:
+        print(f"WARNING: {router.unrouted_sinks()} sinks unreachable")
     if ok:
         valid, err = router.check_routed()
         if not valid:

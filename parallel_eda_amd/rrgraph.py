@@ -84,7 +84,40 @@ def check_rr_graph(g, arch: ArchDef):
         raise RRGraphError("SINK with no in-edges")
     if (indeg[ty == RR_IPIN] == 0).any():
         raise RRGraphError("IPIN with no in-edges")
+    # sampled source->all-sinks connectivity (scipy BFS; catches fabrics
+    # whose W/L ratio leaves the Fs=3 switch digraph reducible — seen at
+    # W=8,L=4 where a stagger class has a single track)
+    check_connectivity(g, arch, samples=3)
     (sw)
+    return True
+
+
+def check_connectivity(g, arch, samples=3):
+    from scipy.sparse import csr_matrix
+    from scipy.sparse.csgraph import breadth_first_order
+    n = g.num_nodes
+    row_ptr = np.asarray(g.row_ptr)
+    dst = np.asarray(g.edge_dst)
+    ty = np.asarray(g.type)
+    m = csr_matrix((np.ones(len(dst), dtype=np.int8), dst, row_ptr),
+                   shape=(n, n))
+    ts = np.asarray(g.tile_source)
+    gy = arch.ny + 2
+    sinks = np.nonzero(ty == RR_SINK)[0]
+    probes = [(1, 1), (arch.nx, arch.ny),
+              (max(1, arch.nx // 2), max(1, arch.ny // 2))][:samples]
+    for (x, y) in probes:
+        src = ts[x * gy + y]
+        if src < 0:
+            continue
+        order = breadth_first_order(m, int(src), return_predecessors=False)
+        seen = np.zeros(n, dtype=bool)
+        seen[order] = True
+        frac = float(seen[sinks].mean())
+        if frac < 1.0:
+            raise RRGraphError(
+                f"fabric not fully connected: SOURCE({x},{y}) reaches "
+                f"{frac:.1%} of sinks (W/L ratio too small for Fs=3?)")
     return True
 
 

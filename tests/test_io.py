@@ -2,7 +2,7 @@ import numpy as np
 import pytest
 
 from parallel_eda_amd.arch.archdef import get_arch
-from parallel_eda_amd.arch.xml_parser import parse_arch_xml, size_grid_for_netlist
+from parallel_eda_amd.arch.xml_parser import parse_arch_xml
 from parallel_eda_amd.io.blif import parse_blif
 from parallel_eda_amd.io.pack import pack_blif
 from parallel_eda_amd.io.place_file import write_place, read_place

@@ -1,7 +1,7 @@
 import numpy as np
 import pytest
 
-from parallel_eda_amd.arch.archdef import get_arch, RR_CHANX, RR_CHANY, RR_SOURCE, RR_SINK
+from parallel_eda_amd.arch.archdef import get_arch, RR_CHANX, RR_CHANY, RR_SINK
 from parallel_eda_amd import rrgraph
 
 
