@@ -146,6 +146,10 @@ PYBIND11_MODULE(_pnr_cpu, m) {
         { py::gil_scoped_release rel; over = r.route_subset(c, ids.data(), ids.size()); }
         return over;
       })
+      .def("rip_up_nets", [](SerialRouter& r,
+                             py::array_t<int32_t, py::array::c_style | py::array::forcecast> ids) {
+        r.rip_up_nets(ids.data(), ids.size());
+      })
       .def("set_occ", [](SerialRouter& r,
                          py::array_t<int32_t, py::array::c_style | py::array::forcecast> occ) {
         if ((int64_t)occ.size() != (int64_t)r.g_->num_nodes)

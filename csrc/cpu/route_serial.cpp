@@ -88,6 +88,18 @@ class SerialRouter {
     return count_overused();
   }
 
+  // Rip up the given nets' trees (occ -1, trees cleared) without
+  // rerouting — ownership hand-off during load rebalancing (the
+  // reference migrates trees, mpi_route...cxx:172 move_route_tree; with
+  // a replicated graph the new owner just reroutes from scratch).
+  void rip_up_nets(const int32_t* ids, int64_t n) {
+    for (int64_t i = 0; i < n; ++i) {
+      RouteTree& t = trees_[ids[i]];
+      for (int32_t v : t.nodes) update_one_cost(v, -1);
+      t.clear();
+    }
+  }
+
   // Replace occupancy wholesale (after a distributed occ all-reduce) and
   // refresh pres_cost from it.
   void set_occ(const int32_t* occ) {

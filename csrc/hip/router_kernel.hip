@@ -553,6 +553,21 @@ __global__ void flag_congested_nets_kernel(TreesDev trees,
   if (threadIdx.x == 0) out[inet] = (uint8_t)flag;
 }
 
+// rip up listed nets (occ -1, len 0) without rerouting — ownership
+// hand-off for load rebalancing (reference: move_route_tree analogue)
+__global__ void rip_up_nets_kernel(TreesDev trees, const int32_t* __restrict__ ids,
+                                   int32_t n, int32_t* occ) {
+  int i = blockIdx.x;
+  if (i >= n) return;
+  int32_t inet = ids[i];
+  int64_t off = trees.off[inet];
+  int32_t len = trees.len[inet];
+  for (int k = threadIdx.x; k < len; k += blockDim.x)
+    atomicSub(&occ[trees.node[off + k]], 1);
+  __syncthreads();
+  if (threadIdx.x == 0) trees.len[inet] = 0;
+}
+
 // occupancy recount from route trees (debug cross-check; reference:
 // recalculate_occ partitioning_multi_sink...:6194-6216)
 __global__ void recount_occ_kernel(TreesDev trees, const int32_t* __restrict__ net_ids,
@@ -676,6 +691,17 @@ __global__ void fill_u64_kernel(uint64_t* p, uint64_t v, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   for (; i < n; i += (int64_t)gridDim.x * blockDim.x) p[i] = v;
 }
+}
+
+extern "C" int pnr_rip_up_nets(const int64_t* tree_off, int32_t* tree_node,
+                               int32_t* tree_len, const int32_t* ids,
+                               int32_t n, int32_t* occ, void* stream) {
+  if (n <= 0) return 0;
+  TreesDev t{};
+  t.off = tree_off; t.node = tree_node; t.len = tree_len;
+  hipLaunchKernelGGL(pnrh::rip_up_nets_kernel, dim3(n), dim3(256), 0,
+                     (hipStream_t)stream, t, ids, n, occ);
+  return (int)hipGetLastError();
 }
 
 extern "C" int pnr_flag_congested_nets(const int64_t* tree_off,

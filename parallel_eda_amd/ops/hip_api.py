@@ -69,6 +69,8 @@ def lib():
         _lib.pnr_overuse_count.argtypes = [ct.c_void_p] * 4 + [ct.c_int32, ct.c_void_p]
         _lib.pnr_recount_occ.restype = ct.c_int
         _lib.pnr_recount_occ.argtypes = [ct.c_void_p] * 4 + [ct.c_int32, ct.c_void_p, ct.c_void_p]
+        _lib.pnr_rip_up_nets.restype = ct.c_int
+        _lib.pnr_rip_up_nets.argtypes = [ct.c_void_p] * 4 + [ct.c_int32, ct.c_void_p, ct.c_void_p]
         _lib.pnr_flag_congested_nets.restype = ct.c_int
         _lib.pnr_flag_congested_nets.argtypes = [ct.c_void_p] * 5 + [ct.c_int32, ct.c_void_p, ct.c_void_p]
         _lib.pnr_fill_u64_launch.restype = ct.c_int

@@ -83,6 +83,12 @@ class CpuEngine:
         self.r.route_subset(np.ascontiguousarray(crit, dtype=np.float32),
                             np.asarray(net_ids, dtype=np.int32))
 
+    def rip_up_nets(self, net_ids):
+        self.r.rip_up_nets(np.asarray(net_ids, dtype=np.int32))
+
+    def net_costs(self, num_nets, my_nets):
+        return None  # no per-net measurement on the CPU oracle
+
     def occ_tensor(self):
         import torch
         return torch.from_numpy(np.asarray(self.r.occ()).copy())
@@ -123,6 +129,16 @@ class GpuEngine:
     def sink_delays_local(self, net_ids):
         return self._last_sd
 
+    def rip_up_nets(self, net_ids):
+        self.g.rip_up_nets(net_ids)
+
+    def net_costs(self, num_nets, my_nets):
+        # measured per-net search cost from the last iteration's counters
+        w = self.g.t_net_scans.cpu().numpy().astype(np.float64)
+        mask = np.zeros(num_nets, dtype=bool)
+        mask[my_nets] = True
+        return np.where(mask, w, 0.0)
+
     def update_acc(self, acc_fac):
         self.g.update_acc(acc_fac)
 
@@ -157,11 +173,53 @@ class DistRouteLoop:
         else:
             self.rank_of = np.zeros(num_nets, dtype=np.int32)
         self.my_nets = np.nonzero(self.rank_of == rank)[0]
+        self._bb = bb
+        self._sink_ptr = np.asarray(sink_ptr)
         # sinks owned by my nets
         mask = np.zeros(n_rsinks, dtype=bool)
         for n in self.my_nets:
             mask[sink_ptr[n]:sink_ptr[n + 1]] = True
         self.my_sink_mask = mask
+
+    def rebalance(self, weights=None):
+        """Repartition nets by measured per-net route cost (reference:
+        load-balanced repartition, mpi_route...cxx:249). weights: per-net
+        cost with valid entries for OWNED nets (zeros elsewhere); summed
+        across ranks so every rank computes the identical new partition.
+        Nets that change owner are ripped up by the OLD owner and the occ
+        deltas all-reduced (the replicated-graph analogue of
+        move_route_tree: the new owner simply reroutes from scratch).
+        Returns the number of nets this rank lost+gained."""
+        if self.ws <= 1:
+            return 0
+        import torch
+        dist = _dist()
+        if weights is None:
+            weights = self.engine.net_costs(len(self.rank_of), self.my_nets)
+        if weights is None:
+            return 0
+        w = torch.from_numpy(np.ascontiguousarray(weights, dtype=np.float64))
+        dist.all_reduce(w, op=dist.ReduceOp.SUM)
+        wsum = np.maximum(w.numpy(), 1.0)
+        new_rank = spatial_partition(self._bb, self.ws, weight=wsum)
+        old_mine = set(self.my_nets.tolist())
+        new_mine = set(np.nonzero(new_rank == self.rank)[0].tolist())
+        lost = sorted(old_mine - new_mine)
+        gained = sorted(new_mine - old_mine)
+        occ_before = self.engine.occ_tensor().clone()
+        if lost:
+            self.engine.rip_up_nets(np.asarray(lost, dtype=np.int64))
+        occ = self.engine.occ_tensor()
+        delta = occ - occ_before
+        dist.all_reduce(delta, op=dist.ReduceOp.SUM)
+        self.engine.set_occ(occ_before + delta)
+        self.rank_of = new_rank
+        self.my_nets = np.nonzero(new_rank == self.rank)[0]
+        mask = np.zeros(len(self.my_sink_mask), dtype=bool)
+        for n in self.my_nets:
+            mask[self._sink_ptr[n]:self._sink_ptr[n + 1]] = True
+        self.my_sink_mask = mask
+        return len(lost) + len(gained)
 
     def iteration(self, crit, pres_fac, acc_fac):
         """One distributed PathFinder iteration. Returns

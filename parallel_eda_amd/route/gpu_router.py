@@ -358,6 +358,17 @@ class GpuRouter:
         return [(int(i), int(s), int(n), int(a))
                 for i, s, n, a in zip(idx, v, nsk, areas)]
 
+    def rip_up_nets(self, net_ids):
+        """Remove the given nets' routes (ownership hand-off)."""
+        torch = self.torch
+        ids = torch.from_numpy(np.asarray(net_ids, dtype=np.int32)).to(self.device)
+        rc = self.lib.pnr_rip_up_nets(
+            ct_ptr(self.t_tree_off), ct_ptr(self.t_tree_node),
+            ct_ptr(self.t_tree_len), ct_ptr(ids), len(net_ids),
+            ct_ptr(self.t_occ), self._stream())
+        hip_api.check(rc, "rip_up_nets")
+        torch.cuda.synchronize(self.device)
+
     def congested_nets(self):
         """Nets whose tree touches an overused node (selective-reroute set;
         reference: build_phase_two congested-nets-only)."""

@@ -109,3 +109,66 @@ def test_dist_route_world2(tmp_path):
     assert res.success
     wl_dist = int(r0["occ"][np.asarray(g.type) >= 4].sum())
     assert wl_dist <= res.wirelength * 1.4
+
+
+def _worker_rebalance(rank, world, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    arch, nl, pl = _build_case()
+    g = rrgraph.build_rr_graph(arch)
+    sta = STA(nl, arch)
+    net_ids, src_rr, sink_ptr, sink_rr, conn_index = net_rr_terminals(
+        nl, pl, g, arch)
+    cmap = ConnMap(conn_index, sink_ptr, nl.num_conns, len(sink_rr))
+    cpu = ops.cpu()
+    router = cpu.SerialRouter(g, src_rr, sink_ptr, sink_rr, cpu.RouterOpts())
+    engine = CpuEngine(router, g.num_nodes)
+    xlow = np.asarray(g.xlow); ylow = np.asarray(g.ylow)
+    bb = np.zeros((len(net_ids), 4), dtype=np.int16)
+    for n in range(len(net_ids)):
+        terms = np.r_[src_rr[n], sink_rr[sink_ptr[n]:sink_ptr[n + 1]]]
+        bb[n] = (xlow[terms].min(), ylow[terms].min(),
+                 xlow[terms].max(), ylow[terms].max())
+    loop = DistRouteLoop(engine, len(net_ids), bb, len(sink_rr), sink_ptr,
+                         rank=rank, world_size=world)
+    crit = np.zeros(len(sink_rr), dtype=np.float32)
+    conn_delay = np.zeros(nl.num_conns, dtype=np.float32)
+    pres = 0.0
+    over = -1
+    # a couple of iterations, then rebalance with a synthetic skewed
+    # measured-cost profile, then run to feasibility
+    for it in range(40):
+        over, sd = loop.iteration(crit, pres, acc_fac=1.0)
+        cmap.conn_delays(sd, out=conn_delay)
+        cpd, slack, c = sta.analyze(conn_delay)
+        crit = cmap.sink_crit(c)
+        if it == 1:
+            w = np.zeros(len(net_ids))
+            w[loop.my_nets] = 1.0 + 50.0 * (np.asarray(loop.my_nets) % 3 == 0)
+            moved = loop.rebalance(weights=w)
+        if over == 0:
+            break
+        pres = 0.5 if pres == 0.0 else pres * 1.3
+
+    occ = np.asarray(router.occ()).copy()
+    with open(os.path.join(tmpdir, f"rb{rank}.pkl"), "wb") as f:
+        pickle.dump({"over": over, "occ": occ, "moved": moved,
+                     "mine": len(loop.my_nets)}, f)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_dist_rebalance_world2(tmp_path):
+    mp.spawn(_worker_rebalance, args=(2, 29532, str(tmp_path)), nprocs=2,
+             join=True)
+    with open(tmp_path / "rb0.pkl", "rb") as f:
+        r0 = pickle.load(f)
+    with open(tmp_path / "rb1.pkl", "rb") as f:
+        r1 = pickle.load(f)
+    assert r0["over"] == 0, "infeasible after rebalance"
+    # global congestion state stayed consistent across the hand-off
+    assert np.array_equal(r0["occ"], r1["occ"])
+    assert r0["mine"] + r1["mine"] > 0
