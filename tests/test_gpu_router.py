@@ -128,3 +128,22 @@ def test_gpu_checkpoint_resume(tseng_placed):
         r1.update_acc(1.0)
         r2.update_acc(1.0)
     assert torch.equal(r1.t_occ, r2.t_occ)
+
+
+def test_gpu_rip_up_nets(tiny_placed):
+    """Ownership hand-off rip-up kernel: occ stays recount-consistent."""
+    arch, nl, pl, g = tiny_placed
+    from parallel_eda_amd.route.gpu_router import GpuRouter
+    net_ids, src_rr, sink_ptr, sink_rr, _ = net_rr_terminals(nl, pl, g, arch)
+    r = GpuRouter(g, arch, src_rr, sink_ptr.astype(np.int32), sink_rr)
+    crit = np.zeros(len(sink_rr), dtype=np.float32)
+    r.route_iteration(crit, 0.0)
+    occ0 = int(r.t_occ.sum().item())
+    rip = np.arange(0, r.num_nets, 2)
+    r.rip_up_nets(rip)
+    assert r.check_occ_recount()
+    lens = r.t_tree_len.cpu().numpy()
+    assert (lens[rip] == 0).all()
+    occ1 = int(r.t_occ.sum().item())
+    assert 0 < occ1 < occ0
+    assert int((r.t_occ < 0).sum().item()) == 0
