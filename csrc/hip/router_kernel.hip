@@ -18,6 +18,13 @@
 #include "pnr_hip.h"
 
 namespace pnrh {
+// EXPERIMENTAL calendar-frontier variant (csrc/hip/router_calendar.hip)
+__global__ void route_nets_cal_kernel(
+    RRDev g, NetsDev nets, TreesDev trees, RouteParams P,
+    const int32_t*, int32_t, const int32_t*, int32_t, int32_t*,
+    int32_t*, const float*, uint64_t*, int64_t, int64_t, int32_t,
+    float4*, int64_t, int64_t, int32_t*, int64_t, int64_t,
+    int32_t*, unsigned long long*, unsigned long long*);
 
 #define WG_THREADS 256
 #define PATH_CAP 4096      // max path nodes per sink backtrack (LDS)
@@ -43,82 +50,6 @@ namespace pnrh {
 #define DBG_LI(li, cap, code) (li)
 #define DBG_NODE(v, code) (v)
 #endif
-
-__device__ __forceinline__ uint32_t f32_bits(float f) {
-  // order-preserving bits for non-negative floats
-  return __float_as_uint(f);
-}
-__device__ __forceinline__ float bits_f32(uint32_t u) { return __uint_as_float(u); }
-
-__device__ __forceinline__ uint64_t pack_state(float back, int32_t prev) {
-  return ((uint64_t)f32_bits(back) << 32) | (uint32_t)prev;
-}
-
-// state[] is updated with device-scope atomicMin, which executes at L2 and
-// BYPASSES the CU's vector L1. A plain load can hit a stale L1 line (e.g.
-// the INF written by the touched-list reset) and miss the atomic's value —
-// the backtrack then extracts prev = -1 and wild-walks (the intermittent
-// memory faults of profiles/README.md's fault-hunt log). Every read of
-// state[] therefore goes through an agent-scope atomic load (L1-bypassing).
-__device__ __forceinline__ uint64_t load_state(const uint64_t* p) {
-  return __hip_atomic_load((const unsigned long long*)p, __ATOMIC_RELAXED,
-                           __HIP_MEMORY_SCOPE_AGENT);
-}
-
-struct SinkCtx {
-  int32_t sink_node;
-  int16_t sx, sy;
-  float crit;
-  float astar_fac;
-};
-
-__device__ __forceinline__ float expected_cost(const RRDev& g, const RouteParams& P,
-                                               int32_t v, const SinkCtx& S) {
-  int8_t ty = g.type[v];
-  if (ty == 1 /*SINK*/) return 0.0f;
-  int tx = S.sx, ty2 = S.sy;
-  int dx = 0, dy = 0;
-  int xl = g.xlow[v], xh = g.xhigh[v], yl = g.ylow[v], yh = g.yhigh[v];
-  if (xl > tx) dx = xl - tx; else if (xh < tx) dx = tx - xh;
-  if (yl > ty2) dy = yl - ty2; else if (yh < ty2) dy = ty2 - yh;
-  int dist = dx + dy;
-  int nseg = (dist + g.L - 1) / g.L;
-  float cong = nseg * P.seg_base * P.cong_mult + P.ipin_base;
-  float del = nseg * P.seg_delay + P.ipin_delay;
-  return S.crit * del + (1.0f - S.crit) * cong;
-}
-
-// congestion cost of entering node v (pres computed from occ on the fly;
-// semantics of congestion.cxx:296 update_one_cost_internal's pres formula)
-__device__ __forceinline__ float cong_cost(const RRDev& g, const RouteParams& P,
-                                           const int32_t* occ, const float* acc,
-                                           int32_t v) {
-  int over = occ[v] + 1 - g.capacity[v];
-  float pres = over > 0 ? 1.0f + over * P.pres_fac : 1.0f;
-  return g.base_cost[g.type[v]] * acc[v] * pres;
-}
-
-__device__ __forceinline__ float hop_delay(const RRDev& g, int8_t sw, int32_t v) {
-  return g.sw_Tdel[sw] + g.C[v] * (g.sw_R[sw] + 0.5f * g.R[v]);
-}
-
-struct LocalIdx {
-  // small class: dense bb-local index; large class: global node id
-  int bx0, by0, bw, bh, npt;
-  bool dense;
-  __device__ __forceinline__ int64_t operator()(const RRDev& g, int32_t v) const {
-    if (!dense) return v;
-    int tx = g.xlow[v] - bx0;
-    int ty = g.ylow[v] - by0;
-    if (tx < 0 || ty < 0 || tx >= bw || ty >= bh) return -1;
-    return ((int64_t)tx * bh + ty) * npt + g.idx_in_tile[v];
-  }
-  __device__ __forceinline__ bool in_bb(const RRDev& g, int32_t v) const {
-    int tx = g.xlow[v] - bx0;
-    int ty = g.ylow[v] - by0;
-    return tx >= 0 && ty >= 0 && tx < bw && ty < bh;
-  }
-};
 
 struct WgShared {
   int dbg;
@@ -622,6 +553,7 @@ struct RouteLaunchArgs {
   int32_t* fail_flags;
   unsigned long long* stats;   // [8] search counters or null
   unsigned long long* net_scans;  // per-net scan counters or null
+  int32_t use_calendar;        // EXPERIMENTAL: calendar-queue frontier
 };
 
 int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
@@ -642,6 +574,20 @@ int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
   P.max_rounds = a->max_rounds;
   P.strict_term = a->strict_term;
   int grid = a->n_small_slots + a->n_large_slots;
+  if (a->use_calendar) {
+    hipLaunchKernelGGL(route_nets_cal_kernel, dim3(grid), dim3(WG_THREADS), 0,
+                       (hipStream_t)stream,
+                       g, nets, trees, P,
+                       a->queue_small, a->n_queue_small,
+                       a->queue_large, a->n_queue_large,
+                       a->q_cursors, a->occ, a->acc,
+                       a->state_base, a->small_cap, a->large_cap,
+                       a->n_small_slots,
+                       a->frontier_base, a->f_cap_small, a->f_cap_large,
+                       a->touched_base, a->t_cap_small, a->t_cap_large,
+                       a->fail_flags, a->stats, a->net_scans);
+    return (int)hipGetLastError();
+  }
   hipLaunchKernelGGL(route_nets_kernel, dim3(grid), dim3(WG_THREADS), 0,
                      (hipStream_t)stream,
                      g, nets, trees, P,

@@ -50,6 +50,7 @@ class RouteLaunchArgs(ct.Structure):
         ("fail_flags", ct.c_void_p),
         ("stats", ct.c_void_p),
         ("net_scans", ct.c_void_p),
+        ("use_calendar", ct.c_int32),
     ]
 
 

@@ -459,6 +459,8 @@ class GpuRouter:
         a.fail_flags = ct_ptr(self.t_fail)
         a.stats = ct_ptr(self.t_stats)
         a.net_scans = ct_ptr(self.t_net_scans)
+        a.use_calendar = 1 if (getattr(self, "use_calendar", False) or
+                               _os.environ.get("PNR_CALENDAR")) else 0
         self._args_keepalive = (t_sink_rr, t_crit, t_sink_orig, q_small,
                                 q_large)
         return a
