@@ -199,8 +199,11 @@ class DistRouteLoop:
         if weights is None:
             return 0
         w = torch.from_numpy(np.ascontiguousarray(weights, dtype=np.float64))
+        occ_dev = self.engine.occ_tensor().device
+        if occ_dev.type == "cuda":
+            w = w.to(occ_dev)   # NCCL requires device tensors
         dist.all_reduce(w, op=dist.ReduceOp.SUM)
-        wsum = np.maximum(w.numpy(), 1.0)
+        wsum = np.maximum(w.cpu().numpy(), 1.0)
         new_rank = spatial_partition(self._bb, self.ws, weight=wsum)
         old_mine = set(self.my_nets.tolist())
         new_mine = set(np.nonzero(new_rank == self.rank)[0].tolist())
