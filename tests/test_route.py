@@ -53,3 +53,68 @@ def test_full_flow_tseng():
     assert res.route.success
     assert res.cpd > 0
     assert res.wirelength > 0
+
+
+def test_oracle_astar_optimal_vs_independent_dijkstra():
+    """The CPU oracle (astar_fac=1 => admissible) must find min-cost paths
+    identical to an independent Python Dijkstra on the same cost surface
+    (crit=1: pure per-hop Elmore delay; congestion term zero)."""
+    import heapq
+    from parallel_eda_amd import ops
+    arch = get_arch("tiny")
+    g = rrgraph.build_rr_graph(arch)
+    row_ptr = np.asarray(g.row_ptr)
+    dst = np.asarray(g.edge_dst)
+    swi = np.asarray(g.edge_sw)
+    ty = np.asarray(g.type)
+    xl = np.asarray(g.xlow); yl = np.asarray(g.ylow)
+    Rn = np.asarray(g.node_R); Cn = np.asarray(g.node_C)
+    swR = np.asarray(g.sw_R); swT = np.asarray(g.sw_Tdel)
+    ts = np.asarray(g.tile_source); tk = np.asarray(g.tile_sink)
+    gy = arch.ny + 2
+
+    def hop(sw, w):
+        return np.float32(swT[sw] + Cn[w] * (swR[sw] + 0.5 * Rn[w]))
+
+    def dijkstra(src, sink):
+        dist = {src: np.float32(0.0)}
+        pq = [(0.0, int(src))]
+        while pq:
+            d, v = heapq.heappop(pq)
+            if v == sink:
+                return d
+            if d > dist.get(v, np.inf):
+                continue
+            for e in range(row_ptr[v], row_ptr[v + 1]):
+                w = int(dst[e])
+                if ty[w] == 1 and w != sink:
+                    continue
+                if ty[w] == 3 and (xl[w] != xl[sink] or yl[w] != yl[sink]):
+                    continue
+                nd = np.float32(d + hop(swi[e], w))
+                if nd < dist.get(w, np.inf):
+                    dist[w] = nd
+                    heapq.heappush(pq, (float(nd), w))
+        return None
+
+    rng = np.random.default_rng(17)
+    cpu = ops.cpu()
+    checked = 0
+    for _ in range(12):
+        sx, sy = 1 + rng.integers(arch.nx), 1 + rng.integers(arch.ny)
+        tx, ty2 = 1 + rng.integers(arch.nx), 1 + rng.integers(arch.ny)
+        if (sx, sy) == (tx, ty2):
+            continue
+        src = int(ts[sx * gy + sy]); sink = int(tk[tx * gy + ty2])
+        opts = cpu.RouterOpts()
+        opts.astar_fac = 1.0  # admissible => optimal
+        r = cpu.SerialRouter(g, np.asarray([src], dtype=np.int32),
+                             np.asarray([0, 1], dtype=np.int64),
+                             np.asarray([sink], dtype=np.int32), opts)
+        r.set_pres_fac(0.0)
+        r.route_iteration(np.asarray([1.0], dtype=np.float32))  # pure delay
+        got = float(r.sink_delays()[0])
+        ref = float(dijkstra(src, sink))
+        assert got == pytest.approx(ref, rel=1e-4), (sx, sy, tx, ty2)
+        checked += 1
+    assert checked >= 8
