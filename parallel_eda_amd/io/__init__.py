@@ -1,3 +1,3 @@
-from .synth import synth_netlist, SynthSpec
+from .synth import synth_netlist, synth_placed_netlist, SynthSpec, NetlistPy
 
-__all__ = ["synth_netlist", "SynthSpec"]
+__all__ = ["synth_netlist", "synth_placed_netlist", "SynthSpec", "NetlistPy"]

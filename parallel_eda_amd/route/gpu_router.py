@@ -108,6 +108,31 @@ class GpuRouter:
         self.t_cap_small = self.small_cap
         self.t_cap_large = self.large_cap
 
+        # sanity-check the planned scratch against free VRAM before
+        # allocating (protects N-GPU runs from silent overcommit)
+        planned = (n_small_slots * self.small_cap + n_large_slots * self.large_cap) * 8 \
+            + (n_small_slots * 2 * self.f_cap_small +
+               n_large_slots * 2 * self.f_cap_large) * 16 \
+            + (n_small_slots * self.t_cap_small +
+               n_large_slots * self.t_cap_large) * 4
+        try:
+            free, total = torch.cuda.mem_get_info(device)
+            if planned > 0.6 * free:
+                import sys
+                print(f"GpuRouter: planned scratch {planned/2**30:.1f} GiB vs "
+                      f"{free/2**30:.1f} GiB free — shrinking slots",
+                      file=sys.stderr)
+                while planned > 0.5 * free and n_small_slots > 128:
+                    n_small_slots //= 2
+                    self.n_small_slots = n_small_slots
+                    planned = (n_small_slots * self.small_cap + n_large_slots * self.large_cap) * 8 \
+                        + (n_small_slots * 2 * self.f_cap_small +
+                           n_large_slots * 2 * self.f_cap_large) * 16 \
+                        + (n_small_slots * self.t_cap_small +
+                           n_large_slots * self.t_cap_large) * 4
+        except Exception:
+            pass
+
         nb = torch.int64
         self.t_state = torch.empty(
             n_small_slots * self.small_cap + n_large_slots * self.large_cap,
