@@ -1,0 +1,45 @@
+"""Round-2 opening GPU checklist — run as ONE gpurun call:
+
+  gpurun --timeout 1800 -- 'python tools/round2_entry.py > gpurun_out/r2entry.log 2>&1'
+
+Covers, in order of information value per GPU-minute:
+  1. full GPU test suite (regressions since round 1)
+  2. calendar-vs-pingpong A/B: quality at tseng, perf at LU32
+  3. bitcoin bench with the winning frontier
+  4. placer anneal at bgm scale (GPU engine quality vs CPU trend)
+"""
+import os, subprocess, sys, time
+from pathlib import Path
+ROOT = Path(__file__).resolve().parent.parent
+os.chdir(ROOT)
+
+def run(name, cmd, env=None, timeout=600):
+    print(f"\n===== {name}: {cmd}", flush=True)
+    e = dict(os.environ); e.update(env or {})
+    t0 = time.time()
+    r = subprocess.run(cmd, shell=True, env=e, timeout=timeout)
+    print(f"===== {name} rc={r.returncode} ({time.time()-t0:.0f}s)", flush=True)
+    return r.returncode
+
+run("gpu-tests", "python -m pytest tests/ -q -m gpu", timeout=900)
+run("lu32-pingpong", "python tools/gpu_sweep_one.py 1.2 3.0 6", timeout=400)
+run("lu32-calendar", "python tools/gpu_sweep_one.py 1.2 3.0 6",
+    env={"PNR_CALENDAR": "1"}, timeout=400)
+run("tseng-quality-calendar",
+    "python -m pytest tests/test_gpu_router.py::test_gpu_route_matches_cpu_quality -q",
+    env={"PNR_CALENDAR": "1"}, timeout=300)
+run("btc-calendar", "python bench.py --config bitcoin_miner --steps 2 "
+    "--warmup 2 --verbose", env={"PNR_CALENDAR": "1"}, timeout=900)
+run("placer-bgm", "python - <<'P'\n"
+    "import sys; sys.path.insert(0, '.')\n"
+    "import time, numpy as np\n"
+    "from parallel_eda_amd.arch.archdef import get_arch\n"
+    "from parallel_eda_amd.io.synth import synth_netlist, spec_for_arch\n"
+    "from parallel_eda_amd.place.gpu_placer import anneal_place_gpu\n"
+    "arch = get_arch('bgm')\n"
+    "nl = synth_netlist(spec_for_arch(arch, fill=0.5, seed=1))\n"
+    "t0 = time.time()\n"
+    "pl = anneal_place_gpu(nl, arch, seed=7, timing_tradeoff=0.0)\n"
+    "print(f'bgm GPU anneal: bb={pl.bb_cost:.0f} temps={pl.stats[\"temps\"]} "
+    "t={time.time()-t0:.0f}s')\n"
+    "P", timeout=900)
