@@ -75,6 +75,13 @@ PYBIND11_MODULE(_pnr_cpu, m) {
     ap.fc_in = a["fc_in"].cast<int>(); ap.fc_out = a["fc_out"].cast<int>();
     ap.clb_in = a["clb_in"].cast<int>(); ap.clb_out = a["clb_out"].cast<int>();
     ap.io_cap = a["io_cap"].cast<int>();
+    auto opt_int = [&](const char* k) {
+      return a.contains(k) ? a[k].cast<int>() : 0;
+    };
+    ap.ram_col_every = opt_int("ram_col_every");
+    ap.dsp_col_every = opt_int("dsp_col_every");
+    ap.ram_in = opt_int("ram_in"); ap.ram_out = opt_int("ram_out");
+    ap.dsp_in = opt_int("dsp_in"); ap.dsp_out = opt_int("dsp_out");
     ap.R_wire = a["R_wire"].cast<float>(); ap.C_wire = a["C_wire"].cast<float>();
     ap.R_sw = a["R_sw"].cast<float>(); ap.C_sw_in = a["C_sw_in"].cast<float>();
     ap.T_sw = a["T_sw"].cast<float>(); ap.T_opin = a["T_opin"].cast<float>();
@@ -189,11 +196,15 @@ PYBIND11_MODULE(_pnr_cpu, m) {
   py::class_<SerialPlacer>(m, "SerialPlacer")
       .def(py::init([](std::shared_ptr<Netlist> nl, int nx, int ny, int io_cap,
                        py::array_t<float, py::array::c_style | py::array::forcecast> delay_mat,
-                       uint64_t seed) {
-        auto p = new SerialPlacer(nl.get(), nx, ny, io_cap, to_vec(delay_mat), seed);
+                       uint64_t seed,
+                       py::array_t<int8_t, py::array::c_style | py::array::forcecast> tile_btype) {
+        auto p = new SerialPlacer(nl.get(), nx, ny, io_cap, to_vec(delay_mat),
+                                  seed, to_vec(tile_btype));
         p->netlist_holder_ = nl;
         return p;
-      }))
+      }), py::arg("nl"), py::arg("nx"), py::arg("ny"), py::arg("io_cap"),
+          py::arg("delay_mat"), py::arg("seed"),
+          py::arg("tile_btype") = py::array_t<int8_t>())
       .def("bb_cost", &SerialPlacer::bb_cost)
       .def("td_cost", &SerialPlacer::td_cost)
       .def("recompute_bb_cost", &SerialPlacer::recompute_bb_cost_from_scratch)
@@ -229,11 +240,14 @@ PYBIND11_MODULE(_pnr_cpu, m) {
       });
 
   py::class_<TimingGraph>(m, "TimingGraph")
-      .def(py::init([](std::shared_ptr<Netlist> nl, float t_clb, float t_out, float t_in) {
-        auto t = new TimingGraph(nl.get(), t_clb, t_out, t_in);
+      .def(py::init([](std::shared_ptr<Netlist> nl, float t_clb, float t_out,
+                       float t_in,
+                       py::array_t<float, py::array::c_style | py::array::forcecast> blk_delay) {
+        auto t = new TimingGraph(nl.get(), t_clb, t_out, t_in, to_vec(blk_delay));
         t->netlist_holder_ = nl;
         return t;
-      }))
+      }), py::arg("nl"), py::arg("t_clb"), py::arg("t_out"), py::arg("t_in"),
+          py::arg("blk_delay") = py::array_t<float>())
       .def("num_levels", &TimingGraph::num_levels)
       .def("analyze", [](TimingGraph& t, py::array_t<float, py::array::c_style | py::array::forcecast> conn_delay) {
         py::ssize_t n = conn_delay.size();

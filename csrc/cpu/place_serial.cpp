@@ -29,14 +29,38 @@ class SerialPlacer {
  public:
   // delay_mat: (nx+2)*(ny+2) delay lookup by |dx|,|dy| (flattened dy-major:
   // delay_mat[dx*(ny+2)+dy]); may be empty for bb-only placement.
+  // tile_btype: (nx+2)*(ny+2) x-major block type per tile (-1 corner,
+  // 0 IO, 1 CLB, 2 RAM, 3 DSP); empty => homogeneous (perimeter IO,
+  // interior CLB), which reproduces the pre-heterogeneous behavior.
   SerialPlacer(const Netlist* nl, int nx, int ny, int io_cap,
-               std::vector<float> delay_mat, uint64_t seed)
+               std::vector<float> delay_mat, uint64_t seed,
+               std::vector<int8_t> tile_btype = {})
       : nl_(nl), nx_(nx), ny_(ny), io_cap_(io_cap),
-        delay_mat_(std::move(delay_mat)), rng_(seed) {
+        delay_mat_(std::move(delay_mat)), rng_(seed),
+        tile_btype_(std::move(tile_btype)) {
     int nb = nl_->num_blocks;
     bx_.assign(nb, -1); by_.assign(nb, -1); bslot_.assign(nb, 0);
     // grid occupancy: per location, list of block ids (size cap)
     gx_ = nx + 2; gy_ = ny + 2;
+    if (tile_btype_.empty()) {
+      tile_btype_.assign((size_t)gx_ * gy_, -1);
+      for (int x = 1; x <= nx_; ++x)
+        for (int y = 1; y <= ny_; ++y) tile_btype_[(size_t)x * gy_ + y] = 1;
+      for (int y = 1; y <= ny_; ++y) {
+        tile_btype_[y] = 0; tile_btype_[(size_t)(gx_ - 1) * gy_ + y] = 0;
+      }
+      for (int x = 1; x <= nx_; ++x) {
+        tile_btype_[(size_t)x * gy_] = 0;
+        tile_btype_[(size_t)x * gy_ + gy_ - 1] = 0;
+      }
+    }
+    if (tile_btype_.size() != (size_t)gx_ * gy_)
+      throw std::runtime_error("tile_btype size mismatch");
+    // per-type column lists for sparse-column (RAM/DSP) move proposals
+    for (int x = 1; x <= nx_; ++x) {
+      int8_t t = tile_btype_[(size_t)x * gy_ + 1];
+      if (t >= 2 && t <= 3) type_cols_[t - 2].push_back(x);
+    }
     grid_.assign((size_t)gx_ * gy_ * std::max(1, io_cap), -1);
     grid_cnt_.assign((size_t)gx_ * gy_, 0);
     build_net_arrays();
@@ -47,11 +71,13 @@ class SerialPlacer {
     recompute_td_all();
   }
 
+  int8_t tile_type(int x, int y) const {
+    return tile_btype_[(size_t)x * gy_ + y];
+  }
   int cap_at(int x, int y) const {
-    bool io = (x == 0 || x == gx_ - 1 || y == 0 || y == gy_ - 1);
-    if (x >= 1 && x <= nx_ && y >= 1 && y <= ny_) return 1;
-    if (io && ((x >= 1 && x <= nx_) || (y >= 1 && y <= ny_))) return io_cap_;
-    return 0;
+    int8_t t = tile_type(x, y);
+    if (t < 0) return 0;
+    return t == 0 ? io_cap_ : 1;
   }
   bool is_io_loc(int x, int y) const {
     return (x == 0 || x == gx_ - 1 || y == 0 || y == gy_ - 1);
@@ -120,8 +146,7 @@ class SerialPlacer {
     for (int b = 0; b < nl_->num_blocks; ++b) {
       int x = bx_[b], y = by_[b];
       if (cap_at(x, y) <= bslot_[b]) { *err = "block in illegal slot"; return false; }
-      bool io = nl_->block_type[b] == 0;
-      if (io != is_io_loc(x, y)) { *err = "type/location mismatch"; return false; }
+      if (tile_type(x, y) != nl_->block_type[b]) { *err = "type/location mismatch"; return false; }
       if (grid_at(x, y, bslot_[b]) != b) { *err = "grid inconsistent"; return false; }
       cnt[(size_t)x * gy_ + y]++;
     }
@@ -142,6 +167,8 @@ class SerialPlacer {
   std::vector<float> delay_mat_;
   std::mt19937_64 rng_;
   std::vector<int32_t> bx_, by_, bslot_;
+  std::vector<int8_t> tile_btype_;       // (gx*gy) x-major; see ctor
+  std::vector<int> type_cols_[2];        // columns of type RAM(0) / DSP(1)
 
  private:
   std::vector<int32_t> grid_;      // (x*gy+y)*cap + slot -> block
@@ -186,26 +213,40 @@ class SerialPlacer {
   }
 
   void initial_placement() {
-    // random legal placement: CLBs into CLB tiles, IOs into IO slots
-    std::vector<std::pair<int, int>> clb_locs, io_locs;
-    for (int x = 1; x <= nx_; ++x) for (int y = 1; y <= ny_; ++y) clb_locs.push_back({x, y});
-    for (int y = 1; y <= ny_; ++y) { io_locs.push_back({0, y}); io_locs.push_back({gx_ - 1, y}); }
-    for (int x = 1; x <= nx_; ++x) { io_locs.push_back({x, 0}); io_locs.push_back({x, gy_ - 1}); }
-    std::shuffle(clb_locs.begin(), clb_locs.end(), rng_);
-    std::shuffle(io_locs.begin(), io_locs.end(), rng_);
-    size_t ci = 0, ii = 0; int io_slot = 0;
+    // random legal placement: each block into a free tile of its type.
+    // Location-list build + shuffle order (CLB, RAM, DSP, then IO) keeps
+    // the rng stream identical to the pre-heterogeneous placer on
+    // homogeneous fabrics (empty lists draw nothing from the rng).
+    std::vector<std::pair<int, int>> locs[4];  // by block type
+    for (int x = 1; x <= nx_; ++x)
+      for (int y = 1; y <= ny_; ++y) {
+        int8_t t = tile_type(x, y);
+        if (t >= 1) locs[t].push_back({x, y});
+      }
+    for (int y = 1; y <= ny_; ++y) { locs[0].push_back({0, y}); locs[0].push_back({gx_ - 1, y}); }
+    for (int x = 1; x <= nx_; ++x) { locs[0].push_back({x, 0}); locs[0].push_back({x, gy_ - 1}); }
+    std::shuffle(locs[1].begin(), locs[1].end(), rng_);
+    std::shuffle(locs[2].begin(), locs[2].end(), rng_);
+    std::shuffle(locs[3].begin(), locs[3].end(), rng_);
+    std::shuffle(locs[0].begin(), locs[0].end(), rng_);
+    size_t cur[4] = {0, 0, 0, 0};
+    int io_slot = 0;
     for (int b = 0; b < nl_->num_blocks; ++b) {
-      if (nl_->block_type[b] == 1) {
-        if (ci >= clb_locs.size()) throw std::runtime_error("too many CLBs for grid");
-        auto [x, y] = clb_locs[ci++];
+      int t = nl_->block_type[b];
+      if (t < 0 || t > 3) throw std::runtime_error("bad block type");
+      if (t != 0) {
+        if (cur[t] >= locs[t].size())
+          throw std::runtime_error("too many blocks of type " +
+                                   std::to_string(t) + " for grid");
+        auto [x, y] = locs[t][cur[t]++];
         bx_[b] = x; by_[b] = y; bslot_[b] = 0;
         grid_at(x, y, 0) = b; grid_cnt_[(size_t)x * gy_ + y]++;
       } else {
-        if (ii >= io_locs.size()) throw std::runtime_error("too many IOs for grid");
-        auto [x, y] = io_locs[ii];
+        if (cur[0] >= locs[0].size()) throw std::runtime_error("too many IOs for grid");
+        auto [x, y] = locs[0][cur[0]];
         bx_[b] = x; by_[b] = y; bslot_[b] = io_slot;
         grid_at(x, y, io_slot) = b; grid_cnt_[(size_t)x * gy_ + y]++;
-        if (++io_slot >= io_cap_) { io_slot = 0; ++ii; }
+        if (++io_slot >= io_cap_) { io_slot = 0; ++cur[0]; }
       }
     }
   }
@@ -318,21 +359,43 @@ class SerialPlacer {
     int nb = nl_->num_blocks;
     int blk = (int)(rng_() % nb);
     int x0 = bx_[blk], y0 = by_[blk];
-    bool io = nl_->block_type[blk] == 0;
+    int btype = nl_->block_type[blk];
     // find_to: range-limited destination of matching type (place.c:1520)
     int irlim = std::max(1, (int)rlim);
     int x1 = -1, y1 = -1, slot1 = 0;
-    for (int attempt = 0; attempt < 12; ++attempt) {
-      int dx = (int)(rng_() % (2 * irlim + 1)) - irlim;
-      int dy = (int)(rng_() % (2 * irlim + 1)) - irlim;
-      int tx = x0 + dx, ty = y0 + dy;
-      if (tx < 0 || tx >= gx_ || ty < 0 || ty >= gy_) continue;
-      if (is_io_loc(tx, ty) != io) continue;
-      if (cap_at(tx, ty) <= 0) continue;
-      if (tx == x0 && ty == y0) continue;
-      x1 = tx; y1 = ty;
-      slot1 = (int)(rng_() % cap_at(tx, ty));
-      break;
+    if (btype >= 2) {
+      // sparse column types (RAM/DSP): draw the target column from this
+      // type's column list clipped to the range window — rejection over
+      // the square window would nearly always miss sparse columns.
+      const std::vector<int>& cols = type_cols_[btype - 2];
+      auto lo = std::lower_bound(cols.begin(), cols.end(), x0 - irlim);
+      auto hi = std::upper_bound(cols.begin(), cols.end(), x0 + irlim);
+      int ncol = (int)(hi - lo);
+      if (ncol > 0) {
+        for (int attempt = 0; attempt < 12; ++attempt) {
+          int tx = *(lo + (int)(rng_() % ncol));
+          int ylo = std::max(1, y0 - irlim), yhi = std::min(ny_, y0 + irlim);
+          int ty = ylo + (int)(rng_() % (yhi - ylo + 1));
+          if (tx == x0 && ty == y0) continue;
+          x1 = tx; y1 = ty; slot1 = 0;
+          break;
+        }
+      }
+    } else {
+      bool io = btype == 0;
+      for (int attempt = 0; attempt < 12; ++attempt) {
+        int dx = (int)(rng_() % (2 * irlim + 1)) - irlim;
+        int dy = (int)(rng_() % (2 * irlim + 1)) - irlim;
+        int tx = x0 + dx, ty = y0 + dy;
+        if (tx < 0 || tx >= gx_ || ty < 0 || ty >= gy_) continue;
+        if (is_io_loc(tx, ty) != io) continue;
+        if (tile_type(tx, ty) != btype) continue;
+        if (cap_at(tx, ty) <= 0) continue;
+        if (tx == x0 && ty == y0) continue;
+        x1 = tx; y1 = ty;
+        slot1 = (int)(rng_() % cap_at(tx, ty));
+        break;
+      }
     }
     if (x1 < 0) return 0;
     int other = grid_at(x1, y1, slot1);

@@ -6,6 +6,7 @@
 // pointer-per-node graph (vpr/SRC/parallel_route/new_rr_graph.h:10-63,
 // graph.h) with a flat CSR layout.
 #pragma once
+#include <algorithm>
 #include <cstdint>
 #include <memory>
 #include <vector>
@@ -39,9 +40,24 @@ struct ArchParams {
   int fc_in, fc_out;
   int clb_in, clb_out;
   int io_cap;
+  // heterogeneous column tiles (0 = none); column phases mirror
+  // parallel_eda_amd/arch/archdef.py col_block_type
+  int ram_col_every = 0, dsp_col_every = 0;
+  int ram_in = 0, ram_out = 0, dsp_in = 0, dsp_out = 0;
   float R_wire, C_wire, R_sw, C_sw_in, T_sw, T_opin, T_ipin;
   float base_cost[6];
 };
+
+// Block type of logic column x (1..nx): 1=CLB 2=RAM 3=DSP.
+inline int col_btype(const ArchParams& ap, int x) {
+  if (ap.ram_col_every > 0 &&
+      x % ap.ram_col_every == std::min(2, ap.ram_col_every - 1))
+    return 2;
+  if (ap.dsp_col_every > 0 &&
+      x % ap.dsp_col_every == std::min(5, ap.dsp_col_every - 1))
+    return 3;
+  return 1;
+}
 
 struct RRGraph {
   int nx = 0, ny = 0, W = 0, L = 0;

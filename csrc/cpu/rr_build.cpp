@@ -107,10 +107,16 @@ RRGraph build_rr_graph(const ArchParams& ap) {
     }
   };
 
-  // CLB tiles
-  for (int x = 1; x <= nx; ++x)
+  // Logic tiles: per-column block type (CLB / RAM column / DSP column;
+  // homogeneous when ram_col_every == dsp_col_every == 0)
+  for (int x = 1; x <= nx; ++x) {
+    int bt = col_btype(ap, x);
+    int t_in = ap.clb_in, t_out = ap.clb_out;
+    if (bt == 2) { t_in = ap.ram_in; t_out = ap.ram_out; }
+    else if (bt == 3) { t_in = ap.dsp_in; t_out = ap.dsp_out; }
     for (int y = 1; y <= ny; ++y)
-      make_tile(x, y, ap.clb_in, ap.clb_out, 1, {0, 1, 2, 3});
+      make_tile(x, y, t_in, t_out, 1, {0, 1, 2, 3});
+  }
   // IO tiles: io_cap slots, each with 1 OPIN + 1 IPIN, one facing side.
   for (int y = 1; y <= ny; ++y) {
     make_tile(0, y, ap.io_cap, ap.io_cap, ap.io_cap, {1});       // left edge faces RIGHT

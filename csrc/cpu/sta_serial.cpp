@@ -16,8 +16,15 @@ namespace pnr {
 
 class TimingGraph {
  public:
-  TimingGraph(const Netlist* nl, float T_clb, float T_seq_out, float T_seq_in)
-      : nl_(nl), T_clb_(T_clb), T_seq_out_(T_seq_out), T_seq_in_(T_seq_in) {
+  // blk_delay: optional per-block combinational propagation delay
+  // (heterogeneous blocks: CLB/RAM/DSP differ); empty => T_clb everywhere.
+  TimingGraph(const Netlist* nl, float T_clb, float T_seq_out, float T_seq_in,
+              std::vector<float> blk_delay = {})
+      : nl_(nl), T_clb_(T_clb), T_seq_out_(T_seq_out), T_seq_in_(T_seq_in),
+        blk_delay_(std::move(blk_delay)) {
+    if (blk_delay_.empty()) blk_delay_.assign(nl_->num_blocks, T_clb_);
+    if (blk_delay_.size() != (size_t)nl_->num_blocks)
+      throw std::runtime_error("blk_delay size mismatch");
     levelize();
   }
 
@@ -32,7 +39,7 @@ class TimingGraph {
       bool seq = nl_->block_is_seq[b];
       if (seq) { t_arr_[b] = T_seq_out_; continue; }
       float a = in_arrival(b, conn_delay);
-      t_arr_[b] = a + T_clb_;
+      t_arr_[b] = a + blk_delay_[b];
     }
     // cpd = max arrival at any sequential/output endpoint input
     float cpd = 0.0f;
@@ -51,7 +58,7 @@ class TimingGraph {
         int64_t conn = out_conn_[k];
         int snk = nl_->net_sinks[conn];
         float req_in = nl_->block_is_seq[snk] ? (cpd - T_seq_in_)
-                                              : (t_req_[snk] - T_clb_);
+                                              : (t_req_[snk] - blk_delay_[snk]);
         float rr = req_in - conn_delay[conn];
         if (rr < r) r = rr;
       }
@@ -63,7 +70,7 @@ class TimingGraph {
       int drv = conn_driver_[c];
       int snk = nl_->net_sinks[c];
       float req_in = nl_->block_is_seq[snk] ? (cpd - T_seq_in_)
-                                            : (t_req_[snk] - T_clb_);
+                                            : (t_req_[snk] - blk_delay_[snk]);
       float s = req_in - (t_arr_[drv] + conn_delay[c]);
       slack[c] = s;
       float cr = 1.0f - s / cpd;
@@ -106,7 +113,7 @@ class TimingGraph {
           if (v > NEG) v += conn_delay[c];
           if (v > m) m = v;
         }
-        a[b] = (m > NEG) ? m + T_clb_ : NEG;
+        a[b] = (m > NEG) ? m + blk_delay_[b] : NEG;
       }
     }
     // backward per sink domain (required time at block outputs)
@@ -124,7 +131,7 @@ class TimingGraph {
           if (nl_->block_is_seq[snk])
             ri = (block_clock[snk] == cj) ? req_ep : POS;
           else
-            ri = (r[snk] < POS) ? r[snk] - T_clb_ : POS;
+            ri = (r[snk] < POS) ? r[snk] - blk_delay_[snk] : POS;
           if (ri < POS) ri -= conn_delay[c];
           if (ri < m) m = ri;
         }
@@ -145,7 +152,7 @@ class TimingGraph {
           if (nl_->block_is_seq[snk])
             ri = (block_clock[snk] == cj) ? constraint - T_seq_in_ : POS;
           else
-            ri = (req[cj][snk] < POS) ? req[cj][snk] - T_clb_ : POS;
+            ri = (req[cj][snk] < POS) ? req[cj][snk] - blk_delay_[snk] : POS;
           if (ri >= POS) continue;
           float s = ri - (arr[ci][drv] + conn_delay[c]);
           if (s < slack[c]) slack[c] = s;
@@ -184,6 +191,7 @@ class TimingGraph {
   std::shared_ptr<Netlist> netlist_holder_;  // lifetime pin for Python bindings
   const Netlist* nl_;
   float T_clb_, T_seq_out_, T_seq_in_;
+  std::vector<float> blk_delay_;
 
  private:
   std::vector<int32_t> topo_, level_;

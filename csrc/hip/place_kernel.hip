@@ -18,7 +18,9 @@ struct PlaceDev {
   const int32_t* net_blks;      //   (driver first, then sinks; may repeat)
   const int32_t* blk_net_ptr;   // [num_blocks+1] CSR: block -> nets (deduped)
   const int32_t* blk_nets;
-  const int8_t* blk_type;       // 0=IO 1=CLB
+  const int8_t* blk_type;       // 0=IO 1=CLB 2=RAM 3=DSP
+  const int8_t* tile_btype;     // [gx*gy] tile block type, -1 corner;
+                                // nullptr => homogeneous (perimeter IO)
   const float* net_q;           // crossing factor per net
   // timing
   const int32_t* net_sink_ptr;  // [num_nets+1] conn ranges
@@ -129,7 +131,10 @@ __global__ void place_propose_kernel(PlaceDev p, MovesDev m, float T,
     int tx = x0 + (int)(r1 % (2 * rlim + 1)) - rlim;
     int ty = y0 + (int)(r2 % (2 * rlim + 1)) - rlim;
     if (tx < 0 || tx >= p.gx || ty < 0 || ty >= p.gy) continue;
-    if (is_io_loc(p, tx, ty) != io) continue;
+    if (p.tile_btype) {
+      // heterogeneous fabric: destination tile must match the block type
+      if (p.tile_btype[tx * p.gy + ty] != p.blk_type[blk]) continue;
+    } else if (is_io_loc(p, tx, ty) != io) continue;
     int c = cap_at(p, tx, ty);
     if (c <= 0) continue;
     if (tx == x0 && ty == y0) continue;
@@ -290,7 +295,7 @@ extern "C" {
 struct PlaceLaunchArgs {
   const int32_t* net_blk_ptr; const int32_t* net_blks;
   const int32_t* blk_net_ptr; const int32_t* blk_nets;
-  const int8_t* blk_type; const float* net_q;
+  const int8_t* blk_type; const int8_t* tile_btype; const float* net_q;
   const int32_t* net_sink_ptr; const float* conn_crit; const float* delay_mat;
   int32_t* bx; int32_t* by; int32_t* bslot; int32_t* grid;
   float* net_cost; float* net_tcost;
@@ -310,7 +315,8 @@ struct PlaceLaunchArgs {
 static void unpack(const PlaceLaunchArgs* a, PlaceDev& p, MovesDev& m) {
   p.net_blk_ptr = a->net_blk_ptr; p.net_blks = a->net_blks;
   p.blk_net_ptr = a->blk_net_ptr; p.blk_nets = a->blk_nets;
-  p.blk_type = a->blk_type; p.net_q = a->net_q;
+  p.blk_type = a->blk_type; p.tile_btype = a->tile_btype;
+  p.net_q = a->net_q;
   p.net_sink_ptr = a->net_sink_ptr; p.conn_crit = a->conn_crit;
   p.delay_mat = a->delay_mat;
   p.bx = a->bx; p.by = a->by; p.bslot = a->bslot; p.grid = a->grid;

@@ -21,6 +21,7 @@ struct StaDev {
   const int32_t* conn_driver;    // per conn
   const int32_t* conn_sink;      // per conn
   const uint8_t* is_seq;
+  const float* blk_delay;        // per-block comb delay; nullptr => T_clb
   float T_clb, T_seq_out, T_seq_in;
   int32_t num_blocks;
   int64_t num_conns;
@@ -41,7 +42,7 @@ __global__ void sta_forward_level(StaDev s, const float* __restrict__ delay,
     float v = s.t_arr[s.conn_driver[c]] + delay[c];
     a = fmaxf(a, v);
   }
-  s.t_arr[b] = a + s.T_clb;
+  s.t_arr[b] = a + (s.blk_delay ? s.blk_delay[b] : s.T_clb);
 }
 
 // cpd = max over seq endpoints of (input arrival + T_seq_in)
@@ -75,8 +76,9 @@ __global__ void sta_backward_level(StaDev s, const float* __restrict__ delay,
   for (int64_t k = s.out_ptr[b]; k < s.out_ptr[b + 1]; ++k) {
     int64_t c = s.out_conn[k];
     int32_t snk = s.conn_sink[c];
-    float req_in = s.is_seq[snk] ? (cpd - s.T_seq_in)
-                                 : (s.t_req[snk] - s.T_clb);
+    float req_in = s.is_seq[snk]
+        ? (cpd - s.T_seq_in)
+        : (s.t_req[snk] - (s.blk_delay ? s.blk_delay[snk] : s.T_clb));
     r = fminf(r, req_in - delay[c]);
   }
   s.t_req[b] = r;
@@ -90,8 +92,9 @@ __global__ void sta_slack_kernel(StaDev s, const float* __restrict__ delay,
   for (; c < s.num_conns; c += (int64_t)gridDim.x * blockDim.x) {
     int32_t drv = s.conn_driver[c];
     int32_t snk = s.conn_sink[c];
-    float req_in = s.is_seq[snk] ? (cpd - s.T_seq_in)
-                                 : (s.t_req[snk] - s.T_clb);
+    float req_in = s.is_seq[snk]
+        ? (cpd - s.T_seq_in)
+        : (s.t_req[snk] - (s.blk_delay ? s.blk_delay[snk] : s.T_clb));
     float sl = req_in - (s.t_arr[drv] + delay[c]);
     slack[c] = sl;
     float cr = 1.0f - sl * inv_cpd;
@@ -111,6 +114,7 @@ struct StaLaunchArgs {
   const int64_t* out_ptr; const int64_t* out_conn;
   const int32_t* conn_driver; const int32_t* conn_sink;
   const uint8_t* is_seq;
+  const float* blk_delay;  // nullptr => scalar T_clb
   float T_clb, T_seq_out, T_seq_in, max_crit;
   int32_t num_blocks, num_levels;
   int64_t num_conns;
@@ -122,11 +126,13 @@ struct StaLaunchArgs {
 int pnr_sta_analyze(const StaLaunchArgs* a, void* stream) {
   StaDev s{a->level_blocks, a->level_start, a->in_ptr, a->in_conn,
            a->out_ptr, a->out_conn, a->conn_driver, a->conn_sink, a->is_seq,
+           a->blk_delay,
            a->T_clb, a->T_seq_out, a->T_seq_in, a->num_blocks, a->num_conns,
            a->t_arr, a->t_req, a->cpd_out};
   hipStream_t st = (hipStream_t)stream;
   const int32_t* ls = a->level_start_host;
-  hipMemsetAsync(a->cpd_out, 0, sizeof(float), st);
+  hipError_t me = hipMemsetAsync(a->cpd_out, 0, sizeof(float), st);
+  if (me != hipSuccess) return (int)me;
   for (int lv = 0; lv < a->num_levels; ++lv) {
     int n = ls[lv + 1] - ls[lv];
     if (n <= 0) continue;

@@ -22,6 +22,10 @@ from dataclasses import dataclass, asdict
 # Block type ids (placement + netlist)
 BLK_IO = 0
 BLK_CLB = 1
+BLK_RAM = 2   # memory-block column tiles (stratixiv M9K / mem32K class)
+BLK_DSP = 3   # DSP/multiplier column tiles
+
+BLK_NAMES = ["io", "clb", "ram", "dsp"]
 
 # RR node types (must match csrc/cpu/pnr_types.h)
 RR_SOURCE = 0
@@ -48,6 +52,21 @@ class ArchDef:
     clb_out: int = 10     # CLB output pins (≡ SOURCE capacity)
     io_cap: int = 8       # IO slots per perimeter tile
 
+    # Heterogeneous column tiles (reference: libarchfpga grid types with
+    # column-repeat fill patterns, physical_types.h grid_loc_def; stratixiv /
+    # k6_frac_N10_mem32K arches place RAM / DSP hard blocks in dedicated
+    # columns). 0 = no such columns (homogeneous CLB fabric).
+    # Column assignment (mirrored in csrc/cpu/rr_build.cpp col_btype):
+    #   RAM  columns: x % ram_col_every == min(2, ram_col_every - 1)
+    #   DSP  columns: x % dsp_col_every == min(5, dsp_col_every - 1),
+    #                 RAM takes precedence on collision.
+    ram_col_every: int = 0
+    dsp_col_every: int = 0
+    ram_in: int = 32      # RAM block input pins
+    ram_out: int = 32     # RAM block output pins
+    dsp_in: int = 36
+    dsp_out: int = 18
+
     # Timing (seconds / ohms / farads); values in the range of 40nm arches.
     R_wire: float = 101.0      # ohm per tile of wire
     C_wire: float = 22.5e-15   # farad per tile of wire
@@ -59,6 +78,8 @@ class ArchDef:
     T_clb: float = 261e-12     # CLB combinational (in->out) delay
     T_seq_out: float = 124e-12 # clock-to-Q
     T_seq_in: float = 66e-12   # setup
+    T_ram: float = 1.5e-9      # RAM block delay (seq: informational)
+    T_dsp: float = 1.2e-9      # DSP block combinational delay
 
     # Congestion base costs by rr type (SOURCE,SINK,OPIN,IPIN,CHANX,CHANY)
     # (reference: rr_graph_indexed_data.c base costs)
@@ -70,6 +91,60 @@ class ArchDef:
 
     def num_io_slots(self):
         return 2 * (self.nx + self.ny) * self.io_cap
+
+    # ---- heterogeneous-tile helpers ----
+    def is_heterogeneous(self):
+        return self.ram_col_every > 0 or self.dsp_col_every > 0
+
+    def col_block_type(self, x: int) -> int:
+        """Block type of logic column x (1..nx). Mirrors rr_build.cpp."""
+        if self.ram_col_every > 0 and \
+                x % self.ram_col_every == min(2, self.ram_col_every - 1):
+            return BLK_RAM
+        if self.dsp_col_every > 0 and \
+                x % self.dsp_col_every == min(5, self.dsp_col_every - 1):
+            return BLK_DSP
+        return BLK_CLB
+
+    def pins_of(self, btype: int):
+        """(n_in, n_out) pin counts of a block type."""
+        if btype == BLK_CLB:
+            return self.clb_in, self.clb_out
+        if btype == BLK_RAM:
+            return self.ram_in, self.ram_out
+        if btype == BLK_DSP:
+            return self.dsp_in, self.dsp_out
+        return self.io_cap, self.io_cap  # IO tile: io_cap 1-pin slots
+
+    def block_delay_of(self, btype: int) -> float:
+        """Combinational propagation delay of a block type (STA)."""
+        if btype == BLK_RAM:
+            return self.T_ram
+        if btype == BLK_DSP:
+            return self.T_dsp
+        return self.T_clb
+
+    def tile_btype_grid(self):
+        """(nx+2)*(ny+2) int8 grid of tile block types, x-major
+        (index x*(ny+2)+y, matching RRGraph.tile_id). -1 = unusable corner."""
+        import numpy as np
+        gx, gy = self.nx + 2, self.ny + 2
+        g = np.full(gx * gy, -1, dtype=np.int8)
+        for x in range(1, self.nx + 1):
+            t = self.col_block_type(x)
+            g[x * gy + 1: x * gy + self.ny + 1] = t
+            g[x * gy + 0] = BLK_IO
+            g[x * gy + self.ny + 1] = BLK_IO
+        for y in range(1, self.ny + 1):
+            g[0 * gy + y] = BLK_IO
+            g[(self.nx + 1) * gy + y] = BLK_IO
+        return g
+
+    def num_tiles_of_type(self, btype: int) -> int:
+        if btype == BLK_IO:
+            return 2 * (self.nx + self.ny)
+        return sum(1 for x in range(1, self.nx + 1)
+                   if self.col_block_type(x) == btype) * self.ny
 
     def to_dict(self):
         return asdict(self)
@@ -100,6 +175,20 @@ BUILTIN_ARCHES = {
     "bitcoin_miner": _scaled("bitcoin_stratixiv", 280, 200, 300, L=4,
                              clb_in=52, clb_out=20, fc_in=12, fc_out=12,
                              io_cap=16),
+    # --- heterogeneous fabrics (RAM/DSP column tiles) ---
+    # unit-test het arch: RAM cols x=2,6; DSP col x=5 (nx=8)
+    "tiny_het": _scaled("tiny_het", 8, 6, 16, L=2, clb_in=6, clb_out=2,
+                        fc_in=4, fc_out=4, io_cap=2,
+                        ram_col_every=4, dsp_col_every=8,
+                        ram_in=8, ram_out=4, dsp_in=8, dsp_out=4),
+    # k6_frac_N10_mem32K_40nm-like: memory column every 8
+    "mem32K": _scaled("k6n10_mem32K", 40, 40, 80, ram_col_every=8,
+                      ram_in=40, ram_out=32),
+    # stratixiv-like with M9K RAM columns + DSP columns at bitcoin scale
+    "bitcoin_miner_het": _scaled("bitcoin_stratixiv_het", 280, 200, 300, L=4,
+                                 clb_in=52, clb_out=20, fc_in=12, fc_out=12,
+                                 io_cap=16, ram_col_every=8, dsp_col_every=16,
+                                 ram_in=64, ram_out=32, dsp_in=72, dsp_out=36),
 }
 
 

@@ -13,7 +13,7 @@ from dataclasses import dataclass
 
 import numpy as np
 
-from ..arch.archdef import ArchDef, BLK_IO, BLK_CLB
+from ..arch.archdef import ArchDef, BLK_IO, BLK_CLB, BLK_RAM, BLK_DSP
 from .. import ops
 
 
@@ -26,15 +26,24 @@ class SynthSpec:
     seq_frac: float = 0.35   # fraction of CLBs that are sequential
     max_fanin: int = 16      # block input-pin budget (arch.clb_in)
     seed: int = 1
+    # heterogeneous blocks (RAM columns / DSP columns); RAM blocks are
+    # sequential (registered outputs), DSP blocks combinational.
+    n_ram: int = 0
+    n_dsp: int = 0
+    ram_fanin: int = 32
+    dsp_fanin: int = 36
 
 
 def spec_for_arch(arch: ArchDef, fill: float = 0.85, seed: int = 1) -> SynthSpec:
-    n_clb = int(arch.nx * arch.ny * fill)
+    n_clb = int(arch.num_tiles_of_type(BLK_CLB) * fill)
+    n_ram = int(arch.num_tiles_of_type(BLK_RAM) * fill)
+    n_dsp = int(arch.num_tiles_of_type(BLK_DSP) * fill)
     n_io = max(4, int(0.12 * n_clb))
     n_io = min(n_io, arch.num_io_slots() // 2 - 2)
     return SynthSpec(n_clb=n_clb, n_in=max(2, n_io // 2),
                      n_out=max(2, n_io // 2), max_fanin=arch.clb_in,
-                     seed=seed)
+                     seed=seed, n_ram=n_ram, n_dsp=n_dsp,
+                     ram_fanin=arch.ram_in, dsp_fanin=arch.dsp_in)
 
 
 class NetlistPy:
@@ -90,23 +99,47 @@ def synth_placed_netlist(arch: ArchDef, fill: float = 0.85,
     from ..place.placer import Placement
     rng = np.random.default_rng(seed)
     nx, ny, io_cap = arch.nx, arch.ny, arch.io_cap
-    n_clb = int(nx * ny * fill)
+    # per-type logic tiles (x-major flat index x_idx*ny + y_idx, matching
+    # the homogeneous generator's tile ordering)
+    col_t = np.asarray([arch.col_block_type(x) for x in range(1, nx + 1)])
+    type_tiles = {}
+    for t in (BLK_CLB, BLK_RAM, BLK_DSP):
+        cols = np.nonzero(col_t == t)[0]          # 0-based column indices
+        if len(cols):
+            type_tiles[t] = (cols[:, None] * ny + np.arange(ny)[None, :]).ravel()
+    n_of = {t: int(len(v) * fill) for t, v in type_tiles.items()}
+    n_clb = n_of.get(BLK_CLB, 0)
+    n_ram = n_of.get(BLK_RAM, 0)
+    n_dsp = n_of.get(BLK_DSP, 0)
+    n_logic = n_clb + n_ram + n_dsp
     n_io_pairs = max(2, min(int(0.06 * n_clb), (nx + ny) * io_cap // 2 - 2))
     n_in = n_out = n_io_pairs
     n_io = n_in + n_out
-    nb = n_io + n_clb
+    nb = n_io + n_logic
 
     block_type = np.full(nb, BLK_CLB, dtype=np.int8)
     block_type[:n_io] = BLK_IO
     block_is_seq = np.zeros(nb, dtype=np.uint8)
     block_is_seq[:n_io] = 1
     clb0 = n_io
-    block_is_seq[clb0:] = (rng.random(n_clb) < seq_frac).astype(np.uint8)
+    ram0 = clb0 + n_clb
+    dsp0 = ram0 + n_ram
+    block_type[ram0:dsp0] = BLK_RAM
+    block_type[dsp0:] = BLK_DSP
+    block_is_seq[clb0:ram0] = (rng.random(n_clb) < seq_frac).astype(np.uint8)
+    block_is_seq[ram0:dsp0] = 1   # RAM: registered outputs
+    # DSP blocks combinational (rank-ordered below)
 
-    # placement: CLBs into random distinct tiles; IOs into perimeter slots
-    clb_tiles = rng.choice(nx * ny, size=n_clb, replace=False)
-    cx = (clb_tiles // ny + 1).astype(np.int32)
-    cy = (clb_tiles % ny + 1).astype(np.int32)
+    # placement: logic blocks into random distinct tiles of their type;
+    # IOs into perimeter slots
+    parts = []
+    for t in (BLK_CLB, BLK_RAM, BLK_DSP):
+        if n_of.get(t, 0):
+            pick = rng.choice(len(type_tiles[t]), size=n_of[t], replace=False)
+            parts.append(type_tiles[t][pick])
+    logic_tiles = np.concatenate(parts) if parts else np.empty(0, np.int64)
+    cx = (logic_tiles // ny + 1).astype(np.int32)
+    cy = (logic_tiles % ny + 1).astype(np.int32)
     io_locs = ([(0, y) for y in range(1, ny + 1)] +
                [(nx + 1, y) for y in range(1, ny + 1)] +
                [(x, 0) for x in range(1, nx + 1)] +
@@ -121,24 +154,24 @@ def synth_placed_netlist(arch: ArchDef, fill: float = 0.85,
         bx[i], by[i], bslot[i] = x, y, s
     bx[clb0:], by[clb0:] = cx, cy
 
-    # spatial index of CLBs for locality draws
-    order = np.lexsort((cy, cx))
-    # nets: every input pad + every CLB drives one net
+    # nets: every input pad + every logic block drives one net
     drivers = np.concatenate([
         np.arange(n_in, dtype=np.int32),
         np.arange(clb0, nb, dtype=np.int32)])
     n_nets = len(drivers)
-    rank = rng.permutation(n_clb)
+    rank = rng.permutation(n_logic)
 
     fanin = np.zeros(nb, dtype=np.int64)
     max_fanin = np.full(nb, arch.clb_in, dtype=np.int64)
     max_fanin[:n_io] = 1
-    # KD-free locality: bucket CLBs per tile-cell grid
+    max_fanin[ram0:dsp0] = arch.ram_in
+    max_fanin[dsp0:] = arch.dsp_in
+    # KD-free locality: bucket logic blocks per tile-cell grid
     cell = max(2, int(locality))
     ncx = (nx + cell - 1) // cell
     ncy = (ny + cell - 1) // cell
     buckets = [[] for _ in range(ncx * ncy)]
-    for i in range(n_clb):
+    for i in range(n_logic):
         bxi = min((cx[i] - 1) // cell, ncx - 1)
         byi = min((cy[i] - 1) // cell, ncy - 1)
         buckets[bxi * ncy + byi].append(i)
@@ -169,7 +202,7 @@ def synth_placed_netlist(arch: ArchDef, fill: float = 0.85,
             # locality: pick a cell near the driver (geometric radius),
             # occasionally a uniform long-range sink
             if rng.random() < 0.08:
-                c = int(rng.integers(n_clb))
+                c = int(rng.integers(n_logic))
             else:
                 r = rng.geometric(1.0 / max(1.0, locality / cell))
                 ang = rng.random() * 2 * np.pi
@@ -191,7 +224,7 @@ def synth_placed_netlist(arch: ArchDef, fill: float = 0.85,
             sinks.add(blk)
             fanin[blk] += 1
         if not sinks:
-            for c in rng.permutation(n_clb)[:64]:
+            for c in rng.permutation(n_logic)[:64]:
                 blk = clb0 + int(c)
                 if blk != drv and (block_is_seq[blk] or drv_rank < 0 or
                                    rank[int(c)] > drv_rank) \
@@ -227,14 +260,22 @@ def synth_netlist(spec: SynthSpec) -> NetlistPy:
     """
     rng = np.random.default_rng(spec.seed)
     n_io = spec.n_in + spec.n_out
-    nb = n_io + spec.n_clb
+    # logic blocks: [clb0, clb0+n_clb) CLB, then RAM, then DSP
+    n_logic = spec.n_clb + spec.n_ram + spec.n_dsp
+    nb = n_io + n_logic
     block_type = np.full(nb, BLK_CLB, dtype=np.int8)
     block_type[:n_io] = BLK_IO
     block_is_seq = np.zeros(nb, dtype=np.uint8)
     block_is_seq[:n_io] = 1  # pads are timing endpoints
     clb0 = n_io
+    ram0 = clb0 + spec.n_clb
+    dsp0 = ram0 + spec.n_ram
+    block_type[ram0:dsp0] = BLK_RAM
+    block_type[dsp0:] = BLK_DSP
     seq_mask = rng.random(spec.n_clb) < spec.seq_frac
-    block_is_seq[clb0:] = seq_mask.astype(np.uint8)
+    block_is_seq[clb0:ram0] = seq_mask.astype(np.uint8)
+    block_is_seq[ram0:dsp0] = 1  # RAM: registered outputs
+    # DSP blocks stay combinational (rank-ordered like comb CLBs)
 
     # Drivers: every input pad and every CLB drives exactly one net.
     drivers = np.concatenate([
@@ -243,10 +284,10 @@ def synth_netlist(spec: SynthSpec) -> NetlistPy:
     ])
     n_nets = len(drivers)
 
-    # Comb rank: random permutation of CLBs; an edge into a COMB clb must
-    # come from a driver with lower rank (or from a seq block / pad).
-    rank = np.empty(spec.n_clb, dtype=np.int64)
-    rank[:] = rng.permutation(spec.n_clb)
+    # Comb rank: random permutation of logic blocks; an edge into a COMB
+    # block must come from a driver with lower rank (or a seq block / pad).
+    rank = np.empty(n_logic, dtype=np.int64)
+    rank[:] = rng.permutation(n_logic)
 
     # Sink candidate pools with block fan-in budgets: a CLB accepts at most
     # max_fanin incoming connections (it has that many input pins); an
@@ -254,6 +295,8 @@ def synth_netlist(spec: SynthSpec) -> NetlistPy:
     fanin = np.zeros(nb, dtype=np.int64)
     max_fanin = np.full(nb, spec.max_fanin, dtype=np.int64)
     max_fanin[:n_io] = 1  # pads: single pin
+    max_fanin[ram0:dsp0] = spec.ram_fanin
+    max_fanin[dsp0:] = spec.dsp_fanin
     sink_lists = []
     out_pad_driven = np.zeros(spec.n_out, dtype=bool)
     for i, drv in enumerate(drivers):
@@ -274,7 +317,7 @@ def synth_netlist(spec: SynthSpec) -> NetlistPy:
                 out_pad_driven[o] = True
                 fanin[blk] += 1
                 continue
-            c = int(rng.integers(spec.n_clb))
+            c = int(rng.integers(n_logic))
             blk = clb0 + c
             if blk == drv or blk in sinks:
                 continue
@@ -297,7 +340,7 @@ def synth_netlist(spec: SynthSpec) -> NetlistPy:
                     placed = True
                     break
             if not placed:
-                for c in rng.permutation(spec.n_clb):
+                for c in rng.permutation(n_logic):
                     blk = clb0 + int(c)
                     if blk != drv and block_is_seq[blk] and fanin[blk] < max_fanin[blk]:
                         sinks.add(blk)

@@ -19,7 +19,8 @@ class PlaceLaunchArgs(ct.Structure):
     _fields_ = [
         ("net_blk_ptr", ct.c_void_p), ("net_blks", ct.c_void_p),
         ("blk_net_ptr", ct.c_void_p), ("blk_nets", ct.c_void_p),
-        ("blk_type", ct.c_void_p), ("net_q", ct.c_void_p),
+        ("blk_type", ct.c_void_p), ("tile_btype", ct.c_void_p),
+        ("net_q", ct.c_void_p),
         ("net_sink_ptr", ct.c_void_p), ("conn_crit", ct.c_void_p),
         ("delay_mat", ct.c_void_p),
         ("bx", ct.c_void_p), ("by", ct.c_void_p), ("bslot", ct.c_void_p),
@@ -104,6 +105,10 @@ class GpuPlacer:
         self.t_blk_net_ptr = up(blk_net_ptr)
         self.t_blk_nets = up(blk_nets)
         self.t_blk_type = up(netlist.block_type)
+        # heterogeneous fabrics carry a tile-type grid; homogeneous pass
+        # nullptr so the validated perimeter-IO fast path runs unchanged
+        self.t_tile_btype = (up(arch.tile_btype_grid())
+                             if arch.is_heterogeneous() else None)
         self.t_net_q = up(q)
         self.t_net_sink_ptr = up(netlist.net_sink_ptr.astype(np.int32))
         self.t_conn_crit = torch.zeros(netlist.num_conns, dtype=torch.float32,
@@ -149,14 +154,31 @@ class GpuPlacer:
         by = np.zeros(nb, dtype=np.int32)
         bslot = np.zeros(nb, dtype=np.int32)
         grid = np.full((self.gx, self.gy, self.cap), -1, dtype=np.int32)
-        clbs = np.nonzero(nl.block_type == 1)[0]
         ios = np.nonzero(nl.block_type == 0)[0]
-        tiles = rng.permutation(arch.nx * arch.ny)[:len(clbs)]
-        if len(tiles) < len(clbs):
-            raise ValueError("too many CLBs for grid")
-        bx[clbs] = tiles // arch.ny + 1
-        by[clbs] = tiles % arch.ny + 1
-        grid[bx[clbs], by[clbs], 0] = clbs
+        if not arch.is_heterogeneous():
+            clbs = np.nonzero(nl.block_type == 1)[0]
+            tiles = rng.permutation(arch.nx * arch.ny)[:len(clbs)]
+            if len(tiles) < len(clbs):
+                raise ValueError("too many CLBs for grid")
+            bx[clbs] = tiles // arch.ny + 1
+            by[clbs] = tiles % arch.ny + 1
+            grid[bx[clbs], by[clbs], 0] = clbs
+        else:
+            col_t = np.asarray([arch.col_block_type(x)
+                                for x in range(1, arch.nx + 1)])
+            for t in (1, 2, 3):
+                blks = np.nonzero(nl.block_type == t)[0]
+                if not len(blks):
+                    continue
+                cols = np.nonzero(col_t == t)[0]
+                cand = (cols[:, None] * arch.ny +
+                        np.arange(arch.ny)[None, :]).ravel()
+                if len(blks) > len(cand):
+                    raise ValueError(f"too many type-{t} blocks for grid")
+                tiles = cand[rng.permutation(len(cand))[:len(blks)]]
+                bx[blks] = tiles // arch.ny + 1
+                by[blks] = tiles % arch.ny + 1
+                grid[bx[blks], by[blks], 0] = blks
         io_locs = ([(0, y) for y in range(1, arch.ny + 1)] +
                    [(self.gx - 1, y) for y in range(1, arch.ny + 1)] +
                    [(x, 0) for x in range(1, arch.nx + 1)] +
@@ -173,7 +195,10 @@ class GpuPlacer:
         a = PlaceLaunchArgs()
         a.net_blk_ptr = ptr(self.t_net_blk_ptr); a.net_blks = ptr(self.t_net_blks)
         a.blk_net_ptr = ptr(self.t_blk_net_ptr); a.blk_nets = ptr(self.t_blk_nets)
-        a.blk_type = ptr(self.t_blk_type); a.net_q = ptr(self.t_net_q)
+        a.blk_type = ptr(self.t_blk_type)
+        a.tile_btype = (ptr(self.t_tile_btype)
+                        if self.t_tile_btype is not None else None)
+        a.net_q = ptr(self.t_net_q)
         a.net_sink_ptr = ptr(self.t_net_sink_ptr)
         a.conn_crit = ptr(self.t_conn_crit)
         a.delay_mat = ptr(self.t_delay_mat) if self.t_delay_mat is not None else None
@@ -252,12 +277,19 @@ class GpuPlacer:
         bslot = self.t_bslot.cpu().numpy()
         grid = self.t_grid.cpu().numpy().reshape(self.gx, self.gy, self.cap)
         nl = self.nl
+        tb = (self.arch.tile_btype_grid().reshape(self.gx, self.gy)
+              if self.arch.is_heterogeneous() else None)
         for b in range(nl.num_blocks):
-            io = nl.block_type[b] == 0
             x, y = bx[b], by[b]
-            on_io = (x == 0 or x == self.gx - 1 or y == 0 or y == self.gy - 1)
-            if io != on_io:
-                return False, f"block {b} type/loc mismatch"
+            if tb is not None:
+                if tb[x, y] != nl.block_type[b]:
+                    return False, f"block {b} type/loc mismatch"
+            else:
+                io = nl.block_type[b] == 0
+                on_io = (x == 0 or x == self.gx - 1 or y == 0 or
+                         y == self.gy - 1)
+                if io != on_io:
+                    return False, f"block {b} type/loc mismatch"
             if grid[x, y, bslot[b]] != b:
                 return False, f"grid inconsistent at block {b}"
         occ = (grid >= 0).sum()

@@ -67,8 +67,10 @@ def anneal_place(netlist, arch: ArchDef, seed: int = 7, timing_tradeoff: float =
             dm = analytic_delay_matrix(arch)
     else:
         dm = np.zeros(0, dtype=np.float32)
+    tb = (arch.tile_btype_grid() if arch.is_heterogeneous()
+          else np.empty(0, dtype=np.int8))
     placer = cpu.SerialPlacer(netlist.cpp(), arch.nx, arch.ny, arch.io_cap,
-                              np.ascontiguousarray(dm.ravel()), seed)
+                              np.ascontiguousarray(dm.ravel()), seed, tb)
     nb = netlist.num_blocks
     move_lim = max(64, int(inner_num * (nb ** 1.3333)))
     rlim = float(max(arch.nx, arch.ny))

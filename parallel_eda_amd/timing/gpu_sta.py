@@ -20,7 +20,7 @@ class StaLaunchArgs(ct.Structure):
         ("in_ptr", ct.c_void_p), ("in_conn", ct.c_void_p),
         ("out_ptr", ct.c_void_p), ("out_conn", ct.c_void_p),
         ("conn_driver", ct.c_void_p), ("conn_sink", ct.c_void_p),
-        ("is_seq", ct.c_void_p),
+        ("is_seq", ct.c_void_p), ("blk_delay", ct.c_void_p),
         ("T_clb", ct.c_float), ("T_seq_out", ct.c_float),
         ("T_seq_in", ct.c_float), ("max_crit", ct.c_float),
         ("num_blocks", ct.c_int32), ("num_levels", ct.c_int32),
@@ -52,8 +52,12 @@ class GpuSTA:
         self.netlist = netlist
         self.max_crit = max_crit
         cpu = ops.cpu()
-        self.tg = cpu.TimingGraph(netlist.cpp(), arch.T_clb, arch.T_seq_out,
-                                  arch.T_seq_in)
+        from .sta import block_delays
+        bd = block_delays(netlist, arch)
+        self.tg = cpu.TimingGraph(
+            netlist.cpp(), arch.T_clb, arch.T_seq_out, arch.T_seq_in,
+            bd if bd is not None else np.empty(0, dtype=np.float32))
+        self._blk_delay_host = bd
         blocks, start = self.tg.level_arrays()
         in_ptr, in_conn, out_ptr, out_conn, conn_driver = self.tg.csr_arrays()
         self.level_start_host = np.ascontiguousarray(start, dtype=np.int32)
@@ -69,6 +73,10 @@ class GpuSTA:
         self.t_conn_driver = up(conn_driver)
         self.t_conn_sink = up(netlist.net_sinks)
         self.t_is_seq = up(netlist.block_is_seq)
+        # heterogeneous: per-block comb delay on device; homogeneous: null
+        # => the validated scalar-T_clb kernel path runs unchanged
+        self.t_blk_delay = (up(self._blk_delay_host)
+                            if self._blk_delay_host is not None else None)
         nb = netlist.num_blocks
         nc = netlist.num_conns
         self.t_arr = torch.zeros(nb, dtype=torch.float32, device=device)
@@ -94,6 +102,8 @@ class GpuSTA:
         a.conn_driver = pt(self.t_conn_driver)
         a.conn_sink = pt(self.t_conn_sink)
         a.is_seq = pt(self.t_is_seq)
+        a.blk_delay = (pt(self.t_blk_delay)
+                       if self.t_blk_delay is not None else None)
         a.T_clb = self.arch.T_clb; a.T_seq_out = self.arch.T_seq_out
         a.T_seq_in = self.arch.T_seq_in; a.max_crit = self.max_crit
         a.num_blocks = self.netlist.num_blocks

@@ -12,13 +12,28 @@ from ..arch.archdef import ArchDef
 from .. import ops
 
 
+def block_delays(netlist, arch: ArchDef):
+    """Per-block combinational delay by block type (CLB/RAM/DSP), or None
+    for a homogeneous arch (scalar T_clb path)."""
+    if not arch.is_heterogeneous():
+        return None
+    bt = np.asarray(netlist.block_type)
+    bd = np.full(len(bt), arch.T_clb, dtype=np.float32)
+    for t in (2, 3):  # BLK_RAM, BLK_DSP
+        bd[bt == t] = arch.block_delay_of(t)
+    return bd
+
+
 class STA:
     def __init__(self, netlist, arch: ArchDef):
         cpu = ops.cpu()
         self.netlist = netlist
         self.arch = arch
+        bd = block_delays(netlist, arch)
+        if bd is None:
+            bd = np.empty(0, dtype=np.float32)
         self.tg = cpu.TimingGraph(netlist.cpp(), arch.T_clb, arch.T_seq_out,
-                                  arch.T_seq_in)
+                                  arch.T_seq_in, bd)
 
     @property
     def num_levels(self):

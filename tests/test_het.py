@@ -1,0 +1,165 @@
+"""Heterogeneous fabric (RAM/DSP column tiles) tests.
+
+Reference scope: libarchfpga grid types with column-repeat fill patterns
+(physical_types.h grid_loc_def) as used by the stratixiv /
+k6_frac_N10_mem32K arches — RAM and DSP hard blocks live in dedicated
+columns with their own pin counts and delays, and the placer may only put
+a block on a tile of its own type.
+"""
+import numpy as np
+import pytest
+
+from parallel_eda_amd.arch.archdef import (get_arch, BLK_IO, BLK_CLB,
+                                           BLK_RAM, BLK_DSP)
+from parallel_eda_amd.io.synth import (synth_netlist, synth_placed_netlist,
+                                       spec_for_arch, NetlistPy)
+from parallel_eda_amd.place.placer import anneal_place
+from parallel_eda_amd.route.router import pathfinder_route
+from parallel_eda_amd.timing.sta import STA, block_delays
+from parallel_eda_amd import rrgraph
+
+
+@pytest.fixture(scope="module")
+def het_arch():
+    return get_arch("tiny_het")
+
+
+def test_column_types(het_arch):
+    a = het_arch
+    assert a.is_heterogeneous()
+    types = [a.col_block_type(x) for x in range(1, a.nx + 1)]
+    assert types.count(BLK_RAM) == a.num_tiles_of_type(BLK_RAM) // a.ny
+    assert types.count(BLK_DSP) == a.num_tiles_of_type(BLK_DSP) // a.ny
+    assert BLK_RAM in types and BLK_DSP in types and BLK_CLB in types
+    # grid agrees with per-column types; perimeter is IO; corners unusable
+    g = het_arch.tile_btype_grid()
+    gy = a.ny + 2
+    assert g[0] == -1 and g[gy - 1] == -1
+    for x in range(1, a.nx + 1):
+        assert g[x * gy + 0] == BLK_IO
+        for y in range(1, a.ny + 1):
+            assert g[x * gy + y] == types[x - 1]
+
+
+def test_rr_graph_het_capacities(het_arch):
+    """Per-column SOURCE/SINK capacities must match the tile type's pins
+    (the C++ builder's column formula must mirror archdef's)."""
+    a = het_arch
+    g = rrgraph.build_rr_graph(a)
+    rrgraph.check_rr_graph(g, a)
+    cap = np.asarray(g.capacity)
+    ts = np.asarray(g.tile_source)
+    tk = np.asarray(g.tile_sink)
+    ty = np.asarray(g.type)
+    row_ptr = np.asarray(g.row_ptr)
+    gy = a.ny + 2
+    for x in range(1, a.nx + 1):
+        n_in, n_out = a.pins_of(a.col_block_type(x))
+        for y in (1, a.ny):
+            src, snk = ts[x * gy + y], tk[x * gy + y]
+            assert cap[src] == n_out, (x, y)
+            assert cap[snk] == n_in, (x, y)
+            # SOURCE drives exactly n_out OPINs
+            assert row_ptr[src + 1] - row_ptr[src] == n_out
+            assert ty[src] == 0 and ty[snk] == 1
+
+
+def test_synth_het_netlist(het_arch):
+    spec = spec_for_arch(het_arch, fill=0.55, seed=5)
+    assert spec.n_ram > 0 and spec.n_dsp > 0
+    nl = synth_netlist(spec)
+    bt = np.asarray(nl.block_type)
+    assert int((bt == BLK_RAM).sum()) == spec.n_ram
+    assert int((bt == BLK_DSP).sum()) == spec.n_dsp
+    # RAM blocks are sequential; acyclicity: TimingGraph levelizes
+    seq = np.asarray(nl.block_is_seq)
+    assert (seq[bt == BLK_RAM] == 1).all()
+    sta = STA(nl, het_arch)  # raises on a combinational cycle
+    assert sta.num_levels >= 1
+    # fan-in budgets respected per type
+    fanin = np.zeros(nl.num_blocks, dtype=np.int64)
+    np.add.at(fanin, nl.net_sinks, 1)
+    for t, budget in ((BLK_RAM, spec.ram_fanin), (BLK_DSP, spec.dsp_fanin),
+                      (BLK_CLB, spec.max_fanin)):
+        assert (fanin[bt == t] <= budget).all(), t
+
+
+def test_block_delays(het_arch):
+    nl = synth_netlist(spec_for_arch(het_arch, fill=0.5, seed=2))
+    bd = block_delays(nl, het_arch)
+    bt = np.asarray(nl.block_type)
+    assert bd is not None
+    assert np.allclose(bd[bt == BLK_CLB], het_arch.T_clb)
+    assert np.allclose(bd[bt == BLK_DSP], het_arch.T_dsp)
+    # homogeneous arch: no per-block array (scalar kernel path)
+    hom = get_arch("tiny")
+    nl2 = synth_netlist(spec_for_arch(hom, fill=0.5, seed=2))
+    assert block_delays(nl2, hom) is None
+
+
+def test_sta_dsp_delay_on_path():
+    """Hand case: pad -> DSP -> pad. cpd must charge T_dsp, not T_clb."""
+    arch = get_arch("tiny_het")
+    # blocks: 0 = in pad (seq), 1 = out pad (seq), 2 = DSP (comb)
+    nl = NetlistPy(
+        block_type=[BLK_IO, BLK_IO, BLK_DSP],
+        block_is_seq=[1, 1, 0],
+        net_driver=[0, 2],
+        net_sink_ptr=[0, 1, 2],
+        net_sinks=[2, 1],
+    )
+    sta = STA(nl, arch)
+    d01, d21 = 1e-9, 2e-9
+    cpd, slack, crit = sta.analyze(np.asarray([d01, d21], dtype=np.float32))
+    expect = arch.T_seq_out + d01 + arch.T_dsp + d21 + arch.T_seq_in
+    assert cpd == pytest.approx(expect, rel=1e-5)
+
+
+def test_anneal_place_het_legality(het_arch):
+    nl = synth_netlist(spec_for_arch(het_arch, fill=0.55, seed=5))
+    sta = STA(nl, het_arch)
+    pl = anneal_place(nl, het_arch, seed=5, timing_tradeoff=0.5, sta=sta)
+    tb = het_arch.tile_btype_grid()
+    gy = het_arch.ny + 2
+    bt = np.asarray(nl.block_type)
+    for b in range(nl.num_blocks):
+        assert tb[pl.x[b] * gy + pl.y[b]] == bt[b], b
+
+
+def test_full_flow_het(het_arch):
+    """Place + route + timing on the heterogeneous fabric."""
+    nl = synth_netlist(spec_for_arch(het_arch, fill=0.55, seed=5))
+    sta = STA(nl, het_arch)
+    pl = anneal_place(nl, het_arch, seed=5, timing_tradeoff=0.5, sta=sta)
+    g = rrgraph.build_rr_graph(het_arch)
+    res = pathfinder_route(nl, pl, g, het_arch, sta=sta, max_iters=60)
+    assert res.success
+    ok, err = res.router.check_routed()
+    assert ok, err
+    # DSP comb delay (1.2 ns) dominates the CLB delay => with a DSP on a
+    # real path the cpd must exceed it
+    assert res.crit_path_delay > het_arch.T_dsp
+
+
+def test_synth_placed_het(het_arch):
+    nl, pl = synth_placed_netlist(het_arch, fill=0.6, seed=3)
+    bt = np.asarray(nl.block_type)
+    assert int((bt == BLK_RAM).sum()) > 0
+    tb = het_arch.tile_btype_grid()
+    gy = het_arch.ny + 2
+    for b in range(nl.num_blocks):
+        assert tb[pl.x[b] * gy + pl.y[b]] == bt[b], b
+    STA(nl, het_arch)  # acyclic
+
+
+def test_mem32K_arch_builds():
+    a = get_arch("mem32K")
+    g = rrgraph.build_rr_graph(a)
+    rrgraph.check_rr_graph(g, a)
+    cap = np.asarray(g.capacity)
+    ts = np.asarray(g.tile_source)
+    gy = a.ny + 2
+    ram_cols = [x for x in range(1, a.nx + 1)
+                if a.col_block_type(x) == BLK_RAM]
+    assert len(ram_cols) == a.nx // 8
+    assert cap[ts[ram_cols[0] * gy + 1]] == a.ram_out
