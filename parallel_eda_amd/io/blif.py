@@ -1,20 +1,42 @@
 """BLIF reader (technology-mapped netlists).
 
 Re-implements the semantics of the reference's read_blif.c (1,981 LoC):
-parses .model/.inputs/.outputs/.names/.latch, sweeps dangling nets, and
-produces a primitive-level netlist (LUTs + latches + IO pads) ready for
-packing (pack.py) into the block-level netlist the placer/router consume.
-.subckt is accepted only for simple single-output cells.
+parses .model/.inputs/.outputs/.names/.latch/.subckt, sweeps dangling
+nets, and produces a primitive-level netlist (LUTs + latches + hard
+blocks + IO pads) ready for packing (pack.py) into the block-level
+netlist the placer/router consume.
+
+.subckt instances are hard blocks (VTR convention: single_port_ram /
+dual_port_ram / multiply / adder ...): kept as one multi-output
+primitive each and mapped by the packer to a RAM or DSP block on a
+heterogeneous fabric (arch/archdef.py column tiles).
 """
 from dataclasses import dataclass, field
 
 
 @dataclass
 class BlifPrimitive:
-    kind: str               # "input", "output", "names", "latch"
+    kind: str               # "input", "output", "names", "latch", "subckt"
     name: str               # output signal name (or pad signal)
     inputs: list = field(default_factory=list)
     clock: str = ""
+    model: str = ""         # subckt model name
+    outputs: list = field(default_factory=list)  # subckt: all output sigs
+
+
+# output-formal prefixes of the VTR primitive models (single_port_ram:
+# "out"; dual_port_ram: "out1"/"out2"; multiply/adder: "out"/"cout"/"sumout")
+_SUBCKT_OUT_PREFIXES = ("out", "q", "dataout", "spo", "dpo", "sum", "cout")
+
+
+def subckt_class(model_name: str) -> str:
+    """Classify a subckt model: "ram", "dsp" or "soft" (unknown cell)."""
+    n = model_name.lower()
+    if "ram" in n or "mem" in n or "rom" in n:
+        return "ram"
+    if "mult" in n or "mac" in n or "dsp" in n or "add" in n:
+        return "dsp"
+    return "soft"
 
 
 @dataclass
@@ -29,7 +51,8 @@ class BlifModel:
         for s in self.inputs:
             drv[s] = ("input", s)
         for p in self.prims:
-            drv[p.name] = (p.kind, p.name)
+            for o in (p.outputs if p.kind == "subckt" else [p.name]):
+                drv[o] = (p.kind, o)
         return drv
 
 
@@ -84,13 +107,26 @@ def parse_blif(text) -> BlifModel:
             clock = toks[4] if len(toks) > 4 else ""
             model.prims.append(BlifPrimitive("latch", out, [inp], clock))
         elif key == ".subckt":
-            # minimal support: treat as a comb primitive with the last
-            # formal=actual as output
-            conns = [t.split("=") for t in toks[2:] if "=" in t]
-            if conns:
-                out = conns[-1][1]
-                ins = [c[1] for c in conns[:-1]]
-                model.prims.append(BlifPrimitive("names", out, ins))
+            # .subckt model formal=actual ... — split formals into
+            # inputs/outputs by the VTR model conventions; clock formals
+            # ("clk"/"clock") are global like latch clocks
+            mdl = toks[1] if len(toks) > 1 else "cell"
+            conns = [t.split("=", 1) for t in toks[2:] if "=" in t]
+            ins, outs, clock = [], [], ""
+            for f, a in conns:
+                fl = f.lower()
+                if fl in ("clk", "clock"):
+                    clock = a
+                elif fl.startswith(_SUBCKT_OUT_PREFIXES):
+                    outs.append(a)
+                else:
+                    ins.append(a)
+            if not outs and conns:  # unknown cell: last formal is output
+                outs = [conns[-1][1]]
+                ins = [a for _, a in conns[:-1]]
+            if outs:
+                model.prims.append(BlifPrimitive(
+                    "subckt", outs[0], ins, clock, mdl, outs))
         elif key == ".end":
             break
     sweep(model)
@@ -105,8 +141,12 @@ def sweep(model: BlifModel):
         used.update(p.inputs)
         if p.clock:
             used.add(p.clock)
-    model.prims = [p for p in model.prims
-                   if p.name in used or p.name in model.outputs]
+
+    def live(p):
+        outs = p.outputs if p.kind == "subckt" else [p.name]
+        return any(o in used for o in outs)
+
+    model.prims = [p for p in model.prims if live(p)]
 
 
 def read_blif(path) -> BlifModel:

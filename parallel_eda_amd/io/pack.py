@@ -11,8 +11,8 @@ from collections import defaultdict
 
 import numpy as np
 
-from ..arch.archdef import ArchDef, BLK_IO, BLK_CLB
-from .blif import BlifModel
+from ..arch.archdef import ArchDef, BLK_IO, BLK_CLB, BLK_RAM, BLK_DSP
+from .blif import BlifModel, subckt_class
 from .synth import NetlistPy
 
 
@@ -23,12 +23,26 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
     """
     prims = model.prims
     np_prims = len(prims)
+
+    def prim_outputs(i):
+        pp = prims[i]
+        return pp.outputs if pp.kind == "subckt" else [pp.name]
+
+    # hard blocks: RAM/DSP subckt instances are not clustered — each is
+    # its own block on a matching column tile (reference: VPR packs
+    # memory/mult molecules into their own block types, cluster.c)
+    hard_prims = [i for i, pp in enumerate(prims)
+                  if pp.kind == "subckt" and
+                  subckt_class(pp.model) in ("ram", "dsp")]
+    hard_set = set(hard_prims)
+
     # signal -> driving primitive index (or input pad)
     drv_of_sig = {}
     for s in model.inputs:
         drv_of_sig[s] = ("pad", s)
-    for i, p in enumerate(prims):
-        drv_of_sig[p.name] = ("prim", i)
+    for i in range(np_prims):
+        for o in prim_outputs(i):
+            drv_of_sig[o] = ("prim", i)
 
     # primitive adjacency (shared signals), for attraction gain
     sig_users = defaultdict(list)
@@ -44,7 +58,7 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
         mem = set(members)
         if cand is not None:
             mem.add(cand)
-        produced = {prims[i].name for i in mem}
+        produced = {o for i in mem for o in prim_outputs(i)}
         ins = set()
         for i in mem:
             for s in prims[i].inputs:
@@ -54,7 +68,7 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
                 pass  # clock nets are global, don't count against pins
         return ins
 
-    unclustered = set(range(np_prims))
+    unclustered = set(range(np_prims)) - hard_set
     while unclustered:
         # seed: primitive with most inputs (hardest to place later)
         seed = max(unclustered, key=lambda i: (len(prims[i].inputs), -i))
@@ -70,9 +84,10 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
                     d = drv_of_sig.get(s)
                     if d and d[0] == "prim" and d[1] in unclustered:
                         gain[d[1]] += 1
-                for j in sig_users.get(p.name, ()):
-                    if j in unclustered:
-                        gain[j] += 1
+                for o in prim_outputs(i):
+                    for j in sig_users.get(o, ()):
+                        if j in unclustered:
+                            gain[j] += 1
             if not gain:
                 break
             # best gain, feasibility-filtered on input pins
@@ -90,10 +105,10 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
             cluster_of[i] = len(clusters) - 1
 
     # ---- block-level netlist ----
-    # blocks: input pads, output pads, clusters
+    # blocks: input pads, output pads, clusters, hard blocks (RAM/DSP)
     n_in = len(model.inputs)
     n_out = len(model.outputs)
-    nb = n_in + n_out + len(clusters)
+    nb = n_in + n_out + len(clusters) + len(hard_prims)
     block_type = np.full(nb, BLK_CLB, dtype=np.int8)
     block_type[:n_in + n_out] = BLK_IO
     block_is_seq = np.zeros(nb, dtype=np.uint8)
@@ -102,6 +117,17 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
              [f"opad:{s}" for s in model.outputs] +
              [f"clb_{k}" for k in range(len(clusters))])
     clb0 = n_in + n_out
+    hard0 = clb0 + len(clusters)
+    blk_of_hard = {}
+    for h, i in enumerate(hard_prims):
+        pp = prims[i]
+        cls = subckt_class(pp.model)
+        blk = hard0 + h
+        blk_of_hard[i] = blk
+        block_type[blk] = BLK_RAM if cls == "ram" else BLK_DSP
+        # RAM outputs are registered; DSP (multiply/adder) is comb
+        block_is_seq[blk] = 1 if cls == "ram" else 0
+        names.append(f"{pp.model}_{h}")
     # a cluster is sequential if it contains any latch; its clock domain
     # is the (majority) latch clock. Clock signals are GLOBAL nets: latch
     # clock pins are not data sinks, so clock-pad nets end up sinkless and
@@ -109,6 +135,12 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
     clock_names = []
     clock_id = {}
     block_clock = np.full(nb, -1, dtype=np.int32)
+    def domain_of(cname):
+        if cname not in clock_id:
+            clock_id[cname] = len(clock_names)
+            clock_names.append(cname)
+        return clock_id[cname]
+
     for k, members in enumerate(clusters):
         latch_clocks = [prims[i].clock for i in members
                         if prims[i].kind == "latch" and prims[i].clock]
@@ -117,10 +149,10 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
         if latch_clocks:
             from collections import Counter
             cname = Counter(latch_clocks).most_common(1)[0][0]
-            if cname not in clock_id:
-                clock_id[cname] = len(clock_names)
-                clock_names.append(cname)
-            block_clock[clb0 + k] = clock_id[cname]
+            block_clock[clb0 + k] = domain_of(cname)
+    for i in hard_prims:
+        if prims[i].clock and block_is_seq[blk_of_hard[i]]:
+            block_clock[blk_of_hard[i]] = domain_of(prims[i].clock)
     # IO pads: assign to domain 0 if any clock exists
     if clock_names:
         block_clock[:n_in + n_out] = 0
@@ -128,20 +160,23 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
     in_pad_of = {s: i for i, s in enumerate(model.inputs)}
     out_pad_of = {s: n_in + i for i, s in enumerate(model.outputs)}
 
-    # nets: one per signal that crosses a cluster boundary (or feeds a pad)
+    def blk_of_prim(i):
+        if i in blk_of_hard:
+            return blk_of_hard[i]
+        return clb0 + cluster_of[i]
+
+    # nets: one per signal that crosses a block boundary (or feeds a pad)
     drivers, sink_lists = [], []
     for sig, d in drv_of_sig.items():
         if d[0] == "pad":
             src_blk = in_pad_of[sig]
-            src_cluster = -1
         else:
-            src_cluster = cluster_of[d[1]]
-            src_blk = clb0 + src_cluster
+            src_blk = blk_of_prim(d[1])
         sinks = set()
         for j in sig_users.get(sig, ()):
-            c = cluster_of[j]
-            if c != src_cluster:
-                sinks.add(clb0 + c)
+            sb = blk_of_prim(j)
+            if sb != src_blk:
+                sinks.add(sb)
         if sig in out_pad_of:
             sinks.add(out_pad_of[sig])
         sinks.discard(src_blk)

@@ -152,6 +152,65 @@ def test_synth_placed_het(het_arch):
     STA(nl, het_arch)  # acyclic
 
 
+HET_BLIF = """
+.model hetero_test
+.inputs a b c d we clk
+.outputs y z
+.names a b n1
+11 1
+.names c d n2
+11 1
+.subckt single_port_ram addr0=n1 addr1=n2 data0=a we=we clk=clk \\
+ out0=m0 out1=m1
+.subckt multiply a0=m0 a1=n1 b0=m1 b1=b out0=p0 out1=p1
+.latch p0 r0 re clk 0
+.names r0 p1 y
+11 1
+.names m0 p1 z
+10 1
+.end
+"""
+
+
+def test_blif_subckt_hard_blocks(het_arch):
+    """.subckt RAM/DSP instances become hard blocks of the matching type
+    (reference: VPR memory/mult molecules, cluster.c; read_blif.c
+    .subckt handling)."""
+    from parallel_eda_amd.io.blif import parse_blif, subckt_class
+    from parallel_eda_amd.io.pack import pack_blif
+    assert subckt_class("single_port_ram") == "ram"
+    assert subckt_class("dual_port_ram") == "ram"
+    assert subckt_class("multiply") == "dsp"
+    assert subckt_class("adder") == "dsp"
+    m = parse_blif(HET_BLIF)
+    sub = [p for p in m.prims if p.kind == "subckt"]
+    assert len(sub) == 2
+    assert sub[0].outputs == ["m0", "m1"]  # multi-output instance kept whole
+    assert "we" in sub[0].inputs and sub[0].clock == "clk"
+    nl, cluster_of, names = pack_blif(m, het_arch, n_ble=4)
+    bt = np.asarray(nl.block_type)
+    assert int((bt == BLK_RAM).sum()) == 1
+    assert int((bt == BLK_DSP).sum()) == 1
+    ram_blk = int(np.nonzero(bt == BLK_RAM)[0][0])
+    assert nl.block_is_seq[ram_blk] == 1       # registered outputs
+    assert nl.block_clock[ram_blk] == 0        # clk domain
+    # both RAM outputs drive nets sourced at the RAM block
+    assert int((nl.net_driver == ram_blk).sum()) == 2
+
+
+def test_blif_subckt_full_flow(het_arch):
+    from parallel_eda_amd.io.blif import parse_blif
+    from parallel_eda_amd.io.pack import pack_blif
+    nl, _, _ = pack_blif(parse_blif(HET_BLIF), het_arch, n_ble=4)
+    sta = STA(nl, het_arch)
+    pl = anneal_place(nl, het_arch, seed=3, timing_tradeoff=0.5, sta=sta)
+    g = rrgraph.build_rr_graph(het_arch)
+    res = pathfinder_route(nl, pl, g, het_arch, sta=sta, max_iters=40)
+    assert res.success
+    ok, err = res.router.check_routed()
+    assert ok, err
+
+
 def test_mem32K_arch_builds():
     a = get_arch("mem32K")
     g = rrgraph.build_rr_graph(a)
