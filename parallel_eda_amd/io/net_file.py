@@ -11,7 +11,7 @@ import xml.etree.ElementTree as ET
 import numpy as np
 
 from .synth import NetlistPy
-from ..arch.archdef import BLK_IO, BLK_CLB
+from ..arch.archdef import BLK_IO, BLK_CLB, BLK_NAMES
 
 
 def write_net(path, netlist: NetlistPy, design_name="top"):
@@ -28,7 +28,7 @@ def write_net(path, netlist: NetlistPy, design_name="top"):
         for s in netlist.net_sinks[netlist.net_sink_ptr[n]:netlist.net_sink_ptr[n + 1]]:
             in_nets[s].append(n)
     for b in range(netlist.num_blocks):
-        ty = "io" if netlist.block_type[b] == BLK_IO else "clb"
+        ty = BLK_NAMES[netlist.block_type[b]]
         el = ET.SubElement(root, "block", name=names[b],
                            instance=f"{ty}[{b}]",
                            mode="seq" if netlist.block_is_seq[b] else "comb")
@@ -47,7 +47,8 @@ def read_net(path) -> NetlistPy:
     for el in root.findall("block"):
         names.append(el.get("name"))
         inst = el.get("instance", "clb[0]")
-        types.append(BLK_IO if inst.startswith("io") else BLK_CLB)
+        kind = inst.split("[", 1)[0]
+        types.append(BLK_NAMES.index(kind) if kind in BLK_NAMES else BLK_CLB)
         seqs.append(1 if el.get("mode") == "seq" else 0)
         ins = (el.findtext("inputs") or "").split()
         outs = (el.findtext("outputs") or "").split()

@@ -211,6 +211,17 @@ def test_blif_subckt_full_flow(het_arch):
     assert ok, err
 
 
+def test_net_file_het_roundtrip(het_arch, tmp_path):
+    from parallel_eda_amd.io.net_file import write_net, read_net, check_netlist
+    nl = synth_netlist(spec_for_arch(het_arch, fill=0.5, seed=4))
+    f = tmp_path / "h.net"
+    write_net(str(f), nl)
+    nl2 = read_net(str(f))
+    assert np.array_equal(np.asarray(nl.block_type), np.asarray(nl2.block_type))
+    errs, _ = check_netlist(nl2)
+    assert not errs
+
+
 def test_mem32K_arch_builds():
     a = get_arch("mem32K")
     g = rrgraph.build_rr_graph(a)
