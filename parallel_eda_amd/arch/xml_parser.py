@@ -88,6 +88,26 @@ def parse_arch_xml(path_or_text, nx=None, ny=None, W=64, name=None) -> ArchDef:
             total += int(p.get("num_pins", "1"))
         return total
 
+    def col_repeat(pb):
+        """<gridlocations><loc type="col" repeat="N"/> (VPR 7 column
+        placement of hard-block types)."""
+        loc = pb.find("gridlocations/loc")
+        if loc is not None and loc.get("type") == "col":
+            try:
+                return max(2, int(loc.get("repeat", "8")))
+            except ValueError:
+                pass
+        return 8
+
+    def block_delay(pb, default):
+        dmx = pb.find(".//delay_constant")
+        if dmx is not None:
+            v = _first_float(dmx, "max", default=None)
+            if v is not None:
+                return v
+        return default
+
+    clb_seen = False
     for pb in root.findall(".//complexblocklist/pb_type"):
         pname = (pb.get("name") or "").lower()
         n_in = pin_count(pb, "input")
@@ -95,7 +115,20 @@ def parse_arch_xml(path_or_text, nx=None, ny=None, W=64, name=None) -> ArchDef:
         if pname in ("io", "inpad", "outpad"):
             cap = int(pb.get("capacity", "8"))
             a.io_cap = max(1, cap)
-        elif n_in > 0 and n_out > 0:
+        elif any(k in pname for k in ("memory", "mem", "ram")):
+            # RAM column hard-block type (e.g. mem32K in
+            # k6_frac_N10_mem32K_40nm)
+            a.ram_in = max(1, n_in)
+            a.ram_out = max(1, n_out)
+            a.ram_col_every = col_repeat(pb)
+            a.T_ram = block_delay(pb, a.T_ram)
+        elif any(k in pname for k in ("mult", "dsp")):
+            a.dsp_in = max(1, n_in)
+            a.dsp_out = max(1, n_out)
+            a.dsp_col_every = col_repeat(pb)
+            a.T_dsp = block_delay(pb, a.T_dsp)
+        elif n_in > 0 and n_out > 0 and not clb_seen:
+            clb_seen = True
             # first real logic block type
             a.clb_in = n_in
             a.clb_out = n_out
@@ -123,11 +156,7 @@ def parse_arch_xml(path_or_text, nx=None, ny=None, W=64, name=None) -> ArchDef:
                 v = _first_float(tcq, "max", "value", default=None)
                 if v is not None:
                     a.T_seq_out = v
-            dmx = pb.find(".//delay_constant")
-            if dmx is not None:
-                v = _first_float(dmx, "max", default=None)
-                if v is not None:
-                    a.T_clb = v
+            a.T_clb = block_delay(pb, a.T_clb)
     a.fc_in = min(a.fc_in, a.W)
     a.fc_out = min(a.fc_out, a.W)
     return a
@@ -135,11 +164,20 @@ def parse_arch_xml(path_or_text, nx=None, ny=None, W=64, name=None) -> ArchDef:
 
 def size_grid_for_netlist(netlist, arch: ArchDef, fill_target=0.8):
     """VPR-style auto grid sizing (reference: SetupGrid.c): smallest square
-    grid fitting the CLBs at fill_target, with enough IO perimeter."""
-    n_clb = int((netlist.block_type == 1).sum())
-    n_io = int((netlist.block_type == 0).sum())
-    side = max(2, math.ceil(math.sqrt(n_clb / fill_target)))
-    while 2 * (side + side) * arch.io_cap < n_io:
+    grid fitting every block type at fill_target (heterogeneous fabrics:
+    enough RAM/DSP column tiles too), with enough IO perimeter."""
+    counts = {t: int((netlist.block_type == t).sum()) for t in (0, 1, 2, 3)}
+    side = max(2, math.ceil(math.sqrt(max(1, counts[1]) / fill_target)))
+    while True:
+        arch.nx = arch.ny = side
+        if 2 * (side + side) * arch.io_cap < counts[0]:
+            side += 1
+            continue
+        ok = True
+        for t in (1, 2, 3):
+            if counts[t] and arch.num_tiles_of_type(t) * fill_target < counts[t]:
+                ok = False
+                break
+        if ok:
+            return arch
         side += 1
-    arch.nx = arch.ny = side
-    return arch

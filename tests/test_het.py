@@ -222,6 +222,64 @@ def test_net_file_het_roundtrip(het_arch, tmp_path):
     assert not errs
 
 
+HET_XML = """
+<architecture>
+ <layout auto="1.0"/>
+ <device/>
+ <switchlist><switch type="mux" name="0" R="551" Cin="0.77e-15"
+   Tdel="58e-12"/></switchlist>
+ <segmentlist><segment length="4" Rmetal="101"
+   Cmetal="22.5e-15"/></segmentlist>
+ <complexblocklist>
+  <pb_type name="io" capacity="8"><input name="outpad" num_pins="1"/>
+    <output name="inpad" num_pins="1"/></pb_type>
+  <pb_type name="clb"><input name="I" num_pins="40"/>
+    <output name="O" num_pins="10"/>
+    <fc default_in_type="frac" default_in_val="0.15"
+        default_out_type="frac" default_out_val="0.1"/>
+  </pb_type>
+  <pb_type name="memory"><input name="addr" num_pins="30"/>
+    <output name="out" num_pins="32"/>
+    <gridlocations><loc type="col" start="2" repeat="8"
+      priority="2"/></gridlocations>
+    <delay_constant max="1.7e-9" in_port="addr" out_port="out"/>
+  </pb_type>
+  <pb_type name="mult_36"><input name="a" num_pins="36"/>
+    <output name="out" num_pins="36"/>
+    <gridlocations><loc type="col" start="4" repeat="16"
+      priority="2"/></gridlocations>
+  </pb_type>
+ </complexblocklist>
+</architecture>
+"""
+
+
+def test_parse_het_arch_xml():
+    """Memory / multiplier pb_types with column gridlocations become
+    heterogeneous column tiles (reference: read_xml_arch_file.c pb_type
+    + grid_loc_def col fill)."""
+    from parallel_eda_amd.arch.xml_parser import parse_arch_xml
+    a = parse_arch_xml(HET_XML, nx=40, ny=40, W=80, name="t")
+    assert a.is_heterogeneous()
+    assert a.ram_col_every == 8 and a.dsp_col_every == 16
+    assert (a.ram_in, a.ram_out) == (30, 32)
+    assert (a.dsp_in, a.dsp_out) == (36, 36)
+    assert a.T_ram == pytest.approx(1.7e-9)
+    assert (a.clb_in, a.clb_out) == (40, 10)  # memory pb didn't clobber CLB
+
+
+def test_size_grid_het():
+    from parallel_eda_amd.arch.xml_parser import parse_arch_xml, \
+        size_grid_for_netlist
+    nl = synth_netlist(spec_for_arch(get_arch("tiny_het"), fill=0.6, seed=1))
+    a = parse_arch_xml(HET_XML, nx=4, ny=4, W=80)
+    size_grid_for_netlist(nl, a, fill_target=0.8)
+    bt = np.asarray(nl.block_type)
+    for t in (1, 2, 3):
+        need = int((bt == t).sum())
+        assert a.num_tiles_of_type(t) * 0.8 >= need or need == 0
+
+
 def test_mem32K_arch_builds():
     a = get_arch("mem32K")
     g = rrgraph.build_rr_graph(a)
