@@ -168,6 +168,12 @@ PYBIND11_MODULE(_pnr_cpu, m) {
       .def("count_overused", &SerialRouter::count_overused)
       .def("unrouted_sinks", &SerialRouter::unrouted_sinks)
       .def("feasible", &SerialRouter::feasible)
+      .def("congested_nets", [](SerialRouter& r) {
+        auto v = r.congested_nets();
+        py::array_t<int32_t> out((py::ssize_t)v.size());
+        std::copy(v.begin(), v.end(), out.mutable_data());
+        return out;
+      })
       .def("total_wirelength", &SerialRouter::total_wirelength)
       .def("heap_pushes", &SerialRouter::heap_pushes)
       .def("heap_pops", &SerialRouter::heap_pops)

@@ -132,6 +132,19 @@ class SerialRouter {
 
   bool feasible() const { return count_overused() == 0; }
 
+  // Nets whose current tree crosses an overused node (reference:
+  // phase-two congested-net selection; GPU analogue:
+  // pnr_flag_congested_nets in csrc/hip/router_kernel.hip). Only
+  // meaningful for nets whose trees this router owns.
+  std::vector<int32_t> congested_nets() const {
+    std::vector<int32_t> out;
+    for (int inet = 0; inet < num_nets(); ++inet) {
+      for (int32_t v : trees_[inet].nodes)
+        if (occ_[v] > g_->capacity[v]) { out.push_back(inet); break; }
+    }
+    return out;
+  }
+
   // Elmore delay at each routed sink (aligned with sinks_).
   void sink_delays(float* out) const {
     for (size_t i = 0; i < sink_delays_.size(); ++i) out[i] = sink_delays_[i];

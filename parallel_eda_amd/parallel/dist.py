@@ -107,6 +107,9 @@ class CpuEngine:
     def num_overused(self):
         return int(self.r.count_overused())
 
+    def congested_nets(self):
+        return np.asarray(self.r.congested_nets())
+
 
 class GpuEngine:
     """GpuRouter adapter for DistRouteLoop."""
@@ -146,6 +149,9 @@ class GpuEngine:
         import torch
         return int((self.g.t_occ > self.g.t_cap.to(torch.int32)).sum().item())
 
+    def congested_nets(self):
+        return np.asarray(self.g.congested_nets())
+
 
 class DistRouteLoop:
     """Distributed PathFinder outer loop over an engine.
@@ -181,14 +187,40 @@ class DistRouteLoop:
             mask[sink_ptr[n]:sink_ptr[n + 1]] = True
         self.my_sink_mask = mask
 
+    def _refresh_sink_mask(self):
+        mask = np.zeros(len(self.my_sink_mask), dtype=bool)
+        for n in self.my_nets:
+            mask[self._sink_ptr[n]:self._sink_ptr[n + 1]] = True
+        self.my_sink_mask = mask
+
+    def set_partition(self, new_rank):
+        """Migrate net ownership to `new_rank` (identical on every rank).
+        Nets that change owner are ripped up by the OLD owner and the occ
+        deltas all-reduced (the replicated-graph analogue of the
+        reference's move_route_tree, mpi_route...cxx:172: the new owner
+        simply reroutes from scratch). Returns nets this rank lost+gained."""
+        dist = _dist()
+        old_mine = set(self.my_nets.tolist())
+        new_mine = set(np.nonzero(new_rank == self.rank)[0].tolist())
+        lost = sorted(old_mine - new_mine)
+        gained = sorted(new_mine - old_mine)
+        occ_before = self.engine.occ_tensor().clone()
+        if lost:
+            self.engine.rip_up_nets(np.asarray(lost, dtype=np.int64))
+        occ = self.engine.occ_tensor()
+        delta = occ - occ_before
+        dist.all_reduce(delta, op=dist.ReduceOp.SUM)
+        self.engine.set_occ(occ_before + delta)
+        self.rank_of = np.asarray(new_rank, dtype=np.int32)
+        self.my_nets = np.nonzero(self.rank_of == self.rank)[0]
+        self._refresh_sink_mask()
+        return len(lost) + len(gained)
+
     def rebalance(self, weights=None):
         """Repartition nets by measured per-net route cost (reference:
         load-balanced repartition, mpi_route...cxx:249). weights: per-net
         cost with valid entries for OWNED nets (zeros elsewhere); summed
         across ranks so every rank computes the identical new partition.
-        Nets that change owner are ripped up by the OLD owner and the occ
-        deltas all-reduced (the replicated-graph analogue of
-        move_route_tree: the new owner simply reroutes from scratch).
         Returns the number of nets this rank lost+gained."""
         if self.ws <= 1:
             return 0
@@ -205,34 +237,64 @@ class DistRouteLoop:
         dist.all_reduce(w, op=dist.ReduceOp.SUM)
         wsum = np.maximum(w.cpu().numpy(), 1.0)
         new_rank = spatial_partition(self._bb, self.ws, weight=wsum)
-        old_mine = set(self.my_nets.tolist())
-        new_mine = set(np.nonzero(new_rank == self.rank)[0].tolist())
-        lost = sorted(old_mine - new_mine)
-        gained = sorted(new_mine - old_mine)
-        occ_before = self.engine.occ_tensor().clone()
-        if lost:
-            self.engine.rip_up_nets(np.asarray(lost, dtype=np.int64))
-        occ = self.engine.occ_tensor()
-        delta = occ - occ_before
-        dist.all_reduce(delta, op=dist.ReduceOp.SUM)
-        self.engine.set_occ(occ_before + delta)
-        self.rank_of = new_rank
-        self.my_nets = np.nonzero(new_rank == self.rank)[0]
-        mask = np.zeros(len(self.my_sink_mask), dtype=bool)
-        for n in self.my_nets:
-            mask[self._sink_ptr[n]:self._sink_ptr[n + 1]] = True
-        self.my_sink_mask = mask
-        return len(lost) + len(gained)
+        return self.set_partition(new_rank)
 
-    def iteration(self, crit, pres_fac, acc_fac):
-        """One distributed PathFinder iteration. Returns
+    def global_congested_mask(self):
+        """Boolean mask over nets whose tree crosses an overused node —
+        identical on every rank (each rank reports its OWNED nets, then a
+        max all-reduce unions them). Reference: phase-two congested-net
+        selection, here made globally consistent for the shrink decision."""
+        import torch
+        mask = np.zeros(len(self.rank_of), dtype=np.float32)
+        local = np.intersect1d(np.asarray(self.engine.congested_nets(),
+                                          dtype=np.int64),
+                               self.my_nets)
+        mask[local] = 1.0
+        if self.ws > 1:
+            t = torch.from_numpy(mask)
+            occ_dev = self.engine.occ_tensor().device
+            if occ_dev.type == "cuda":
+                t = t.to(occ_dev)
+            _dist().all_reduce(t, op=_dist().ReduceOp.MAX)
+            mask = t.cpu().numpy()
+        return mask > 0.5
+
+    def shrink_active(self, active_mask, k=1):
+        """Elastic comm-shrink analogue (reference: mpi_comm_shrink — the
+        MPI router drops to fewer ranks when the contested endgame no
+        longer fills the machine). Consolidates the ACTIVE nets onto the
+        first k ranks; frozen nets keep their owners (their trees never
+        move). After this, call iteration(..., active_mask=...) so idle
+        ranks route nothing but still join the (cheap) collectives.
+        Returns nets moved by this rank."""
+        if self.ws <= 1:
+            return 0
+        active = np.nonzero(active_mask)[0]
+        new_rank = self.rank_of.copy()
+        if len(active):
+            nsk = np.diff(self._sink_ptr)[active]
+            semi = ((self._bb[active, 2] - self._bb[active, 0]).astype(np.int64)
+                    + (self._bb[active, 3] - self._bb[active, 1]).astype(np.int64)
+                    + 2)
+            new_rank[active] = spatial_partition(self._bb[active], k,
+                                                 weight=nsk * semi)
+        return self.set_partition(new_rank)
+
+    def iteration(self, crit, pres_fac, acc_fac, active_mask=None):
+        """One distributed PathFinder iteration. active_mask: optional
+        bool mask over nets (selective reroute); only owned ACTIVE nets
+        are routed, but every rank joins the collectives. Returns
         (overused_global, sink_delays_global)."""
         import torch
         dist = _dist() if self.ws > 1 else None
         eng = self.engine
+        nets = self.my_nets
+        if active_mask is not None:
+            nets = nets[np.asarray(active_mask)[nets]]
         if self.ws > 1:
             occ_before = eng.occ_tensor().clone()
-        eng.route_subset(crit, pres_fac, self.my_nets)
+        if len(nets):
+            eng.route_subset(crit, pres_fac, nets)
         sd = eng.sink_delays_local(self.my_nets)
         if self.ws > 1:
             occ = eng.occ_tensor()

@@ -172,3 +172,80 @@ def test_dist_rebalance_world2(tmp_path):
     # global congestion state stayed consistent across the hand-off
     assert np.array_equal(r0["occ"], r1["occ"])
     assert r0["mine"] + r1["mine"] > 0
+
+
+def _worker_shrink(rank, world, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    arch, nl, pl = _build_case()
+    g = rrgraph.build_rr_graph(arch)
+    sta = STA(nl, arch)
+    net_ids, src_rr, sink_ptr, sink_rr, conn_index = net_rr_terminals(
+        nl, pl, g, arch)
+    cmap = ConnMap(conn_index, sink_ptr, nl.num_conns, len(sink_rr))
+    cpu = ops.cpu()
+    router = cpu.SerialRouter(g, src_rr, sink_ptr, sink_rr, cpu.RouterOpts())
+    engine = CpuEngine(router, g.num_nodes)
+    xlow = np.asarray(g.xlow); ylow = np.asarray(g.ylow)
+    bb = np.zeros((len(net_ids), 4), dtype=np.int16)
+    for n in range(len(net_ids)):
+        terms = np.r_[src_rr[n], sink_rr[sink_ptr[n]:sink_ptr[n + 1]]]
+        bb[n] = (xlow[terms].min(), ylow[terms].min(),
+                 xlow[terms].max(), ylow[terms].max())
+    loop = DistRouteLoop(engine, len(net_ids), bb, len(sink_rr), sink_ptr,
+                         rank=rank, world_size=world)
+    crit = np.zeros(len(sink_rr), dtype=np.float32)
+    conn_delay = np.zeros(nl.num_conns, dtype=np.float32)
+    pres = 0.0
+    over = -1
+    shrunk = False
+    active = None
+    n_active_at_shrink = -1
+    for it in range(60):
+        over, sd = loop.iteration(crit, pres, acc_fac=1.0,
+                                  active_mask=active)
+        cmap.conn_delays(sd, out=conn_delay)
+        cpd, slack, c = sta.analyze(conn_delay)
+        crit = cmap.sink_crit(c)
+        if over == 0:
+            break
+        if it >= 2 and not shrunk:
+            # contested endgame: consolidate active nets onto rank 0
+            # (elastic comm-shrink analogue)
+            mask = loop.global_congested_mask()
+            n_active_at_shrink = int(mask.sum())
+            if 0 < n_active_at_shrink:
+                loop.shrink_active(mask, k=1)
+                active = mask
+                shrunk = True
+        pres = 0.5 if pres == 0.0 else pres * 1.3
+
+    occ = np.asarray(router.occ()).copy()
+    mine_active = (len(loop.my_nets) if active is None
+                   else int(np.asarray(active)[loop.my_nets].sum()))
+    with open(os.path.join(tmpdir, f"sh{rank}.pkl"), "wb") as f:
+        pickle.dump({"over": over, "occ": occ, "shrunk": shrunk,
+                     "mine_active": mine_active,
+                     "n_active": n_active_at_shrink}, f)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_dist_shrink_world2(tmp_path):
+    """Elastic comm-shrink analogue: the contested endgame consolidates
+    onto rank 0 (reference: mpi_comm_shrink), frozen nets keep their
+    owners, and the global congestion state stays rank-identical."""
+    mp.spawn(_worker_shrink, args=(2, 29533, str(tmp_path)), nprocs=2,
+             join=True)
+    with open(tmp_path / "sh0.pkl", "rb") as f:
+        r0 = pickle.load(f)
+    with open(tmp_path / "sh1.pkl", "rb") as f:
+        r1 = pickle.load(f)
+    assert r0["over"] == 0, "infeasible after shrink"
+    assert np.array_equal(r0["occ"], r1["occ"])
+    assert r0["shrunk"], "shrink path not exercised (converged too early)"
+    assert r1["mine_active"] == 0          # rank 1 went idle
+    assert r0["mine_active"] == r0["n_active"]
