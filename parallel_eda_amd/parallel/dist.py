@@ -312,3 +312,62 @@ class DistRouteLoop:
         over = eng.num_overused()
         eng.update_acc(acc_fac)
         return over, sd
+
+
+def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
+                          pres_fac_mult=1.3, acc_fac=1.0,
+                          shrink_threshold=128, verbose=False):
+    """Distributed PathFinder outer loop (flow-level driver).
+
+    Mirrors route.gpu_router.pathfinder_route_gpu's schedule — iteration 1
+    routes everything congestion-blind, later iterations re-route only the
+    congested nets (reference: build_phase_two) — with the distributed
+    twists from SURVEY §5.8: occ/delay all-reduce every iteration (inside
+    loop.iteration) and the elastic comm-shrink analogue: once the active
+    set drops below shrink_threshold nets, the endgame consolidates onto
+    rank 0 (reference: mpi_comm_shrink) so the contested tail is routed by
+    one engine with bb-disjoint wave scheduling instead of oscillating
+    across ranks.
+
+    loop: DistRouteLoop; cmap: route.router.ConnMap; sta: timing.sta.STA
+    (or None for congestion-only). Returns a dict with success/overused/
+    cpd/iters/history.
+    """
+    n_rsinks = len(loop.my_sink_mask)
+    crit = np.zeros(n_rsinks, dtype=np.float32)
+    conn_delay = None
+    if sta is not None:
+        conn_delay = np.zeros(cmap.num_conns, dtype=np.float32)
+    pres = 0.0
+    active = None
+    shrunk = False
+    cpd = 0.0
+    history = []
+    overused = -1
+    it = 0
+    for it in range(1, max_iters + 1):
+        overused, sd = loop.iteration(crit, pres, acc_fac,
+                                      active_mask=active)
+        if sta is not None:
+            cmap.conn_delays(sd, out=conn_delay)
+            cpd, _slack, c = sta.analyze(conn_delay)
+            crit = cmap.sink_crit(c)
+        n_active = None
+        if overused > 0:
+            active = loop.global_congested_mask()
+            n_active = int(active.sum())
+            if not n_active:
+                active = None
+            elif loop.ws > 1 and not shrunk and n_active < shrink_threshold:
+                loop.shrink_active(active, k=1)
+                shrunk = True
+        history.append(dict(iter=it, overused=int(overused),
+                            active=n_active, cpd=cpd, shrunk=shrunk))
+        if verbose and loop.rank == 0:
+            print(f"[dist] iter {it}: overused={overused} "
+                  f"active={n_active} cpd={cpd*1e9:.2f}ns shrunk={shrunk}")
+        if overused == 0:
+            break
+        pres = pres_fac_init if pres == 0.0 else pres * pres_fac_mult
+    return dict(success=overused == 0, overused=int(overused), cpd=cpd,
+                iters=it, shrunk=shrunk, history=history)
