@@ -21,6 +21,8 @@ struct PlaceDev {
   const int8_t* blk_type;       // 0=IO 1=CLB 2=RAM 3=DSP
   const int8_t* tile_btype;     // [gx*gy] tile block type, -1 corner;
                                 // nullptr => homogeneous (perimeter IO)
+  const int32_t* type_cols;     // sorted column lists: RAM cols then DSP
+  const int32_t* type_col_ptr;  // [3] bounds into type_cols (RAM, DSP)
   const float* net_q;           // crossing factor per net
   // timing
   const int32_t* net_sink_ptr;  // [num_nets+1] conn ranges
@@ -125,22 +127,50 @@ __global__ void place_propose_kernel(PlaceDev p, MovesDev m, float T,
   bool io = p.blk_type[blk] == 0;
   int x0 = p.bx[blk], y0 = p.by[blk];
   int x1 = -1, y1 = -1, slot1 = 0;
-  for (int att = 0; att < 8; ++att) {
-    uint32_t r1 = rng_hash(seed, batch, i * 131 + 7 * att + 1);
-    uint32_t r2 = rng_hash(seed, batch, i * 131 + 7 * att + 2);
-    int tx = x0 + (int)(r1 % (2 * rlim + 1)) - rlim;
-    int ty = y0 + (int)(r2 % (2 * rlim + 1)) - rlim;
-    if (tx < 0 || tx >= p.gx || ty < 0 || ty >= p.gy) continue;
-    if (p.tile_btype) {
-      // heterogeneous fabric: destination tile must match the block type
-      if (p.tile_btype[tx * p.gy + ty] != p.blk_type[blk]) continue;
-    } else if (is_io_loc(p, tx, ty) != io) continue;
-    int c = cap_at(p, tx, ty);
-    if (c <= 0) continue;
-    if (tx == x0 && ty == y0) continue;
-    x1 = tx; y1 = ty;
-    slot1 = (int)(rng_hash(seed, batch, i * 131 + 7 * att + 3) % c);
-    break;
+  int bt = p.blk_type[blk];
+  if (p.tile_btype && bt >= 2 && p.type_cols) {
+    // sparse column type (RAM/DSP): draw the target column from this
+    // type's sorted column list clipped to the range window — rejection
+    // over the square window would nearly always miss sparse columns
+    // (mirrors the CPU placer's find_to for column types).
+    int c0 = p.type_col_ptr[bt - 2], c1 = p.type_col_ptr[bt - 1];
+    int lo = c0, hi = c1;              // first col >= x0 - rlim
+    while (lo < hi) { int m = (lo + hi) >> 1;
+                      if (p.type_cols[m] < x0 - rlim) lo = m + 1; else hi = m; }
+    int lo2 = lo, hi2 = c1;            // first col > x0 + rlim
+    while (lo2 < hi2) { int m = (lo2 + hi2) >> 1;
+                        if (p.type_cols[m] <= x0 + rlim) lo2 = m + 1; else hi2 = m; }
+    int ncol = lo2 - lo;
+    if (ncol > 0) {
+      int ylo = max(1, y0 - rlim), yhi = min(p.ny, y0 + rlim);
+      for (int att = 0; att < 8; ++att) {
+        uint32_t r1 = rng_hash(seed, batch, i * 131 + 7 * att + 1);
+        uint32_t r2 = rng_hash(seed, batch, i * 131 + 7 * att + 2);
+        int tx = p.type_cols[lo + (int)(r1 % ncol)];
+        int ty = ylo + (int)(r2 % (yhi - ylo + 1));
+        if (tx == x0 && ty == y0) continue;
+        x1 = tx; y1 = ty; slot1 = 0;
+        break;
+      }
+    }
+  } else {
+    for (int att = 0; att < 8; ++att) {
+      uint32_t r1 = rng_hash(seed, batch, i * 131 + 7 * att + 1);
+      uint32_t r2 = rng_hash(seed, batch, i * 131 + 7 * att + 2);
+      int tx = x0 + (int)(r1 % (2 * rlim + 1)) - rlim;
+      int ty = y0 + (int)(r2 % (2 * rlim + 1)) - rlim;
+      if (tx < 0 || tx >= p.gx || ty < 0 || ty >= p.gy) continue;
+      if (p.tile_btype) {
+        // heterogeneous fabric: destination tile must match the block type
+        if (p.tile_btype[tx * p.gy + ty] != p.blk_type[blk]) continue;
+      } else if (is_io_loc(p, tx, ty) != io) continue;
+      int c = cap_at(p, tx, ty);
+      if (c <= 0) continue;
+      if (tx == x0 && ty == y0) continue;
+      x1 = tx; y1 = ty;
+      slot1 = (int)(rng_hash(seed, batch, i * 131 + 7 * att + 3) % c);
+      break;
+    }
   }
   if (x1 < 0) return;
   int32_t other = p.grid[((int64_t)x1 * p.gy + y1) * p.cap + slot1];
@@ -295,7 +325,9 @@ extern "C" {
 struct PlaceLaunchArgs {
   const int32_t* net_blk_ptr; const int32_t* net_blks;
   const int32_t* blk_net_ptr; const int32_t* blk_nets;
-  const int8_t* blk_type; const int8_t* tile_btype; const float* net_q;
+  const int8_t* blk_type; const int8_t* tile_btype;
+  const int32_t* type_cols; const int32_t* type_col_ptr;
+  const float* net_q;
   const int32_t* net_sink_ptr; const float* conn_crit; const float* delay_mat;
   int32_t* bx; int32_t* by; int32_t* bslot; int32_t* grid;
   float* net_cost; float* net_tcost;
@@ -316,6 +348,7 @@ static void unpack(const PlaceLaunchArgs* a, PlaceDev& p, MovesDev& m) {
   p.net_blk_ptr = a->net_blk_ptr; p.net_blks = a->net_blks;
   p.blk_net_ptr = a->blk_net_ptr; p.blk_nets = a->blk_nets;
   p.blk_type = a->blk_type; p.tile_btype = a->tile_btype;
+  p.type_cols = a->type_cols; p.type_col_ptr = a->type_col_ptr;
   p.net_q = a->net_q;
   p.net_sink_ptr = a->net_sink_ptr; p.conn_crit = a->conn_crit;
   p.delay_mat = a->delay_mat;

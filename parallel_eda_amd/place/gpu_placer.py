@@ -20,6 +20,7 @@ class PlaceLaunchArgs(ct.Structure):
         ("net_blk_ptr", ct.c_void_p), ("net_blks", ct.c_void_p),
         ("blk_net_ptr", ct.c_void_p), ("blk_nets", ct.c_void_p),
         ("blk_type", ct.c_void_p), ("tile_btype", ct.c_void_p),
+        ("type_cols", ct.c_void_p), ("type_col_ptr", ct.c_void_p),
         ("net_q", ct.c_void_p),
         ("net_sink_ptr", ct.c_void_p), ("conn_crit", ct.c_void_p),
         ("delay_mat", ct.c_void_p),
@@ -109,6 +110,19 @@ class GpuPlacer:
         # nullptr so the validated perimeter-IO fast path runs unchanged
         self.t_tile_btype = (up(arch.tile_btype_grid())
                              if arch.is_heterogeneous() else None)
+        self.t_type_cols = self.t_type_col_ptr = None
+        if arch.is_heterogeneous():
+            # sorted column lists (RAM then DSP) for the kernel's
+            # sparse-column move proposals
+            ram_cols = [x for x in range(1, arch.nx + 1)
+                        if arch.col_block_type(x) == 2]
+            dsp_cols = [x for x in range(1, arch.nx + 1)
+                        if arch.col_block_type(x) == 3]
+            self.t_type_cols = up(np.asarray(ram_cols + dsp_cols,
+                                             dtype=np.int32))
+            self.t_type_col_ptr = up(np.asarray(
+                [0, len(ram_cols), len(ram_cols) + len(dsp_cols)],
+                dtype=np.int32))
         self.t_net_q = up(q)
         self.t_net_sink_ptr = up(netlist.net_sink_ptr.astype(np.int32))
         self.t_conn_crit = torch.zeros(netlist.num_conns, dtype=torch.float32,
@@ -198,6 +212,10 @@ class GpuPlacer:
         a.blk_type = ptr(self.t_blk_type)
         a.tile_btype = (ptr(self.t_tile_btype)
                         if self.t_tile_btype is not None else None)
+        a.type_cols = (ptr(self.t_type_cols)
+                       if self.t_type_cols is not None else None)
+        a.type_col_ptr = (ptr(self.t_type_col_ptr)
+                          if self.t_type_col_ptr is not None else None)
         a.net_q = ptr(self.t_net_q)
         a.net_sink_ptr = ptr(self.t_net_sink_ptr)
         a.conn_crit = ptr(self.t_conn_crit)

@@ -291,3 +291,14 @@ def test_mem32K_arch_builds():
                 if a.col_block_type(x) == BLK_RAM]
     assert len(ram_cols) == a.nx // 8
     assert cap[ts[ram_cols[0] * gy + 1]] == a.ram_out
+
+
+def test_full_flow_mem32K():
+    """mem32K-scale heterogeneous flow (40x40, RAM column every 8):
+    place + route + timing end-to-end on the CPU oracle."""
+    from parallel_eda_amd.flow import run_flow
+    res = run_flow("mem32K", seed=2, timing_driven=True, fill=0.45,
+                   max_route_iters=60)
+    assert res.route.success
+    assert res.wirelength > 10000     # nontrivial design routed
+    assert 1e-9 < res.cpd < 1e-7

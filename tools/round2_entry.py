@@ -30,6 +30,26 @@ run("tseng-quality-calendar",
     env={"PNR_CALENDAR": "1"}, timeout=300)
 run("btc-calendar", "python bench.py --config bitcoin_miner --steps 2 "
     "--warmup 2 --verbose", env={"PNR_CALENDAR": "1"}, timeout=900)
+run("het-gpu", "python - <<'P'\n"
+    "import sys; sys.path.insert(0, '.')\n"
+    "import numpy as np\n"
+    "from parallel_eda_amd.arch.archdef import get_arch\n"
+    "from parallel_eda_amd.io.synth import synth_netlist, spec_for_arch\n"
+    "from parallel_eda_amd.place.gpu_placer import anneal_place_gpu\n"
+    "from parallel_eda_amd.place.placer import anneal_place\n"
+    "# heterogeneous GPU placer (dormant round-1 path): legality + quality\n"
+    "arch = get_arch('mem32K')\n"
+    "nl = synth_netlist(spec_for_arch(arch, fill=0.45, seed=2))\n"
+    "pl = anneal_place_gpu(nl, arch, seed=7, timing_tradeoff=0.0)\n"
+    "tb = arch.tile_btype_grid(); gy = arch.ny + 2\n"
+    "bt = np.asarray(nl.block_type)\n"
+    "bad = sum(1 for b in range(nl.num_blocks)\n"
+    "          if tb[pl.x[b] * gy + pl.y[b]] != bt[b])\n"
+    "cpu = anneal_place(nl, arch, seed=7, timing_tradeoff=0.0)\n"
+    "print(f'het GPU anneal: bb={pl.bb_cost:.0f} vs CPU {cpu.bb_cost:.0f} '\n"
+    "      f'type-violations={bad}')\n"
+    "assert bad == 0\n"
+    "P", timeout=600)
 run("placer-bgm", "python - <<'P'\n"
     "import sys; sys.path.insert(0, '.')\n"
     "import time, numpy as np\n"
