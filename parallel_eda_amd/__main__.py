@@ -57,6 +57,10 @@ def build_parser():
                    help="write a critical-path report here after routing")
     p.add_argument("--out_place", type=str, default=None)
     p.add_argument("--out_route", type=str, default=None)
+    p.add_argument("--out_verilog", type=str, default=None,
+                   help="post-route structural verilog")
+    p.add_argument("--out_sdf", type=str, default=None,
+                   help="post-route SDF timing annotation")
     p.add_argument("--stats_dir", type=str, default=None)
     p.add_argument("--verbose", "-v", action="store_true")
     return p
@@ -187,6 +191,22 @@ def main(argv=None):
             ok_sdc = all(wp <= p_ * (1 + 1e-6) for p_ in periods[:1])
             print(f"SDC analysis: worst achieved period {wp*1e9:.3f} ns "
                   f"across {max(1, len(periods))} clock domain(s)")
+        if args.out_verilog or args.out_sdf:
+            from .io.verilog import write_verilog, write_sdf
+            if args.out_verilog:
+                write_verilog(args.out_verilog, netlist)
+                print(f"wrote {args.out_verilog}")
+            if args.out_sdf:
+                from .route.router import ConnMap
+                _, _s4, sink_ptr4, sink_rr4, ci4 = net_rr_terminals(
+                    netlist, placement, g, arch)
+                cmap4 = ConnMap(ci4, sink_ptr4, netlist.num_conns,
+                                len(sink_rr4))
+                sd4 = np.asarray(res.router.sink_delays()) \
+                    if hasattr(res.router, "sink_delays") \
+                    else res.router.t_sink_delay.cpu().numpy()
+                write_sdf(args.out_sdf, netlist, arch, cmap4.conn_delays(sd4))
+                print(f"wrote {args.out_sdf}")
         if args.timing_report and sta is not None:
             from .route.router import ConnMap
             from .timing.report import write_timing_report

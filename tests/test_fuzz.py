@@ -11,18 +11,29 @@ from parallel_eda_amd.timing.sta import STA
 from parallel_eda_amd import rrgraph
 
 
-def random_arch(rng):
+def random_arch(rng, het_prob=0.0):
     # realistic W/L: at least 2 tracks per (direction, stagger) class,
     # else the Fs=3 switch digraph can be reducible (validator rejects)
     L = int(rng.integers(1, 5))
     W = 2 * int(rng.integers(max(4, 2 * L), 17))
+    kw = {}
+    # het_prob=0 must not consume rng draws (seed-stable homogeneous fuzz)
+    if het_prob > 0 and rng.random() < het_prob:
+        # heterogeneous columns: random RAM (and sometimes DSP) spacing
+        kw["ram_col_every"] = int(rng.integers(3, 6))
+        if rng.random() < 0.5:
+            kw["dsp_col_every"] = int(rng.integers(6, 10))
+        kw["ram_in"] = int(rng.integers(4, 12))
+        kw["ram_out"] = int(rng.integers(2, 6))
+        kw["dsp_in"] = int(rng.integers(4, 12))
+        kw["dsp_out"] = int(rng.integers(2, 6))
     return ArchDef(
         name=f"fuzz", nx=int(rng.integers(3, 9)), ny=int(rng.integers(3, 9)),
         W=W, L=L,
         fc_in=int(rng.integers(2, min(9, W + 1))),
         fc_out=int(rng.integers(2, min(9, W + 1))),
         clb_in=int(rng.integers(4, 12)), clb_out=int(rng.integers(1, 4)),
-        io_cap=int(rng.integers(1, 4)))
+        io_cap=int(rng.integers(1, 4)), **kw)
 
 
 @pytest.mark.parametrize("seed", [11, 23, 37, 51, 68])
@@ -63,3 +74,35 @@ def test_random_flow_end_to_end(seed):
         assert res.crit_path_delay > 0
         ok, err = res.router.check_routed()
         assert ok, err
+
+
+@pytest.mark.parametrize("seed", [7, 19, 43, 71])
+def test_random_het_fabric_flow(seed):
+    """Fuzz heterogeneous fabrics: random RAM/DSP column spacings through
+    the validator + a placed/routed flow with per-type legality."""
+    from parallel_eda_amd.io.synth import spec_for_arch
+    rng = np.random.default_rng(seed)
+    arch = random_arch(rng, het_prob=1.0)
+    g = rrgraph.build_rr_graph(arch)
+    try:
+        rrgraph.check_rr_graph(g, arch)
+    except rrgraph.RRGraphError as e:
+        pytest.skip(f"fuzzed fabric rejected by validator: {e}")
+    spec = spec_for_arch(arch, fill=0.4, seed=seed)
+    if spec.n_clb < 2:
+        pytest.skip("degenerate fabric")
+    nl = synth_netlist(spec)
+    if int((nl.block_type == 0).sum()) > arch.num_io_slots():
+        pytest.skip("io overflow for this random arch")
+    pl = anneal_place(nl, arch, seed=seed, timing_tradeoff=0.0)
+    tb = arch.tile_btype_grid()
+    gy = arch.ny + 2
+    bt = np.asarray(nl.block_type)
+    for b in range(nl.num_blocks):
+        assert tb[pl.x[b] * gy + pl.y[b]] == bt[b], b
+    res = pathfinder_route(nl, pl, g, arch, sta=STA(nl, arch), max_iters=70)
+    if res.success:
+        ok, err = res.router.check_routed()
+        assert ok, err
+    else:
+        assert res.overused > 0 or res.router.unrouted_sinks() > 0

@@ -302,3 +302,14 @@ def test_full_flow_mem32K():
     assert res.route.success
     assert res.wirelength > 10000     # nontrivial design routed
     assert 1e-9 < res.cpd < 1e-7
+
+
+def test_cli_het_flow_with_sdf(tmp_path):
+    from parallel_eda_amd.__main__ import main
+    v = tmp_path / "h.v"
+    s = tmp_path / "h.sdf"
+    rc = main(["--synth", "tiny_het", "--fill", "0.5", "--seed", "5",
+               "--out_verilog", str(v), "--out_sdf", str(s)])
+    assert rc == 0
+    assert "ram_seq" in v.read_text()
+    assert '(CELLTYPE "ram")' in s.read_text()
