@@ -1,0 +1,164 @@
+"""Lookahead-quality simulator (round-2 scan-bottleneck study).
+
+The bitcoin-scale router is scan-bound (339k scans/sink, MEASUREMENTS.md);
+the congestion-scaled lookahead regressed 5.5x (inflating h breaks the
+bucket structure). The remaining lever is a TIGHTER ADMISSIBLE h: a
+map-based lookahead table (VPR 8's router lookahead: min cost-to-target
+per (dx, dy) offset, computed by backward Dijkstra over the uncongested
+cost surface) instead of the analytic segment-count formula. Congestion
+only multiplies base costs by pres >= 1, so the uncongested table stays
+admissible under congestion.
+
+This sim A/Bs, on the real rr graph with synthetic congestion, using the
+calendar frontier (the round-2 default):
+    analytic h x astar 1.2  (current production setting)
+    analytic h x astar 1.0  (admissible analytic)
+    table    h x astar 1.0  (admissible, tight)
+    table    h x astar 1.2  (inflated table)
+and verifies each lands within the delta-stepping envelope of exact
+Dijkstra. Run: python tools/lookahead_sim.py [arch] [n_pairs]
+"""
+import heapq
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+import numpy as np
+
+from parallel_eda_amd.arch.archdef import get_arch
+from parallel_eda_amd import rrgraph
+from tools.frontier_sim import Cost, dijkstra_ref, calendar
+
+
+def build_lookahead_table(g, arch):
+    """min uncongested cost-to-sink per (|dx|, |dy|) offset, from a
+    backward Dijkstra rooted at a center-tile SINK. dist(v) = min cost of
+    a v->sink path, where entering node w costs base[type(w)] (the
+    kernel's congestion-only relaxation at occ=0)."""
+    row_ptr = np.asarray(g.row_ptr)
+    dst = np.asarray(g.edge_dst)
+    ty = np.asarray(g.type)
+    base = np.asarray(g.base_cost)
+    xl = np.asarray(g.xlow); yl = np.asarray(g.ylow)
+    xh = np.asarray(g.xhigh); yh = np.asarray(g.yhigh)
+    n = g.num_nodes
+    # reversed adjacency
+    indeg = np.zeros(n + 1, dtype=np.int64)
+    np.add.at(indeg, dst + 1, 1)
+    rptr = np.cumsum(indeg)
+    rsrc = np.empty(len(dst), dtype=np.int32)
+    cur = rptr[:-1].copy()
+    src_of_edge = np.repeat(np.arange(n), np.diff(row_ptr))
+    for e in range(len(dst)):
+        rsrc[cur[dst[e]]] = src_of_edge[e]
+        cur[dst[e]] += 1
+
+    cx, cy = arch.nx // 2 + 1, arch.ny // 2 + 1
+    gy = arch.ny + 2
+    sink = int(np.asarray(g.tile_sink)[cx * gy + cy])
+    INF = np.float32(np.inf)
+    dist = np.full(n, INF, dtype=np.float32)
+    dist[sink] = 0.0
+    pq = [(0.0, sink)]
+    while pq:
+        d, v = heapq.heappop(pq)
+        if d > dist[v]:
+            continue
+        for k in range(rptr[v], rptr[v + 1]):
+            u = int(rsrc[k])
+            nd = np.float32(d + base[ty[v]])  # entering v costs base[v]
+            if nd < dist[u]:
+                dist[u] = nd
+                heapq.heappush(pq, (float(nd), u))
+    # bin by closest-point offset to the sink tile (the kernel's dx/dy)
+    table = np.full((arch.nx + 2, arch.ny + 2), INF, dtype=np.float32)
+    chan = (ty == 4) | (ty == 5)
+    dx = np.maximum(np.maximum(xl - cx, cx - xh), 0)
+    dy = np.maximum(np.maximum(yl - cy, cy - yh), 0)
+    ok = chan & np.isfinite(dist)
+    np.minimum.at(table, (dx[ok], dy[ok]), dist[ok])
+    # monotone fill for offsets never seen (edge effects): nearest smaller
+    for i in range(table.shape[0]):
+        for j in range(table.shape[1]):
+            if not np.isfinite(table[i, j]):
+                cands = []
+                if i:
+                    cands.append(table[i - 1, j])
+                if j:
+                    cands.append(table[i, j - 1])
+                table[i, j] = max([c for c in cands if np.isfinite(c)],
+                                  default=0.0)
+    table[0, 0] = 0.0
+    return table
+
+
+class TableCost(Cost):
+    def __init__(self, g, arch, table, astar, **kw):
+        super().__init__(g, arch, **kw)
+        self.table = table
+        self.astar = astar
+
+    def h(self, v, sink):
+        tx, ty2 = self.xl[sink], self.yl[sink]
+        dx = max(self.xl[v] - tx, tx - self.xh[v], 0)
+        dy = max(self.yl[v] - ty2, ty2 - self.yh[v], 0)
+        return np.float32(self.astar * self.table[dx, dy])
+
+
+def main():
+    arch_name = sys.argv[1] if len(sys.argv) > 1 else "tseng"
+    n_pairs = int(sys.argv[2]) if len(sys.argv) > 2 else 40
+    arch = get_arch(arch_name)
+    g = rrgraph.build_rr_graph(arch)
+    table = build_lookahead_table(g, arch)
+    rng = np.random.default_rng(3)
+    ts = np.asarray(g.tile_source)
+    tk = np.asarray(g.tile_sink)
+    gy = arch.ny + 2
+    occ = rng.integers(0, 3, g.num_nodes).astype(np.int32)
+    variants = {}
+    for name, use_table, astar in (("analytic-1.2", False, 1.2),
+                                   ("analytic-1.0", False, 1.0),
+                                   ("table-1.0", True, 1.0),
+                                   ("table-1.2", True, 1.2)):
+        if use_table:
+            c = TableCost(g, arch, table, astar, occ=occ, pres_fac=1.5)
+        else:
+            c = Cost(g, arch, occ=occ, pres_fac=1.5)
+            c.astar = astar
+        variants[name] = c
+    delta = 3.0 * variants["analytic-1.2"].seg_base
+    totals = {k: dict(scans=0, appends=0, cost=0.0, mism=0)
+              for k in variants}
+    pairs = []
+    for _ in range(n_pairs):
+        sx, sy = 1 + rng.integers(arch.nx), 1 + rng.integers(arch.ny)
+        tx, ty2 = 1 + rng.integers(arch.nx), 1 + rng.integers(arch.ny)
+        if (sx, sy) != (tx, ty2):
+            pairs.append((int(ts[sx * gy + sy]), int(tk[tx * gy + ty2])))
+    for src, sink in pairs:
+        ref = dijkstra_ref(variants["analytic-1.0"], src, sink)
+        for name, c in variants.items():
+            got, st = calendar(c, src, sink, delta)
+            t = totals[name]
+            t["scans"] += st["scans"]
+            t["appends"] += st["appends"]
+            t["cost"] += float(got)
+            # admissible variants must stay in the delta envelope; the
+            # inflated ones within astar x optimum (bounded suboptimality)
+            bound = (ref + delta + 1e-4 if c.astar <= 1.0
+                     else 1.2 * ref + delta + 1e-4)
+            if got > bound:
+                t["mism"] += 1
+    print(f"pairs={len(pairs)} delta={delta:.2f} arch={arch_name}")
+    base_scans = totals["analytic-1.2"]["scans"]
+    for name, t in totals.items():
+        print(f"{name:13s} scans={t['scans']:8d} "
+              f"({base_scans / max(1, t['scans']):.2f}x vs prod) "
+              f"appends={t['appends']:8d} sum_cost={t['cost']:.1f} "
+              f"mism={t['mism']}")
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
