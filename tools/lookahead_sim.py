@@ -160,5 +160,162 @@ def main():
     return 0
 
 
-if __name__ == "__main__":
+if __name__ == "__main__" and "timing" not in sys.argv:
     sys.exit(main())
+
+
+# ---- timing-driven extension (crit > 0): does a delay table help? ----
+
+class TimingCost(Cost):
+    """Kernel's timing-driven relaxation: crit*hop_delay + (1-crit)*cong."""
+
+    def __init__(self, g, arch, crit, astar, delay_table=None, **kw):
+        super().__init__(g, arch, **kw)
+        self.crit = crit
+        self.astar = astar
+        self.swr = np.asarray(g.sw_R)
+        self.swt = np.asarray(g.sw_Tdel)
+        self.Rn = np.asarray(g.node_R)
+        self.Cn = np.asarray(g.node_C)
+        self.esw = np.asarray(g.edge_sw)
+        # analytic per-seg delay (placer.py's seg_delay formula)
+        self.seg_delay = float(arch.T_sw + arch.C_wire * arch.L *
+                               (arch.R_sw + 0.5 * arch.R_wire * arch.L))
+        self.ipin_delay = float(arch.T_ipin)
+        self.delay_table = delay_table
+        self._edge_of = {}
+        rp = self.row_ptr
+        for v in range(g.num_nodes):
+            for e in range(rp[v], rp[v + 1]):
+                self._edge_of[(v, int(self.dst[e]))] = e
+
+    def edge_cost_vw(self, v, w):
+        e = self._edge_of[(v, w)]
+        sw = self.esw[e]
+        d = self.swt[sw] + self.Cn[w] * (self.swr[sw] + 0.5 * self.Rn[w])
+        return np.float32(self.crit * d +
+                          (1.0 - self.crit) * Cost.edge_cost(self, w))
+
+    def h(self, v, sink):
+        tx, ty2 = self.xl[sink], self.yl[sink]
+        dx = max(self.xl[v] - tx, tx - self.xh[v], 0)
+        dy = max(self.yl[v] - ty2, ty2 - self.yh[v], 0)
+        nseg = -(-(dx + dy) // self.L)
+        cong = nseg * self.seg_base + 0.95
+        if self.delay_table is not None:
+            dly = self.delay_table[dx, dy]
+        else:
+            dly = nseg * self.seg_delay + self.ipin_delay
+        return np.float32(self.astar *
+                          (self.crit * dly + (1.0 - self.crit) * cong))
+
+
+def timing_variant(c, src, sink, delta):
+    """calendar() clone that relaxes with edge_cost_vw (needs v)."""
+    import heapq as _h
+    INF = np.float32(np.inf)
+    best = {src: (np.float32(0.0), src)}
+    pq = [(float(c.h(src, sink)), 0.0, src, src)]
+    scans = 0
+    best_sink = INF
+    while pq:
+        tot, back, v, prev = _h.heappop(pq)
+        scans += 1
+        if best.get(v, (INF, -1)) != (np.float32(back), prev):
+            continue
+        if np.float32(tot) >= best_sink:
+            break
+        if v == sink:
+            best_sink = min(best_sink, np.float32(back))
+            continue
+        for w in c.edges(v, sink):
+            nb = np.float32(back + c.edge_cost_vw(v, w))
+            if (nb, v) < best.get(w, (INF, -1)):
+                best[w] = (nb, v)
+                _h.heappush(pq, (float(nb + c.h(w, sink)), float(nb), w, v))
+    return best_sink, scans
+
+
+def timing_study(arch_name="tseng", n_pairs=30, crit=0.99):
+    """High-crit sink search: analytic delay h vs delay table (built from
+    the same backward Dijkstra, weights = hop delays)."""
+    arch = get_arch(arch_name)
+    g = rrgraph.build_rr_graph(arch)
+    rng = np.random.default_rng(3)
+    occ = rng.integers(0, 3, g.num_nodes).astype(np.int32)
+    # delay table: backward Dijkstra with hop-delay weights
+    row_ptr = np.asarray(g.row_ptr); dst = np.asarray(g.edge_dst)
+    esw = np.asarray(g.edge_sw); ty = np.asarray(g.type)
+    swr = np.asarray(g.sw_R); swt = np.asarray(g.sw_Tdel)
+    Rn = np.asarray(g.node_R); Cn = np.asarray(g.node_C)
+    xl = np.asarray(g.xlow); yl = np.asarray(g.ylow)
+    xh = np.asarray(g.xhigh); yh = np.asarray(g.yhigh)
+    n = g.num_nodes
+    cx, cy = arch.nx // 2 + 1, arch.ny // 2 + 1
+    gy = arch.ny + 2
+    sink = int(np.asarray(g.tile_sink)[cx * gy + cy])
+    # reversed edges with forward hop-delay weight of entering dst
+    import heapq as _h
+    radj = [[] for _ in range(n)]
+    src_of_edge = np.repeat(np.arange(n), np.diff(row_ptr))
+    for e in range(len(dst)):
+        w = int(dst[e]); u = int(src_of_edge[e])
+        d = float(swt[esw[e]] + Cn[w] * (swr[esw[e]] + 0.5 * Rn[w]))
+        radj[w].append((u, d))
+    INF = np.float32(np.inf)
+    dist = np.full(n, np.inf)
+    dist[sink] = 0.0
+    pq = [(0.0, sink)]
+    while pq:
+        d, v = _h.heappop(pq)
+        if d > dist[v]:
+            continue
+        for (u, wgt) in radj[v]:
+            nd = d + wgt
+            if nd < dist[u]:
+                dist[u] = nd
+                _h.heappush(pq, (nd, u))
+    table = np.full((arch.nx + 2, arch.ny + 2), np.inf, dtype=np.float32)
+    chan = (ty == 4) | (ty == 5)
+    dxs = np.maximum(np.maximum(xl - cx, cx - xh), 0)
+    dys = np.maximum(np.maximum(yl - cy, cy - yh), 0)
+    ok = chan & np.isfinite(dist)
+    np.minimum.at(table, (dxs[ok], dys[ok]), dist[ok].astype(np.float32))
+    for i in range(table.shape[0]):
+        for j in range(table.shape[1]):
+            if not np.isfinite(table[i, j]):
+                cands = [table[i - 1, j]] if i else []
+                cands += [table[i, j - 1]] if j else []
+                table[i, j] = max([c for c in cands if np.isfinite(c)],
+                                  default=0.0)
+    table[0, 0] = 0.0
+
+    ts = np.asarray(g.tile_source); tk = np.asarray(g.tile_sink)
+    variants = {
+        "analytic": TimingCost(g, arch, crit, 1.2, occ=occ, pres_fac=1.5),
+        "delay-table": TimingCost(g, arch, crit, 1.2, delay_table=table,
+                                  occ=occ, pres_fac=1.5),
+        "delay-table-1.0": TimingCost(g, arch, crit, 1.0, delay_table=table,
+                                      occ=occ, pres_fac=1.5),
+    }
+    delta = 3.0 * variants["analytic"].seg_base
+    tot = {k: [0, 0.0] for k in variants}
+    for _ in range(n_pairs):
+        sx, sy = 1 + rng.integers(arch.nx), 1 + rng.integers(arch.ny)
+        tx2, ty3 = 1 + rng.integers(arch.nx), 1 + rng.integers(arch.ny)
+        if (sx, sy) == (tx2, ty3):
+            continue
+        s0 = int(ts[sx * gy + sy]); k0 = int(tk[tx2 * gy + ty3])
+        for name, c in variants.items():
+            got, scans = timing_variant(c, s0, k0, delta)
+            tot[name][0] += scans
+            tot[name][1] += float(got)
+    print(f"timing-driven crit={crit} arch={arch_name} pairs={n_pairs}")
+    for name, (scans, cost) in tot.items():
+        print(f"  {name:16s} scans={scans:8d} sum_cost={cost:.4e}")
+
+
+if __name__ == "__main__" and "timing" in sys.argv:
+    timing_study(sys.argv[2] if len(sys.argv) > 2 else "tseng",
+                 int(sys.argv[3]) if len(sys.argv) > 3 else 30)
+    sys.exit(0)
