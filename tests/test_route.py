@@ -118,3 +118,42 @@ def test_oracle_astar_optimal_vs_independent_dijkstra():
         assert got == pytest.approx(ref, rel=1e-4), (sx, sy, tx, ty2)
         checked += 1
     assert checked >= 8
+
+
+def test_virtual_net_split():
+    """route/vnet.py: spatial sink clustering for sink-parallel routing
+    (reference: create_virtual_nets). Every conn appears in exactly one
+    vnet of its parent; clusters respect the size bound and are spatially
+    coherent (bb area sum <= whole-net bb area per axis split)."""
+    from parallel_eda_amd.route.vnet import split_virtual_nets, cluster_sinks
+    arch = get_arch("tseng")
+    from parallel_eda_amd.io.synth import SynthSpec
+    spec = spec_for_arch(arch, fill=0.4, seed=7)
+    spec.avg_fanout = 10.0
+    nl = synth_netlist(spec)
+    pl = anneal_place(nl, arch, seed=7, timing_tradeoff=0.0)
+    g = rrgraph.build_rr_graph(arch)
+    net_ids, src_rr, sink_ptr, sink_rr, _ = net_rr_terminals(nl, pl, g, arch)
+    xl = np.asarray(g.xlow); yl = np.asarray(g.ylow)
+    parents, vptr, vconns = split_virtual_nets(sink_ptr, sink_rr, xl, yl,
+                                               max_sinks=8)
+    # partition property
+    assert len(vconns) == len(sink_rr)
+    assert sorted(vconns.tolist()) == list(range(len(sink_rr)))
+    sizes = np.diff(vptr)
+    assert (sizes >= 1).all() and (sizes <= 8).all()
+    # each vnet's conns belong to its parent net's range
+    for i in range(len(parents)):
+        lo, hi = sink_ptr[parents[i]], sink_ptr[parents[i] + 1]
+        cs = vconns[vptr[i]:vptr[i + 1]]
+        assert ((cs >= lo) & (cs < hi)).all()
+    # wide nets actually split
+    fan = np.diff(sink_ptr)
+    assert (fan > 8).any()
+    assert len(parents) > len(net_ids)
+    # cluster_sinks determinism
+    rng = np.random.default_rng(1)
+    xs = rng.integers(0, 30, 40); ys = rng.integers(0, 30, 40)
+    a = cluster_sinks(xs, ys, 6)
+    b = cluster_sinks(xs, ys, 6)
+    assert all(np.array_equal(p, q) for p, q in zip(a, b))
