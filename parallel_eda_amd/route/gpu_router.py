@@ -17,6 +17,31 @@ def _torch():
     return torch
 
 
+def schedule_bb_waves(bb, net_ids, areas, nx, ny, cell=8):
+    """Greedy bb-disjoint wave schedule (ParaDRo-style: reference builds a
+    bb-overlap graph and colors it, partitioning_multi_sink...:3598; we
+    rasterize bbs onto a coarse cell grid and greedily assign each net the
+    first wave whose cells are all free — deterministic, O(nets x cells)).
+    Nets in the same wave have disjoint cell footprints, hence disjoint
+    bbs, hence disjoint search state and congestion reads. Largest-area
+    first keeps the giant nets from serializing the tail. Pure host-side
+    numpy — unit-tested on CPU (tests/test_route.py)."""
+    ncx = (nx + 2 + cell - 1) // cell
+    ncy = (ny + 2 + cell - 1) // cell
+    next_free = np.zeros(ncx * ncy, dtype=np.int32)
+    order = net_ids[np.argsort(-areas[net_ids], kind="stable")]
+    wave_of = np.zeros(len(order), dtype=np.int32)
+    for i, n in enumerate(order):
+        cx0 = bb[n, 0] // cell; cy0 = bb[n, 1] // cell
+        cx1 = bb[n, 2] // cell; cy1 = bb[n, 3] // cell
+        blockv = next_free[:].reshape(ncx, ncy)[cx0:cx1 + 1, cy0:cy1 + 1]
+        w = int(blockv.max())
+        blockv[:] = np.maximum(blockv, w + 1)
+        wave_of[i] = w
+    n_waves = int(wave_of.max()) + 1 if len(order) else 0
+    return [order[wave_of == w] for w in range(n_waves)]
+
+
 class GpuRouter:
     def __init__(self, g, arch: ArchDef, src_rr, sink_ptr, sink_rr,
                  device="cuda:0", astar_fac=1.2, n_small_slots=1024,
@@ -230,24 +255,9 @@ class GpuRouter:
         return waves
 
     def _schedule_waves_impl(self, net_ids):
-        cell = 8
-        ncx = (self.arch.nx + 2 + cell - 1) // cell
-        ncy = (self.arch.ny + 2 + cell - 1) // cell
-        next_free = np.zeros(ncx * ncy, dtype=np.int32)
         areas = self._bb_areas(self.bb)
-        order = net_ids[np.argsort(-areas[net_ids], kind="stable")]
-        wave_of = np.zeros(len(order), dtype=np.int32)
-        bb = self.bb
-        for i, n in enumerate(order):
-            cx0 = bb[n, 0] // cell; cy0 = bb[n, 1] // cell
-            cx1 = bb[n, 2] // cell; cy1 = bb[n, 3] // cell
-            blockv = next_free[:].reshape(ncx, ncy)[cx0:cx1 + 1, cy0:cy1 + 1]
-            w = int(blockv.max())
-            blockv[:] = np.maximum(blockv, w + 1)
-            wave_of[i] = w
-        n_waves = int(wave_of.max()) + 1 if len(order) else 0
-        waves = [order[wave_of == w] for w in range(n_waves)]
-        return waves
+        return schedule_bb_waves(self.bb, net_ids, areas,
+                                 self.arch.nx, self.arch.ny)
 
     # ---- one PathFinder iteration ----
     def route_iteration(self, crit, pres_fac, net_subset=None, fail_ok=False):

@@ -157,3 +157,39 @@ def test_virtual_net_split():
     a = cluster_sinks(xs, ys, 6)
     b = cluster_sinks(xs, ys, 6)
     assert all(np.array_equal(p, q) for p, q in zip(a, b))
+
+
+def test_schedule_bb_waves():
+    """The GPU router's wave scheduler is pure host numpy — validate its
+    invariants on CPU: every net scheduled exactly once; nets sharing a
+    wave have disjoint coarse-cell footprints (=> disjoint bbs, disjoint
+    search state); deterministic across calls."""
+    from parallel_eda_amd.route.gpu_router import schedule_bb_waves
+    rng = np.random.default_rng(11)
+    n = 400
+    nx = ny = 60
+    x0 = rng.integers(0, nx, n); y0 = rng.integers(0, ny, n)
+    w = rng.integers(1, 20, n); h = rng.integers(1, 20, n)
+    bb = np.stack([x0, y0, np.minimum(x0 + w, nx + 1),
+                   np.minimum(y0 + h, ny + 1)], axis=1).astype(np.int32)
+    areas = ((bb[:, 2] - bb[:, 0] + 1) *
+             (bb[:, 3] - bb[:, 1] + 1)).astype(np.int64)
+    ids = np.arange(n, dtype=np.int64)
+    waves = schedule_bb_waves(bb, ids, areas, nx, ny)
+    flat = np.concatenate(waves)
+    assert sorted(flat.tolist()) == list(range(n))
+    cell = 8
+    for wv in waves:
+        seen = set()
+        for net in wv:
+            cells = {(cx, cy)
+                     for cx in range(bb[net, 0] // cell, bb[net, 2] // cell + 1)
+                     for cy in range(bb[net, 1] // cell, bb[net, 3] // cell + 1)}
+            assert not (cells & seen), "overlapping nets in one wave"
+            seen |= cells
+    waves2 = schedule_bb_waves(bb, ids, areas, nx, ny)
+    assert all(np.array_equal(a, b) for a, b in zip(waves, waves2))
+    # subset scheduling too
+    sub = ids[::3]
+    wsub = schedule_bb_waves(bb, sub, areas, nx, ny)
+    assert sorted(np.concatenate(wsub).tolist()) == sorted(sub.tolist())
