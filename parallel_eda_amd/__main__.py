@@ -62,12 +62,37 @@ def build_parser():
     p.add_argument("--out_sdf", type=str, default=None,
                    help="post-route SDF timing annotation")
     p.add_argument("--stats_dir", type=str, default=None)
+    p.add_argument("--settings", type=str, default=None,
+                   help="TOML settings file: keys = CLI flag names; "
+                        "explicit CLI flags win (reference: read_settings.c)")
     p.add_argument("--verbose", "-v", action="store_true")
     return p
 
 
+def apply_settings(parser, args, argv):
+    """Overlay a TOML settings file under explicit CLI flags (reference:
+    base/read_settings.c — file supplies defaults, command line wins)."""
+    import tomli
+    with open(args.settings, "rb") as f:
+        cfg = tomli.load(f)
+    given = set()
+    for tok in (argv if argv is not None else sys.argv[1:]):
+        if tok.startswith("--"):
+            given.add(tok[2:].split("=", 1)[0])
+    for k, v in cfg.items():
+        if k in given:
+            continue  # explicit flag wins
+        if not hasattr(args, k):
+            raise SystemExit(f"settings: unknown option {k!r}")
+        setattr(args, k, v)
+    return args
+
+
 def main(argv=None):
-    args = build_parser().parse_args(argv)
+    parser = build_parser()
+    args = parser.parse_args(argv)
+    if args.settings:
+        args = apply_settings(parser, args, argv)
     from .arch.archdef import get_arch
     from .arch.xml_parser import parse_arch_xml, size_grid_for_netlist
     from .io.synth import synth_netlist, spec_for_arch

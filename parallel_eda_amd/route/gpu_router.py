@@ -544,6 +544,13 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
                             else router.num_nets))
         if verbose:
             print(f"[gpu] iter {it}: overused={overused} cpd={cpd*1e9:.2f}ns")
+        import os as _os
+        if _os.environ.get("PNR_CHECK_OCC"):
+            # reference: occ==recalc cross-check EVERY iteration
+            # (partitioning_multi_sink...:6194) — catches lost congestion
+            # updates; off by default (costs one kernel + compare)
+            if not router.check_occ_recount():
+                raise RuntimeError(f"occ recount mismatch at iteration {it}")
         if sta is not None:
             cmap.conn_delays(sink_delays, out=conn_delay)
             cpd, slack, c = sta.analyze(conn_delay)

@@ -38,6 +38,18 @@ class StatsWriter:
             f.write(f"{it} {t:.3f} {overused} {rerouted} {heap_pops} "
                     f"{cpd*1e9:.4f}\n")
 
+    def net_costs(self, it, costs, rank=0):
+        """Per-net measured route cost for one iteration (reference:
+        net_route_time_iter_N_rank_R.txt, mpi_route...cxx:1171): one
+        '<net> <cost>' line per net, nonzero entries only."""
+        path = os.path.join(self.dir,
+                            f"net_cost_iter_{it}_rank_{rank}.txt")
+        c = list(costs)
+        with open(path, "w") as f:
+            for n, v in enumerate(c):
+                if v:
+                    f.write(f"{n} {v}\n")
+
     def final(self, success, wirelength, cpd, extra=None):
         total = time.perf_counter() - self._t0
         data = {

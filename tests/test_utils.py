@@ -151,3 +151,26 @@ def test_sdf_writer(tmp_path):
     # DSP comb delay annotated
     d = arch.T_dsp * 1e12
     assert f"({d:.1f}:{d:.1f}:{d:.1f})" in s
+
+
+def test_stats_net_costs(tmp_path):
+    from parallel_eda_amd.utils.stats import StatsWriter
+    sw = StatsWriter(str(tmp_path))
+    sw.net_costs(3, [0, 5, 0, 7], rank=1)
+    body = (tmp_path / "net_cost_iter_3_rank_1.txt").read_text()
+    assert body == "1 5\n3 7\n"
+
+
+def test_cli_settings_file(tmp_path):
+    from parallel_eda_amd.__main__ import main
+    cfg = tmp_path / "s.toml"
+    cfg.write_text('fill = 0.4\nmax_router_iterations = 50\n')
+    rc = main(["--synth", "tiny", "--seed", "3",
+               "--settings", str(cfg)])
+    assert rc == 0
+    # unknown keys rejected
+    bad = tmp_path / "b.toml"
+    bad.write_text('no_such_option = 1\n')
+    import pytest as _pt
+    with _pt.raises(SystemExit):
+        main(["--synth", "tiny", "--settings", str(bad)])
