@@ -48,6 +48,9 @@ def build_parser():
     p.add_argument("--bb_factor", type=int, default=3)
     p.add_argument("--min_channel_width", action="store_true",
                    help="binary search the minimum routable W")
+    p.add_argument("--rip_up_always", action="store_true",
+                   help="re-route every net every iteration instead of "
+                        "congested-only selective reroute (GPU engine)")
     p.add_argument("--deterministic", action="store_true",
                    help="force the fixed wave schedule (GPU)")
     # outputs
@@ -173,7 +176,8 @@ def main(argv=None):
             pres_fac_init=args.initial_pres_fac,
             pres_fac_mult=args.pres_fac_mult, acc_fac=args.acc_fac,
             astar_fac=args.astar_fac, verbose=args.verbose,
-            engine=args.engine)
+            engine=args.engine, rip_up_always=args.rip_up_always,
+            deterministic=args.deterministic)
         rt = time.perf_counter() - t0
         if not res.success:
             print(f"ROUTING FAILED: {res.overused} overused nodes after "

@@ -118,14 +118,20 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
                      max_iters: int = 60, pres_fac_init: float = 0.5,
                      pres_fac_mult: float = 1.3, acc_fac: float = 1.0,
                      astar_fac: float = 1.2, verbose: bool = False,
-                     engine: str = "cpu"):
-    """Timing-driven PathFinder: route all nets to feasibility."""
+                     engine: str = "cpu", rip_up_always: bool = False,
+                     deterministic: bool = False):
+    """Timing-driven PathFinder: route all nets to feasibility.
+
+    rip_up_always / deterministic reach the GPU engine (the CPU oracle
+    rips all nets every iteration by construction and is serial, hence
+    already deterministic)."""
     if engine == "gpu":
         from .gpu_router import pathfinder_route_gpu
         return pathfinder_route_gpu(
             netlist, placement, g, arch, sta=sta, max_iters=max_iters,
             pres_fac_init=pres_fac_init, pres_fac_mult=pres_fac_mult,
-            acc_fac=acc_fac, astar_fac=astar_fac, verbose=verbose)
+            acc_fac=acc_fac, astar_fac=astar_fac, verbose=verbose,
+            rip_up_always=rip_up_always, deterministic=deterministic)
     cpu = ops.cpu()
     net_ids, src_rr, sink_ptr, sink_rr, conn_index = net_rr_terminals(
         netlist, placement, g, arch)
