@@ -177,7 +177,7 @@ def main(argv=None):
             pres_fac_mult=args.pres_fac_mult, acc_fac=args.acc_fac,
             astar_fac=args.astar_fac, verbose=args.verbose,
             engine=args.engine, rip_up_always=args.rip_up_always,
-            deterministic=args.deterministic)
+            deterministic=args.deterministic, bb_factor=args.bb_factor)
         rt = time.perf_counter() - t0
         if not res.success:
             print(f"ROUTING FAILED: {res.overused} overused nodes after "

@@ -508,14 +508,18 @@ def ct_ptr(t):
 def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
                          pres_fac_init=0.5, pres_fac_mult=1.3, acc_fac=1.0,
                          astar_fac=1.2, verbose=False, device="cuda:0",
-                         rip_up_always=False, deterministic=False):
-    """GPU PathFinder outer loop — mirrors route.router.pathfinder_route."""
+                         rip_up_always=False, deterministic=False,
+                         bb_factor=4):
+    """GPU PathFinder outer loop — mirrors route.router.pathfinder_route.
+
+    bb_factor: initial per-net bounding-box margin in tiles (reference:
+    -bb_factor route option); grown automatically on route failure."""
     net_ids, src_rr, sink_ptr, sink_rr, conn_index = net_rr_terminals(
         netlist, placement, g, arch)
     from .router import ConnMap
     router = GpuRouter(g, arch, src_rr, sink_ptr.astype(np.int32), sink_rr,
                        device=device, astar_fac=astar_fac,
-                       deterministic=deterministic)
+                       deterministic=deterministic, bb_margin=bb_factor)
     n_rsinks = len(sink_rr)
     cmap = ConnMap(conn_index, sink_ptr, netlist.num_conns, n_rsinks)
     crit = np.zeros(n_rsinks, dtype=np.float32)

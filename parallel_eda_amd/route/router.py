@@ -119,7 +119,7 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
                      pres_fac_mult: float = 1.3, acc_fac: float = 1.0,
                      astar_fac: float = 1.2, verbose: bool = False,
                      engine: str = "cpu", rip_up_always: bool = False,
-                     deterministic: bool = False):
+                     deterministic: bool = False, bb_factor: int = 4):
     """Timing-driven PathFinder: route all nets to feasibility.
 
     rip_up_always / deterministic reach the GPU engine (the CPU oracle
@@ -131,7 +131,8 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
             netlist, placement, g, arch, sta=sta, max_iters=max_iters,
             pres_fac_init=pres_fac_init, pres_fac_mult=pres_fac_mult,
             acc_fac=acc_fac, astar_fac=astar_fac, verbose=verbose,
-            rip_up_always=rip_up_always, deterministic=deterministic)
+            rip_up_always=rip_up_always, deterministic=deterministic,
+            bb_factor=bb_factor)
     cpu = ops.cpu()
     net_ids, src_rr, sink_ptr, sink_rr, conn_index = net_rr_terminals(
         netlist, placement, g, arch)
