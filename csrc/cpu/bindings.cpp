@@ -239,6 +239,13 @@ PYBIND11_MODULE(_pnr_cpu, m) {
                                py::array_t<int32_t, py::array::c_style | py::array::forcecast> s) {
         p.set_placement(x.data(), y.data(), s.data());
       })
+      .def("fix_blocks", [](SerialPlacer& p,
+                            py::array_t<int32_t, py::array::c_style | py::array::forcecast> ids,
+                            py::array_t<int32_t, py::array::c_style | py::array::forcecast> x,
+                            py::array_t<int32_t, py::array::c_style | py::array::forcecast> y,
+                            py::array_t<int32_t, py::array::c_style | py::array::forcecast> s) {
+        p.fix_blocks(ids.data(), x.data(), y.data(), s.data(), ids.size());
+      })
       .def("check_place", [](SerialPlacer& p) {
         std::string err;
         bool ok = p.check_place(&err);

@@ -160,6 +160,35 @@ class SerialPlacer {
 
   uint64_t rand_u64() { return rng_(); }
 
+  // Pin blocks to fixed locations (reference: -pad_loc_file / fix_pins,
+  // place.c initial_placement_location with pad constraints): teleport
+  // each block to its location, then exclude it from all moves.
+  void fix_blocks(const int32_t* ids, const int32_t* fx, const int32_t* fy,
+                  const int32_t* fslot, int64_t n) {
+    if (fixed_.empty()) fixed_.assign(nl_->num_blocks, 0);
+    for (int64_t i = 0; i < n; ++i) {
+      int b = ids[i];
+      int x = fx[i], y = fy[i], sl = fslot[i];
+      if (tile_type(x, y) != nl_->block_type[b])
+        throw std::runtime_error("fix_blocks: type/location mismatch");
+      if (sl < 0 || sl >= cap_at(x, y))
+        throw std::runtime_error("fix_blocks: bad slot");
+      int32_t occ = grid_at(x, y, sl);
+      if (occ == b) { fixed_[b] = 1; continue; }
+      // displaced occupant swaps into b's old slot
+      grid_at(bx_[b], by_[b], bslot_[b]) = occ;
+      if (occ >= 0) {
+        if (fixed_[occ]) throw std::runtime_error("fix_blocks: collision");
+        bx_[occ] = bx_[b]; by_[occ] = by_[b]; bslot_[occ] = bslot_[b];
+      }
+      bx_[b] = x; by_[b] = y; bslot_[b] = sl;
+      grid_at(x, y, sl) = b;
+      fixed_[b] = 1;
+    }
+    recompute_bb_all();
+    recompute_td_all();
+  }
+
  public:
   std::shared_ptr<Netlist> netlist_holder_;  // lifetime pin for Python bindings
   const Netlist* nl_;
@@ -169,6 +198,7 @@ class SerialPlacer {
   std::vector<int32_t> bx_, by_, bslot_;
   std::vector<int8_t> tile_btype_;       // (gx*gy) x-major; see ctor
   std::vector<int> type_cols_[2];        // columns of type RAM(0) / DSP(1)
+  std::vector<uint8_t> fixed_;           // empty => nothing fixed
 
  private:
   std::vector<int32_t> grid_;      // (x*gy+y)*cap + slot -> block
@@ -358,6 +388,7 @@ class SerialPlacer {
                double bb_norm, double td_norm) {
     int nb = nl_->num_blocks;
     int blk = (int)(rng_() % nb);
+    if (!fixed_.empty() && fixed_[blk]) return 0;
     int x0 = bx_[blk], y0 = by_[blk];
     int btype = nl_->block_type[blk];
     // find_to: range-limited destination of matching type (place.c:1520)
@@ -400,6 +431,7 @@ class SerialPlacer {
     if (x1 < 0) return 0;
     int other = grid_at(x1, y1, slot1);
     if (other == blk) return 0;
+    if (other >= 0 && !fixed_.empty() && fixed_[other]) return 0;
 
     // save + compute delta over affected nets
     saved_bbs_.clear(); saved_costs_.clear(); saved_ids_.clear();

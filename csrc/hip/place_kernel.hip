@@ -23,6 +23,7 @@ struct PlaceDev {
                                 // nullptr => homogeneous (perimeter IO)
   const int32_t* type_cols;     // sorted column lists: RAM cols then DSP
   const int32_t* type_col_ptr;  // [3] bounds into type_cols (RAM, DSP)
+  const uint8_t* fixed;         // per-block pin-down mask; nullptr = none
   const float* net_q;           // crossing factor per net
   // timing
   const int32_t* net_sink_ptr;  // [num_nets+1] conn ranges
@@ -124,6 +125,7 @@ __global__ void place_propose_kernel(PlaceDev p, MovesDev m, float T,
   // pick block + destination
   uint32_t r0 = rng_hash(seed ^ 0x5BD1E995u, batch, i);
   int32_t blk = r0 % p.num_blocks;
+  if (p.fixed && p.fixed[blk]) return;   // pinned (pad_loc_file)
   bool io = p.blk_type[blk] == 0;
   int x0 = p.bx[blk], y0 = p.by[blk];
   int x1 = -1, y1 = -1, slot1 = 0;
@@ -175,6 +177,7 @@ __global__ void place_propose_kernel(PlaceDev p, MovesDev m, float T,
   if (x1 < 0) return;
   int32_t other = p.grid[((int64_t)x1 * p.gy + y1) * p.cap + slot1];
   if (other == blk) return;
+  if (other >= 0 && p.fixed && p.fixed[other]) return;
   atomicAdd(&m.counters[0], 1);  // valid proposals only
 
   // collect affected nets (dedup)
@@ -327,6 +330,7 @@ struct PlaceLaunchArgs {
   const int32_t* blk_net_ptr; const int32_t* blk_nets;
   const int8_t* blk_type; const int8_t* tile_btype;
   const int32_t* type_cols; const int32_t* type_col_ptr;
+  const uint8_t* fixed;
   const float* net_q;
   const int32_t* net_sink_ptr; const float* conn_crit; const float* delay_mat;
   int32_t* bx; int32_t* by; int32_t* bslot; int32_t* grid;
@@ -349,6 +353,7 @@ static void unpack(const PlaceLaunchArgs* a, PlaceDev& p, MovesDev& m) {
   p.blk_net_ptr = a->blk_net_ptr; p.blk_nets = a->blk_nets;
   p.blk_type = a->blk_type; p.tile_btype = a->tile_btype;
   p.type_cols = a->type_cols; p.type_col_ptr = a->type_col_ptr;
+  p.fixed = a->fixed;
   p.net_q = a->net_q;
   p.net_sink_ptr = a->net_sink_ptr; p.conn_crit = a->conn_crit;
   p.delay_mat = a->delay_mat;

@@ -30,6 +30,9 @@ def build_parser():
                    choices=["timing_driven", "breadth_first"],
                    default="timing_driven")
     # placer opts (reference: s_placer_opts)
+    p.add_argument("--pad_loc_file", type=str, default=None,
+                   help="pin IO pads (or any named blocks) to fixed "
+                        "locations: lines of 'name x y subblk'")
     p.add_argument("--place_file", type=str, default=None,
                    help="read placement instead of annealing")
     p.add_argument("--timing_tradeoff", type=float, default=0.5)
@@ -147,11 +150,16 @@ def main(argv=None):
         placement = read_place(args.place_file, netlist)
         print(f"read placement from {args.place_file}")
     else:
+        fixed = None
+        if args.pad_loc_file:
+            from .io.place_file import read_pad_loc
+            fixed = read_pad_loc(args.pad_loc_file, netlist)
+            print(f"pinned {len(fixed[0])} blocks from {args.pad_loc_file}")
         placement = anneal_place(
             netlist, arch, seed=args.seed,
             timing_tradeoff=args.timing_tradeoff if timing else 0.0,
             inner_num=args.inner_num, sta=sta, verbose=args.verbose,
-            engine=args.engine, delay_matrix=args.delay_matrix)
+            engine=args.engine, delay_matrix=args.delay_matrix, fixed=fixed)
         print(f"placement: bb_cost={placement.bb_cost:.1f} "
               f"({time.perf_counter()-t0:.2f}s)")
     if args.out_place:

@@ -57,3 +57,29 @@ def read_place(path, netlist):
         raise ValueError(f".place file has {seen} blocks, netlist has {nb}")
     from ..place.placer import Placement
     return Placement(x, y, slot)
+
+
+def read_pad_loc(path, netlist):
+    """VPR pad location file (reference: -pad_loc_file, read_place.c
+    read_user_pad_loc): lines of "blockname x y subblk"; '#' comments.
+    Returns (ids, x, y, slot) int32 arrays for blocks found by name."""
+    import numpy as np
+    names = {n: i for i, n in enumerate(netlist.names or [])}
+    ids, xs, ys, ss = [], [], [], []
+    with open(path) as f:
+        for raw in f:
+            line = raw.split("#", 1)[0].strip()
+            if not line:
+                continue
+            toks = line.split()
+            if len(toks) < 3:
+                raise ValueError(f"bad pad_loc line: {raw!r}")
+            name = toks[0]
+            if name not in names:
+                raise ValueError(f"pad_loc: unknown block {name!r}")
+            ids.append(names[name])
+            xs.append(int(toks[1]))
+            ys.append(int(toks[2]))
+            ss.append(int(toks[3]) if len(toks) > 3 else 0)
+    return (np.asarray(ids, dtype=np.int32), np.asarray(xs, dtype=np.int32),
+            np.asarray(ys, dtype=np.int32), np.asarray(ss, dtype=np.int32))

@@ -21,6 +21,7 @@ class PlaceLaunchArgs(ct.Structure):
         ("blk_net_ptr", ct.c_void_p), ("blk_nets", ct.c_void_p),
         ("blk_type", ct.c_void_p), ("tile_btype", ct.c_void_p),
         ("type_cols", ct.c_void_p), ("type_col_ptr", ct.c_void_p),
+        ("fixed", ct.c_void_p),
         ("net_q", ct.c_void_p),
         ("net_sink_ptr", ct.c_void_p), ("conn_crit", ct.c_void_p),
         ("delay_mat", ct.c_void_p),
@@ -61,7 +62,7 @@ def ptr(t):
 
 class GpuPlacer:
     def __init__(self, netlist, arch: ArchDef, seed=7, timing=False,
-                 device="cuda:0", n_moves=None):
+                 device="cuda:0", n_moves=None, fixed=None):
         import torch
         self.torch = torch
         self.device = device
@@ -124,6 +125,13 @@ class GpuPlacer:
                 [0, len(ram_cols), len(ram_cols) + len(dsp_cols)],
                 dtype=np.int32))
         self.t_net_q = up(q)
+        # pinned blocks (pad_loc_file): mask on device, teleport at init
+        self._fixed = fixed
+        self.t_fixed = None
+        if fixed is not None:
+            mask = np.zeros(nb, dtype=np.uint8)
+            mask[np.asarray(fixed[0], dtype=np.int64)] = 1
+            self.t_fixed = up(mask)
         self.t_net_sink_ptr = up(netlist.net_sink_ptr.astype(np.int32))
         self.t_conn_crit = torch.zeros(netlist.num_conns, dtype=torch.float32,
                                        device=device)
@@ -167,6 +175,28 @@ class GpuPlacer:
         bx = np.zeros(nb, dtype=np.int32)
         by = np.zeros(nb, dtype=np.int32)
         bslot = np.zeros(nb, dtype=np.int32)
+        if self._fixed is not None:
+            bx2, by2, bs2, grid = self._initial_placement_free(rng)
+            ids, fx, fy, fs = [np.asarray(a) for a in self._fixed]
+            for b, x, y, sl in zip(ids, fx, fy, fs):
+                occ = grid[x, y, sl]
+                if occ == b:
+                    continue
+                grid[bx2[b], by2[b], bs2[b]] = occ
+                if occ >= 0:
+                    bx2[occ], by2[occ], bs2[occ] = bx2[b], by2[b], bs2[b]
+                bx2[b], by2[b], bs2[b] = x, y, sl
+                grid[x, y, sl] = b
+            return bx2, by2, bs2, grid.reshape(-1)
+        return self._initial_placement_free(rng, flat=True)
+
+    def _initial_placement_free(self, rng, flat=False):
+        arch = self.arch
+        nl = self.nl
+        nb = nl.num_blocks
+        bx = np.zeros(nb, dtype=np.int32)
+        by = np.zeros(nb, dtype=np.int32)
+        bslot = np.zeros(nb, dtype=np.int32)
         grid = np.full((self.gx, self.gy, self.cap), -1, dtype=np.int32)
         ios = np.nonzero(nl.block_type == 0)[0]
         if not arch.is_heterogeneous():
@@ -203,7 +233,8 @@ class GpuPlacer:
             x, y, s = slots[k]
             bx[b], by[b], bslot[b] = x, y, s
             grid[x, y, s] = b
-        return bx, by, bslot, grid.reshape(-1)
+        return (bx, by, bslot, grid.reshape(-1)) if flat \
+            else (bx, by, bslot, grid)
 
     def _args(self, T=0.0, rlim=1, tt=0.0, inv_bb=1.0, inv_td=1.0):
         a = PlaceLaunchArgs()
@@ -216,6 +247,7 @@ class GpuPlacer:
                        if self.t_type_cols is not None else None)
         a.type_col_ptr = (ptr(self.t_type_col_ptr)
                           if self.t_type_col_ptr is not None else None)
+        a.fixed = ptr(self.t_fixed) if self.t_fixed is not None else None
         a.net_q = ptr(self.t_net_q)
         a.net_sink_ptr = ptr(self.t_net_sink_ptr)
         a.conn_crit = ptr(self.t_conn_crit)
@@ -327,11 +359,11 @@ def _cross_count(n):
 
 def anneal_place_gpu(netlist, arch, seed=7, timing_tradeoff=0.5, inner_num=1.0,
                      sta=None, crit_exp=1.0, verbose=False, device="cuda:0",
-                     n_moves=None):
+                     n_moves=None, fixed=None):
     """GPU anneal with the same adaptive schedule as the CPU oracle."""
     timing = sta is not None and timing_tradeoff > 0
     placer = GpuPlacer(netlist, arch, seed=seed, timing=timing, device=device,
-                       n_moves=n_moves)
+                       n_moves=n_moves, fixed=fixed)
     nb = netlist.num_blocks
     move_lim = max(256, int(inner_num * (nb ** 1.3333)))
     rlim = float(max(arch.nx, arch.ny))

@@ -46,18 +46,21 @@ def analytic_delay_matrix(arch: ArchDef) -> np.ndarray:
 def anneal_place(netlist, arch: ArchDef, seed: int = 7, timing_tradeoff: float = 0.5,
                  inner_num: float = 1.0, sta=None, crit_exp: float = 1.0,
                  verbose: bool = False, engine: str = "cpu",
-                 delay_matrix: str = "analytic") -> Placement:
+                 delay_matrix: str = "analytic", fixed=None) -> Placement:
     """Run the full SA schedule; returns final Placement.
 
     sta: optional TimingGraph wrapper (timing.sta.STA) for criticality
     refresh each temperature; None => pure bounding-box placement.
+    fixed: optional (ids, x, y, slot) arrays pinning blocks to locations
+    (reference: -pad_loc_file / fix_pins); pinned blocks never move.
     """
     if engine == "gpu":
         from .gpu_placer import anneal_place_gpu
         return anneal_place_gpu(netlist, arch, seed=seed,
                                 timing_tradeoff=timing_tradeoff,
                                 inner_num=inner_num, sta=sta,
-                                crit_exp=crit_exp, verbose=verbose)
+                                crit_exp=crit_exp, verbose=verbose,
+                                fixed=fixed)
     cpu = ops.cpu()
     if sta is not None and timing_tradeoff > 0:
         if delay_matrix == "routed":
@@ -71,6 +74,12 @@ def anneal_place(netlist, arch: ArchDef, seed: int = 7, timing_tradeoff: float =
           else np.empty(0, dtype=np.int8))
     placer = cpu.SerialPlacer(netlist.cpp(), arch.nx, arch.ny, arch.io_cap,
                               np.ascontiguousarray(dm.ravel()), seed, tb)
+    if fixed is not None:
+        ids, fx, fy, fs = fixed
+        placer.fix_blocks(np.asarray(ids, dtype=np.int32),
+                          np.asarray(fx, dtype=np.int32),
+                          np.asarray(fy, dtype=np.int32),
+                          np.asarray(fs, dtype=np.int32))
     nb = netlist.num_blocks
     move_lim = max(64, int(inner_num * (nb ** 1.3333)))
     rlim = float(max(arch.nx, arch.ny))
