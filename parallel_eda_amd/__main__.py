@@ -36,6 +36,10 @@ def build_parser():
     p.add_argument("--place_file", type=str, default=None,
                    help="read placement instead of annealing")
     p.add_argument("--timing_tradeoff", type=float, default=0.5)
+    p.add_argument("--criticality_exp", type=float, default=1.0,
+                   help="criticality exponent (reference: -place_cost_exp"
+                        " / criticality_exp router opt)")
+    p.add_argument("--max_criticality", type=float, default=0.99)
     p.add_argument("--delay_matrix", choices=["analytic", "routed"],
                    default="analytic",
                    help="placer delay LUT: analytic Elmore or router-measured"
@@ -159,7 +163,8 @@ def main(argv=None):
             netlist, arch, seed=args.seed,
             timing_tradeoff=args.timing_tradeoff if timing else 0.0,
             inner_num=args.inner_num, sta=sta, verbose=args.verbose,
-            engine=args.engine, delay_matrix=args.delay_matrix, fixed=fixed)
+            engine=args.engine, delay_matrix=args.delay_matrix, fixed=fixed,
+            crit_exp=args.criticality_exp)
         print(f"placement: bb_cost={placement.bb_cost:.1f} "
               f"({time.perf_counter()-t0:.2f}s)")
     if args.out_place:
@@ -185,7 +190,9 @@ def main(argv=None):
             pres_fac_mult=args.pres_fac_mult, acc_fac=args.acc_fac,
             astar_fac=args.astar_fac, verbose=args.verbose,
             engine=args.engine, rip_up_always=args.rip_up_always,
-            deterministic=args.deterministic, bb_factor=args.bb_factor)
+            deterministic=args.deterministic, bb_factor=args.bb_factor,
+            crit_exp=args.criticality_exp,
+            max_criticality=args.max_criticality)
         rt = time.perf_counter() - t0
         if not res.success:
             print(f"ROUTING FAILED: {res.overused} overused nodes after "

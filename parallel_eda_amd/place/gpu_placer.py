@@ -62,7 +62,8 @@ def ptr(t):
 
 class GpuPlacer:
     def __init__(self, netlist, arch: ArchDef, seed=7, timing=False,
-                 device="cuda:0", n_moves=None, fixed=None):
+                 device="cuda:0", n_moves=None, fixed=None,
+                 delay_matrix="analytic"):
         import torch
         self.torch = torch
         self.device = device
@@ -136,9 +137,15 @@ class GpuPlacer:
         self.t_conn_crit = torch.zeros(netlist.num_conns, dtype=torch.float32,
                                        device=device)
         if timing:
-            dm = analytic_delay_matrix(arch)
+            if delay_matrix == "routed":
+                from .delay_matrix import routed_delay_matrix
+                dm = routed_delay_matrix(arch)
+            else:
+                dm = analytic_delay_matrix(arch)
+            self._dm = dm
             self.t_delay_mat = up(dm.ravel())
         else:
+            self._dm = None
             self.t_delay_mat = None
 
         # initial placement (host, deterministic)
@@ -359,11 +366,11 @@ def _cross_count(n):
 
 def anneal_place_gpu(netlist, arch, seed=7, timing_tradeoff=0.5, inner_num=1.0,
                      sta=None, crit_exp=1.0, verbose=False, device="cuda:0",
-                     n_moves=None, fixed=None):
+                     n_moves=None, fixed=None, delay_matrix="analytic"):
     """GPU anneal with the same adaptive schedule as the CPU oracle."""
     timing = sta is not None and timing_tradeoff > 0
     placer = GpuPlacer(netlist, arch, seed=seed, timing=timing, device=device,
-                       n_moves=n_moves, fixed=fixed)
+                       n_moves=n_moves, fixed=fixed, delay_matrix=delay_matrix)
     nb = netlist.num_blocks
     move_lim = max(256, int(inner_num * (nb ** 1.3333)))
     rlim = float(max(arch.nx, arch.ny))
@@ -376,7 +383,7 @@ def anneal_place_gpu(netlist, arch, seed=7, timing_tradeoff=0.5, inner_num=1.0,
         bx = placer.t_bx.cpu().numpy(); by = placer.t_by.cpu().numpy()
         drv = netlist.net_driver
         d_per_conn = np.empty(netlist.num_conns, dtype=np.float32)
-        dm = analytic_delay_matrix(arch)
+        dm = placer._dm if placer._dm is not None else analytic_delay_matrix(arch)
         net_of_conn = np.repeat(np.arange(netlist.num_nets),
                                 np.diff(netlist.net_sink_ptr))
         dx = np.abs(bx[netlist.net_sinks] - bx[drv[net_of_conn]])

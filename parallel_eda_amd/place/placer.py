@@ -60,7 +60,7 @@ def anneal_place(netlist, arch: ArchDef, seed: int = 7, timing_tradeoff: float =
                                 timing_tradeoff=timing_tradeoff,
                                 inner_num=inner_num, sta=sta,
                                 crit_exp=crit_exp, verbose=verbose,
-                                fixed=fixed)
+                                fixed=fixed, delay_matrix=delay_matrix)
     cpu = ops.cpu()
     if sta is not None and timing_tradeoff > 0:
         if delay_matrix == "routed":

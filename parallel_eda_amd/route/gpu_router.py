@@ -509,7 +509,7 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
                          pres_fac_init=0.5, pres_fac_mult=1.3, acc_fac=1.0,
                          astar_fac=1.2, verbose=False, device="cuda:0",
                          rip_up_always=False, deterministic=False,
-                         bb_factor=4):
+                         bb_factor=4, crit_exp=1.0, max_criticality=0.99):
     """GPU PathFinder outer loop — mirrors route.router.pathfinder_route.
 
     bb_factor: initial per-net bounding-box margin in tiles (reference:
@@ -558,7 +558,8 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
         if sta is not None:
             cmap.conn_delays(sink_delays, out=conn_delay)
             cpd, slack, c = sta.analyze(conn_delay)
-            crit = cmap.sink_crit(c)
+            crit = cmap.sink_crit(c, max_crit=max_criticality,
+                                  crit_exp=crit_exp)
         if overused == 0:
             break
         pres_fac = pres_fac_init if pres_fac == 0.0 else pres_fac * pres_fac_mult

@@ -119,7 +119,8 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
                      pres_fac_mult: float = 1.3, acc_fac: float = 1.0,
                      astar_fac: float = 1.2, verbose: bool = False,
                      engine: str = "cpu", rip_up_always: bool = False,
-                     deterministic: bool = False, bb_factor: int = 4):
+                     deterministic: bool = False, bb_factor: int = 4,
+                     crit_exp: float = 1.0, max_criticality: float = 0.99):
     """Timing-driven PathFinder: route all nets to feasibility.
 
     rip_up_always / deterministic reach the GPU engine (the CPU oracle
@@ -132,7 +133,8 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
             pres_fac_init=pres_fac_init, pres_fac_mult=pres_fac_mult,
             acc_fac=acc_fac, astar_fac=astar_fac, verbose=verbose,
             rip_up_always=rip_up_always, deterministic=deterministic,
-            bb_factor=bb_factor)
+            bb_factor=bb_factor, crit_exp=crit_exp,
+            max_criticality=max_criticality)
     cpu = ops.cpu()
     net_ids, src_rr, sink_ptr, sink_rr, conn_index = net_rr_terminals(
         netlist, placement, g, arch)
@@ -161,7 +163,8 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
             # net delays -> connection delays -> STA -> criticality
             cmap.conn_delays(router.sink_delays(), out=conn_delay)
             cpd, slack, c = sta.analyze(conn_delay)
-            crit = cmap.sink_crit(c)
+            crit = cmap.sink_crit(c, max_crit=max_criticality,
+                              crit_exp=crit_exp)
         if overused == 0 and router.unrouted_sinks() == 0:
             break
         pres_fac = pres_fac_init if it == 1 else pres_fac * pres_fac_mult
