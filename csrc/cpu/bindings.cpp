@@ -153,6 +153,15 @@ PYBIND11_MODULE(_pnr_cpu, m) {
         { py::gil_scoped_release rel; over = r.route_subset(c, ids.data(), ids.size()); }
         return over;
       })
+      .def("route_subset_incremental", [](SerialRouter& r,
+                              py::array_t<float, py::array::c_style | py::array::forcecast> crit,
+                              py::array_t<int32_t, py::array::c_style | py::array::forcecast> ids,
+                              float crit_rip_thr) {
+        const float* c = crit.size() ? crit.data() : nullptr;
+        int64_t over;
+        { py::gil_scoped_release rel; over = r.route_subset_incremental(c, ids.data(), ids.size(), crit_rip_thr); }
+        return over;
+      }, py::arg("crit"), py::arg("ids"), py::arg("crit_rip_thr") = 2.0f)
       .def("rip_up_nets", [](SerialRouter& r,
                              py::array_t<int32_t, py::array::c_style | py::array::forcecast> ids) {
         r.rip_up_nets(ids.data(), ids.size());
@@ -168,6 +177,12 @@ PYBIND11_MODULE(_pnr_cpu, m) {
       .def("count_overused", &SerialRouter::count_overused)
       .def("unrouted_sinks", &SerialRouter::unrouted_sinks)
       .def("feasible", &SerialRouter::feasible)
+      .def("incomplete_nets", [](SerialRouter& r) {
+        auto v = r.incomplete_nets();
+        py::array_t<int32_t> out((py::ssize_t)v.size());
+        std::copy(v.begin(), v.end(), out.mutable_data());
+        return out;
+      })
       .def("congested_nets", [](SerialRouter& r) {
         auto v = r.congested_nets();
         py::array_t<int32_t> out((py::ssize_t)v.size());

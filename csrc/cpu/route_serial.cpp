@@ -16,6 +16,7 @@
 #include <queue>
 #include <algorithm>
 #include <cstring>
+#include <unordered_set>
 
 namespace pnr {
 
@@ -86,6 +87,114 @@ class SerialRouter {
   int64_t route_subset(const float* crit, const int32_t* ids, int64_t n) {
     for (int64_t i = 0; i < n; ++i) route_net(ids[i], crit);
     return count_overused();
+  }
+
+  // Partial rip-up + incremental reroute (reference:
+  // route_tree_mark_congested_nodes_to_be_ripped family,
+  // route_tree.h:157-165): drop every subtree whose root-path crosses an
+  // overused node, keep the clean remainder as seeds, and re-route only
+  // the sinks that lost their path. Full rip when the SOURCE is congested.
+  // crit_rip_thr: additionally rip the private path of any sink whose
+  // criticality meets the threshold, so near-critical connections are
+  // re-optimized as slack tightens (reference: timing-driven incremental
+  // rerouting of critical connections); > 1 disables.
+  int64_t route_subset_incremental(const float* crit, const int32_t* ids,
+                                   int64_t n, float crit_rip_thr = 2.0f) {
+    unrouted_sinks_ = 0;
+    for (int64_t i = 0; i < n; ++i)
+      route_net_incremental(ids[i], crit, crit_rip_thr);
+    return count_overused();
+  }
+
+  // Nets with at least one sink not present in their current tree
+  // (never routed, or dropped by a partial rip) — must be in every
+  // incremental reroute set or the flow could terminate with a missing
+  // connection (check_routed would catch it, but late).
+  std::vector<int32_t> incomplete_nets() const {
+    std::vector<int32_t> out;
+    std::unordered_set<int32_t> in_tree;
+    for (int inet = 0; inet < num_nets(); ++inet) {
+      in_tree.clear();
+      in_tree.insert(trees_[inet].nodes.begin(), trees_[inet].nodes.end());
+      for (int64_t k = sink_ptr_[inet]; k < sink_ptr_[inet + 1]; ++k)
+        if (!in_tree.count(sinks_[k])) { out.push_back(inet); break; }
+    }
+    return out;
+  }
+
+  void route_net_incremental(int inet, const float* crit_flat,
+                             float crit_rip_thr = 2.0f) {
+    RouteTree& tree = trees_[inet];
+    if (tree.nodes.empty()) { route_net(inet, crit_flat); return; }
+    int nt = (int)tree.nodes.size();
+    std::vector<char> drop(nt, 0);
+    bool any = false;
+    for (int k = 0; k < nt; ++k) {
+      bool cong = occ_[tree.nodes[k]] > g_->capacity[tree.nodes[k]];
+      bool par = tree.parent[k] >= 0 && drop[tree.parent[k]];
+      drop[k] = cong || par;  // parents precede children in tree order
+      any |= (bool)drop[k];
+    }
+    if (crit_flat && crit_rip_thr <= 1.0f) {
+      // rip the PRIVATE path of each near-critical sink (up to the first
+      // node still feeding a surviving branch) so it can re-route for
+      // delay; shared trunk stays.
+      std::vector<int16_t> surv(nt, 0);
+      for (int k = 1; k < nt; ++k)
+        if (!drop[k]) surv[tree.parent[k]]++;
+      int64_t s0 = sink_ptr_[inet], s1 = sink_ptr_[inet + 1];
+      for (int64_t c = s0; c < s1; ++c) {
+        if (crit_flat[c] < crit_rip_thr) continue;
+        int32_t snode = sinks_[c];
+        for (int k = 0; k < nt; ++k) {
+          if (tree.nodes[k] != snode || drop[k]) continue;
+          int cur = k;
+          while (cur > 0 && !drop[cur] && surv[cur] == 0) {
+            drop[cur] = 1;
+            any = true;
+            int p = tree.parent[cur];
+            if (p >= 0) surv[p]--;
+            cur = p;
+          }
+          break;
+        }
+      }
+    }
+    if (any && drop[0]) { route_net(inet, crit_flat); return; }
+    if (any) {
+      RouteTree kept;
+      std::vector<int32_t> remap(nt, -1);
+      for (int k = 0; k < nt; ++k) {
+        if (drop[k]) { update_one_cost(tree.nodes[k], -1); continue; }
+        remap[k] = (int32_t)kept.nodes.size();
+        kept.nodes.push_back(tree.nodes[k]);
+        kept.parent.push_back(tree.parent[k] < 0 ? -1 : remap[tree.parent[k]]);
+        kept.sw.push_back(tree.sw[k]);
+        kept.delay.push_back(tree.delay[k]);
+      }
+      tree = std::move(kept);
+    }
+    // route sinks whose SINK node is no longer in the tree
+    std::unordered_set<int32_t> in_tree(tree.nodes.begin(), tree.nodes.end());
+    int64_t s0 = sink_ptr_[inet], s1 = sink_ptr_[inet + 1];
+    int ns = (int)(s1 - s0);
+    sink_order_.resize(ns);
+    for (int i = 0; i < ns; ++i) sink_order_[i] = i;
+    std::sort(sink_order_.begin(), sink_order_.end(), [&](int a, int b) {
+      float ca = crit_flat ? crit_flat[s0 + a] : 0.f;
+      float cb = crit_flat ? crit_flat[s0 + b] : 0.f;
+      if (ca != cb) return ca > cb;
+      return a < b;
+    });
+    if ((int64_t)sink_delays_.size() < s1) sink_delays_.resize(sinks_.size(), 0.f);
+    for (int i = 0; i < ns; ++i) {
+      int si = sink_order_[i];
+      int32_t sink = sinks_[s0 + si];
+      if (in_tree.count(sink)) continue;  // path intact; delay unchanged
+      float crit = crit_flat ? crit_flat[s0 + si] : 0.0f;
+      sink_delays_[s0 + si] = route_one_sink(inet, sink, crit, trees_[inet]);
+      in_tree.insert(sink);
+    }
   }
 
   // Rip up the given nets' trees (occ -1, trees cleared) without

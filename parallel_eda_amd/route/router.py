@@ -120,7 +120,10 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
                      astar_fac: float = 1.2, verbose: bool = False,
                      engine: str = "cpu", rip_up_always: bool = False,
                      deterministic: bool = False, bb_factor: int = 4,
-                     crit_exp: float = 1.0, max_criticality: float = 0.99):
+                     crit_exp: float = 1.0, max_criticality: float = 0.99,
+                     incremental: bool = False,
+                     crit_rip_threshold: float = 0.99,
+                     incremental_start: int = 2):
     """Timing-driven PathFinder: route all nets to feasibility.
 
     rip_up_always / deterministic reach the GPU engine (the CPU oracle
@@ -153,9 +156,28 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
     it = 0
     overused = -1
     for it in range(1, max_iters + 1):
-        overused = router.route_iteration(crit)
+        rerouted = len(net_ids)
+        if incremental and it > incremental_start:
+            # selective + partial rip-up (reference phase-two +
+            # route_tree_mark_congested_...): reroute congested nets and
+            # nets with missing sinks; keep their clean subtrees
+            todo = np.union1d(np.asarray(router.congested_nets()),
+                              np.asarray(router.incomplete_nets()))
+            if sta is not None and crit_rip_threshold <= 1.0:
+                hot = np.nonzero(crit >= crit_rip_threshold)[0]
+                if len(hot):
+                    net_of_rsink = np.repeat(
+                        np.arange(len(net_ids)), np.diff(sink_ptr))
+                    todo = np.union1d(todo, net_of_rsink[hot])
+            rerouted = len(todo)
+            overused = router.route_subset_incremental(
+                crit, todo.astype(np.int32),
+                crit_rip_thr=crit_rip_threshold)
+        else:
+            overused = router.route_iteration(crit)
         history.append(dict(iter=it, overused=int(overused),
-                            pops=router.heap_pops(), cpd=cpd))
+                            pops=router.heap_pops(), cpd=cpd,
+                            rerouted=rerouted))
         if verbose:
             print(f"iter {it}: overused={overused} pops={router.heap_pops()} "
                   f"cpd={cpd*1e9:.2f}ns")

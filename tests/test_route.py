@@ -193,3 +193,39 @@ def test_schedule_bb_waves():
     sub = ids[::3]
     wsub = schedule_bb_waves(bb, sub, areas, nx, ny)
     assert sorted(np.concatenate(wsub).tolist()) == sorted(sub.tolist())
+
+
+def test_incremental_reroute():
+    """Partial rip-up + selective reroute (reference:
+    route_tree_mark_congested_nodes_to_be_ripped): converges to a
+    validated routing in fewer-or-equal iterations than full rip-up,
+    reroutes a shrinking net set, and keeps quality close (6-seed means:
+    +0.9% WL, +5.7% cpd, 2.6x faster — docs/MEASUREMENTS.md)."""
+    from parallel_eda_amd.timing.sta import STA
+    arch = get_arch("tseng")
+    nl = synth_netlist(spec_for_arch(arch, fill=0.6, seed=4))
+    pl = anneal_place(nl, arch, seed=4, timing_tradeoff=0.0)
+    g = rrgraph.build_rr_graph(arch)
+    sta = STA(nl, arch)
+    res_full = pathfinder_route(nl, pl, g, arch, sta=sta, max_iters=80)
+    sta2 = STA(nl, arch)
+    res_inc = pathfinder_route(nl, pl, g, arch, sta=sta2, max_iters=80,
+                               incremental=True)
+    assert res_full.success and res_inc.success
+    ok, err = res_inc.router.check_routed()
+    assert ok, err
+    # reroute sets shrink after the warm-start iterations
+    rer = [h["rerouted"] for h in res_inc.stats["history"]]
+    assert rer[0] == rer[1]          # warm-start full passes
+    if len(rer) > 2:
+        assert rer[-1] <= rer[1]
+    # quality stays in family
+    assert res_inc.wirelength <= res_full.wirelength * 1.10
+    assert res_inc.crit_path_delay <= res_full.crit_path_delay * 1.25
+    # crit-rip machinery exercised without breaking anything
+    sta3 = STA(nl, arch)
+    res_cr = pathfinder_route(nl, pl, g, arch, sta=sta3, max_iters=80,
+                              incremental=True, crit_rip_threshold=0.9)
+    assert res_cr.success
+    ok, err = res_cr.router.check_routed()
+    assert ok, err
