@@ -177,6 +177,15 @@ PYBIND11_MODULE(_pnr_cpu, m) {
       .def("count_overused", &SerialRouter::count_overused)
       .def("unrouted_sinks", &SerialRouter::unrouted_sinks)
       .def("feasible", &SerialRouter::feasible)
+      .def("route_net_sink_parallel", [](SerialRouter& r, int inet,
+                              py::array_t<float, py::array::c_style | py::array::forcecast> crit,
+                              py::array_t<int32_t, py::array::c_style | py::array::forcecast> grp_ptr,
+                              py::array_t<int32_t, py::array::c_style | py::array::forcecast> grp) {
+        const float* c = crit.size() ? crit.data() : nullptr;
+        py::gil_scoped_release rel;
+        r.route_net_sink_parallel(inet, c, grp_ptr.data(), grp.data(),
+                                  (int)grp_ptr.size() - 1);
+      })
       .def("incomplete_nets", [](SerialRouter& r) {
         auto v = r.incomplete_nets();
         py::array_t<int32_t> out((py::ssize_t)v.size());

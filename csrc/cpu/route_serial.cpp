@@ -106,6 +106,78 @@ class SerialRouter {
     return count_overused();
   }
 
+  // Sink-parallel (virtual-net) routing oracle for the round-2 GPU
+  // kernel (reference: MultiSinkParallelRouter :975-1064 routes >16-sink
+  // nets' sinks independently from the source and merges the local
+  // trees, merge :880). Each cluster routes BLIND — it sees other nets'
+  // congestion but not its siblings' paths, exactly like concurrent
+  // workgroups would — then the cluster trees merge with first-wins
+  // parents and single-counted occupancy.
+  // grp_ptr/grp: CSR of sink positions (0-based within the net).
+  void route_net_sink_parallel(int inet, const float* crit_flat,
+                               const int32_t* grp_ptr, const int32_t* grp,
+                               int n_grp) {
+    RouteTree& tree = trees_[inet];
+    for (int32_t v : tree.nodes) update_one_cost(v, -1);
+    tree.clear();
+    int32_t src = net_src_[inet];
+    int64_t s0 = sink_ptr_[inet];
+    if ((int64_t)sink_delays_.size() < sink_ptr_[inet + 1])
+      sink_delays_.resize(sinks_.size(), 0.f);
+    // 1) each cluster routes from a private source-only tree; its own
+    //    occupancy is withdrawn afterwards so siblings cannot see it
+    std::vector<RouteTree> sub((size_t)n_grp);
+    for (int gi = 0; gi < n_grp; ++gi) {
+      RouteTree& t = sub[gi];
+      t.nodes.push_back(src); t.parent.push_back(-1);
+      t.sw.push_back(-1); t.delay.push_back(0.0f);
+      update_one_cost(src, +1);
+      for (int32_t k = grp_ptr[gi]; k < grp_ptr[gi + 1]; ++k) {
+        int si = grp[k];
+        int32_t sink = sinks_[s0 + si];
+        float crit = crit_flat ? crit_flat[s0 + si] : 0.0f;
+        sink_delays_[s0 + si] = route_one_sink(inet, sink, crit, t);
+      }
+      for (int32_t v : t.nodes) update_one_cost(v, -1);
+    }
+    // 2) merge with first-wins parents (tree order guarantees a node's
+    //    parent is already merged), occupancy once per union node
+    std::unordered_map<int32_t, int32_t> pos;
+    tree.nodes.push_back(src); tree.parent.push_back(-1);
+    tree.sw.push_back(-1); tree.delay.push_back(0.0f);
+    update_one_cost(src, +1);
+    pos.emplace(src, 0);
+    for (int gi = 0; gi < n_grp; ++gi) {
+      const RouteTree& t = sub[gi];
+      for (size_t k = 1; k < t.nodes.size(); ++k) {
+        int32_t n = t.nodes[k];
+        if (pos.count(n)) continue;
+        int32_t p = t.nodes[t.parent[k]];
+        auto it = pos.find(p);
+        // parent precedes child within the cluster tree, so it is merged
+        int32_t pi = it->second;
+        pos.emplace(n, (int32_t)tree.nodes.size());
+        tree.nodes.push_back(n);
+        tree.parent.push_back(pi);
+        tree.sw.push_back(t.sw[k]);
+        tree.delay.push_back(0.0f);  // recomputed below
+        update_one_cost(n, +1);
+      }
+    }
+    // 3) delays along the merged topology (a grafted subtree's upstream
+    //    path may differ from its cluster-of-origin's)
+    for (size_t k = 1; k < tree.nodes.size(); ++k) {
+      int32_t n = tree.nodes[k];
+      int8_t sw = tree.sw[k];
+      float Tdel = g_->sw_Tdel[sw] + g_->C[n] * (g_->sw_R[sw] + 0.5f * g_->R[n]);
+      tree.delay[k] = tree.delay[tree.parent[k]] + Tdel;
+    }
+    for (int64_t c = s0; c < sink_ptr_[inet + 1]; ++c) {
+      auto it = pos.find(sinks_[c]);
+      if (it != pos.end()) sink_delays_[c] = tree.delay[it->second];
+    }
+  }
+
   // Nets with at least one sink not present in their current tree
   // (never routed, or dropped by a partial rip) — must be in every
   // incremental reroute set or the flow could terminate with a missing
