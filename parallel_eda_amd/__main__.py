@@ -55,6 +55,9 @@ def build_parser():
     p.add_argument("--bb_factor", type=int, default=3)
     p.add_argument("--min_channel_width", action="store_true",
                    help="binary search the minimum routable W")
+    p.add_argument("--route_incremental", action="store_true",
+                   help="CPU engine: partial rip-up + selective reroute "
+                        "after a full warm-start pass (2.6x at ~1%% WL)")
     p.add_argument("--rip_up_always", action="store_true",
                    help="re-route every net every iteration instead of "
                         "congested-only selective reroute (GPU engine)")
@@ -192,7 +195,8 @@ def main(argv=None):
             engine=args.engine, rip_up_always=args.rip_up_always,
             deterministic=args.deterministic, bb_factor=args.bb_factor,
             crit_exp=args.criticality_exp,
-            max_criticality=args.max_criticality)
+            max_criticality=args.max_criticality,
+            incremental=args.route_incremental)
         rt = time.perf_counter() - t0
         if not res.success:
             print(f"ROUTING FAILED: {res.overused} overused nodes after "
