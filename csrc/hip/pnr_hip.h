@@ -75,6 +75,10 @@ struct RouteParams {
                              // inflate real edge costs far above base
   int32_t max_rounds;        // safety bound on delta-stepping rounds
   int32_t strict_term;       // deterministic mode: process the == bucket
+  int32_t partial;           // partial rip-up: keep clean subtrees, skip
+                             // still-connected sinks (reference:
+                             // route_tree_mark_congested_...; CPU oracle
+                             // route_net_incremental)
 };
 
 // ---- shared device helpers (router kernels) ----

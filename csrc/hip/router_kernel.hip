@@ -146,27 +146,79 @@ __global__ void route_nets_kernel(
     if (tid == 0 && (old_len < 0 || old_len > tcap))
       printf("rip-up: net %d len %d cap %d\n", inet, old_len, tcap);
 #endif
-    for (int k = tid; k < old_len && k < tcap; k += WG_THREADS) {
-      int32_t rv = t_node[k];
-#ifdef PNR_DEBUG_BOUNDS
-      if (rv < 0 || rv >= g.num_nodes) {
-        printf("rip-up: net %d k %d bad node %d\n", inet, k, rv);
-        continue;
-      }
-#endif
-      atomicSub(&occ[rv], 1);
-    }
-    __syncthreads();
-    // ---- new tree root ----
+    // partial mode (P.partial, reference route_tree_mark_congested_...):
+    // keep subtrees whose root-path avoids overused nodes; fall back to
+    // a full rip when the tree is empty/oversized or the root itself is
+    // congested. Decided by thread 0 (uniform branch for the workgroup).
     if (tid == 0) {
-      t_node[0] = src; t_parent[0] = -1; t_sw[0] = -1; t_delay[0] = 0.0f;
-      trees.len[inet] = 1;
-      atomicAdd(&occ[src], 1);
-      sh.fail = 0;
-      sh.dbg = 0;
+      int ok = 0;
+      if (P.partial && old_len > 0 && old_len <= tcap &&
+          old_len <= (int32_t)t_cap) {
+        int32_t r = t_node[0];
+        if (r >= 0 && r < g.num_nodes && occ[r] <= g.capacity[r]) ok = 1;
+      }
+      sh.fcnt[1] = ok;
     }
     __syncthreads();
-    int tree_len = 1;
+    const bool do_partial = sh.fcnt[1] != 0;
+    __syncthreads();
+    int tree_len;
+    if (!do_partial) {
+      for (int k = tid; k < old_len && k < tcap; k += WG_THREADS) {
+        int32_t rv = t_node[k];
+#ifdef PNR_DEBUG_BOUNDS
+        if (rv < 0 || rv >= g.num_nodes) {
+          printf("rip-up: net %d k %d bad node %d\n", inet, k, rv);
+          continue;
+        }
+#endif
+        atomicSub(&occ[rv], 1);
+      }
+      __syncthreads();
+      // ---- new tree root ----
+      if (tid == 0) {
+        t_node[0] = src; t_parent[0] = -1; t_sw[0] = -1; t_delay[0] = 0.0f;
+        trees.len[inet] = 1;
+        atomicAdd(&occ[src], 1);
+        sh.fail = 0;
+        sh.dbg = 0;
+      }
+      __syncthreads();
+      tree_len = 1;
+    } else {
+      // serial drop + in-place compaction on thread 0 (endgame trees are
+      // short; touched[] doubles as the old-index -> new-index remap)
+      if (tid == 0) {
+        int32_t keep = 0;
+        for (int32_t k = 0; k < old_len; ++k) {
+          int32_t v = t_node[k];
+          int32_t par = t_parent[k];
+          int8_t swk = t_sw[k];
+          float dl = t_delay[k];
+          bool pdrop = (k > 0) && (par >= 0) && (touched[par] < 0);
+          bool cong = (v < 0 || v >= g.num_nodes) ? true
+                      : (occ[v] > g.capacity[v]);
+          if ((k > 0 && (pdrop || cong)) ) {
+            touched[k] = -1;
+            if (v >= 0 && v < g.num_nodes) atomicSub(&occ[v], 1);
+            continue;
+          }
+          touched[k] = keep;
+          t_node[keep] = v;
+          t_parent[keep] = (k == 0) ? -1 : touched[par];
+          t_sw[keep] = swk;
+          t_delay[keep] = dl;
+          ++keep;
+        }
+        trees.len[inet] = keep;
+        sh.fcnt[0] = keep;
+        sh.fail = 0;
+        sh.dbg = 0;
+      }
+      __syncthreads();
+      tree_len = sh.fcnt[0];
+      __syncthreads();
+    }
 
     // ---- route each sink (pre-ordered by criticality on host) ----
     for (int32_t si = s0; si < s1; ++si) {
@@ -176,6 +228,17 @@ __global__ void route_nets_kernel(
       S.sy = g.ylow[S.sink_node];
       S.crit = nets.crit[si];
       S.astar_fac = P.astar_fac;
+      if (do_partial) {
+        // skip sinks whose kept subtree still reaches them
+        if (tid == 0) sh.fcnt[1] = 0;
+        __syncthreads();
+        for (int k = tid; k < tree_len; k += WG_THREADS)
+          if (t_node[k] == S.sink_node) atomicOr(&sh.fcnt[1], 1);
+        __syncthreads();
+        const bool connected = sh.fcnt[1] != 0;
+        __syncthreads();
+        if (connected) continue;
+      }
       // per-sink bucket width: delta_fac edge-steps in this sink's cost
       // units (wider buckets = fewer, fatter delta-stepping rounds;
       // PathFinder tolerates the relaxed expansion order)
@@ -554,6 +617,7 @@ struct RouteLaunchArgs {
   unsigned long long* stats;   // [8] search counters or null
   unsigned long long* net_scans;  // per-net scan counters or null
   int32_t use_calendar;        // EXPERIMENTAL: calendar-queue frontier
+  int32_t partial;             // partial rip-up (see RouteParams)
 };
 
 int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
@@ -573,6 +637,7 @@ int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
   P.cong_mult = a->cong_mult < 1.0f ? 1.0f : a->cong_mult;
   P.max_rounds = a->max_rounds;
   P.strict_term = a->strict_term;
+  P.partial = a->partial;
   int grid = a->n_small_slots + a->n_large_slots;
   if (a->use_calendar) {
     hipLaunchKernelGGL(route_nets_cal_kernel, dim3(grid), dim3(WG_THREADS), 0,

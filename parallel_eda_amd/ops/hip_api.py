@@ -51,6 +51,7 @@ class RouteLaunchArgs(ct.Structure):
         ("stats", ct.c_void_p),
         ("net_scans", ct.c_void_p),
         ("use_calendar", ct.c_int32),
+        ("partial", ct.c_int32),
     ]
 
 

@@ -260,14 +260,19 @@ class GpuRouter:
                                  self.arch.nx, self.arch.ny)
 
     # ---- one PathFinder iteration ----
-    def route_iteration(self, crit, pres_fac, net_subset=None, fail_ok=False):
+    def route_iteration(self, crit, pres_fac, net_subset=None, fail_ok=False,
+                        partial=False):
         """crit: per-sink criticality aligned with sink_rr (original order).
         Returns (overused_count, sink_delays aligned with original order).
         net_subset: optional array of net ids to (re)route; others keep
         their route trees (multi-GPU partitioning / selective reroute).
         fail_ok: exhausted retries leave the stragglers partially routed
         instead of raising (multi-rank benches must not kill a rank
-        mid-collective)."""
+        mid-collective).
+        partial: partial rip-up (keep clean subtrees, skip connected
+        sinks; reference route_tree_mark_congested_...; EXPERIMENTAL,
+        pending GPU validation — not honored by the calendar kernel)."""
+        self._partial = partial
         torch = self.torch
         # order sinks by decreasing criticality within each net
         net_of_sink = np.repeat(np.arange(self.num_nets), np.diff(self.sink_ptr))
@@ -496,6 +501,7 @@ class GpuRouter:
         a.net_scans = ct_ptr(self.t_net_scans)
         a.use_calendar = 1 if (getattr(self, "use_calendar", False) or
                                _os.environ.get("PNR_CALENDAR")) else 0
+        a.partial = 1 if getattr(self, "_partial", False) else 0
         self._args_keepalive = (t_sink_rr, t_crit, t_sink_orig, q_small,
                                 q_large)
         return a
@@ -509,7 +515,8 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
                          pres_fac_init=0.5, pres_fac_mult=1.3, acc_fac=1.0,
                          astar_fac=1.2, verbose=False, device="cuda:0",
                          rip_up_always=False, deterministic=False,
-                         bb_factor=4, crit_exp=1.0, max_criticality=0.99):
+                         bb_factor=4, crit_exp=1.0, max_criticality=0.99,
+                         incremental=False):
     """GPU PathFinder outer loop — mirrors route.router.pathfinder_route.
 
     bb_factor: initial per-net bounding-box margin in tiles (reference:
@@ -539,8 +546,9 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
             if len(subset) == 0:
                 subset = None
         router.reset_search_stats()
-        overused, sink_delays = router.route_iteration(crit, pres_fac,
-                                                       net_subset=subset)
+        overused, sink_delays = router.route_iteration(
+            crit, pres_fac, net_subset=subset,
+            partial=incremental and subset is not None)
         st = router.search_stats()
         history.append(dict(iter=it, overused=int(overused), cpd=cpd,
                             rounds=st["rounds"], scanned=st["scanned"],

@@ -137,7 +137,7 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
             acc_fac=acc_fac, astar_fac=astar_fac, verbose=verbose,
             rip_up_always=rip_up_always, deterministic=deterministic,
             bb_factor=bb_factor, crit_exp=crit_exp,
-            max_criticality=max_criticality)
+            max_criticality=max_criticality, incremental=incremental)
     cpu = ops.cpu()
     net_ids, src_rr, sink_ptr, sink_rr, conn_index = net_rr_terminals(
         netlist, placement, g, arch)
