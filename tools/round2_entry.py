@@ -30,9 +30,8 @@ run("tseng-quality-calendar",
     env={"PNR_CALENDAR": "1"}, timeout=300)
 run("btc-calendar", "python bench.py --config bitcoin_miner --steps 2 "
     "--warmup 2 --verbose", env={"PNR_CALENDAR": "1"}, timeout=900)
-run("partial-rip-gpu",
-    "python -m pytest tests/test_gpu_router.py -q -k 'quality or ripup'",
-    env={"PNR_PARTIAL_AB": "1"}, timeout=400)
+run("partial-rip-ab", "python tools/gpu_partial_ab.py tseng 0.6 && "
+    "python tools/gpu_partial_ab.py LU32PEEng 0.55", timeout=700)
 run("het-gpu", "python - <<'P'\n"
     "import sys; sys.path.insert(0, '.')\n"
     "import numpy as np\n"
