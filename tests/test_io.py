@@ -195,3 +195,22 @@ def test_two_clock_blif_flow(tmp_path):
     rc = main([str(blif), str(xml), "--route_chan_width", "20",
                "--sdc", str(sdc)])
     assert rc == 0
+
+
+def test_bench_flow_tool(tmp_path):
+    """tools/bench_flow.py (place+route wall-clock, the BASELINE metric)
+    emits a valid JSON line and succeeds on the CPU path."""
+    import json
+    import subprocess
+    import sys
+    from pathlib import Path
+    root = Path(__file__).resolve().parent.parent
+    r = subprocess.run([sys.executable, str(root / "tools" / "bench_flow.py"),
+                        "tseng", "--placer", "cpu", "--router", "cpu"],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr
+    data = json.loads(r.stdout.strip().splitlines()[-1])
+    assert data["success"] and data["value"] > 0
+    assert data["higher_is_better"] is False
+    assert set(data["phase_s"]) == {"netlist_synth", "place", "rr_build",
+                                    "route"}

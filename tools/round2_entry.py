@@ -32,6 +32,9 @@ run("btc-calendar", "python bench.py --config bitcoin_miner --steps 2 "
     "--warmup 2 --verbose", env={"PNR_CALENDAR": "1"}, timeout=900)
 run("partial-rip-ab", "python tools/gpu_partial_ab.py tseng 0.6 && "
     "python tools/gpu_partial_ab.py LU32PEEng 0.55", timeout=700)
+run("flow-bench", "python tools/bench_flow.py LU32PEEng > "
+    "gpurun_out/flow_lu32.json && cat gpurun_out/flow_lu32.json",
+    timeout=900)
 run("het-gpu", "python - <<'P'\n"
     "import sys; sys.path.insert(0, '.')\n"
     "import numpy as np\n"
