@@ -124,20 +124,66 @@ __global__ void route_nets_cal_kernel(
     L.bh = nets.bb[4 * inet + 3] - L.by0 + 1;
     L.npt = g.npt;
 
-    // rip-up + root (same as production kernel)
+    // rip-up + root (same as production kernel; partial mode mirrors
+    // router_kernel.hip's P.partial branches)
     int32_t old_len = trees.len[inet];
-    for (int k = tid; k < old_len && k < tcap; k += CWG_THREADS)
-      atomicSub(&occ[t_node[k]], 1);
-    __syncthreads();
     if (tid == 0) {
-      t_node[0] = src; t_parent[0] = -1; t_sw[0] = -1; t_delay[0] = 0.0f;
-      trees.len[inet] = 1;
-      atomicAdd(&occ[src], 1);
-      sh.fail = 0;
-      sh.tree_len = 1;
+      int ok = 0;
+      if (P.partial && old_len > 0 && old_len <= tcap &&
+          old_len <= (int32_t)t_cap) {
+        int32_t r = t_node[0];
+        if (r >= 0 && r < g.num_nodes && occ[r] <= g.capacity[r]) ok = 1;
+      }
+      sh.qi = ok;
     }
     __syncthreads();
-    int tree_len = 1;
+    const bool do_partial = sh.qi != 0;
+    __syncthreads();
+    int tree_len;
+    if (!do_partial) {
+      for (int k = tid; k < old_len && k < tcap; k += CWG_THREADS)
+        atomicSub(&occ[t_node[k]], 1);
+      __syncthreads();
+      if (tid == 0) {
+        t_node[0] = src; t_parent[0] = -1; t_sw[0] = -1; t_delay[0] = 0.0f;
+        trees.len[inet] = 1;
+        atomicAdd(&occ[src], 1);
+        sh.fail = 0;
+        sh.tree_len = 1;
+      }
+      __syncthreads();
+      tree_len = 1;
+    } else {
+      if (tid == 0) {
+        int32_t keep = 0;
+        for (int32_t k = 0; k < old_len; ++k) {
+          int32_t v = t_node[k];
+          int32_t par = t_parent[k];
+          int8_t swk = t_sw[k];
+          float dl = t_delay[k];
+          bool pdrop = (k > 0) && (par >= 0) && (touched[par] < 0);
+          bool cong = (v < 0 || v >= g.num_nodes) ? true
+                      : (occ[v] > g.capacity[v]);
+          if (k > 0 && (pdrop || cong)) {
+            touched[k] = -1;
+            if (v >= 0 && v < g.num_nodes) atomicSub(&occ[v], 1);
+            continue;
+          }
+          touched[k] = keep;
+          t_node[keep] = v;
+          t_parent[keep] = (k == 0) ? -1 : touched[par];
+          t_sw[keep] = swk;
+          t_delay[keep] = dl;
+          ++keep;
+        }
+        trees.len[inet] = keep;
+        sh.tree_len = keep;
+        sh.fail = 0;
+      }
+      __syncthreads();
+      tree_len = sh.tree_len;
+      __syncthreads();
+    }
 
     for (int32_t si = s0; si < s1; ++si) {
       SinkCtx S;
@@ -146,6 +192,16 @@ __global__ void route_nets_cal_kernel(
       S.sy = g.ylow[S.sink_node];
       S.crit = nets.crit[si];
       S.astar_fac = P.astar_fac;
+      if (do_partial) {
+        if (tid == 0) sh.qi = 0;
+        __syncthreads();
+        for (int k = tid; k < tree_len; k += CWG_THREADS)
+          if (t_node[k] == S.sink_node) atomicOr(&sh.qi, 1);
+        __syncthreads();
+        const bool connected = sh.qi != 0;
+        __syncthreads();
+        if (connected) continue;   // kept subtree still reaches this sink
+      }
       const float delta = P.delta_fac * (S.crit * P.seg_delay +
                                          (1.0f - S.crit) * P.seg_base);
       const float inv_delta = 1.0f / delta;
