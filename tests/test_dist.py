@@ -301,3 +301,20 @@ def test_dist_flow_selective_world2(tmp_path):
     # NOTE: trees are split across ranks, so check_routed (which expects
     # every net) only holds on whichever rank owns each net — per-rank
     # validation runs inside the worker only when it owns all nets.
+
+
+def test_dist_flow_world4(tmp_path):
+    """4-rank gloo flow: 4-strip partition + shrink consolidation from 4
+    ranks, rank-identical occ. (The driver's 8-GPU bench uses the same
+    collective logic at world 8 over RCCL.)"""
+    mp.spawn(_worker_flow, args=(4, 29535, str(tmp_path)), nprocs=4,
+             join=True)
+    rs = []
+    for rank in range(4):
+        with open(tmp_path / f"fl{rank}.pkl", "rb") as f:
+            rs.append(pickle.load(f))
+    assert rs[0]["res"]["success"]
+    for r in rs[1:]:
+        assert np.array_equal(rs[0]["occ"], r["occ"])
+        assert rs[0]["res"]["cpd"] == pytest.approx(r["res"]["cpd"],
+                                                    rel=1e-6)
