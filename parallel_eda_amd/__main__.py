@@ -75,6 +75,10 @@ def build_parser():
     p.add_argument("--out_sdf", type=str, default=None,
                    help="post-route SDF timing annotation")
     p.add_argument("--stats_dir", type=str, default=None)
+    p.add_argument("--echo_routes", action="store_true",
+                   help="with --stats_dir: dump the full .route traceback "
+                        "after routing (reference: write_routes "
+                        "routes_iter_N.txt)")
     p.add_argument("--settings", type=str, default=None,
                    help="TOML settings file: keys = CLI flag names; "
                         "explicit CLI flags win (reference: read_settings.c)")
@@ -220,6 +224,13 @@ def main(argv=None):
             write_route(args.out_route, g, arch, net_ids,
                         lambda k: res.router.tree(k), netlist=netlist)
             print(f"wrote {args.out_route}")
+        if args.echo_routes and args.stats_dir:
+            import os as _os
+            ep = _os.path.join(args.stats_dir,
+                               f"routes_iter_{res.iterations}.txt")
+            write_route(ep, g, arch, net_ids,
+                        lambda k: res.router.tree(k), netlist=netlist)
+            print(f"wrote {ep}")
         if sdc_clocks and getattr(netlist, "block_clock", None) is not None \
                 and len(sdc_clocks) >= 1 and sta is not None:
             # multi-clock analysis against the SDC constraints

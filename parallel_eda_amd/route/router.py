@@ -198,6 +198,17 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
     if ok:
         valid, err = router.check_routed()
         if not valid:
+            # forensics dump (reference: check_route_tree writes
+            # error.dot on failure, router.cxx:145-200)
+            import re
+            from ..utils.debug_dump import write_tree_dot
+            m = re.search(r"net (\d+)", err)
+            if m:
+                inet = int(m.group(1))
+                nodes, parents, _sw, _d = router.tree(inet)
+                write_tree_dot("error.dot", nodes, parents, g=g,
+                               label=f"net {inet}: {err}")
+                err += " (tree dumped to error.dot)"
             raise RuntimeError(f"check_route failed: {err}")
     return RouteResult(success=ok, iterations=it, overused=int(overused),
                        wirelength=int(router.total_wirelength()),

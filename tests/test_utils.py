@@ -174,3 +174,11 @@ def test_cli_settings_file(tmp_path):
     import pytest as _pt
     with _pt.raises(SystemExit):
         main(["--synth", "tiny", "--settings", str(bad)])
+
+
+def test_tree_dot_dump(tmp_path):
+    from parallel_eda_amd.utils.debug_dump import write_tree_dot
+    f = tmp_path / "t.dot"
+    write_tree_dot(str(f), [10, 11, 12], [-1, 0, 0], label="x")
+    body = f.read_text()
+    assert "digraph" in body and "n0 -> n1" in body and "n0 -> n2" in body
