@@ -68,6 +68,10 @@ def build_parser():
                    help="SDC constraints (create_clock -period, subset)")
     p.add_argument("--timing_report", type=str, default=None,
                    help="write a critical-path report here after routing")
+    p.add_argument("--place_only", action="store_true",
+                   help="stop after placement (reference: -place_only)")
+    p.add_argument("--out_net", type=str, default=None,
+                   help="write the packed .net netlist")
     p.add_argument("--out_place", type=str, default=None)
     p.add_argument("--out_route", type=str, default=None)
     p.add_argument("--out_verilog", type=str, default=None,
@@ -174,9 +178,16 @@ def main(argv=None):
             crit_exp=args.criticality_exp)
         print(f"placement: bb_cost={placement.bb_cost:.1f} "
               f"({time.perf_counter()-t0:.2f}s)")
+    if args.out_net:
+        from .io.net_file import write_net
+        write_net(args.out_net, netlist)
+        print(f"wrote {args.out_net}")
     if args.out_place:
         write_place(args.out_place, placement, netlist, arch)
         print(f"wrote {args.out_place}")
+    if args.place_only:
+        print(f"entire flow took {time.perf_counter()-t_start:.2f}s")
+        return 0
 
     # ---- routing ----
     if args.min_channel_width:

@@ -75,13 +75,17 @@ class CpuEngine:
         self.r = router
         self.num_nodes = num_nodes
 
-    def route_subset(self, crit, pres_fac, net_ids):
+    def route_subset(self, crit, pres_fac, net_ids, partial=False):
         import numpy as _np
         self.r.set_pres_fac(pres_fac)
         # refresh pres array from occ under the new pres_fac
         self.r.set_occ(_np.asarray(self.r.occ()))
-        self.r.route_subset(np.ascontiguousarray(crit, dtype=np.float32),
-                            np.asarray(net_ids, dtype=np.int32))
+        c = np.ascontiguousarray(crit, dtype=np.float32)
+        ids = np.asarray(net_ids, dtype=np.int32)
+        if partial:
+            self.r.route_subset_incremental(c, ids)
+        else:
+            self.r.route_subset(c, ids)
 
     def rip_up_nets(self, net_ids):
         self.r.rip_up_nets(np.asarray(net_ids, dtype=np.int32))
@@ -118,9 +122,9 @@ class GpuEngine:
         self.g = router
         self._last_sd = np.zeros(router.n_sinks, dtype=np.float32)
 
-    def route_subset(self, crit, pres_fac, net_ids):
+    def route_subset(self, crit, pres_fac, net_ids, partial=False):
         _, sd = self.g.route_iteration(crit, pres_fac, net_subset=net_ids,
-                                       fail_ok=True)
+                                       fail_ok=True, partial=partial)
         self._last_sd = sd
 
     def occ_tensor(self):
@@ -280,10 +284,12 @@ class DistRouteLoop:
                                                  weight=nsk * semi)
         return self.set_partition(new_rank)
 
-    def iteration(self, crit, pres_fac, acc_fac, active_mask=None):
+    def iteration(self, crit, pres_fac, acc_fac, active_mask=None,
+                  partial=False):
         """One distributed PathFinder iteration. active_mask: optional
         bool mask over nets (selective reroute); only owned ACTIVE nets
-        are routed, but every rank joins the collectives. Returns
+        are routed, but every rank joins the collectives. partial:
+        partial rip-up on the engine (keep clean subtrees). Returns
         (overused_global, sink_delays_global)."""
         import torch
         dist = _dist() if self.ws > 1 else None
@@ -294,7 +300,7 @@ class DistRouteLoop:
         if self.ws > 1:
             occ_before = eng.occ_tensor().clone()
         if len(nets):
-            eng.route_subset(crit, pres_fac, nets)
+            eng.route_subset(crit, pres_fac, nets, partial=partial)
         sd = eng.sink_delays_local(self.my_nets)
         if self.ws > 1:
             occ = eng.occ_tensor()
@@ -316,7 +322,8 @@ class DistRouteLoop:
 
 def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
                           pres_fac_mult=1.3, acc_fac=1.0,
-                          shrink_threshold=128, verbose=False):
+                          shrink_threshold=128, verbose=False,
+                          incremental=False):
     """Distributed PathFinder outer loop (flow-level driver).
 
     Mirrors route.gpu_router.pathfinder_route_gpu's schedule — iteration 1
@@ -347,7 +354,9 @@ def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
     it = 0
     for it in range(1, max_iters + 1):
         overused, sd = loop.iteration(crit, pres, acc_fac,
-                                      active_mask=active)
+                                      active_mask=active,
+                                      partial=incremental and
+                                      active is not None)
         if sta is not None:
             cmap.conn_delays(sd, out=conn_delay)
             cpd, _slack, c = sta.analyze(conn_delay)

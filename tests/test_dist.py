@@ -275,7 +275,8 @@ def _worker_flow(rank, world, port, tmpdir):
                  xlow[terms].max(), ylow[terms].max())
     loop = DistRouteLoop(engine, len(net_ids), bb, len(sink_rr), sink_ptr,
                          rank=rank, world_size=world)
-    res = pathfinder_route_dist(loop, cmap, sta, max_iters=60)
+    res = pathfinder_route_dist(loop, cmap, sta, max_iters=60,
+                                incremental=True)
     ok, err = (router.check_routed() if res["success"] else (True, ""))
     occ = np.asarray(router.occ()).copy()
     with open(os.path.join(tmpdir, f"fl{rank}.pkl"), "wb") as f:
