@@ -319,3 +319,19 @@ def test_dist_flow_world4(tmp_path):
         assert np.array_equal(rs[0]["occ"], r["occ"])
         assert rs[0]["res"]["cpd"] == pytest.approx(r["res"]["cpd"],
                                                     rel=1e-6)
+
+
+def test_dist_flow_deterministic(tmp_path):
+    """The whole distributed flow is deterministic: identical seeds give
+    bit-identical global congestion across independent runs (integer
+    occ all-reduce + deterministic shrink/partition + serial engines)."""
+    a = tmp_path / "a"; b = tmp_path / "b"
+    a.mkdir(); b.mkdir()
+    mp.spawn(_worker_flow, args=(2, 29536, str(a)), nprocs=2, join=True)
+    mp.spawn(_worker_flow, args=(2, 29537, str(b)), nprocs=2, join=True)
+    with open(a / "fl0.pkl", "rb") as f:
+        ra = pickle.load(f)
+    with open(b / "fl0.pkl", "rb") as f:
+        rb = pickle.load(f)
+    assert np.array_equal(ra["occ"], rb["occ"])
+    assert ra["res"]["cpd"] == rb["res"]["cpd"]
