@@ -40,6 +40,17 @@ def parse_arch_xml(path_or_text, nx=None, ny=None, W=64, name=None) -> ArchDef:
 
     a = ArchDef()
     a.name = name or f"xml:{src}"
+    # fixed layout from the XML (reference: <layout width= height=>);
+    # explicit nx/ny arguments override, auto layouts defer to the flow
+    lay = root.find("layout")
+    if lay is not None:
+        w_ = lay.get("width"); h_ = lay.get("height")
+        if w_ is not None and h_ is not None:
+            try:
+                a.nx, a.ny = max(2, int(float(w_))), max(2, int(float(h_)))
+                a._fixed_layout = True
+            except ValueError:
+                pass
     if nx:
         a.nx = nx
     if ny:
@@ -165,7 +176,10 @@ def parse_arch_xml(path_or_text, nx=None, ny=None, W=64, name=None) -> ArchDef:
 def size_grid_for_netlist(netlist, arch: ArchDef, fill_target=0.8):
     """VPR-style auto grid sizing (reference: SetupGrid.c): smallest square
     grid fitting every block type at fill_target (heterogeneous fabrics:
-    enough RAM/DSP column tiles too), with enough IO perimeter."""
+    enough RAM/DSP column tiles too), with enough IO perimeter. A fixed
+    <layout width height> in the XML wins over auto sizing."""
+    if getattr(arch, "_fixed_layout", False):
+        return arch
     counts = {t: int((netlist.block_type == t).sum()) for t in (0, 1, 2, 3)}
     side = max(2, math.ceil(math.sqrt(max(1, counts[1]) / fill_target)))
     while True:

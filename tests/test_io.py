@@ -214,3 +214,21 @@ def test_bench_flow_tool(tmp_path):
     assert data["higher_is_better"] is False
     assert set(data["phase_s"]) == {"netlist_synth", "place", "rr_build",
                                     "route"}
+
+
+def test_fixed_layout_xml():
+    """<layout width height> fixes the grid; auto layouts size from the
+    netlist (reference: SetupGrid.c fixed vs auto)."""
+    from parallel_eda_amd.arch.xml_parser import (parse_arch_xml,
+                                                  size_grid_for_netlist)
+    from parallel_eda_amd.io.synth import synth_netlist, spec_for_arch
+    from parallel_eda_amd.arch.archdef import get_arch
+    a = parse_arch_xml('<architecture><layout width="20" height="10"/>'
+                       '</architecture>')
+    assert (a.nx, a.ny) == (20, 10)
+    nl = synth_netlist(spec_for_arch(get_arch("tiny"), fill=0.5, seed=1))
+    size_grid_for_netlist(nl, a)
+    assert (a.nx, a.ny) == (20, 10)     # fixed layout wins
+    b = parse_arch_xml('<architecture><layout auto="1.0"/></architecture>')
+    size_grid_for_netlist(nl, b)
+    assert b.nx >= 2                    # auto-sized
