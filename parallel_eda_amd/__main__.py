@@ -142,7 +142,7 @@ def main(argv=None):
             return 2
         model = read_blif(args.circuit)
         arch = parse_arch_xml(args.arch, W=args.route_chan_width or 64)
-        netlist, _, _ = pack_blif(model, arch)
+        netlist, _, _ = pack_blif(model, arch, n_ble=arch.clb_n_ble)
         size_grid_for_netlist(netlist, arch)
         print(f"read {args.circuit}: {len(model.prims)} primitives -> "
               f"{netlist.num_blocks} blocks on {arch.nx}x{arch.ny} grid")

@@ -50,6 +50,7 @@ class ArchDef:
     fc_out: int = 8       # wire-starts each OPIN can drive (absolute count)
     clb_in: int = 40      # CLB input pins (≡ SINK capacity)
     clb_out: int = 10     # CLB output pins (≡ SOURCE capacity)
+    clb_n_ble: int = 10   # BLEs per CLB (packer cluster size, arch <pb_type num_pb>)
     io_cap: int = 8       # IO slots per perimeter tile
 
     # Heterogeneous column tiles (reference: libarchfpga grid types with

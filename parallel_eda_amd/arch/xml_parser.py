@@ -140,6 +140,13 @@ def parse_arch_xml(path_or_text, nx=None, ny=None, W=64, name=None) -> ArchDef:
             a.T_dsp = block_delay(pb, a.T_dsp)
         elif n_in > 0 and n_out > 0 and not clb_seen:
             clb_seen = True
+            # cluster size: the inner BLE pb_type's num_pb
+            inner = pb.find(".//pb_type[@num_pb]")
+            if inner is not None:
+                try:
+                    a.clb_n_ble = max(1, int(inner.get("num_pb")))
+                except (TypeError, ValueError):
+                    pass
             # first real logic block type
             a.clb_in = n_in
             a.clb_out = n_out
