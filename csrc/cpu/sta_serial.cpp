@@ -4,8 +4,9 @@
 // (vpr/SRC/timing/path_delay.c:1994 do_timing_analysis_new, levelization
 //  path_delay2.c:81): block-granularity timing graph, T_arr max-plus forward
 //  sweep per level, T_req min-minus backward, per-connection slack and
-//  criticality = 1 - slack/cpd. Single clock domain (the reference's
-//  multi-domain loop collapses to one domain on our netlists).
+//  criticality = 1 - slack/cpd; analyze_domains implements the
+//  reference's multi-clock (src,sink)-domain-pair loop. Heterogeneous
+//  blocks carry per-block combinational delays (blk_delay_).
 //
 // The level arrays built here are uploaded to HBM for the GPU STA kernels
 // (csrc/hip/sta_kernel.hip), which run the same sweeps level-synchronously.
