@@ -330,7 +330,8 @@ def synth_netlist(spec: SynthSpec) -> NetlistPy:
             sinks.add(blk)
             fanin[blk] += 1
         if not sinks:
-            # fall back: an undriven output pad, else any seq CLB with room
+            # fall back: an undriven output pad, else any block with room
+            # (seq, or comb with a legal rank)
             placed = False
             for o in range(spec.n_out):
                 if not out_pad_driven[o]:
@@ -342,14 +343,28 @@ def synth_netlist(spec: SynthSpec) -> NetlistPy:
             if not placed:
                 for c in rng.permutation(n_logic):
                     blk = clb0 + int(c)
-                    if blk != drv and block_is_seq[blk] and fanin[blk] < max_fanin[blk]:
-                        sinks.add(blk)
-                        fanin[blk] += 1
-                        placed = True
-                        break
+                    if blk == drv or fanin[blk] >= max_fanin[blk]:
+                        continue
+                    if not block_is_seq[blk] and drv_rank >= 0                             and rank[int(c)] <= drv_rank:
+                        continue
+                    sinks.add(blk)
+                    fanin[blk] += 1
+                    placed = True
+                    break
             if not placed:
-                raise RuntimeError("synth: no legal sink available")
+                # every fan-in budget saturated: the driver's output is
+                # unused — drop the net (the reference's sweep removes
+                # dangling nets the same way)
+                sink_lists.append([])
+                continue
         sink_lists.append(sorted(sinks))
+
+    # drop empty (swept) nets
+    keep = [i for i, sl in enumerate(sink_lists) if sl]
+    if len(keep) != len(drivers):
+        drivers = drivers[np.asarray(keep, dtype=np.int64)]
+        sink_lists = [sink_lists[i] for i in keep]
+        n_nets = len(drivers)
 
     # ensure every output pad is driven by exactly one net
     for o in range(spec.n_out):
