@@ -271,7 +271,7 @@ class GpuRouter:
         mid-collective).
         partial: partial rip-up (keep clean subtrees, skip connected
         sinks; reference route_tree_mark_congested_...; EXPERIMENTAL,
-        pending GPU validation — not honored by the calendar kernel)."""
+        pending round-2 GPU validation; both kernels honor it)."""
         self._partial = partial
         torch = self.torch
         # order sinks by decreasing criticality within each net
