@@ -123,7 +123,8 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
                      crit_exp: float = 1.0, max_criticality: float = 0.99,
                      incremental: bool = False,
                      crit_rip_threshold: float = 0.99,
-                     incremental_start: int = 2):
+                     incremental_start: int = 2,
+                     full_resync_every: int = 2):
     """Timing-driven PathFinder: route all nets to feasibility.
 
     rip_up_always / deterministic reach the GPU engine (the CPU oracle
@@ -157,7 +158,9 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
     overused = -1
     for it in range(1, max_iters + 1):
         rerouted = len(net_ids)
-        if incremental and it > incremental_start:
+        resync = (full_resync_every > 0 and it > incremental_start and
+                  (it - incremental_start) % full_resync_every == 0)
+        if incremental and it > incremental_start and not resync:
             # selective + partial rip-up (reference phase-two +
             # route_tree_mark_congested_...): reroute congested nets and
             # nets with missing sinks; keep their clean subtrees
