@@ -541,7 +541,12 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
         # iteration 1 routes everything; later iterations re-route only
         # nets whose trees touch overused nodes
         subset = None
-        if it > 1 and not rip_up_always:
+        # incremental mode mirrors the CPU flow's schedule: full passes
+        # through iteration 2, then partial-rip selective iterations with
+        # a full resync every 2nd (recovers the timing the pure
+        # incremental mode leaves; docs/MEASUREMENTS.md)
+        resync = incremental and it > 2 and (it - 2) % 2 == 0
+        if it > 1 and not rip_up_always and not resync:
             subset = router.congested_nets()
             if len(subset) == 0:
                 subset = None
