@@ -353,10 +353,13 @@ def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
     overused = -1
     it = 0
     for it in range(1, max_iters + 1):
+        # incremental resync cadence mirrors the single-GPU flows
+        resync = incremental and it > 2 and (it - 2) % 2 == 0
+        mask = None if resync else active
         overused, sd = loop.iteration(crit, pres, acc_fac,
-                                      active_mask=active,
+                                      active_mask=mask,
                                       partial=incremental and
-                                      active is not None)
+                                      mask is not None)
         if sta is not None:
             cmap.conn_delays(sd, out=conn_delay)
             cpd, _slack, c = sta.analyze(conn_delay)
