@@ -288,7 +288,7 @@ __global__ void route_nets_cal_kernel(
           __syncthreads();
           for (int i = lo + tid; i < hi; i += CWG_THREADS) {
             float4 e = buckets[(int64_t)cur * bcap + i];
-            float tot = e.x, back = e.y;
+            float back = e.y;
             int32_t v = __float_as_int(e.z);
             int32_t prev = __float_as_int(e.w);
             int64_t li = L(g, v);
