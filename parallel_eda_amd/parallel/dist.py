@@ -114,6 +114,9 @@ class CpuEngine:
     def congested_nets(self):
         return np.asarray(self.r.congested_nets())
 
+    def incomplete_nets(self):
+        return np.asarray(self.r.incomplete_nets())
+
 
 class GpuEngine:
     """GpuRouter adapter for DistRouteLoop."""
@@ -155,6 +158,11 @@ class GpuEngine:
 
     def congested_nets(self):
         return np.asarray(self.g.congested_nets())
+
+    def incomplete_nets(self):
+        # the GPU router raises (or reports) on unroutable sinks itself;
+        # no cheap device-side missing-sink query yet (round 2)
+        return np.zeros(0, dtype=np.int64)
 
 
 class DistRouteLoop:
@@ -243,15 +251,12 @@ class DistRouteLoop:
         new_rank = spatial_partition(self._bb, self.ws, weight=wsum)
         return self.set_partition(new_rank)
 
-    def global_congested_mask(self):
-        """Boolean mask over nets whose tree crosses an overused node —
-        identical on every rank (each rank reports its OWNED nets, then a
-        max all-reduce unions them). Reference: phase-two congested-net
-        selection, here made globally consistent for the shrink decision."""
+    def _global_owned_mask(self, ids):
+        """Union of per-rank OWNED net id sets into a rank-identical
+        boolean mask (max all-reduce)."""
         import torch
         mask = np.zeros(len(self.rank_of), dtype=np.float32)
-        local = np.intersect1d(np.asarray(self.engine.congested_nets(),
-                                          dtype=np.int64),
+        local = np.intersect1d(np.asarray(ids, dtype=np.int64),
                                self.my_nets)
         mask[local] = 1.0
         if self.ws > 1:
@@ -262,6 +267,19 @@ class DistRouteLoop:
             _dist().all_reduce(t, op=_dist().ReduceOp.MAX)
             mask = t.cpu().numpy()
         return mask > 0.5
+
+    def global_congested_mask(self):
+        """Boolean mask over nets whose tree crosses an overused node —
+        identical on every rank (each rank reports its OWNED nets, then a
+        max all-reduce unions them). Reference: phase-two congested-net
+        selection, here made globally consistent for the shrink decision.
+        Nets with missing sinks (never routed / partial-ripped) are
+        included so selective reroute can never strand a connection."""
+        m = self._global_owned_mask(self.engine.congested_nets())
+        inc = self.engine.incomplete_nets()
+        if len(inc):
+            m |= self._global_owned_mask(inc)
+        return m
 
     def shrink_active(self, active_mask, k=1):
         """Elastic comm-shrink analogue (reference: mpi_comm_shrink — the
