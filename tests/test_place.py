@@ -125,3 +125,62 @@ def test_read_pad_loc(tmp_path):
     assert ids.tolist() == [0, 1]
     assert x.tolist() == [0, 0] and y.tolist() == [1, 2]
     assert s.tolist() == [0, 0]
+
+
+def test_gpu_placer_initial_placement_host_logic():
+    """The GPU placer's HOST-side initial placement (deterministic by
+    seed; refactored for het/fixed support) must stay bit-identical to
+    the round-1-validated homogeneous algorithm — checked on CPU so no
+    GPU budget is spent guarding the refactor."""
+    import numpy as np
+    from parallel_eda_amd.arch.archdef import get_arch
+    from parallel_eda_amd.io.synth import synth_netlist, spec_for_arch
+    from parallel_eda_amd.place.gpu_placer import GpuPlacer
+
+    arch = get_arch("tiny")
+    nl = synth_netlist(spec_for_arch(arch, fill=0.6, seed=2))
+    gp = GpuPlacer.__new__(GpuPlacer)
+    gp.arch = arch; gp.nl = nl
+    gp.gx, gp.gy = arch.nx + 2, arch.ny + 2
+    gp.cap = max(1, arch.io_cap)
+    gp._fixed = None
+    bx, by, bs, grid = gp._initial_placement(7)
+
+    rng = np.random.default_rng(7)
+    nb = nl.num_blocks
+    bx2 = np.zeros(nb, dtype=np.int32); by2 = np.zeros(nb, dtype=np.int32)
+    bs2 = np.zeros(nb, dtype=np.int32)
+    grid2 = np.full((gp.gx, gp.gy, gp.cap), -1, dtype=np.int32)
+    clbs = np.nonzero(nl.block_type == 1)[0]
+    ios = np.nonzero(nl.block_type == 0)[0]
+    tiles = rng.permutation(arch.nx * arch.ny)[:len(clbs)]
+    bx2[clbs] = tiles // arch.ny + 1
+    by2[clbs] = tiles % arch.ny + 1
+    grid2[bx2[clbs], by2[clbs], 0] = clbs
+    io_locs = ([(0, y) for y in range(1, arch.ny + 1)] +
+               [(gp.gx - 1, y) for y in range(1, arch.ny + 1)] +
+               [(x, 0) for x in range(1, arch.nx + 1)] +
+               [(x, gp.gy - 1) for x in range(1, arch.nx + 1)])
+    slots = [(x, y, s) for (x, y) in io_locs for s in range(arch.io_cap)]
+    sel = rng.permutation(len(slots))[:len(ios)]
+    for b, k in zip(ios, sel):
+        x, y, s = slots[k]
+        bx2[b], by2[b], bs2[b] = x, y, s
+        grid2[x, y, s] = b
+    assert np.array_equal(bx, bx2) and np.array_equal(by, by2)
+    assert np.array_equal(bs, bs2)
+    assert np.array_equal(grid, grid2.reshape(-1))
+
+    # het + fixed variants at least produce legal type placements
+    arch_h = get_arch("tiny_het")
+    nl_h = synth_netlist(spec_for_arch(arch_h, fill=0.5, seed=3))
+    gh = GpuPlacer.__new__(GpuPlacer)
+    gh.arch = arch_h; gh.nl = nl_h
+    gh.gx, gh.gy = arch_h.nx + 2, arch_h.ny + 2
+    gh.cap = max(1, arch_h.io_cap)
+    gh._fixed = None
+    hx, hy, hs, hgrid = gh._initial_placement(3)
+    tb = arch_h.tile_btype_grid()
+    bt = np.asarray(nl_h.block_type)
+    for b in range(nl_h.num_blocks):
+        assert tb[hx[b] * gh.gy + hy[b]] == bt[b], b
