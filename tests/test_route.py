@@ -295,3 +295,35 @@ def test_sink_parallel_merge_oracle():
     assert ok, err          # merged trees: connected, rooted, sinks hit,
     #                         occ recount exact (single-counted merge)
     assert r_v.total_wirelength() <= r_ref.total_wirelength() * 1.10
+
+
+def test_gpu_router_bb_host_logic():
+    """Pin the GPU router's HOST-side per-net bounding boxes on CPU:
+    every terminal inside the bb, margin applied, grid-clipped; growth
+    doubles the margin like the retry path."""
+    from parallel_eda_amd.route.gpu_router import GpuRouter
+    arch = get_arch("tseng")
+    nl = synth_netlist(spec_for_arch(arch, fill=0.5, seed=6))
+    pl = anneal_place(nl, arch, seed=6, timing_tradeoff=0.0)
+    g = rrgraph.build_rr_graph(arch)
+    net_ids, src_rr, sink_ptr, sink_rr, _ = net_rr_terminals(nl, pl, g, arch)
+    r = GpuRouter.__new__(GpuRouter)
+    r.g = g; r.arch = arch
+    r.src_rr = src_rr; r.sink_ptr = sink_ptr; r.sink_rr = sink_rr
+    r.num_nets = len(net_ids)
+    r.bb_margin_per_net = np.full(r.num_nets, 4, dtype=np.int32)
+    bb = r._compute_bbs()
+    xl = np.asarray(g.xlow); yl = np.asarray(g.ylow)
+    for n in range(r.num_nets):
+        terms = np.r_[src_rr[n], sink_rr[sink_ptr[n]:sink_ptr[n + 1]]]
+        assert (xl[terms] >= bb[n, 0]).all() and (xl[terms] <= bb[n, 2]).all()
+        assert (yl[terms] >= bb[n, 1]).all() and (yl[terms] <= bb[n, 3]).all()
+        assert 0 <= bb[n, 0] and bb[n, 2] <= arch.nx + 1
+        assert 0 <= bb[n, 1] and bb[n, 3] <= arch.ny + 1
+        # margin honored where the grid allows
+        assert bb[n, 0] <= max(0, xl[terms].min() - 4)
+        assert bb[n, 2] >= min(arch.nx + 1, xl[terms].max() + 4)
+    # growth path: doubling the margin widens (until clipped)
+    r.bb_margin_per_net[:] = 12
+    bb2 = r._compute_bbs()
+    assert (bb2[:, 0] <= bb[:, 0]).all() and (bb2[:, 2] >= bb[:, 2]).all()
