@@ -263,6 +263,10 @@ PYBIND11_MODULE(_pnr_cpu, m) {
                                py::array_t<int32_t, py::array::c_style | py::array::forcecast> s) {
         p.set_placement(x.data(), y.data(), s.data());
       })
+      .def("set_move_region", &SerialPlacer::set_move_region)
+      .def("reseed", &SerialPlacer::reseed)
+      .def("last_valid_attempts", &SerialPlacer::last_valid_attempts)
+      .def("last_accepts", &SerialPlacer::last_accepts)
       .def("fix_blocks", [](SerialPlacer& p,
                             py::array_t<int32_t, py::array::c_style | py::array::forcecast> ids,
                             py::array_t<int32_t, py::array::c_style | py::array::forcecast> x,

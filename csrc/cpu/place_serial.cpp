@@ -101,8 +101,10 @@ class SerialPlacer {
                    double bb_norm, double td_norm) {
     int64_t acc = 0;
     delta_sum_ = delta_sq_sum_ = 0; delta_n_ = 0;
+    att_valid_ = acc_cnt_ = 0;
     for (int64_t m = 0; m < nmoves; ++m)
       acc += try_swap(T, rlim, timing_tradeoff, bb_norm, td_norm);
+    acc_cnt_ = acc;
     return (double)acc / std::max<int64_t>(1, nmoves);
   }
 
@@ -160,6 +162,17 @@ class SerialPlacer {
 
   uint64_t rand_u64() { return rng_(); }
 
+  // Restrict moves to a column range [x0, x1] (inclusive), the grid-shard
+  // domain decomposition of SURVEY section 7 step 6: a rank only moves
+  // blocks currently inside its strip, and only within the strip, so
+  // concurrent ranks touch disjoint grid cells and their placements
+  // merge conflict-free. x0 = -1 disables.
+  void set_move_region(int x0, int x1) { rx0_ = x0; rx1_ = x1; }
+  // Per-rank stream divergence after a shared-seed initial placement.
+  void reseed(uint64_t seed) { rng_.seed(seed); }
+  int64_t last_valid_attempts() const { return att_valid_; }
+  int64_t last_accepts() const { return acc_cnt_; }
+
   // Pin blocks to fixed locations (reference: -pad_loc_file / fix_pins,
   // place.c initial_placement_location with pad constraints): teleport
   // each block to its location, then exclude it from all moves.
@@ -199,6 +212,8 @@ class SerialPlacer {
   std::vector<int8_t> tile_btype_;       // (gx*gy) x-major; see ctor
   std::vector<int> type_cols_[2];        // columns of type RAM(0) / DSP(1)
   std::vector<uint8_t> fixed_;           // empty => nothing fixed
+  int rx0_ = -1, rx1_ = -1;              // move region (column strip)
+  int64_t att_valid_ = 0, acc_cnt_ = 0;
 
  private:
   std::vector<int32_t> grid_;      // (x*gy+y)*cap + slot -> block
@@ -390,7 +405,9 @@ class SerialPlacer {
     int blk = (int)(rng_() % nb);
     if (!fixed_.empty() && fixed_[blk]) return 0;
     int x0 = bx_[blk], y0 = by_[blk];
+    if (rx0_ >= 0 && (x0 < rx0_ || x0 > rx1_)) return 0;  // not my shard
     int btype = nl_->block_type[blk];
+    ++att_valid_;
     // find_to: range-limited destination of matching type (place.c:1520)
     int irlim = std::max(1, (int)rlim);
     int x1 = -1, y1 = -1, slot1 = 0;
@@ -405,6 +422,7 @@ class SerialPlacer {
       if (ncol > 0) {
         for (int attempt = 0; attempt < 12; ++attempt) {
           int tx = *(lo + (int)(rng_() % ncol));
+          if (rx0_ >= 0 && (tx < rx0_ || tx > rx1_)) continue;
           int ylo = std::max(1, y0 - irlim), yhi = std::min(ny_, y0 + irlim);
           int ty = ylo + (int)(rng_() % (yhi - ylo + 1));
           if (tx == x0 && ty == y0) continue;
@@ -419,6 +437,7 @@ class SerialPlacer {
         int dy = (int)(rng_() % (2 * irlim + 1)) - irlim;
         int tx = x0 + dx, ty = y0 + dy;
         if (tx < 0 || tx >= gx_ || ty < 0 || ty >= gy_) continue;
+        if (rx0_ >= 0 && (tx < rx0_ || tx > rx1_)) continue;
         if (is_io_loc(tx, ty) != io) continue;
         if (tile_type(tx, ty) != btype) continue;
         if (cap_at(tx, ty) <= 0) continue;

@@ -1,0 +1,170 @@
+"""Multi-GPU / multi-process simulated-annealing placement.
+
+SURVEY §7 step 6's second half: the SA placer reuses the router's
+decomposition — the grid is cut into per-rank COLUMN STRIPS, each rank
+anneals only blocks inside its strip (moves confined to the strip, so
+concurrent ranks touch disjoint grid cells), and placements fuse with
+one masked all-reduce per temperature. Strip boundaries alternate
+between two offsets so blocks migrate across cuts over temperatures
+(the reference's MPI router rotates work the same way; the serial
+placer semantics transfer because within a strip this IS the serial
+placer).
+
+Deterministic: the shared seed gives every rank the identical initial
+placement, per-rank move streams only touch owned blocks, the masked
+sum reconstructs the identical merged placement everywhere, and the
+adaptive schedule consumes the all-reduced global acceptance rate.
+
+Engine: the CPU oracle (SerialPlacer.set_move_region). The GPU placer
+runs the same driver with its own region mask in round 2 — the fusion
+and schedule logic here is engine-independent.
+"""
+import numpy as np
+
+
+def _strips(nx, ws, phase):
+    """Partition columns 0..nx+1 into ws contiguous strips; phase 1
+    shifts the cuts by half a strip (first/last strips shorter)."""
+    total = nx + 2
+    w = max(1, total // ws)
+    cuts = [0]
+    off = (w // 2) if phase else 0
+    for r in range(1, ws):
+        cuts.append(min(total - 1, r * w + off))
+    cuts.append(total)
+    return [(cuts[r], cuts[r + 1] - 1) for r in range(ws)]
+
+
+def anneal_place_dist(netlist, arch, rank=0, world_size=1, seed=7,
+                      timing_tradeoff=0.0, sta=None, inner_num=1.0,
+                      crit_exp=1.0, verbose=False):
+    """Distributed SA anneal. Returns the (rank-identical) Placement."""
+    import torch
+    import torch.distributed as dist
+    from ..place.placer import Placement, analytic_delay_matrix
+    from .. import ops
+
+    cpu = ops.cpu()
+    tt = timing_tradeoff if sta is not None else 0.0
+    if sta is not None and tt > 0:
+        dm = analytic_delay_matrix(arch)
+    else:
+        dm = np.zeros(0, dtype=np.float32)
+    tb = (arch.tile_btype_grid() if arch.is_heterogeneous()
+          else np.empty(0, dtype=np.int8))
+    placer = cpu.SerialPlacer(netlist.cpp(), arch.nx, arch.ny, arch.io_cap,
+                              np.ascontiguousarray(dm.ravel()), seed, tb)
+    # identical initial placement everywhere (same seed), then diverge
+    # the move streams per rank
+    placer.reseed((seed + 1) * 1_000_003 + rank)
+    nb = netlist.num_blocks
+    ws = world_size
+    move_lim = max(64, int(inner_num * (nb ** 1.3333)))
+    rlim = float(max(arch.nx, arch.ny))
+
+    crit = np.zeros(netlist.num_conns, dtype=np.float32)
+
+    def refresh_crit():
+        nonlocal crit
+        if sta is None or tt <= 0:
+            return
+        cpd, slack, c = sta.analyze(placer.conn_delays())
+        crit = (np.asarray(c) ** crit_exp).astype(np.float32)
+        placer.set_crit(crit)
+
+    def fuse(owner_lo, owner_hi):
+        """Masked all-reduce merge: every block is owned by exactly one
+        rank (by its pre-move column), so summing owned positions
+        reconstructs the full placement identically on every rank."""
+        if ws <= 1:
+            return
+        x, y, sl = [np.asarray(a, dtype=np.int64)
+                    for a in placer.placement()]
+        own = (x0_pre >= owner_lo) & (x0_pre <= owner_hi)
+        buf = torch.from_numpy(np.concatenate([
+            np.where(own, x, 0), np.where(own, y, 0),
+            np.where(own, sl, 0)]))
+        dist.all_reduce(buf, op=dist.ReduceOp.SUM)
+        m = buf.numpy()
+        placer.set_placement(m[:nb].astype(np.int32),
+                             m[nb:2 * nb].astype(np.int32),
+                             m[2 * nb:].astype(np.int32))
+
+    def global_srate():
+        acc = placer.last_accepts()
+        att = placer.last_valid_attempts()
+        if ws > 1:
+            t = torch.tensor([acc, att], dtype=torch.int64)
+            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+            acc, att = int(t[0]), int(t[1])
+        return acc / max(1, att), att
+
+    refresh_crit()
+    norm_mode = tt > 0
+
+    def norms():
+        if norm_mode:
+            return (max(placer.bb_cost(), 1e-12), max(placer.td_cost(), 1e-30))
+        return (1.0, 1.0)
+
+    # starting T from a probe pass on the own strip
+    x0_pre = np.asarray(placer.placement()[0])
+    lo, hi = _strips(arch.nx, ws, 0)[rank] if ws > 1 else (-1, -1)
+    placer.set_move_region(lo, hi)
+    bb_norm, td_norm = norms()
+    placer.run_moves(1e30, rlim, max(64, nb), tt, bb_norm, td_norm)
+    t = 20.0 * placer.last_delta_std()
+    if ws > 1:
+        tt_t = torch.tensor([t], dtype=torch.float64)
+        dist.all_reduce(tt_t, op=dist.ReduceOp.MAX)
+        t = float(tt_t[0])
+    fuse(lo, hi)
+    if t <= 0:
+        t = 1.0
+
+    itemp = 0
+    history = []
+    while True:
+        refresh_crit()
+        bb_norm, td_norm = norms()
+        x0_pre = np.asarray(placer.placement()[0])
+        if ws > 1:
+            lo, hi = _strips(arch.nx, ws, itemp % 2)[rank]
+        placer.set_move_region(lo, hi)
+        placer.run_moves(t, rlim, move_lim, tt, bb_norm, td_norm)
+        srate, att = global_srate()
+        fuse(lo, hi)
+        cost = placer.bb_cost()
+        history.append((t, cost, srate, rlim))
+        if verbose and rank == 0:
+            print(f"[dist-sa] T={t:.3e} bb={cost:.1f} acc={srate:.2f} "
+                  f"att={att} rlim={rlim:.1f}")
+        if srate > 0.96:
+            t *= 0.5
+        elif srate > 0.8:
+            t *= 0.9
+        elif srate > 0.15 and rlim > 1:
+            t *= 0.95
+        else:
+            t *= 0.8
+        rlim = min(max(rlim * (1.0 - 0.44 + srate), 1.0),
+                   float(max(arch.nx, arch.ny)))
+        itemp += 1
+        exit_cost = 1.0 if norm_mode else cost
+        if t < 0.005 * exit_cost / max(1, netlist.num_nets):
+            break
+        if itemp > 500:
+            break
+    # quench on own strip, then final fuse
+    x0_pre = np.asarray(placer.placement()[0])
+    placer.run_moves(0.0, 1.0, move_lim, tt, bb_norm, td_norm)
+    fuse(lo, hi)
+    placer.set_move_region(-1, -1)
+    ok, err = placer.check_place()
+    if not ok:
+        raise RuntimeError(f"dist check_place failed: {err}")
+    x, y, sl = placer.placement()
+    return Placement(np.asarray(x), np.asarray(y), np.asarray(sl),
+                     bb_cost=placer.bb_cost(), td_cost=placer.td_cost(),
+                     stats={"temps": itemp, "move_lim": move_lim,
+                            "history": history, "engine": f"dist-cpu x{ws}"})

@@ -335,3 +335,46 @@ def test_dist_flow_deterministic(tmp_path):
         rb = pickle.load(f)
     assert np.array_equal(ra["occ"], rb["occ"])
     assert ra["res"]["cpd"] == rb["res"]["cpd"]
+
+
+def _worker_place(rank, world, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+    from parallel_eda_amd.parallel.dist_place import anneal_place_dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    arch, nl, _ = _build_case()
+    pl = anneal_place_dist(nl, arch, rank=rank, world_size=world, seed=9)
+    with open(os.path.join(tmpdir, f"pl{rank}.pkl"), "wb") as f:
+        pickle.dump({"x": np.asarray(pl.x), "y": np.asarray(pl.y),
+                     "slot": np.asarray(pl.slot),
+                     "bb": pl.bb_cost, "temps": pl.stats["temps"]}, f)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_dist_place_world2(tmp_path):
+    """Strip-sharded distributed SA (SURVEY step-6 placer half): ranks
+    produce the IDENTICAL fused placement, it is legal, and quality
+    stays in family with the serial anneal."""
+    mp.spawn(_worker_place, args=(2, 29538, str(tmp_path)), nprocs=2,
+             join=True)
+    with open(tmp_path / "pl0.pkl", "rb") as f:
+        r0 = pickle.load(f)
+    with open(tmp_path / "pl1.pkl", "rb") as f:
+        r1 = pickle.load(f)
+    assert np.array_equal(r0["x"], r1["x"])
+    assert np.array_equal(r0["y"], r1["y"])
+    assert np.array_equal(r0["slot"], r1["slot"])
+    assert r0["bb"] == pytest.approx(r1["bb"], rel=1e-9)
+    # quality: within 25% of the serial anneal on the same case
+    arch, nl, _ = _build_case()
+    serial = anneal_place(nl, arch, seed=9, timing_tradeoff=0.0)
+    assert r0["bb"] <= serial.bb_cost * 1.25, (r0["bb"], serial.bb_cost)
+    # the fused placement routes
+    from parallel_eda_amd.route.router import pathfinder_route
+    from parallel_eda_amd.place.placer import Placement
+    g = rrgraph.build_rr_graph(arch)
+    pl = Placement(r0["x"], r0["y"], r0["slot"])
+    res = pathfinder_route(nl, pl, g, arch, sta=None, max_iters=60)
+    assert res.success
