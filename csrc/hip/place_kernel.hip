@@ -37,6 +37,9 @@ struct PlaceDev {
   float* net_cost;              // current bb cost per net
   float* net_tcost;             // current timing cost per net
   int32_t num_blocks, num_nets, gx, gy, cap, nx, ny, io_cap;
+  int32_t rx0, rx1;             // move region (column strip) for the
+                                // distributed strip-sharded anneal
+                                // (parallel/dist_place.py); rx0 < 0 = off
 };
 
 struct MovesDev {
@@ -126,6 +129,8 @@ __global__ void place_propose_kernel(PlaceDev p, MovesDev m, float T,
   uint32_t r0 = rng_hash(seed ^ 0x5BD1E995u, batch, i);
   int32_t blk = r0 % p.num_blocks;
   if (p.fixed && p.fixed[blk]) return;   // pinned (pad_loc_file)
+  if (p.rx0 >= 0 && (p.bx[blk] < p.rx0 || p.bx[blk] > p.rx1))
+    return;                              // not this rank's strip
   bool io = p.blk_type[blk] == 0;
   int x0 = p.bx[blk], y0 = p.by[blk];
   int x1 = -1, y1 = -1, slot1 = 0;
@@ -149,6 +154,7 @@ __global__ void place_propose_kernel(PlaceDev p, MovesDev m, float T,
         uint32_t r1 = rng_hash(seed, batch, i * 131 + 7 * att + 1);
         uint32_t r2 = rng_hash(seed, batch, i * 131 + 7 * att + 2);
         int tx = p.type_cols[lo + (int)(r1 % ncol)];
+        if (p.rx0 >= 0 && (tx < p.rx0 || tx > p.rx1)) continue;
         int ty = ylo + (int)(r2 % (yhi - ylo + 1));
         if (tx == x0 && ty == y0) continue;
         x1 = tx; y1 = ty; slot1 = 0;
@@ -162,6 +168,7 @@ __global__ void place_propose_kernel(PlaceDev p, MovesDev m, float T,
       int tx = x0 + (int)(r1 % (2 * rlim + 1)) - rlim;
       int ty = y0 + (int)(r2 % (2 * rlim + 1)) - rlim;
       if (tx < 0 || tx >= p.gx || ty < 0 || ty >= p.gy) continue;
+      if (p.rx0 >= 0 && (tx < p.rx0 || tx > p.rx1)) continue;
       if (p.tile_btype) {
         // heterogeneous fabric: destination tile must match the block type
         if (p.tile_btype[tx * p.gy + ty] != p.blk_type[blk]) continue;
@@ -336,6 +343,7 @@ struct PlaceLaunchArgs {
   int32_t* bx; int32_t* by; int32_t* bslot; int32_t* grid;
   float* net_cost; float* net_tcost;
   int32_t num_blocks, num_nets, gx, gy, cap, nx, ny, io_cap;
+  int32_t rx0, rx1;
   // moves
   int32_t* mv_blk; int32_t* mv_to; int32_t* mv_other;
   float* mv_dbb; float* mv_dtd; uint8_t* mv_flags;
@@ -362,6 +370,7 @@ static void unpack(const PlaceLaunchArgs* a, PlaceDev& p, MovesDev& m) {
   p.num_blocks = a->num_blocks; p.num_nets = a->num_nets;
   p.gx = a->gx; p.gy = a->gy; p.cap = a->cap;
   p.nx = a->nx; p.ny = a->ny; p.io_cap = a->io_cap;
+  p.rx0 = a->rx0; p.rx1 = a->rx1;
   m.mv_blk = a->mv_blk; m.mv_to = a->mv_to; m.mv_other = a->mv_other;
   m.mv_dbb = a->mv_dbb; m.mv_dtd = a->mv_dtd; m.mv_flags = a->mv_flags;
   m.net_claim = a->net_claim; m.loc_claim = a->loc_claim;

@@ -31,6 +31,7 @@ class PlaceLaunchArgs(ct.Structure):
         ("num_blocks", ct.c_int32), ("num_nets", ct.c_int32),
         ("gx", ct.c_int32), ("gy", ct.c_int32), ("cap", ct.c_int32),
         ("nx", ct.c_int32), ("ny", ct.c_int32), ("io_cap", ct.c_int32),
+        ("rx0", ct.c_int32), ("rx1", ct.c_int32),
         ("mv_blk", ct.c_void_p), ("mv_to", ct.c_void_p), ("mv_other", ct.c_void_p),
         ("mv_dbb", ct.c_void_p), ("mv_dtd", ct.c_void_p), ("mv_flags", ct.c_void_p),
         ("net_claim", ct.c_void_p), ("loc_claim", ct.c_void_p),
@@ -265,6 +266,7 @@ class GpuPlacer:
         a.num_blocks = self.nl.num_blocks; a.num_nets = self.nl.num_nets
         a.gx = self.gx; a.gy = self.gy; a.cap = self.cap
         a.nx = self.arch.nx; a.ny = self.arch.ny; a.io_cap = self.arch.io_cap
+        a.rx0, a.rx1 = getattr(self, "move_region", (-1, -1))
         a.mv_blk = ptr(self.t_mv_blk); a.mv_to = ptr(self.t_mv_to)
         a.mv_other = ptr(self.t_mv_other); a.mv_dbb = ptr(self.t_mv_dbb)
         a.mv_dtd = ptr(self.t_mv_dtd); a.mv_flags = ptr(self.t_mv_flags)
@@ -317,6 +319,11 @@ class GpuPlacer:
         # claim-conflict losers are neither accepts nor rejects
         srate = acc / max(1, att)
         return srate, att
+
+    def set_move_region(self, x0, x1):
+        """Confine moves to columns [x0, x1] (strip-sharded distributed
+        anneal, parallel/dist_place.py); (-1, -1) disables."""
+        self.move_region = (x0, x1)
 
     def set_crit(self, conn_crit):
         t = self.torch.from_numpy(np.ascontiguousarray(conn_crit,
