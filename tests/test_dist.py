@@ -378,3 +378,37 @@ def test_dist_place_world2(tmp_path):
     pl = Placement(r0["x"], r0["y"], r0["slot"])
     res = pathfinder_route(nl, pl, g, arch, sta=None, max_iters=60)
     assert res.success
+
+
+def _worker_full_flow(rank, world, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+    from parallel_eda_amd.parallel.full_flow import run_flow_dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    arch, nl, _ = _build_case()
+    res = run_flow_dist(nl, arch, rank=rank, world_size=world, seed=9,
+                        timing_driven=True, max_route_iters=60)
+    with open(os.path.join(tmpdir, f"ff{rank}.pkl"), "wb") as f:
+        pickle.dump({"success": res["success"], "wl": res["wirelength"],
+                     "cpd": res["cpd"],
+                     "px": np.asarray(res["place"].x)}, f)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_dist_full_flow_world2(tmp_path):
+    """BASELINE config-4 shape end to end on gloo: distributed SA
+    placement + distributed PathFinder + replicated STA; every rank
+    returns the identical successful result."""
+    mp.spawn(_worker_full_flow, args=(2, 29539, str(tmp_path)), nprocs=2,
+             join=True)
+    with open(tmp_path / "ff0.pkl", "rb") as f:
+        r0 = pickle.load(f)
+    with open(tmp_path / "ff1.pkl", "rb") as f:
+        r1 = pickle.load(f)
+    assert r0["success"] and r1["success"]
+    assert r0["wl"] == r1["wl"]
+    assert r0["cpd"] == pytest.approx(r1["cpd"], rel=1e-9)
+    assert np.array_equal(r0["px"], r1["px"])
+    assert r0["wl"] > 0 and r0["cpd"] > 0
