@@ -66,6 +66,11 @@ def build_parser():
     # outputs
     p.add_argument("--sdc", type=str, default=None,
                    help="SDC constraints (create_clock -period, subset)")
+    p.add_argument("--power_report", type=str, default=None,
+                   help="post-route power estimate (reference: power.c)")
+    p.add_argument("--activity_file", type=str, default=None,
+                   help="per-net switching activities (.act lines: "
+                        "'netname activity')")
     p.add_argument("--timing_report", type=str, default=None,
                    help="write a critical-path report here after routing")
     p.add_argument("--place_only", action="store_true",
@@ -277,6 +282,16 @@ def main(argv=None):
                     else res.router.t_sink_delay.cpu().numpy()
                 write_sdf(args.out_sdf, netlist, arch, cmap4.conn_delays(sd4))
                 print(f"wrote {args.out_sdf}")
+        if args.power_report and hasattr(res.router, "tree"):
+            from .utils.power import (estimate_power, read_activity_file,
+                                      write_power_report)
+            act = (read_activity_file(args.activity_file, netlist)
+                   if args.activity_file else None)
+            pw = estimate_power(netlist, arch, g, res.router,
+                                activities=act)
+            write_power_report(args.power_report, pw)
+            print(f"wrote {args.power_report} "
+                  f"(total {pw['total_W']*1e3:.3f} mW)")
         if args.timing_report and sta is not None:
             from .route.router import ConnMap
             from .timing.report import write_timing_report

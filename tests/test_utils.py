@@ -182,3 +182,43 @@ def test_tree_dot_dump(tmp_path):
     write_tree_dot(str(f), [10, 11, 12], [-1, 0, 0], label="x")
     body = f.read_text()
     assert "digraph" in body and "n0 -> n1" in body and "n0 -> n2" in body
+
+
+def test_power_estimate(tmp_path):
+    """Block-level power model (reference: power/power.c power_total —
+    dynamic + leakage from activities): components positive, total adds
+    up, routing power scales with activity, .act file honored."""
+    import numpy as np
+    from parallel_eda_amd.arch.archdef import get_arch
+    from parallel_eda_amd.io.synth import synth_netlist, spec_for_arch
+    from parallel_eda_amd.place.placer import anneal_place
+    from parallel_eda_amd.route.router import pathfinder_route
+    from parallel_eda_amd.utils.power import (estimate_power,
+                                              read_activity_file,
+                                              write_power_report)
+    from parallel_eda_amd import rrgraph
+
+    arch = get_arch("tiny")
+    nl = synth_netlist(spec_for_arch(arch, fill=0.5, seed=3))
+    pl = anneal_place(nl, arch, seed=3, timing_tradeoff=0.0)
+    g = rrgraph.build_rr_graph(arch)
+    res = pathfinder_route(nl, pl, g, arch, sta=None, max_iters=40)
+    assert res.success
+    p = estimate_power(nl, arch, g, res.router)
+    assert p["total_W"] > 0
+    assert abs(p["total_W"] - (p["routing_W"] + p["logic_W"] +
+                               p["clock_W"] + p["leakage_W"])) < 1e-12
+    # doubling activity doubles routing power
+    p2 = estimate_power(nl, arch, g, res.router,
+                        activities=np.full(nl.num_nets, 0.30))
+    assert p2["routing_W"] == pytest.approx(2 * p["routing_W"], rel=1e-6)
+    # activity file
+    nl.names = [f"b{i}" for i in range(nl.num_blocks)]
+    f = tmp_path / "x.act"
+    drv0 = f"b{int(nl.net_driver[0])}"
+    f.write_text(f"{drv0} 0.5\n")
+    act = read_activity_file(str(f), nl)
+    assert act[0] == 0.5 and act[1] == 0.15
+    rpt = tmp_path / "p.rpt"
+    write_power_report(str(rpt), p)
+    assert "total_W" in rpt.read_text()
