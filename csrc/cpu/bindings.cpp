@@ -214,7 +214,10 @@ PYBIND11_MODULE(_pnr_cpu, m) {
         bool ok = r.check_routed(&err);
         return py::make_tuple(ok, err);
       })
+      .def("num_nets", &SerialRouter::num_nets)
       .def("tree", [](SerialRouter& r, int inet) {
+        if (inet < 0 || inet >= r.num_nets())
+          throw py::index_error("net id out of range");
         const RouteTree& t = r.tree(inet);
         return py::make_tuple(
             py::array_t<int32_t>((py::ssize_t)t.nodes.size(), t.nodes.data()),

@@ -66,6 +66,10 @@ def build_parser():
     # outputs
     p.add_argument("--sdc", type=str, default=None,
                    help="SDC constraints (create_clock -period, subset)")
+    p.add_argument("--draw_place", type=str, default=None,
+                   help="render the placement to SVG (headless draw.c)")
+    p.add_argument("--draw_route", type=str, default=None,
+                   help="render the routing to SVG")
     p.add_argument("--power_report", type=str, default=None,
                    help="post-route power estimate (reference: power.c)")
     p.add_argument("--activity_file", type=str, default=None,
@@ -190,6 +194,10 @@ def main(argv=None):
     if args.out_place:
         write_place(args.out_place, placement, netlist, arch)
         print(f"wrote {args.out_place}")
+    if args.draw_place:
+        from .utils.draw import write_placement_svg
+        write_placement_svg(args.draw_place, placement, netlist, arch)
+        print(f"wrote {args.draw_place}")
     if args.place_only:
         print(f"entire flow took {time.perf_counter()-t_start:.2f}s")
         return 0
@@ -282,6 +290,11 @@ def main(argv=None):
                     else res.router.t_sink_delay.cpu().numpy()
                 write_sdf(args.out_sdf, netlist, arch, cmap4.conn_delays(sd4))
                 print(f"wrote {args.out_sdf}")
+        if args.draw_route and hasattr(res.router, "tree"):
+            from .utils.draw import write_routing_svg
+            write_routing_svg(args.draw_route, g, arch, res.router,
+                              net_ids=net_ids)
+            print(f"wrote {args.draw_route}")
         if args.power_report and hasattr(res.router, "tree"):
             from .utils.power import (estimate_power, read_activity_file,
                                       write_power_report)

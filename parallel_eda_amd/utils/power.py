@@ -53,12 +53,9 @@ def estimate_power(netlist, arch, g, router, activities=None,
     half_fv2 = 0.5 * f_clk * vdd * vdd
 
     p_route = 0.0
-    n_routed = min(nn, 1 << 62)
+    n_routed = min(nn, int(router.num_nets()))
     for n in range(n_routed):
-        try:
-            nodes, parents, sw, delay = router.tree(n)
-        except Exception:
-            break
+        nodes, parents, sw, delay = router.tree(n)
         nodes = np.asarray(nodes)
         if not len(nodes):
             continue

@@ -222,3 +222,27 @@ def test_power_estimate(tmp_path):
     rpt = tmp_path / "p.rpt"
     write_power_report(str(rpt), p)
     assert "total_W" in rpt.read_text()
+
+
+def test_svg_rendering(tmp_path):
+    """Headless placement/routing SVG dumps (reference: graphics.c/draw.c
+    interactive view -> SURVEY's dump-to-image)."""
+    from parallel_eda_amd.arch.archdef import get_arch
+    from parallel_eda_amd.io.synth import synth_netlist, spec_for_arch
+    from parallel_eda_amd.place.placer import anneal_place
+    from parallel_eda_amd.route.router import pathfinder_route
+    from parallel_eda_amd.utils.draw import (write_placement_svg,
+                                             write_routing_svg)
+    from parallel_eda_amd import rrgraph
+    arch = get_arch("tiny_het")
+    nl = synth_netlist(spec_for_arch(arch, fill=0.5, seed=5))
+    pl = anneal_place(nl, arch, seed=5, timing_tradeoff=0.0)
+    g = rrgraph.build_rr_graph(arch)
+    res = pathfinder_route(nl, pl, g, arch, sta=None, max_iters=60)
+    f1 = tmp_path / "p.svg"; f2 = tmp_path / "r.svg"
+    write_placement_svg(str(f1), pl, nl, arch)
+    write_routing_svg(str(f2), g, arch, res.router)
+    a, b = f1.read_text(), f2.read_text()
+    assert a.startswith("<svg") and a.rstrip().endswith("</svg>")
+    assert b.count("<line") > 10           # wires drawn
+    assert a.count("<rect") >= nl.num_blocks
