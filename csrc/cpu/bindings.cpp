@@ -266,6 +266,14 @@ PYBIND11_MODULE(_pnr_cpu, m) {
                                py::array_t<int32_t, py::array::c_style | py::array::forcecast> s) {
         p.set_placement(x.data(), y.data(), s.data());
       })
+      .def("set_macros", [](SerialPlacer& p,
+                            py::array_t<int64_t, py::array::c_style | py::array::forcecast> ptr,
+                            py::array_t<int32_t, py::array::c_style | py::array::forcecast> blk,
+                            py::array_t<int32_t, py::array::c_style | py::array::forcecast> dx,
+                            py::array_t<int32_t, py::array::c_style | py::array::forcecast> dy) {
+        p.set_macros(ptr.data(), blk.data(), dx.data(), dy.data(),
+                     (int)ptr.size() - 1);
+      })
       .def("set_move_region", &SerialPlacer::set_move_region)
       .def("reseed", &SerialPlacer::reseed)
       .def("last_valid_attempts", &SerialPlacer::last_valid_attempts)

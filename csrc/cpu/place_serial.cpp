@@ -152,6 +152,7 @@ class SerialPlacer {
       if (grid_at(x, y, bslot_[b]) != b) { *err = "grid inconsistent"; return false; }
       cnt[(size_t)x * gy_ + y]++;
     }
+    if (!macro_ptr_.empty() && !macro_members_consistent(err)) return false;
     double fresh = const_cast<SerialPlacer*>(this)->net_cost_sum_check();
     if (std::abs(fresh - bb_cost_) > 1e-3 * std::max(1.0, fresh)) {
       *err = "bb cost drift: " + std::to_string(fresh) + " vs " + std::to_string(bb_cost_);
@@ -172,6 +173,54 @@ class SerialPlacer {
   void reseed(uint64_t seed) { rng_.seed(seed); }
   int64_t last_valid_attempts() const { return att_valid_; }
   int64_t last_accepts() const { return acc_cnt_; }
+
+  // Placement macros (reference: place/place_macro.c — carry chains):
+  // groups of blocks at fixed relative offsets from a head, moved
+  // atomically by try_swap. member_ptr/member_blk/dx/dy: CSR, first
+  // member of each macro is the head at offset (0,0). Re-runs the
+  // initial placement so macros start legal.
+  void set_macros(const int64_t* member_ptr, const int32_t* member_blk,
+                  const int32_t* mdx, const int32_t* mdy, int n_macros) {
+    int nb = nl_->num_blocks;
+    macro_ptr_.assign(member_ptr, member_ptr + n_macros + 1);
+    int64_t total = member_ptr[n_macros];
+    macro_blk_.assign(member_blk, member_blk + total);
+    macro_dx_.assign(mdx, mdx + total);
+    macro_dy_.assign(mdy, mdy + total);
+    macro_of_.assign(nb, -1);
+    for (int m = 0; m < n_macros; ++m) {
+      if (macro_dx_[macro_ptr_[m]] != 0 || macro_dy_[macro_ptr_[m]] != 0)
+        throw std::runtime_error("macro head must be at offset (0,0)");
+      for (int64_t k = macro_ptr_[m]; k < macro_ptr_[m + 1]; ++k) {
+        int b = macro_blk_[k];
+        if (nl_->block_type[b] == 0)
+          throw std::runtime_error("IO blocks cannot join macros");
+        if (macro_of_[b] >= 0)
+          throw std::runtime_error("block in two macros");
+        macro_of_[b] = m;
+      }
+    }
+    std::fill(grid_.begin(), grid_.end(), -1);
+    std::fill(grid_cnt_.begin(), grid_cnt_.end(), 0);
+    initial_placement_with_macros();
+    recompute_bb_all();
+    recompute_td_all();
+  }
+
+  bool macro_members_consistent(std::string* err) const {
+    for (size_t m = 0; m + 1 < macro_ptr_.size(); ++m) {
+      int head = macro_blk_[macro_ptr_[m]];
+      for (int64_t k = macro_ptr_[m]; k < macro_ptr_[m + 1]; ++k) {
+        int b = macro_blk_[k];
+        if (bx_[b] != bx_[head] + macro_dx_[k] ||
+            by_[b] != by_[head] + macro_dy_[k]) {
+          *err = "macro " + std::to_string(m) + " member offset broken";
+          return false;
+        }
+      }
+    }
+    return true;
+  }
 
   // Pin blocks to fixed locations (reference: -pad_loc_file / fix_pins,
   // place.c initial_placement_location with pad constraints): teleport
@@ -214,6 +263,9 @@ class SerialPlacer {
   std::vector<uint8_t> fixed_;           // empty => nothing fixed
   int rx0_ = -1, rx1_ = -1;              // move region (column strip)
   int64_t att_valid_ = 0, acc_cnt_ = 0;
+  std::vector<int64_t> macro_ptr_;       // empty => no macros
+  std::vector<int32_t> macro_blk_, macro_dx_, macro_dy_;
+  std::vector<int32_t> macro_of_;
 
  private:
   std::vector<int32_t> grid_;      // (x*gy+y)*cap + slot -> block
@@ -295,6 +347,196 @@ class SerialPlacer {
       }
     }
   }
+
+  void initial_placement_with_macros() {
+    int nb = nl_->num_blocks;
+    std::fill(bx_.begin(), bx_.end(), -1);
+    std::fill(by_.begin(), by_.end(), -1);
+    std::fill(bslot_.begin(), bslot_.end(), 0);
+    // 1) macros: scan shuffled head locations where every member offset
+    //    lands on a free tile of the member's type
+    std::vector<std::pair<int, int>> cells;
+    for (int x = 1; x <= nx_; ++x)
+      for (int y = 1; y <= ny_; ++y) cells.push_back({x, y});
+    std::shuffle(cells.begin(), cells.end(), rng_);
+    int n_macros = (int)macro_ptr_.size() - 1;
+    for (int m = 0; m < n_macros; ++m) {
+      bool placed = false;
+      for (auto [hx, hy] : cells) {
+        bool ok = true;
+        for (int64_t k = macro_ptr_[m]; k < macro_ptr_[m + 1] && ok; ++k) {
+          int x = hx + macro_dx_[k], y = hy + macro_dy_[k];
+          if (x < 1 || x > nx_ || y < 1 || y > ny_ ||
+              tile_type(x, y) != nl_->block_type[macro_blk_[k]] ||
+              grid_at(x, y, 0) >= 0)
+            ok = false;
+        }
+        if (!ok) continue;
+        for (int64_t k = macro_ptr_[m]; k < macro_ptr_[m + 1]; ++k) {
+          int b = macro_blk_[k];
+          bx_[b] = hx + macro_dx_[k]; by_[b] = hy + macro_dy_[k];
+          bslot_[b] = 0;
+          grid_at(bx_[b], by_[b], 0) = b;
+          grid_cnt_[(size_t)bx_[b] * gy_ + by_[b]]++;
+        }
+        placed = true;
+        break;
+      }
+      if (!placed)
+        throw std::runtime_error("no legal location for macro " +
+                                 std::to_string(m));
+    }
+    // 2) the rest: free tiles of the right type / IO slots
+    std::vector<std::pair<int, int>> locs[4];
+    for (int x = 1; x <= nx_; ++x)
+      for (int y = 1; y <= ny_; ++y) {
+        int8_t t = tile_type(x, y);
+        if (t >= 1 && grid_at(x, y, 0) < 0) locs[t].push_back({x, y});
+      }
+    for (int y = 1; y <= ny_; ++y) { locs[0].push_back({0, y}); locs[0].push_back({gx_ - 1, y}); }
+    for (int x = 1; x <= nx_; ++x) { locs[0].push_back({x, 0}); locs[0].push_back({x, gy_ - 1}); }
+    for (int t = 0; t < 4; ++t) std::shuffle(locs[t].begin(), locs[t].end(), rng_);
+    size_t cur[4] = {0, 0, 0, 0};
+    int io_slot = 0;
+    for (int b = 0; b < nb; ++b) {
+      if (!macro_of_.empty() && macro_of_[b] >= 0) continue;
+      int t = nl_->block_type[b];
+      if (t != 0) {
+        if (cur[t] >= locs[t].size())
+          throw std::runtime_error("too many blocks of type " +
+                                   std::to_string(t) + " for grid");
+        auto [x, y] = locs[t][cur[t]++];
+        bx_[b] = x; by_[b] = y; bslot_[b] = 0;
+        grid_at(x, y, 0) = b; grid_cnt_[(size_t)x * gy_ + y]++;
+      } else {
+        if (cur[0] >= locs[0].size()) throw std::runtime_error("too many IOs for grid");
+        auto [x, y] = locs[0][cur[0]];
+        bx_[b] = x; by_[b] = y; bslot_[b] = io_slot;
+        grid_at(x, y, io_slot) = b; grid_cnt_[(size_t)x * gy_ + y]++;
+        if (++io_slot >= io_cap_) { io_slot = 0; ++cur[0]; }
+      }
+    }
+  }
+
+  // Atomic macro move (reference: find_affected_blocks place.c:1192):
+  // shift every member by (dx,dy); member targets must be type-legal and
+  // hold either nothing or single (non-macro) blocks, which swap into
+  // the vacated member cells.
+  int try_macro_move(int m, double T, double rlim, double timing_tradeoff,
+                     double bb_norm, double td_norm) {
+    int irlim = std::max(1, (int)rlim);
+    int dx = (int)(rng_() % (2 * irlim + 1)) - irlim;
+    int dy = (int)(rng_() % (2 * irlim + 1)) - irlim;
+    if (dx == 0 && dy == 0) return 0;
+    int64_t k0 = macro_ptr_[m], k1 = macro_ptr_[m + 1];
+    // validate all targets
+    mm_moves_.clear();
+    for (int64_t k = k0; k < k1; ++k) {
+      int b = macro_blk_[k];
+      int tx = bx_[b] + dx, ty = by_[b] + dy;
+      if (tx < 1 || tx > nx_ || ty < 1 || ty > ny_) return 0;
+      if (rx0_ >= 0 && (tx < rx0_ || tx > rx1_)) return 0;
+      if (tile_type(tx, ty) != nl_->block_type[b]) return 0;
+      int occ = grid_at(tx, ty, 0);
+      if (occ >= 0) {
+        if (macro_of_[occ] >= 0) return 0;   // another macro's member
+        if (!fixed_.empty() && fixed_[occ]) return 0;
+        mm_moves_.push_back({occ, bx_[b], by_[b]});  // displaced -> vacated
+      }
+      mm_moves_.push_back({b, tx, ty});
+    }
+    ++att_valid_;
+    // collect affected nets
+    saved_bbs_.clear(); saved_costs_.clear(); saved_ids_.clear();
+    for (auto& mv : mm_moves_) {
+      int b = mv.b;
+      for (int64_t kk = blk_net_ptr_[b]; kk < blk_net_ptr_[b + 1]; ++kk) {
+        int n = blk_nets_[kk];
+        if (!mark_net(n)) continue;
+        saved_ids_.push_back(n);
+        saved_bbs_.push_back(bbs_[n]);
+        saved_costs_.push_back(net_cost_[n]);
+      }
+    }
+    double before = 0, td_before = 0;
+    for (int n : saved_ids_) before += net_cost_[n];
+    if (timing_tradeoff > 0) td_before = td_of_move_blocks();
+    // apply: clear all source cells, then write all targets
+    mm_save_.clear();
+    for (auto& mv : mm_moves_) mm_save_.push_back({mv.b, bx_[mv.b], by_[mv.b]});
+    for (auto& mv : mm_moves_) grid_at(bx_[mv.b], by_[mv.b], 0) = -1;
+    for (auto& mv : mm_moves_) {
+      bx_[mv.b] = mv.x; by_[mv.b] = mv.y; bslot_[mv.b] = 0;
+      grid_at(mv.x, mv.y, 0) = mv.b;
+    }
+    for (int n : saved_ids_) net_cost_from_scratch(n);
+    double after = 0, td_after = 0;
+    for (int n : saved_ids_) after += net_cost_[n];
+    if (timing_tradeoff > 0) td_after = td_of_move_blocks();
+    double d_bb = after - before, d_td = td_after - td_before;
+    double delta = (1.0 - timing_tradeoff) * d_bb / bb_norm +
+                   timing_tradeoff * d_td / td_norm;
+    delta_sum_ += delta; delta_sq_sum_ += delta * delta; ++delta_n_;
+    bool accept;
+    if (delta <= 0) accept = true;
+    else if (T <= 0) accept = false;
+    else {
+      double u = (double)(rng_() % (1ull << 53)) / (double)(1ull << 53);
+      accept = u < std::exp(-delta / T);
+    }
+    if (accept) {
+      bb_cost_ += d_bb; td_cost_ += d_td;
+      for (int n : saved_ids_) unmark_net(n);
+      update_conn_delays_of_move();
+      return 1;
+    }
+    // revert
+    for (auto& mv : mm_moves_) grid_at(bx_[mv.b], by_[mv.b], 0) = -1;
+    for (auto& sv : mm_save_) {
+      bx_[sv.b] = sv.x; by_[sv.b] = sv.y;
+      grid_at(sv.x, sv.y, 0) = sv.b;
+    }
+    for (size_t i = 0; i < saved_ids_.size(); ++i) {
+      bbs_[saved_ids_[i]] = saved_bbs_[i];
+      net_cost_[saved_ids_[i]] = saved_costs_[i];
+      unmark_net(saved_ids_[i]);
+    }
+    return 0;
+  }
+
+  double td_of_move_blocks() {
+    double t = 0;
+    for (auto& mv : mm_moves_) {
+      for (int64_t k = blk_net_ptr_[mv.b]; k < blk_net_ptr_[mv.b + 1]; ++k) {
+        int n = blk_nets_[k];
+        if (!mark_net2(n)) continue;
+        int drv = nl_->net_driver[n];
+        for (int64_t c2 = nl_->net_sink_ptr[n]; c2 < nl_->net_sink_ptr[n + 1]; ++c2)
+          t += (double)crit_[c2] * conn_delay(drv, nl_->net_sinks[c2]);
+      }
+    }
+    for (int n : marked2_) net_mark2_[n] = 0;
+    marked2_.clear();
+    return t;
+  }
+
+  void update_conn_delays_of_move() {
+    if (delay_mat_.empty()) return;
+    for (auto& mv : mm_moves_) {
+      for (int64_t k = blk_net_ptr_[mv.b]; k < blk_net_ptr_[mv.b + 1]; ++k) {
+        int n = blk_nets_[k];
+        if (!mark_net2(n)) continue;
+        int drv = nl_->net_driver[n];
+        for (int64_t c2 = nl_->net_sink_ptr[n]; c2 < nl_->net_sink_ptr[n + 1]; ++c2)
+          conn_delay_[c2] = conn_delay(drv, nl_->net_sinks[c2]);
+      }
+    }
+    for (int n : marked2_) net_mark2_[n] = 0;
+    marked2_.clear();
+  }
+
+  struct MacroMove { int b, x, y; };
+  std::vector<MacroMove> mm_moves_, mm_save_;
 
   float net_cost_from_scratch(int n) {
     Bb& b = bbs_[n];
@@ -406,6 +648,9 @@ class SerialPlacer {
     if (!fixed_.empty() && fixed_[blk]) return 0;
     int x0 = bx_[blk], y0 = by_[blk];
     if (rx0_ >= 0 && (x0 < rx0_ || x0 > rx1_)) return 0;  // not my shard
+    if (!macro_of_.empty() && macro_of_[blk] >= 0)
+      return try_macro_move(macro_of_[blk], T, rlim, timing_tradeoff,
+                            bb_norm, td_norm);
     int btype = nl_->block_type[blk];
     ++att_valid_;
     // find_to: range-limited destination of matching type (place.c:1520)
@@ -451,6 +696,8 @@ class SerialPlacer {
     int other = grid_at(x1, y1, slot1);
     if (other == blk) return 0;
     if (other >= 0 && !fixed_.empty() && fixed_[other]) return 0;
+    if (other >= 0 && !macro_of_.empty() && macro_of_[other] >= 0)
+      return 0;   // don't break a macro by swapping with its member
 
     // save + compute delta over affected nets
     saved_bbs_.clear(); saved_costs_.clear(); saved_ids_.clear();

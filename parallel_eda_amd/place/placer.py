@@ -46,13 +46,17 @@ def analytic_delay_matrix(arch: ArchDef) -> np.ndarray:
 def anneal_place(netlist, arch: ArchDef, seed: int = 7, timing_tradeoff: float = 0.5,
                  inner_num: float = 1.0, sta=None, crit_exp: float = 1.0,
                  verbose: bool = False, engine: str = "cpu",
-                 delay_matrix: str = "analytic", fixed=None) -> Placement:
+                 delay_matrix: str = "analytic", fixed=None,
+                 macros=None) -> Placement:
     """Run the full SA schedule; returns final Placement.
 
     sta: optional TimingGraph wrapper (timing.sta.STA) for criticality
     refresh each temperature; None => pure bounding-box placement.
     fixed: optional (ids, x, y, slot) arrays pinning blocks to locations
     (reference: -pad_loc_file / fix_pins); pinned blocks never move.
+    macros: optional list of [(blk, dx, dy), ...] groups placed and moved
+    atomically at fixed relative offsets (reference: place_macro.c carry
+    chains); the first member is the head at (0, 0). CPU engine only.
     """
     if engine == "gpu":
         from .gpu_placer import anneal_place_gpu
@@ -74,6 +78,19 @@ def anneal_place(netlist, arch: ArchDef, seed: int = 7, timing_tradeoff: float =
           else np.empty(0, dtype=np.int8))
     placer = cpu.SerialPlacer(netlist.cpp(), arch.nx, arch.ny, arch.io_cap,
                               np.ascontiguousarray(dm.ravel()), seed, tb)
+    if macros:
+        if engine != "cpu":
+            raise NotImplementedError("macros: CPU engine only (round 2: GPU)")
+        ptr = [0]
+        mb, mdx, mdy = [], [], []
+        for grp in macros:
+            for (b, dx, dy) in grp:
+                mb.append(b); mdx.append(dx); mdy.append(dy)
+            ptr.append(len(mb))
+        placer.set_macros(np.asarray(ptr, dtype=np.int64),
+                          np.asarray(mb, dtype=np.int32),
+                          np.asarray(mdx, dtype=np.int32),
+                          np.asarray(mdy, dtype=np.int32))
     if fixed is not None:
         ids, fx, fy, fs = fixed
         placer.fix_blocks(np.asarray(ids, dtype=np.int32),

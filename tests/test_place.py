@@ -184,3 +184,66 @@ def test_gpu_placer_initial_placement_host_logic():
     bt = np.asarray(nl_h.block_type)
     for b in range(nl_h.num_blocks):
         assert tb[hx[b] * gh.gy + hy[b]] == bt[b], b
+
+
+def test_placement_macros():
+    """Carry-chain macros (reference: place_macro.c + find_affected_blocks
+    place.c:1192): members hold fixed relative offsets through the whole
+    anneal, moves are atomic, and the placement stays legal."""
+    import numpy as np
+    from parallel_eda_amd.arch.archdef import get_arch
+    from parallel_eda_amd.io.synth import synth_netlist, spec_for_arch
+    from parallel_eda_amd.place.placer import anneal_place
+    arch = get_arch("tseng")
+    nl = synth_netlist(spec_for_arch(arch, fill=0.5, seed=8))
+    clbs = np.nonzero(np.asarray(nl.block_type) == 1)[0]
+    # two vertical 3-chains + one horizontal pair
+    macros = [
+        [(int(clbs[0]), 0, 0), (int(clbs[1]), 0, 1), (int(clbs[2]), 0, 2)],
+        [(int(clbs[3]), 0, 0), (int(clbs[4]), 0, 1), (int(clbs[5]), 0, 2)],
+        [(int(clbs[6]), 0, 0), (int(clbs[7]), 1, 0)],
+    ]
+    pl = anneal_place(nl, arch, seed=8, timing_tradeoff=0.0, macros=macros)
+    for grp in macros:
+        hb, _, _ = grp[0]
+        for (b, dx, dy) in grp:
+            assert pl.x[b] == pl.x[hb] + dx, (b, grp)
+            assert pl.y[b] == pl.y[hb] + dy, (b, grp)
+    # no cell double-occupied (logic area, slot 0)
+    cells = set()
+    bt = np.asarray(nl.block_type)
+    for b in range(nl.num_blocks):
+        if bt[b] != 0:
+            key = (int(pl.x[b]), int(pl.y[b]))
+            assert key not in cells
+            cells.add(key)
+    # quality sanity: still in family with the unconstrained anneal
+    free = anneal_place(nl, arch, seed=8, timing_tradeoff=0.0)
+    assert pl.bb_cost <= free.bb_cost * 1.3
+
+
+def test_placement_macros_timing():
+    """Macros compose with timing-driven mode and heterogeneous fabrics."""
+    import numpy as np
+    from parallel_eda_amd.arch.archdef import get_arch
+    from parallel_eda_amd.io.synth import synth_netlist, spec_for_arch
+    from parallel_eda_amd.place.placer import anneal_place
+    from parallel_eda_amd.timing.sta import STA
+    arch = get_arch("tiny_het")
+    nl = synth_netlist(spec_for_arch(arch, fill=0.5, seed=5))
+    bt = np.asarray(nl.block_type)
+    clbs = np.nonzero(bt == 1)[0]
+    rams = np.nonzero(bt == 2)[0]
+    macros = [[(int(clbs[0]), 0, 0), (int(clbs[1]), 0, 1)]]
+    if len(rams) >= 2:
+        macros.append([(int(rams[0]), 0, 0), (int(rams[1]), 0, 1)])
+    sta = STA(nl, arch)
+    pl = anneal_place(nl, arch, seed=5, timing_tradeoff=0.5, sta=sta,
+                      macros=macros)
+    tb = arch.tile_btype_grid()
+    gy = arch.ny + 2
+    for grp in macros:
+        hb = grp[0][0]
+        for (b, dx, dy) in grp:
+            assert pl.x[b] == pl.x[hb] + dx and pl.y[b] == pl.y[hb] + dy
+            assert tb[pl.x[b] * gy + pl.y[b]] == bt[b]
