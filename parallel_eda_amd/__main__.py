@@ -179,12 +179,20 @@ def main(argv=None):
             from .io.place_file import read_pad_loc
             fixed = read_pad_loc(args.pad_loc_file, netlist)
             print(f"pinned {len(fixed[0])} blocks from {args.pad_loc_file}")
+        macros = getattr(netlist, "macros", None) or None
+        if macros and args.engine != "cpu":
+            print(f"note: {len(macros)} carry chains ignored on the "
+                  f"{args.engine} engine (macros: CPU placer only)")
+            macros = None
+        elif macros:
+            print(f"carry chains: {len(macros)} macros "
+                  f"({sum(len(m) for m in macros)} blocks)")
         placement = anneal_place(
             netlist, arch, seed=args.seed,
             timing_tradeoff=args.timing_tradeoff if timing else 0.0,
             inner_num=args.inner_num, sta=sta, verbose=args.verbose,
             engine=args.engine, delay_matrix=args.delay_matrix, fixed=fixed,
-            crit_exp=args.criticality_exp)
+            crit_exp=args.criticality_exp, macros=macros)
         print(f"placement: bb_cost={placement.bb_cost:.1f} "
               f"({time.perf_counter()-t0:.2f}s)")
     if args.out_net:

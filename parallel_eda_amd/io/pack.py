@@ -200,7 +200,53 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
                    names=names)
     nl.block_clock = block_clock
     nl.clock_names = clock_names
+    # carry-chain macros (reference: place_macro.c follows the carry
+    # links between chain primitives): adder/DSP hard blocks whose cout
+    # actual feeds another's cin form a vertical macro, placed and moved
+    # atomically by the placer.
+    nl.macros = _detect_carry_chains(prims, hard_prims, blk_of_hard)
     return nl, cluster_of, names
+
+
+def _detect_carry_chains(prims, hard_prims, blk_of_hard):
+    cin_of = {}    # cin actual signal -> hard prim index
+    cout_of = {}   # hard prim index -> cout actual signal
+    for i in hard_prims:
+        pp = prims[i]
+        if subckt_class(pp.model) != "dsp":
+            continue
+        # blif.py stores actuals only; carry linkage = an output of one
+        # instance used as an input of another DSP instance
+        cout_of[i] = set(pp.outputs)
+    users = {}
+    for i in hard_prims:
+        pp = prims[i]
+        if subckt_class(pp.model) != "dsp":
+            continue
+        for a in pp.inputs:
+            users.setdefault(a, []).append(i)
+    nxt = {}
+    prev = {}
+    for i, outs in cout_of.items():
+        for a in outs:
+            for j in users.get(a, []):
+                if j != i and i not in nxt and j not in prev:
+                    nxt[i] = j
+                    prev[j] = i
+                    break
+            if i in nxt:
+                break
+    macros = []
+    for i in cout_of:
+        if i in prev:
+            continue  # not a chain head
+        chain = [i]
+        while chain[-1] in nxt:
+            chain.append(nxt[chain[-1]])
+        if len(chain) >= 2:
+            macros.append([(blk_of_hard[c], 0, k)
+                           for k, c in enumerate(chain)])
+    return macros
 
 
 def _break_comb_cycles(block_is_seq, drivers, sink_ptr, net_sinks, nb):

@@ -59,6 +59,7 @@ class NetlistPy:
         self.names = names
         self.block_clock = None   # optional per-block clock domain ids
         self.clock_names = []     # domain id -> clock name
+        self.macros = []          # optional carry-chain macro groups
         self._cpp = None
 
     @property

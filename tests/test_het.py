@@ -313,3 +313,40 @@ def test_cli_het_flow_with_sdf(tmp_path):
     assert rc == 0
     assert "ram_seq" in v.read_text()
     assert '(CELLTYPE "ram")' in s.read_text()
+
+
+CHAIN_BLIF = """
+.model chain
+.inputs a b c d clk
+.outputs y
+.names a b n1
+11 1
+.subckt adder a0=n1 b0=b cin=c sumout=s0 cout=c0
+.subckt adder a0=s0 b0=d cin=c0 sumout=s1 cout=c1
+.subckt adder a0=s1 b0=a cin=c1 sumout=s2 cout=c2
+.names s2 y
+1 1
+.end
+"""
+
+
+def test_carry_chain_macros_from_blif(het_arch):
+    """Carry chains inferred from .subckt adder cout->cin links become
+    placement macros (reference: place_macro.c) and survive the anneal
+    as vertical runs."""
+    from parallel_eda_amd.io.blif import parse_blif
+    from parallel_eda_amd.io.pack import pack_blif
+    nl, _, _ = pack_blif(parse_blif(CHAIN_BLIF), het_arch, n_ble=4)
+    assert len(nl.macros) == 1
+    assert [(dx, dy) for (_b, dx, dy) in nl.macros[0]] == \
+        [(0, 0), (0, 1), (0, 2)]
+    pl = anneal_place(nl, het_arch, seed=4, timing_tradeoff=0.0,
+                      macros=nl.macros)
+    hb = nl.macros[0][0][0]
+    for (b, dx, dy) in nl.macros[0]:
+        assert pl.x[b] == pl.x[hb] + dx and pl.y[b] == pl.y[hb] + dy
+    # chain members are DSP blocks on DSP tiles
+    tb = het_arch.tile_btype_grid()
+    gy = het_arch.ny + 2
+    for (b, _dx, _dy) in nl.macros[0]:
+        assert tb[pl.x[b] * gy + pl.y[b]] == BLK_DSP
