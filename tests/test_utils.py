@@ -246,3 +246,22 @@ def test_svg_rendering(tmp_path):
     assert a.startswith("<svg") and a.rstrip().endswith("</svg>")
     assert b.count("<line") > 10           # wires drawn
     assert a.count("<rect") >= nl.num_blocks
+
+
+def test_hip_build_artifacts_present():
+    """Both gfx950 kernel libraries (release + debug-bounds) must be
+    built in-tree — the GPU box receives them via the repo snapshot, and
+    ops.hip() refuses to run without them (no silent CPU fallback)."""
+    from parallel_eda_amd import ops
+    import parallel_eda_amd as pkg
+    from pathlib import Path
+    root = Path(pkg.__file__).parent
+    assert (root / "libpnr_hip.so").exists()
+    assert (root / "libpnr_hip_dbg.so").exists()
+    assert ops.hip_lib_path().name == "libpnr_hip.so"
+    import os
+    os.environ["PNR_HIP_DEBUG"] = "1"
+    try:
+        assert ops.hip_lib_path().name == "libpnr_hip_dbg.so"
+    finally:
+        del os.environ["PNR_HIP_DEBUG"]
