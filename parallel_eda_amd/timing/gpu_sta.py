@@ -31,6 +31,16 @@ class StaLaunchArgs(ct.Structure):
     ]
 
 
+class StaDomainsArgs(ct.Structure):
+    _fields_ = [
+        ("base", StaLaunchArgs),
+        ("block_clock", ct.c_void_p), ("periods", ct.c_void_p),
+        ("K", ct.c_int32),
+        ("arr_all", ct.c_void_p), ("req_all", ct.c_void_p),
+        ("worst_bits", ct.c_void_p),
+    ]
+
+
 def _lib():
     lib = hip_api.lib()
     if not hasattr(lib, "_sta_ready"):
@@ -39,6 +49,12 @@ def _lib():
         lib.pnr_sta_args_sizeof.restype = ct.c_int64
         if lib.pnr_sta_args_sizeof() != ct.sizeof(StaLaunchArgs):
             raise RuntimeError("StaLaunchArgs ABI mismatch")
+        lib.pnr_sta_analyze_domains.restype = ct.c_int
+        lib.pnr_sta_analyze_domains.argtypes = [ct.POINTER(StaDomainsArgs),
+                                                ct.c_void_p]
+        lib.pnr_sta_domains_args_sizeof.restype = ct.c_int64
+        if lib.pnr_sta_domains_args_sizeof() != ct.sizeof(StaDomainsArgs):
+            raise RuntimeError("StaDomainsArgs ABI mismatch")
         lib._sta_ready = True
     return lib
 
@@ -118,6 +134,69 @@ class GpuSTA:
         rc = self.lib.pnr_sta_analyze(ct.byref(a), stream)
         hip_api.check(rc, "sta_analyze")
         return self.t_cpd, self.t_slack, self.t_crit
+
+    def analyze_domains(self, conn_delay, block_clock, periods):
+        """Multi-clock analysis on the device (reference:
+        do_timing_analysis_new's (src,sink)-domain-pair loop;
+        EXPERIMENTAL — pending round-2 GPU validation against the CPU
+        TimingGraph.analyze_domains). Returns (worst_achieved_period,
+        slack[], crit[]); the per-conn slack/crit arrays are exact, the
+        worst-period scalar is tracked with a benign race and should be
+        treated as advisory on the GPU path."""
+        torch = self.torch
+        K = len(periods)
+        nb = self.netlist.num_blocks
+        t_delay = torch.from_numpy(
+            np.ascontiguousarray(conn_delay, dtype=np.float32)).to(self.device)
+        t_bc = torch.from_numpy(
+            np.ascontiguousarray(block_clock, dtype=np.int32)).to(self.device)
+        t_per = torch.from_numpy(
+            np.ascontiguousarray(periods, dtype=np.float32)).to(self.device)
+        t_arr_all = torch.zeros(K * nb, dtype=torch.float32,
+                                device=self.device)
+        t_req_all = torch.zeros(K * nb, dtype=torch.float32,
+                                device=self.device)
+        t_worst = torch.zeros(2, dtype=torch.int32, device=self.device)
+        d = StaDomainsArgs()
+        d.base = self._base_args(t_delay)
+        pt = lambda t: ct.c_void_p(t.data_ptr())
+        d.block_clock = pt(t_bc); d.periods = pt(t_per); d.K = K
+        d.arr_all = pt(t_arr_all); d.req_all = pt(t_req_all)
+        d.worst_bits = pt(t_worst)
+        stream = self.torch.cuda.current_stream().cuda_stream
+        rc = self.lib.pnr_sta_analyze_domains(ct.byref(d), stream)
+        hip_api.check(rc, "sta_analyze_domains")
+        torch.cuda.synchronize(self.device)
+        worst = float(np.frombuffer(
+            t_worst[1:2].cpu().numpy().tobytes(), dtype=np.float32)[0])
+        if worst == 0.0:
+            worst = float(max(periods))
+        return (worst, self.t_slack.cpu().numpy(),
+                self.t_crit.cpu().numpy())
+
+    def _base_args(self, t_conn_delay):
+        a = StaLaunchArgs()
+        pt = lambda t: ct.c_void_p(t.data_ptr())
+        a.level_blocks = pt(self.t_level_blocks)
+        a.level_start = pt(self.t_level_start)
+        a.in_ptr = pt(self.t_in_ptr); a.in_conn = pt(self.t_in_conn)
+        a.out_ptr = pt(self.t_out_ptr); a.out_conn = pt(self.t_out_conn)
+        a.conn_driver = pt(self.t_conn_driver)
+        a.conn_sink = pt(self.t_conn_sink)
+        a.is_seq = pt(self.t_is_seq)
+        a.blk_delay = (pt(self.t_blk_delay)
+                       if self.t_blk_delay is not None else None)
+        a.T_clb = self.arch.T_clb; a.T_seq_out = self.arch.T_seq_out
+        a.T_seq_in = self.arch.T_seq_in; a.max_crit = self.max_crit
+        a.num_blocks = self.netlist.num_blocks
+        a.num_levels = self.num_levels
+        a.num_conns = self.netlist.num_conns
+        a.t_arr = pt(self.t_arr); a.t_req = pt(self.t_req)
+        a.cpd_out = pt(self.t_cpd)
+        a.delay = pt(t_conn_delay); a.slack = pt(self.t_slack)
+        a.crit = pt(self.t_crit)
+        a.level_start_host = self.level_start_host.ctypes.data_as(ct.c_void_p)
+        return a
 
     def analyze(self, conn_delay):
         """Host-compatible API (numpy in/out), mirroring timing.sta.STA

@@ -32,6 +32,27 @@ run("btc-calendar", "python bench.py --config bitcoin_miner --steps 2 "
     "--warmup 2 --verbose", env={"PNR_CALENDAR": "1"}, timeout=900)
 run("partial-rip-ab", "python tools/gpu_partial_ab.py tseng 0.6 && "
     "python tools/gpu_partial_ab.py LU32PEEng 0.55", timeout=700)
+run("sta-domains-gpu", "python - <<'P'\n"
+    "import sys; sys.path.insert(0, '.')\n"
+    "import numpy as np\n"
+    "from parallel_eda_amd.arch.archdef import get_arch\n"
+    "from parallel_eda_amd.io.synth import synth_netlist, spec_for_arch\n"
+    "from parallel_eda_amd.timing.sta import STA\n"
+    "from parallel_eda_amd.timing.gpu_sta import GpuSTA\n"
+    "arch = get_arch('tseng')\n"
+    "nl = synth_netlist(spec_for_arch(arch, fill=0.5, seed=3))\n"
+    "rng = np.random.default_rng(3)\n"
+    "bc = np.where(np.asarray(nl.block_is_seq) > 0,\n"
+    "              rng.integers(0, 2, nl.num_blocks), -1).astype(np.int32)\n"
+    "per = np.asarray([5e-9, 8e-9], dtype=np.float32)\n"
+    "dly = rng.uniform(0.1e-9, 2e-9, nl.num_conns).astype(np.float32)\n"
+    "wp_c, sl_c, cr_c = STA(nl, arch).analyze_domains(dly, bc, per)\n"
+    "g = GpuSTA(nl, arch)\n"
+    "wp_g, sl_g, cr_g = g.analyze_domains(dly, bc, per)\n"
+    "print('slack match:', np.allclose(sl_c, sl_g, rtol=1e-4, atol=1e-12))\n"
+    "print('crit match:', np.allclose(np.minimum(cr_c, 0.99), cr_g, rtol=1e-4))\n"
+    "assert np.allclose(sl_c, sl_g, rtol=1e-4, atol=1e-12)\n"
+    "P", timeout=300)
 run("flow-bench", "python tools/bench_flow.py LU32PEEng > "
     "gpurun_out/flow_lu32.json && cat gpurun_out/flow_lu32.json",
     timeout=900)
