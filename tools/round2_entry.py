@@ -1,19 +1,27 @@
-"""Round-2 opening GPU checklist — run as ONE gpurun call:
+"""Round-2 opening GPU checklist, STAGED to the 90-GPU-minute budget:
 
-  gpurun --timeout 1800 -- 'python tools/round2_entry.py > gpurun_out/r2entry.log 2>&1'
+  # core (~30-40 min incl. box setup): regressions + the two biggest A/Bs
+  gpurun --timeout 2100 -- 'python tools/round2_entry.py core > gpurun_out/r2a.log 2>&1'
+  # full tail (separate call, after reading r2a): bench + placer scale
+  gpurun --timeout 2400 -- 'python tools/round2_entry.py full > gpurun_out/r2b.log 2>&1'
 
-Covers, in order of information value per GPU-minute:
-  1. full GPU test suite (regressions since round 1)
-  2. calendar-vs-pingpong A/B: quality at tseng, perf at LU32
-  3. bitcoin bench with the winning frontier
-  4. placer anneal at bgm scale (GPU engine quality vs CPU trend)
+Order of information value per GPU-minute:
+  core: gpu test suite -> calendar A/B (tseng quality, LU32 perf)
+        -> partial-rip A/B (tseng) -> multi-domain STA -> het placer
+  full: bitcoin bench w/ winning frontier -> LU32 partial A/B
+        -> LU32 flow bench -> bgm-scale GPU anneal
+Every stage keeps its own subprocess timeout so one hang can't eat the
+whole call.
 """
 import os, subprocess, sys, time
 from pathlib import Path
 ROOT = Path(__file__).resolve().parent.parent
 os.chdir(ROOT)
+STAGE = sys.argv[1] if len(sys.argv) > 1 else "core"
 
-def run(name, cmd, env=None, timeout=600):
+def run(name, cmd, env=None, timeout=600, stage="core"):
+    if STAGE != stage and STAGE != "all":
+        return 0
     print(f"\n===== {name}: {cmd}", flush=True)
     e = dict(os.environ); e.update(env or {})
     t0 = time.time()
@@ -29,9 +37,12 @@ run("tseng-quality-calendar",
     "python -m pytest tests/test_gpu_router.py::test_gpu_route_matches_cpu_quality -q",
     env={"PNR_CALENDAR": "1"}, timeout=300)
 run("btc-calendar", "python bench.py --config bitcoin_miner --steps 2 "
-    "--warmup 2 --verbose", env={"PNR_CALENDAR": "1"}, timeout=900)
-run("partial-rip-ab", "python tools/gpu_partial_ab.py tseng 0.6 && "
-    "python tools/gpu_partial_ab.py LU32PEEng 0.55", timeout=700)
+    "--warmup 2 --verbose", env={"PNR_CALENDAR": "1"}, timeout=900,
+    stage="full")
+run("partial-rip-tseng", "python tools/gpu_partial_ab.py tseng 0.6",
+    timeout=300)
+run("partial-rip-lu32", "python tools/gpu_partial_ab.py LU32PEEng 0.55",
+    timeout=600, stage="full")
 run("sta-domains-gpu", "python - <<'P'\n"
     "import sys; sys.path.insert(0, '.')\n"
     "import numpy as np\n"
@@ -55,7 +66,7 @@ run("sta-domains-gpu", "python - <<'P'\n"
     "P", timeout=300)
 run("flow-bench", "python tools/bench_flow.py LU32PEEng > "
     "gpurun_out/flow_lu32.json && cat gpurun_out/flow_lu32.json",
-    timeout=900)
+    timeout=900, stage="full")
 run("het-gpu", "python - <<'P'\n"
     "import sys; sys.path.insert(0, '.')\n"
     "import numpy as np\n"
@@ -88,4 +99,4 @@ run("placer-bgm", "python - <<'P'\n"
     "pl = anneal_place_gpu(nl, arch, seed=7, timing_tradeoff=0.0)\n"
     "print(f'bgm GPU anneal: bb={pl.bb_cost:.0f} temps={pl.stats[\"temps\"]} "
     "t={time.time()-t0:.0f}s')\n"
-    "P", timeout=900)
+    "P", timeout=900, stage="full")
