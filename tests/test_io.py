@@ -232,3 +232,37 @@ def test_fixed_layout_xml():
     b = parse_arch_xml('<architecture><layout auto="1.0"/></architecture>')
     size_grid_for_netlist(nl, b)
     assert b.nx >= 2                    # auto-sized
+
+
+def test_cli_kitchen_sink(tmp_path):
+    """Integration smoke: the widest CLI surface in one flow — synth het
+    arch, settings file, pad pinning, criticality knobs, incremental
+    routing, and every output artifact at once (catches option
+    interaction regressions)."""
+    from parallel_eda_amd.__main__ import main
+    cfg = tmp_path / "s.toml"
+    cfg.write_text("fill = 0.5\nmax_router_iterations = 60\n")
+    out = {k: tmp_path / k for k in
+           ("o.net", "o.place", "o.route", "o.v", "o.sdf", "p.rpt",
+            "pl.svg", "rt.svg", "t.rpt")}
+    stats = tmp_path / "stats"
+    rc = main(["--synth", "tiny_het", "--seed", "5",
+               "--settings", str(cfg),
+               "--criticality_exp", "1.0", "--max_criticality", "0.98",
+               "--route_incremental",
+               "--out_net", str(out["o.net"]),
+               "--out_place", str(out["o.place"]),
+               "--out_route", str(out["o.route"]),
+               "--out_verilog", str(out["o.v"]),
+               "--out_sdf", str(out["o.sdf"]),
+               "--power_report", str(out["p.rpt"]),
+               "--draw_place", str(out["pl.svg"]),
+               "--draw_route", str(out["rt.svg"]),
+               "--timing_report", str(out["t.rpt"]),
+               "--stats_dir", str(stats), "--echo_routes"])
+    assert rc == 0
+    for k, f in out.items():
+        assert f.exists() and f.stat().st_size > 0, k
+    assert (stats / "iter_stats.txt").exists()
+    assert any(p.name.startswith("routes_iter_")
+               for p in stats.iterdir())
