@@ -17,6 +17,46 @@ def _torch():
     return torch
 
 
+def xcd_interleaved_order(bb, order, n_xcd=8):
+    """Reorder a concurrent-launch queue so each XCD's workgroups pull
+    spatially-clustered nets. MI355X dispatches consecutive workgroups
+    round-robin across the 8 XCDs (each with a private L2); with the
+    atomic-cursor queue, consecutive pops therefore land on DIFFERENT
+    XCDs. Sorting nets by bb-center Morton code clusters them spatially,
+    and an 8-way deal (net i -> position (i%%8)*ceil(n/8) + i//8) maps
+    each spatial cluster onto ONE XCD's pop sequence — so an XCD's L2
+    sees one region of the rr graph instead of all of them. Flag-gated
+    (PNR_XCD_ORDER=1) pending a round-2 rocprofv3 L2-hit A/B; a pure
+    permutation, so results are unchanged in concurrent mode either way
+    (same net set, same atomics)."""
+    cx = (bb[order, 0].astype(np.int64) + bb[order, 2])
+    cy = (bb[order, 1].astype(np.int64) + bb[order, 3])
+
+    def spread(v):
+        v = (v | (v << 8)) & 0x00FF00FF
+        v = (v | (v << 4)) & 0x0F0F0F0F
+        v = (v | (v << 2)) & 0x33333333
+        v = (v | (v << 1)) & 0x55555555
+        return v
+
+    morton = spread(cx & 0xFFFF) | (spread(cy & 0xFFFF) << 1)
+    spatial = order[np.argsort(morton, kind="stable")]
+    n = len(spatial)
+    chunk = (n + n_xcd - 1) // n_xcd
+    # gather: queue position p (popped by XCD p % n_xcd) takes from
+    # spatial chunk (p % n_xcd) — so each XCD walks one Morton run
+    pos = np.arange(n)
+    src = (pos % n_xcd) * chunk + pos // n_xcd
+    ok = src < n
+    out = np.empty(n, dtype=spatial.dtype)
+    out[pos[ok]] = spatial[src[ok]]
+    if not ok.all():
+        used = np.zeros(n, dtype=bool)
+        used[src[ok]] = True
+        out[pos[~ok]] = spatial[~used]
+    return out
+
+
 def schedule_bb_waves(bb, net_ids, areas, nx, ny, cell=8):
     """Greedy bb-disjoint wave schedule (ParaDRo-style: reference builds a
     bb-overlap graph and colors it, partitioning_multi_sink...:3598; we
@@ -322,7 +362,10 @@ class GpuRouter:
                 # one concurrent launch; biggest work first for load balance
                 areas_t = self._bb_areas(self.bb)[todo]
                 nsk = (self.sink_ptr[todo + 1] - self.sink_ptr[todo]).astype(np.int64)
-                waves = [todo[np.argsort(-(areas_t * nsk), kind="stable")]]
+                order = todo[np.argsort(-(areas_t * nsk), kind="stable")]
+                if _os.environ.get("PNR_XCD_ORDER"):
+                    order = xcd_interleaved_order(self.bb, order)
+                waves = [order]
             else:
                 waves = self._schedule_waves(todo)
             areas = self._bb_areas(self.bb)
