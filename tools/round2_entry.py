@@ -64,6 +64,8 @@ run("sta-domains-gpu", "python - <<'P'\n"
     "print('crit match:', np.allclose(np.minimum(cr_c, 0.99), cr_g, rtol=1e-4))\n"
     "assert np.allclose(sl_c, sl_g, rtol=1e-4, atol=1e-12)\n"
     "P", timeout=300)
+run("xcd-order-ab", "python tools/gpu_sweep_one.py 1.2 3.0 6",
+    env={"PNR_XCD_ORDER": "1"}, timeout=400, stage="full")
 run("flow-bench", "python tools/bench_flow.py LU32PEEng > "
     "gpurun_out/flow_lu32.json && cat gpurun_out/flow_lu32.json",
     timeout=900, stage="full")
