@@ -429,6 +429,14 @@ class SerialPlacer {
     int dy = (int)(rng_() % (2 * irlim + 1)) - irlim;
     if (dx == 0 && dy == 0) return 0;
     int64_t k0 = macro_ptr_[m], k1 = macro_ptr_[m + 1];
+    // strip-sharded mode: the WHOLE macro must live inside this rank's
+    // region — a boundary-straddling macro would mutate another rank's
+    // cells and break the conflict-free fusion (parallel/dist_place.py)
+    if (rx0_ >= 0)
+      for (int64_t k = k0; k < k1; ++k) {
+        int xb = bx_[macro_blk_[k]];
+        if (xb < rx0_ || xb > rx1_) return 0;
+      }
     // validate all targets
     mm_moves_.clear();
     for (int64_t k = k0; k < k1; ++k) {
