@@ -37,8 +37,13 @@ def _strips(nx, ws, phase):
 
 def anneal_place_dist(netlist, arch, rank=0, world_size=1, seed=7,
                       timing_tradeoff=0.0, sta=None, inner_num=1.0,
-                      crit_exp=1.0, verbose=False):
-    """Distributed SA anneal. Returns the (rank-identical) Placement."""
+                      crit_exp=1.0, verbose=False, macros=None):
+    """Distributed SA anneal. Returns the (rank-identical) Placement.
+
+    macros: carry-chain groups (see anneal_place); applied BEFORE the
+    per-rank reseed so every rank starts from the identical macro-legal
+    placement, and boundary-straddling macros simply sit out the
+    temperatures whose strips split them (C++ region guard)."""
     import torch
     import torch.distributed as dist
     from ..place.placer import Placement, analytic_delay_matrix
@@ -54,8 +59,19 @@ def anneal_place_dist(netlist, arch, rank=0, world_size=1, seed=7,
           else np.empty(0, dtype=np.int8))
     placer = cpu.SerialPlacer(netlist.cpp(), arch.nx, arch.ny, arch.io_cap,
                               np.ascontiguousarray(dm.ravel()), seed, tb)
-    # identical initial placement everywhere (same seed), then diverge
-    # the move streams per rank
+    if macros:
+        ptr = [0]
+        mb, mdx, mdy = [], [], []
+        for grp in macros:
+            for (b, dx, dy) in grp:
+                mb.append(b); mdx.append(dx); mdy.append(dy)
+            ptr.append(len(mb))
+        placer.set_macros(np.asarray(ptr, dtype=np.int64),
+                          np.asarray(mb, dtype=np.int32),
+                          np.asarray(mdx, dtype=np.int32),
+                          np.asarray(mdy, dtype=np.int32))
+    # identical initial placement everywhere (same seed; set_macros
+    # consumed the same draws on every rank), then diverge the streams
     placer.reseed((seed + 1) * 1_000_003 + rank)
     nb = netlist.num_blocks
     ws = world_size

@@ -412,3 +412,40 @@ def test_dist_full_flow_world2(tmp_path):
     assert r0["cpd"] == pytest.approx(r1["cpd"], rel=1e-9)
     assert np.array_equal(r0["px"], r1["px"])
     assert r0["wl"] > 0 and r0["cpd"] > 0
+
+
+def _worker_place_macros(rank, world, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+    from parallel_eda_amd.parallel.dist_place import anneal_place_dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    arch, nl, _ = _build_case()
+    clbs = np.nonzero(np.asarray(nl.block_type) == 1)[0]
+    macros = [[(int(clbs[0]), 0, 0), (int(clbs[1]), 0, 1),
+               (int(clbs[2]), 0, 2)]]
+    pl = anneal_place_dist(nl, arch, rank=rank, world_size=world, seed=9,
+                           macros=macros)
+    with open(os.path.join(tmpdir, f"pm{rank}.pkl"), "wb") as f:
+        pickle.dump({"x": np.asarray(pl.x), "y": np.asarray(pl.y),
+                     "macros": macros, "bb": pl.bb_cost}, f)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_dist_place_macros_world2(tmp_path):
+    """Carry-chain macros compose with the strip-sharded anneal: ranks
+    agree bit-identically and the chain offsets survive."""
+    mp.spawn(_worker_place_macros, args=(2, 29540, str(tmp_path)), nprocs=2,
+             join=True)
+    with open(tmp_path / "pm0.pkl", "rb") as f:
+        r0 = pickle.load(f)
+    with open(tmp_path / "pm1.pkl", "rb") as f:
+        r1 = pickle.load(f)
+    assert np.array_equal(r0["x"], r1["x"])
+    assert np.array_equal(r0["y"], r1["y"])
+    for grp in r0["macros"]:
+        hb = grp[0][0]
+        for (b, dx, dy) in grp:
+            assert r0["x"][b] == r0["x"][hb] + dx
+            assert r0["y"][b] == r0["y"][hb] + dy
