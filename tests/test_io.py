@@ -266,3 +266,74 @@ def test_cli_kitchen_sink(tmp_path):
     assert (stats / "iter_stats.txt").exists()
     assert any(p.name.startswith("routes_iter_")
                for p in stats.iterdir())
+
+
+def test_big_blif_end_to_end(tmp_path):
+    """Scale integration for the front end: ~1.6k primitives (names,
+    latches, RAM subckts, an 8-adder carry chain) through parse -> pack
+    (cluster size from the arch num_pb) -> auto grid sizing -> macro-
+    aware placement -> routing, via the CLI's positional path."""
+    import numpy as np
+    from parallel_eda_amd.__main__ import main
+    rng = np.random.default_rng(99)
+    n_in, n_out = 24, 12
+    sigs = [f"i{k}" for k in range(n_in)]
+    lines = [".model big", ".inputs " + " ".join(sigs),
+             ".outputs " + " ".join(f"o{k}" for k in range(n_out))]
+    for k in range(2000):
+        nin = int(rng.integers(2, 5))
+        pick = rng.choice(len(sigs), nin, replace=False)
+        lines += [".names " + " ".join(sigs[int(i)] for i in pick)
+                  + f" n{k}", "1" * nin + " 1"]
+        sigs.append(f"n{k}")
+    for k in range(200):
+        src = sigs[int(rng.integers(len(sigs)))]
+        lines.append(f".latch {src} q{k} re clk 0")
+        sigs.append(f"q{k}")
+    for k in range(30):
+        a = sigs[int(rng.integers(len(sigs)))]
+        b = sigs[int(rng.integers(len(sigs)))]
+        lines.append(f".subckt single_port_ram addr0={a} data0={b} "
+                     f"clk=clk out0=m{k}")
+        sigs.append(f"m{k}")
+    prev = None
+    for k in range(8):
+        a = sigs[int(rng.integers(len(sigs)))]
+        cin = f"cin=c{k-1} " if prev else ""
+        lines.append(f".subckt adder a0={a} {cin}sumout=as{k} cout=c{k}")
+        sigs.append(f"as{k}")
+        prev = f"c{k}"
+    for k in range(n_out):
+        lines += [f".names {sigs[int(rng.integers(len(sigs)))]} o{k}",
+                  "1 1"]
+    lines.append(".end")
+    xml = """<architecture><layout auto="1.0"/>
+ <switchlist><switch type="mux" name="0" R="551" Cin="0.77e-15"
+   Tdel="58e-12"/></switchlist>
+ <segmentlist><segment length="4" Rmetal="101"
+   Cmetal="22.5e-15"/></segmentlist>
+ <complexblocklist>
+  <pb_type name="io" capacity="8"><input name="o" num_pins="1"/>
+    <output name="i" num_pins="1"/></pb_type>
+  <pb_type name="clb"><input name="I" num_pins="22"/>
+    <output name="O" num_pins="8"/>
+    <pb_type name="ble" num_pb="8"><input name="in" num_pins="6"/>
+      <output name="out" num_pins="1"/></pb_type>
+    <fc default_in_type="frac" default_in_val="0.15"
+        default_out_type="frac" default_out_val="0.1"/></pb_type>
+  <pb_type name="memory"><input name="addr" num_pins="16"/>
+    <output name="out" num_pins="8"/>
+    <gridlocations><loc type="col" start="2" repeat="6"
+      priority="2"/></gridlocations></pb_type>
+  <pb_type name="mult_36"><input name="a" num_pins="16"/>
+    <output name="out" num_pins="8"/>
+    <gridlocations><loc type="col" start="4" repeat="9"
+      priority="2"/></gridlocations></pb_type>
+ </complexblocklist></architecture>"""
+    bf = tmp_path / "big.blif"
+    xf = tmp_path / "arch.xml"
+    bf.write_text("\n".join(lines))
+    xf.write_text(xml)
+    rc = main([str(bf), str(xf), "--route_chan_width", "80",
+               "--max_router_iterations", "80"])
+    assert rc == 0
