@@ -10,6 +10,11 @@ if str(ROOT) not in sys.path:
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: needs an AMD GPU (MI355X)")
+    # hang insurance for the driver's -x runs: no single test may stall
+    # the round (pytest-timeout is installed in this image)
+    if getattr(config.option, "timeout", None) in (None, 0):
+        config.option.timeout = 600
+        config.option.timeout_method = "thread"
 
 
 def _have_gpu():
