@@ -276,9 +276,11 @@ class DistRouteLoop:
         Nets with missing sinks (never routed / partial-ripped) are
         included so selective reroute can never strand a connection."""
         m = self._global_owned_mask(self.engine.congested_nets())
-        inc = self.engine.incomplete_nets()
-        if len(inc):
-            m |= self._global_owned_mask(inc)
+        # UNCONDITIONAL second union: the incomplete set is rank-LOCAL,
+        # so gating the collective on len(inc) desynchronizes ranks (one
+        # all-reduces, the other doesn't — gloo aborts with a size
+        # mismatch; found by tools/soak.py dist fuzzing, seed 90006).
+        m |= self._global_owned_mask(self.engine.incomplete_nets())
         return m
 
     def shrink_active(self, active_mask, k=1):

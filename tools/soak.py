@@ -124,15 +124,69 @@ def soak_blif(seeds):
           f"{time.time()-t0:.0f}s")
 
 
+def _dist_worker(rank, world, port, seed, tmpdir):
+    import os, pickle
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+    from tests.test_fuzz import random_arch
+    from parallel_eda_amd.io.synth import synth_netlist, spec_for_arch
+    from parallel_eda_amd.parallel.full_flow import run_flow_dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    rng = np.random.default_rng(seed)
+    arch = random_arch(rng, het_prob=0.5)
+    arch.W = max(arch.W, 8 * arch.L + 8)
+    spec = spec_for_arch(arch, fill=0.35, seed=seed)
+    res = {"skip": True}
+    if spec.n_clb >= 4:
+        nl = synth_netlist(spec)
+        if int((nl.block_type == 0).sum()) <= arch.num_io_slots():
+            out = run_flow_dist(nl, arch, rank=rank, world_size=world,
+                                seed=seed, timing_driven=True,
+                                max_route_iters=70)
+            res = {"skip": False, "success": out["success"],
+                   "wl": out["wirelength"], "cpd": out["cpd"]}
+    with open(os.path.join(tmpdir, f"r{rank}.pkl"), "wb") as f:
+        pickle.dump(res, f)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def soak_dist(seeds, world=2):
+    """Random-fabric DISTRIBUTED full flows: catches rank-divergent
+    control flow (collectives must match exactly across ranks) — found
+    the conditional-all-reduce bug at seed 90006."""
+    import tempfile, os, pickle
+    import torch.multiprocessing as mp
+    t0 = time.time()
+    ran = ok = 0
+    for i, seed in enumerate(seeds):
+        with tempfile.TemporaryDirectory() as d:
+            mp.spawn(_dist_worker, args=(world, 29700 + (i % 200), seed, d),
+                     nprocs=world, join=True)
+            rs = [pickle.load(open(os.path.join(d, f"r{r}.pkl"), "rb"))
+                  for r in range(world)]
+        if rs[0]["skip"]:
+            continue
+        ran += 1
+        for r in rs[1:]:
+            assert r == rs[0], (seed, rs)
+        ok += rs[0]["success"]
+    print(f"DIST SOAK OK: {ran} random world-{world} flows, {ok} routed, "
+          f"all rank-identical ({time.time()-t0:.0f}s)")
+
+
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("mode", choices=["flows", "big", "blif"])
+    ap.add_argument("mode", choices=["flows", "big", "blif", "dist"])
     ap.add_argument("--seeds", default="1000:1100")
     ap.add_argument("--het", type=float, default=0.4)
     args = ap.parse_args()
     seeds = seed_range(args.seeds)
     if args.mode == "blif":
         soak_blif(seeds)
+    elif args.mode == "dist":
+        soak_dist(seeds)
     else:
         soak_flows(seeds, args.het, big=args.mode == "big")
     return 0
