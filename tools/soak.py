@@ -181,12 +181,13 @@ def main():
     ap.add_argument("mode", choices=["flows", "big", "blif", "dist"])
     ap.add_argument("--seeds", default="1000:1100")
     ap.add_argument("--het", type=float, default=0.4)
+    ap.add_argument("--world", type=int, default=2)
     args = ap.parse_args()
     seeds = seed_range(args.seeds)
     if args.mode == "blif":
         soak_blif(seeds)
     elif args.mode == "dist":
-        soak_dist(seeds)
+        soak_dist(seeds, world=args.world)
     else:
         soak_flows(seeds, args.het, big=args.mode == "big")
     return 0

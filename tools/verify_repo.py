@@ -42,6 +42,8 @@ def main():
         "assert r.route.success"])
     run("soak-sample", [py, "tools/soak.py", "flows",
                         "--seeds", "9000:9030"])
+    run("dist-soak-sample", [py, "tools/soak.py", "dist",
+                             "--seeds", "9100:9106"])
     run("flow-bench-cpu", [py, "tools/bench_flow.py", "tseng",
                            "--placer", "cpu", "--router", "cpu"])
     print("verify_repo: ALL GREEN")
