@@ -250,6 +250,13 @@ def _detect_carry_chains(prims, hard_prims, blk_of_hard):
 
 
 def _break_comb_cycles(block_is_seq, drivers, sink_ptr, net_sinks, nb):
+    """Flip one member of each combinational cycle to sequential so the
+    netlist levelizes. This CHANGES timing semantics (the flipped block
+    becomes a clocked endpoint), so every conversion is reported with a
+    warning and the list of converted blocks is returned (reference VPR
+    errors out on combinational loops during timing-graph build;
+    path_delay.c levelization)."""
+    converted = []
     while True:
         indeg = np.zeros(nb, dtype=np.int64)
         adj = defaultdict(list)
@@ -273,5 +280,13 @@ def _break_comb_cycles(block_is_seq, drivers, sink_ptr, net_sinks, nb):
                     q.append(s)
         stuck = np.nonzero(~seen)[0]
         if len(stuck) == 0:
-            return
+            if converted:
+                import warnings
+                warnings.warn(
+                    f"netlist has combinational cycle(s): converted "
+                    f"block(s) {converted} to sequential endpoints to "
+                    f"break them — timing through these loops is NOT "
+                    f"analyzed as combinational", stacklevel=3)
+            return converted
+        converted.append(int(stuck[0]))
         block_is_seq[stuck[0]] = 1  # break one cycle member, re-check

@@ -160,9 +160,11 @@ class GpuEngine:
         return np.asarray(self.g.congested_nets())
 
     def incomplete_nets(self):
-        # the GPU router raises (or reports) on unroutable sinks itself;
-        # no cheap device-side missing-sink query yet (round 2)
-        return np.zeros(0, dtype=np.int64)
+        """Nets with unreached sinks, tracked by the router across
+        iterations (retry exhaustion under fail_ok, or rip-up without
+        reroute). ADVICE r1: without this the distributed flow could
+        report success with stranded connections."""
+        return self.g.incomplete_nets()
 
 
 class DistRouteLoop:
@@ -384,22 +386,26 @@ def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
             cmap.conn_delays(sd, out=conn_delay)
             cpd, _slack, c = sta.analyze(conn_delay)
             crit = cmap.sink_crit(c)
-        n_active = None
-        if overused > 0:
-            active = loop.global_congested_mask()
-            n_active = int(active.sum())
-            if not n_active:
-                active = None
-            elif loop.ws > 1 and not shrunk and n_active < shrink_threshold:
-                loop.shrink_active(active, k=1)
-                shrunk = True
+        # the active mask is congested ∪ incomplete (global, collective on
+        # every rank) — success requires BOTH overused == 0 and no net with
+        # unreached sinks (fail_ok engines can leave stranded
+        # connections that never show up as overuse)
+        active = loop.global_congested_mask()
+        n_active = int(active.sum())
+        done = overused == 0 and n_active == 0
+        if not n_active:
+            active = None
+        elif (overused > 0 and loop.ws > 1 and not shrunk and
+              n_active < shrink_threshold):
+            loop.shrink_active(active, k=1)
+            shrunk = True
         history.append(dict(iter=it, overused=int(overused),
                             active=n_active, cpd=cpd, shrunk=shrunk))
         if verbose and loop.rank == 0:
             print(f"[dist] iter {it}: overused={overused} "
                   f"active={n_active} cpd={cpd*1e9:.2f}ns shrunk={shrunk}")
-        if overused == 0:
+        if done:
             break
         pres = pres_fac_init if pres == 0.0 else pres * pres_fac_mult
-    return dict(success=overused == 0, overused=int(overused), cpd=cpd,
+    return dict(success=done, overused=int(overused), cpd=cpd,
                 iters=it, shrunk=shrunk, history=history)

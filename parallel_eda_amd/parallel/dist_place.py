@@ -159,7 +159,7 @@ def anneal_place_dist(netlist, arch, rank=0, world_size=1, seed=7,
             t *= 0.5
         elif srate > 0.8:
             t *= 0.9
-        elif srate > 0.15 and rlim > 1:
+        elif srate > 0.15 or rlim > 1:
             t *= 0.95
         else:
             t *= 0.8

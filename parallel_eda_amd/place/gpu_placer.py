@@ -441,7 +441,7 @@ def anneal_place_gpu(netlist, arch, seed=7, timing_tradeoff=0.5, inner_num=1.0,
             t *= 0.5
         elif srate > 0.8:
             t *= 0.9
-        elif srate > 0.15 and rlim > 1:
+        elif srate > 0.15 or rlim > 1:
             t *= 0.95
         else:
             t *= 0.8

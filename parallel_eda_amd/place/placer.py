@@ -147,7 +147,7 @@ def anneal_place(netlist, arch: ArchDef, seed: int = 7, timing_tradeoff: float =
             t *= 0.5
         elif srate > 0.8:
             t *= 0.9
-        elif srate > 0.15 and rlim > 1:
+        elif srate > 0.15 or rlim > 1:
             t *= 0.95
         else:
             t *= 0.8

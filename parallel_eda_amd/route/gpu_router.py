@@ -232,6 +232,11 @@ class GpuRouter:
             self.t_occ = occ
         self.t_acc = torch.ones(g.num_nodes, dtype=torch.float32, device=device)
         self.t_fail = torch.zeros(self.num_nets, dtype=torch.int32, device=device)
+        # per-net completeness: True until the net has been routed with all
+        # sinks reached (the distributed driver must never call
+        # a flow successful while a net exhausted its bb-growth retries or
+        # was ripped for ownership hand-off and never rerouted)
+        self.incomplete = np.ones(self.num_nets, dtype=bool)
         self.t_cursors = torch.zeros(2, dtype=torch.int32, device=device)
         self.t_stats = torch.zeros(8, dtype=torch.int64, device=device)
         self.t_net_scans = torch.zeros(self.num_nets, dtype=torch.int64,
@@ -323,6 +328,7 @@ class GpuRouter:
 
         todo = (np.arange(self.num_nets, dtype=np.int64) if net_subset is None
                 else np.asarray(net_subset, dtype=np.int64))
+        attempted = todo
         attempts = 0
         self.last_retries = []
         # congestion-aware lookahead: scale the heuristic's wire cost by
@@ -394,6 +400,7 @@ class GpuRouter:
             fail = self.t_fail.cpu().numpy()
             failed = np.nonzero(fail)[0]
             if len(failed) == 0:
+                self.incomplete[attempted] = False
                 break
             attempts += 1
             self.last_retries.append(
@@ -404,6 +411,8 @@ class GpuRouter:
                     print(f"router: giving up on {len(failed)} nets after "
                           f"retries (codes {np.unique(fail[failed])})",
                           file=sys.stderr, flush=True)
+                    self.incomplete[attempted] = False
+                    self.incomplete[failed] = True
                     break
                 raise RuntimeError(
                     f"router: {len(failed)} nets failed after retries "
@@ -441,9 +450,15 @@ class GpuRouter:
         return [(int(i), int(s), int(n), int(a))
                 for i, s, n, a in zip(idx, v, nsk, areas)]
 
+    def incomplete_nets(self):
+        """Nets with unreached sinks: exhausted bb-growth retries under
+        fail_ok, or ripped for ownership hand-off and not yet rerouted."""
+        return np.nonzero(self.incomplete)[0]
+
     def rip_up_nets(self, net_ids):
         """Remove the given nets' routes (ownership hand-off)."""
         torch = self.torch
+        self.incomplete[np.asarray(net_ids, dtype=np.int64)] = True
         ids = torch.from_numpy(np.asarray(net_ids, dtype=np.int32)).to(self.device)
         rc = self.lib.pnr_rip_up_nets(
             ct_ptr(self.t_tree_off), ct_ptr(self.t_tree_node),
@@ -574,6 +589,7 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
     cmap = ConnMap(conn_index, sink_ptr, netlist.num_conns, n_rsinks)
     crit = np.zeros(n_rsinks, dtype=np.float32)
     conn_delay = np.zeros(netlist.num_conns, dtype=np.float32)
+    intra_delay = float(arch.T_opin + arch.T_ipin)
     pres_fac = 0.0   # first iteration: congestion-blind (VPR style)
     cpd = 0.0
     history = []
@@ -612,7 +628,7 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
             if not router.check_occ_recount():
                 raise RuntimeError(f"occ recount mismatch at iteration {it}")
         if sta is not None:
-            cmap.conn_delays(sink_delays, out=conn_delay)
+            cmap.conn_delays(sink_delays, out=conn_delay, fill=intra_delay)
             cpd, slack, c = sta.analyze(conn_delay)
             crit = cmap.sink_crit(c, max_crit=max_criticality,
                                   crit_exp=crit_exp)
@@ -621,7 +637,7 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
         pres_fac = pres_fac_init if pres_fac == 0.0 else pres_fac * pres_fac_mult
         router.update_acc(acc_fac)
 
-    ok = overused == 0
+    ok = overused == 0 and not router.incomplete.any()
     if ok and not router.check_occ_recount():
         raise RuntimeError("GPU route: occ recount mismatch")
     return RouteResult(success=ok, iterations=it, overused=int(overused),

@@ -79,9 +79,15 @@ class ConnMap:
         self.num_conns = num_conns
         self.n_rsinks = n_rsinks
 
-    def conn_delays(self, sink_delays, out=None):
+    def conn_delays(self, sink_delays, out=None, fill=0.0):
+        """Scatter routed-sink delays to connection delays. Connections
+        with no routed sink (driver and sink share a tile) get `fill` —
+        pass the intra-tile constant (arch.T_opin + arch.T_ipin) so STA
+        sees the same delay the placer's delay model charges for
+        same-tile connections (placer.analytic_delay_matrix d[0,0])."""
         if out is None:
             out = np.zeros(self.num_conns, dtype=np.float32)
+        out[:] = fill
         out[self.conn] = sink_delays[self.rsink]
         return out
 
@@ -150,7 +156,10 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
     cmap = ConnMap(conn_index, sink_ptr, netlist.num_conns, n_rsinks)
     crit = np.zeros(n_rsinks, dtype=np.float32)
     conn_delay = np.zeros(netlist.num_conns, dtype=np.float32)
-    pres_fac = pres_fac_init
+    intra_delay = float(arch.T_opin + arch.T_ipin)
+    # iteration 1 is congestion-blind (VPR first_iter_pres_fac = 0;
+    # matches the C++ oracle's documented schedule and the GPU driver)
+    pres_fac = 0.0
     router.set_pres_fac(pres_fac)
     cpd = 0.0
     history = []
@@ -186,7 +195,8 @@ def pathfinder_route(netlist, placement, g, arch: ArchDef, sta=None,
                   f"cpd={cpd*1e9:.2f}ns")
         if sta is not None:
             # net delays -> connection delays -> STA -> criticality
-            cmap.conn_delays(router.sink_delays(), out=conn_delay)
+            cmap.conn_delays(router.sink_delays(), out=conn_delay,
+                             fill=intra_delay)
             cpd, slack, c = sta.analyze(conn_delay)
             crit = cmap.sink_crit(c, max_crit=max_criticality,
                               crit_exp=crit_exp)
