@@ -337,13 +337,21 @@ __global__ void route_nets_cal_kernel(
           continue;
         }
         // current bucket exhausted: terminate / advance / redistribute.
-        // Remaining entries all have tot >= f0 + delta (next buckets).
-        // remaining entries all have tot >= f0 + delta; strict mode must
-        // still process ties at exactly that bound
+        // Remaining CALENDAR entries all have tot >= f0 + delta (later
+        // buckets) — but the OVERFLOW region also holds bucket-full
+        // spills whose tot may be BELOW that (ov_min tracks them), so the
+        // termination bound is min(f0 + delta, ov_min). Round-1 shipped
+        // without the ov_min term and terminated past cheap spilled
+        // entries under the pres_fac=0 tie explosion: LU32 A/B measured
+        // +27% wirelength and a 20k-overused plateau. Strict mode must
+        // still process ties at exactly the bound.
+        float term_bound = sh.f0 + delta;
+        if (sh.ov_cnt[sh.ov_cur] > 0)
+          term_bound = fminf(term_bound, bits_f32(sh.ov_min));
         bool done = sh.best_sink_back != 0xffffffffu &&
                     (P.strict_term
-                         ? bits_f32(sh.best_sink_back) < sh.f0 + delta
-                         : bits_f32(sh.best_sink_back) <= sh.f0 + delta);
+                         ? bits_f32(sh.best_sink_back) < term_bound
+                         : bits_f32(sh.best_sink_back) <= term_bound);
         if (done) break;
         // any entries left in the calendar?
         bool any = false;

@@ -345,7 +345,7 @@ class DistRouteLoop:
 def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
                           pres_fac_mult=1.3, acc_fac=1.0,
                           shrink_threshold=128, verbose=False,
-                          incremental=False):
+                          incremental=False, intra_delay=0.0):
     """Distributed PathFinder outer loop (flow-level driver).
 
     Mirrors route.gpu_router.pathfinder_route_gpu's schedule — iteration 1
@@ -383,7 +383,7 @@ def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
                                       partial=incremental and
                                       mask is not None)
         if sta is not None:
-            cmap.conn_delays(sd, out=conn_delay)
+            cmap.conn_delays(sd, out=conn_delay, fill=intra_delay)
             cpd, _slack, c = sta.analyze(conn_delay)
             crit = cmap.sink_crit(c)
         # the active mask is congested ∪ incomplete (global, collective on

@@ -1,5 +1,5 @@
-"""Distributed full place-and-route flow (BASELINE config 4's shape:
-"bgm full place+route on 8 GPUs").
+"""Full place-and-route flows (BASELINE configs 4-5: full flow on 1-8
+GPUs; the bench.py headline step).
 
 Composes the two distributed drivers over one process group:
   1. anneal_place_dist — strip-sharded SA with per-temperature fusion
@@ -7,11 +7,76 @@ Composes the two distributed drivers over one process group:
      occ all-reduce, selective reroute, elastic shrink
 plus replicated STA. Every rank returns the identical result.
 
-CPU engines under gloo here (world-2/4 tested); the same drivers take
-the GPU engines over RCCL in round 2 (GpuPlacer.set_move_region and
-GpuEngine are already wired).
+run_flow_dist = CPU engines (gloo, world-2/4 tested); run_flow_gpu =
+the same drivers over the CDNA4 kernels (GpuPlacer/GpuRouter), RCCL
+collectives when world_size > 1.
 """
+import time
+
 import numpy as np
+
+
+def run_flow_gpu(netlist, arch, g, dev_graph=None, rank=0, world_size=1,
+                 device="cuda:0", seed=7, timing_driven=True,
+                 max_route_iters=80, incremental=True, verbose=False,
+                 inner_num=1.0, check=False):
+    """One complete GPU place+route flow to feasibility (the headline
+    benchmark step). The rr graph g (and optionally its device-resident
+    upload dev_graph) is the fixed device model — passed in, like the
+    FPGA itself. Everything netlist-dependent (placement anneal, net
+    terminals, route trees, routing to 0 overused, STA) happens here.
+
+    Returns dict(place, route, wirelength, cpd, success, phase_s) —
+    rank-identical."""
+    from ..timing.sta import STA
+    from ..route.router import net_rr_terminals, ConnMap
+    from ..route.gpu_router import GpuRouter, DevGraph
+    from ..place.gpu_placer import anneal_place_gpu
+    from .dist import GpuEngine, DistRouteLoop, pathfinder_route_dist
+    from .dist_place import anneal_place_dist
+
+    sta = STA(netlist, arch) if timing_driven else None
+    tt = 0.5 if timing_driven else 0.0
+
+    t0 = time.perf_counter()
+    if world_size > 1:
+        pl = anneal_place_dist(netlist, arch, rank=rank,
+                               world_size=world_size, seed=seed,
+                               timing_tradeoff=tt, sta=sta, engine="gpu",
+                               device=device, verbose=verbose,
+                               inner_num=inner_num)
+    else:
+        pl = anneal_place_gpu(netlist, arch, seed=seed, timing_tradeoff=tt,
+                              sta=sta, device=device, verbose=verbose,
+                              inner_num=inner_num)
+    t_place = time.perf_counter() - t0
+
+    t0 = time.perf_counter()
+    net_ids, src_rr, sink_ptr, sink_rr, conn_index = net_rr_terminals(
+        netlist, pl, g, arch)
+    cmap = ConnMap(conn_index, sink_ptr, netlist.num_conns, len(sink_rr))
+    if dev_graph is None:
+        dev_graph = DevGraph(g, arch, device)
+    router = GpuRouter(g, arch, src_rr, sink_ptr.astype(np.int32), sink_rr,
+                       device=device, dev_graph=dev_graph)
+    engine = GpuEngine(router)
+    loop = DistRouteLoop(engine, len(net_ids), router.bb, len(sink_rr),
+                         sink_ptr, rank=rank, world_size=world_size)
+    t_setup = time.perf_counter() - t0
+
+    t0 = time.perf_counter()
+    res = pathfinder_route_dist(
+        loop, cmap, sta, max_iters=max_route_iters, incremental=incremental,
+        verbose=verbose, intra_delay=float(arch.T_opin + arch.T_ipin))
+    t_route = time.perf_counter() - t0
+    if check and res["success"]:
+        if not router.check_occ_recount():
+            raise RuntimeError("flow: occ recount mismatch after routing")
+    wl = router.wirelength()
+    return dict(place=pl, route=res, wirelength=wl, cpd=res["cpd"],
+                success=res["success"], router=router,
+                phase_s={"place": t_place, "route_setup": t_setup,
+                         "route": t_route})
 
 
 def run_flow_dist(netlist, arch, rank=0, world_size=1, seed=7,

@@ -88,16 +88,17 @@ class GpuPlacer:
             np.arange(net_blk_ptr[i] + 1, net_blk_ptr[i + 1])
             for i in range(nn)]) if nn else np.zeros(0, dtype=np.int64)
         net_blks[sidx] = netlist.net_sinks
-        # block -> nets CSR (deduped per block)
-        blk_lists = [[] for _ in range(nb)]
-        for n in range(nn):
-            for b in set(net_blks[net_blk_ptr[n]:net_blk_ptr[n + 1]].tolist()):
-                blk_lists[b].append(n)
+        # block -> nets CSR (deduped per block), vectorized: sort (block,
+        # net) membership pairs and drop duplicates within a block
+        net_of_member = np.repeat(np.arange(nn, dtype=np.int64),
+                                  np.diff(net_blk_ptr))
+        key = net_blks.astype(np.int64) * nn + net_of_member
+        key = np.unique(key)   # sorted by (block, net), deduped
+        mb = (key // nn).astype(np.int32)
+        blk_nets = (key % nn).astype(np.int32)
         blk_net_ptr = np.zeros(nb + 1, dtype=np.int32)
-        for b in range(nb):
-            blk_net_ptr[b + 1] = blk_net_ptr[b] + len(blk_lists[b])
-        blk_nets = np.concatenate([np.asarray(l, dtype=np.int32)
-                                   for l in blk_lists]) if nb else np.zeros(0, np.int32)
+        np.add.at(blk_net_ptr, mb + 1, 1)
+        blk_net_ptr = np.cumsum(blk_net_ptr, dtype=np.int64).astype(np.int32)
         q = np.array([_cross_count(int(c)) for c in counts], dtype=np.float32)
 
         def up(a, dtype=None):

@@ -82,12 +82,55 @@ def schedule_bb_waves(bb, net_ids, areas, nx, ny, cell=8):
     return [order[wave_of == w] for w in range(n_waves)]
 
 
+class DevGraph:
+    """Device-resident SoA rr graph, shared across GpuRouter instances
+    (the graph depends only on the arch; repeated flows re-place and
+    re-route but never re-upload the 12.7M-node/78M-edge CSR)."""
+
+    def __init__(self, g, arch: ArchDef, device="cuda:0"):
+        torch = _torch()
+        if g.num_edges >= 2**31:
+            raise ValueError("edge count exceeds int32 CSR limit")
+        self.g = g
+        self.device = device
+
+        def up(a):
+            return torch.from_numpy(np.ascontiguousarray(a)).to(device)
+
+        self.t_type = up(np.asarray(g.type))
+        xlow = np.asarray(g.xlow); ylow = np.asarray(g.ylow)
+        self.t_xlow = up(xlow); self.t_ylow = up(ylow)
+        self.t_xhigh = up(np.asarray(g.xhigh))
+        self.t_yhigh = up(np.asarray(g.yhigh))
+        self.t_cap = up(np.asarray(g.capacity))
+        self.t_R = up(np.asarray(g.node_R)); self.t_C = up(np.asarray(g.node_C))
+        self.t_row_ptr = up(np.asarray(g.row_ptr).astype(np.int32))
+        self.t_edge_dst = up(np.asarray(g.edge_dst))
+        self.t_edge_sw = up(np.asarray(g.edge_sw))
+        self.t_sw_R = up(np.asarray(g.sw_R))
+        self.t_sw_Tdel = up(np.asarray(g.sw_Tdel))
+        self.t_base_cost = up(np.asarray(g.base_cost))
+
+        # dense per-tile node index (anchor = (xlow, ylow))
+        gy = arch.ny + 2
+        tile = xlow.astype(np.int64) * gy + ylow.astype(np.int64)
+        order = np.argsort(tile, kind="stable")
+        sorted_tile = tile[order]
+        starts = np.r_[0, np.nonzero(np.diff(sorted_tile))[0] + 1]
+        counts = np.diff(np.r_[starts, len(tile)])
+        idx_in_tile = np.empty(len(tile), dtype=np.int32)
+        pos = np.arange(len(tile)) - np.repeat(starts, counts)
+        idx_in_tile[order] = pos.astype(np.int32)
+        self.npt = int(counts.max())
+        self.t_idx_in_tile = up(idx_in_tile)
+
+
 class GpuRouter:
     def __init__(self, g, arch: ArchDef, src_rr, sink_ptr, sink_rr,
                  device="cuda:0", astar_fac=1.2, n_small_slots=1024,
                  n_large_slots=64, bb_margin=4, max_rounds=200000,
                  delta_fac=3.0, deterministic=False,
-                 concurrent_threshold=768, occ=None):
+                 concurrent_threshold=768, occ=None, dev_graph=None):
         torch = _torch()
         self.torch = torch
         self.device = device
@@ -104,40 +147,22 @@ class GpuRouter:
         self.deterministic = deterministic
         self.concurrent_threshold = concurrent_threshold
 
-        if g.num_edges >= 2**31:
-            raise ValueError("edge count exceeds int32 CSR limit")
+        # ---- graph (SoA, device-resident, shareable) ----
+        dg = dev_graph if dev_graph is not None else DevGraph(g, arch, device)
+        if dg.g is not g or dg.device != device:
+            raise ValueError("dev_graph built for a different graph/device")
+        self.dev_graph = dg
+        for name in ("t_type", "t_xlow", "t_ylow", "t_xhigh", "t_yhigh",
+                     "t_cap", "t_R", "t_C", "t_row_ptr", "t_edge_dst",
+                     "t_edge_sw", "t_sw_R", "t_sw_Tdel", "t_base_cost",
+                     "t_idx_in_tile", "npt"):
+            setattr(self, name, getattr(dg, name))
 
-        # ---- graph upload (SoA, untransformed) ----
         def up(a, dtype=None):
             t = torch.from_numpy(np.ascontiguousarray(a))
             if dtype is not None:
                 t = t.to(dtype)
             return t.to(device)
-
-        self.t_type = up(np.asarray(g.type))
-        xlow = np.asarray(g.xlow); ylow = np.asarray(g.ylow)
-        self.t_xlow = up(xlow); self.t_ylow = up(ylow)
-        self.t_xhigh = up(np.asarray(g.xhigh)); self.t_yhigh = up(np.asarray(g.yhigh))
-        self.t_cap = up(np.asarray(g.capacity))
-        self.t_R = up(np.asarray(g.node_R)); self.t_C = up(np.asarray(g.node_C))
-        self.t_row_ptr = up(np.asarray(g.row_ptr).astype(np.int32))
-        self.t_edge_dst = up(np.asarray(g.edge_dst))
-        self.t_edge_sw = up(np.asarray(g.edge_sw))
-        self.t_sw_R = up(np.asarray(g.sw_R)); self.t_sw_Tdel = up(np.asarray(g.sw_Tdel))
-        self.t_base_cost = up(np.asarray(g.base_cost))
-
-        # dense per-tile node index (anchor = (xlow, ylow))
-        gy = arch.ny + 2
-        tile = xlow.astype(np.int64) * gy + ylow.astype(np.int64)
-        order = np.argsort(tile, kind="stable")
-        sorted_tile = tile[order]
-        starts = np.r_[0, np.nonzero(np.diff(sorted_tile))[0] + 1]
-        counts = np.diff(np.r_[starts, len(tile)])
-        idx_in_tile = np.empty(len(tile), dtype=np.int32)
-        pos = np.arange(len(tile)) - np.repeat(starts, counts)
-        idx_in_tile[order] = pos.astype(np.int32)
-        self.npt = int(counts.max())
-        self.t_idx_in_tile = up(idx_in_tile)
 
         # ---- nets ----
         self.src_rr = np.asarray(src_rr, dtype=np.int32)
@@ -557,8 +582,12 @@ class GpuRouter:
         a.fail_flags = ct_ptr(self.t_fail)
         a.stats = ct_ptr(self.t_stats)
         a.net_scans = ct_ptr(self.t_net_scans)
-        a.use_calendar = 1 if (getattr(self, "use_calendar", False) or
-                               _os.environ.get("PNR_CALENDAR")) else 0
+        # deterministic mode always uses the ping-pong kernel: its fixed
+        # wave schedule + strict-termination semantics are the validated
+        # bit-identical path (checkpoint-resume test)
+        a.use_calendar = 1 if (not self.deterministic and
+                               (getattr(self, "use_calendar", False) or
+                                _os.environ.get("PNR_CALENDAR"))) else 0
         a.partial = 1 if getattr(self, "_partial", False) else 0
         self._args_keepalive = (t_sink_rr, t_crit, t_sink_orig, q_small,
                                 q_large)
