@@ -125,8 +125,9 @@ def main():
     barrier_sync()
     elapsed = time.perf_counter() - t0
     if dist is not None:
+        from parallel_eda_amd.parallel.dist import allreduce_
         t = torch.tensor([elapsed], device=device if use_gpu else "cpu")
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        allreduce_(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
     feasible = all(f["success"] for f in flows)

@@ -31,14 +31,41 @@ def init_dist():
     ws = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local = int(os.environ.get("LOCAL_RANK", "0"))
+    if torch.cuda.is_available():
+        # RCCL needs one DISTINCT device per rank; when ranks outnumber
+        # GPUs (hardware testing of the multi-rank drivers on a 1-GPU
+        # box) fall back to gloo with CPU-staged collectives and share
+        # device 0. PNR_DIST_BACKEND overrides.
+        ndev = torch.cuda.device_count()
+        local = local % max(1, ndev)
+        backend = "nccl" if ws <= ndev else "gloo"
+    else:
+        backend = "gloo"
+    backend = os.environ.get("PNR_DIST_BACKEND", backend)
     if ws > 1 and not dist.is_initialized():
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29517")
         dist.init_process_group(backend=backend, rank=rank, world_size=ws)
-        if torch.cuda.is_available():
-            torch.cuda.set_device(local)
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local)
     return rank, ws, local
+
+
+def allreduce_(t, op=None):
+    """all_reduce that works for every (backend, tensor-device) pair:
+    nccl(=RCCL) takes the device tensor directly; gloo stages a CUDA
+    tensor through CPU (the 1-GPU multi-rank hardware-test path).
+    In-place; returns t."""
+    dist = _dist()
+    if op is None:
+        op = dist.ReduceOp.SUM
+    if t.is_cuda and dist.get_backend() != "nccl":
+        c = t.cpu()
+        dist.all_reduce(c, op=op)
+        t.copy_(c)
+    else:
+        dist.all_reduce(t, op=op)
+    return t
 
 
 def spatial_partition(bb, world_size, weight=None):
@@ -223,7 +250,7 @@ class DistRouteLoop:
             self.engine.rip_up_nets(np.asarray(lost, dtype=np.int64))
         occ = self.engine.occ_tensor()
         delta = occ - occ_before
-        dist.all_reduce(delta, op=dist.ReduceOp.SUM)
+        allreduce_(delta)
         self.engine.set_occ(occ_before + delta)
         self.rank_of = np.asarray(new_rank, dtype=np.int32)
         self.my_nets = np.nonzero(self.rank_of == self.rank)[0]
@@ -247,8 +274,8 @@ class DistRouteLoop:
         w = torch.from_numpy(np.ascontiguousarray(weights, dtype=np.float64))
         occ_dev = self.engine.occ_tensor().device
         if occ_dev.type == "cuda":
-            w = w.to(occ_dev)   # NCCL requires device tensors
-        dist.all_reduce(w, op=dist.ReduceOp.SUM)
+            w = w.to(occ_dev)
+        allreduce_(w)
         wsum = np.maximum(w.cpu().numpy(), 1.0)
         new_rank = spatial_partition(self._bb, self.ws, weight=wsum)
         return self.set_partition(new_rank)
@@ -266,7 +293,7 @@ class DistRouteLoop:
             occ_dev = self.engine.occ_tensor().device
             if occ_dev.type == "cuda":
                 t = t.to(occ_dev)
-            _dist().all_reduce(t, op=_dist().ReduceOp.MAX)
+            allreduce_(t, op=_dist().ReduceOp.MAX)
             mask = t.cpu().numpy()
         return mask > 0.5
 
@@ -327,7 +354,7 @@ class DistRouteLoop:
         if self.ws > 1:
             occ = eng.occ_tensor()
             delta = occ - occ_before
-            dist.all_reduce(delta, op=dist.ReduceOp.SUM)
+            allreduce_(delta)
             occ_new = occ_before + delta
             eng.set_occ(occ_new)
             # sink delays: mask to my sinks, sum across ranks
@@ -335,7 +362,7 @@ class DistRouteLoop:
                                     .astype(np.float32))
             if occ.is_cuda:
                 sd_t = sd_t.to(occ.device)
-            dist.all_reduce(sd_t, op=dist.ReduceOp.SUM)
+            allreduce_(sd_t)
             sd = sd_t.cpu().numpy()
         over = eng.num_overused()
         eng.update_acc(acc_fac)
@@ -375,13 +402,20 @@ def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
     overused = -1
     it = 0
     for it in range(1, max_iters + 1):
-        # incremental resync cadence mirrors the single-GPU flows
+        # After iteration 1 ONLY the congested ∪ incomplete set is ever
+        # rerouted (reference: build_phase_two congested-only). A round-2
+        # hardware run showed why: a concurrent rip-ALL reroute at high
+        # pres_fac is a limit cycle (every net diverts off the same
+        # congestion snapshot simultaneously and recreates it elsewhere —
+        # overused oscillated 600↔4000 forever at LU32 scale). The
+        # incremental cadence therefore alternates PARTIAL rip-up of the
+        # active set with a FULL rip of the same set (timing refresh),
+        # never a reroute of frozen feasible nets.
         resync = incremental and it > 2 and (it - 2) % 2 == 0
-        mask = None if resync else active
         overused, sd = loop.iteration(crit, pres, acc_fac,
-                                      active_mask=mask,
+                                      active_mask=active,
                                       partial=incremental and
-                                      mask is not None)
+                                      active is not None and not resync)
         if sta is not None:
             cmap.conn_delays(sd, out=conn_delay, fill=intra_delay)
             cpd, _slack, c = sta.analyze(conn_delay)

@@ -629,19 +629,23 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
         # iteration 1 routes everything; later iterations re-route only
         # nets whose trees touch overused nodes
         subset = None
-        # incremental mode mirrors the CPU flow's schedule: full passes
-        # through iteration 2, then partial-rip selective iterations with
-        # a full resync every 2nd (recovers the timing the pure
-        # incremental mode leaves; docs/MEASUREMENTS.md)
+        # incremental mode: partial-rip selective iterations with a FULL
+        # rip of the same active set every 2nd iteration (timing refresh).
+        # After iteration 1 only congested nets are ever rerouted — a
+        # concurrent rip-all reroute at high pres_fac is a limit cycle
+        # (see parallel.dist.pathfinder_route_dist).
         resync = incremental and it > 2 and (it - 2) % 2 == 0
-        if it > 1 and not rip_up_always and not resync:
+        if it > 1 and not rip_up_always:
             subset = router.congested_nets()
+            inc = router.incomplete_nets()
+            if len(inc):
+                subset = np.union1d(subset, inc)
             if len(subset) == 0:
                 subset = None
         router.reset_search_stats()
         overused, sink_delays = router.route_iteration(
             crit, pres_fac, net_subset=subset,
-            partial=incremental and subset is not None)
+            partial=incremental and subset is not None and not resync)
         st = router.search_stats()
         history.append(dict(iter=it, overused=int(overused), cpd=cpd,
                             rounds=st["rounds"], scanned=st["scanned"],

@@ -57,10 +57,11 @@ def test_gpu_route_matches_cpu_quality(tseng_placed):
     res_gpu = pathfinder_route(nl, pl, g, arch, sta=sta_g, max_iters=60,
                                engine="gpu")
     assert res_cpu.success and res_gpu.success
-    # iso-quality: wirelength within 15%, critical path within 15%
-    assert res_gpu.wirelength <= res_cpu.wirelength * 1.15, (
+    # iso-quality: wirelength within 10%, critical path within 10%
+    # (tightened from 15% per VERDICT r1; the at-scale gate below is 8%)
+    assert res_gpu.wirelength <= res_cpu.wirelength * 1.10, (
         f"GPU WL {res_gpu.wirelength} vs CPU {res_cpu.wirelength}")
-    assert res_gpu.crit_path_delay <= res_cpu.crit_path_delay * 1.15, (
+    assert res_gpu.crit_path_delay <= res_cpu.crit_path_delay * 1.10, (
         f"GPU cpd {res_gpu.crit_path_delay} vs CPU {res_cpu.crit_path_delay}")
 
 
@@ -147,3 +148,30 @@ def test_gpu_rip_up_nets(tiny_placed):
     occ1 = int(r.t_occ.sum().item())
     assert 0 < occ1 < occ0
     assert int((r.t_occ < 0).sum().item()) == 0
+
+
+@pytest.mark.parametrize("config,fill,bound", [
+    ("stereovision2", 0.5, 1.08),
+    ("LU32PEEng", 0.55, 1.08),
+])
+def test_gpu_route_quality_at_scale(config, fill, bound):
+    """GPU vs CPU-oracle quality parity AT SCALE (the
+    12x12 tseng gate would pass a degraded engine). Both engines route
+    the same placed netlist to feasibility; wirelength and critical path
+    must match within 8%; the GPU result passes the occ recount
+    cross-check (reference: check_route.c:27 + recalculate_occ)."""
+    from parallel_eda_amd.io.synth import synth_placed_netlist
+    arch = get_arch(config)
+    nl, pl = synth_placed_netlist(arch, fill=fill, seed=11)
+    g = rrgraph.build_rr_graph(arch)
+    res_cpu = pathfinder_route(nl, pl, g, arch, sta=STA(nl, arch),
+                               max_iters=60, incremental=True)
+    res_gpu = pathfinder_route(nl, pl, g, arch, sta=STA(nl, arch),
+                               max_iters=60, engine="gpu")
+    assert res_cpu.success and res_gpu.success, (
+        f"cpu over={res_cpu.overused} gpu over={res_gpu.overused}")
+    assert res_gpu.router.check_occ_recount()
+    assert res_gpu.wirelength <= res_cpu.wirelength * bound, (
+        f"GPU WL {res_gpu.wirelength} vs CPU {res_cpu.wirelength}")
+    assert res_gpu.crit_path_delay <= res_cpu.crit_path_delay * bound, (
+        f"GPU cpd {res_gpu.crit_path_delay} vs CPU {res_cpu.crit_path_delay}")
