@@ -121,7 +121,8 @@ def main():
         flows.append(r)
         log(f"step {k}: success={r['success']} wl={r['wirelength']} "
             f"cpd={r['cpd']*1e9:.2f}ns iters={r['route']['iters']} "
-            f"phases={r.get('phase_s')}")
+            f"phases={r.get('phase_s')} prof={r['route'].get('prof')} "
+            f"rprof={getattr(r.get('router'), 'prof', None)}")
     barrier_sync()
     elapsed = time.perf_counter() - t0
     if dist is not None:

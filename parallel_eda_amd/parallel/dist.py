@@ -389,8 +389,10 @@ def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
     (or None for congestion-only). Returns a dict with success/overused/
     cpd/iters/history.
     """
+    import time as _time
     n_rsinks = len(loop.my_sink_mask)
     crit = np.zeros(n_rsinks, dtype=np.float32)
+    prof = dict(route=0.0, sta=0.0, mask=0.0)
     conn_delay = None
     if sta is not None:
         conn_delay = np.zeros(cmap.num_conns, dtype=np.float32)
@@ -412,20 +414,27 @@ def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
         # active set with a FULL rip of the same set (timing refresh),
         # never a reroute of frozen feasible nets.
         resync = incremental and it > 2 and (it - 2) % 2 == 0
+        _t0 = _time.perf_counter()
         overused, sd = loop.iteration(crit, pres, acc_fac,
                                       active_mask=active,
                                       partial=incremental and
                                       active is not None and not resync)
+        _t1 = _time.perf_counter()
         if sta is not None:
             cmap.conn_delays(sd, out=conn_delay, fill=intra_delay)
             cpd, _slack, c = sta.analyze(conn_delay)
             crit = cmap.sink_crit(c)
+        _t2 = _time.perf_counter()
         # the active mask is congested ∪ incomplete (global, collective on
         # every rank) — success requires BOTH overused == 0 and no net with
         # unreached sinks (fail_ok engines can leave stranded
         # connections that never show up as overuse)
         active = loop.global_congested_mask()
         n_active = int(active.sum())
+        _t3 = _time.perf_counter()
+        prof["route"] += _t1 - _t0
+        prof["sta"] += _t2 - _t1
+        prof["mask"] += _t3 - _t2
         done = overused == 0 and n_active == 0
         if not n_active:
             active = None
@@ -437,9 +446,11 @@ def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
                             active=n_active, cpd=cpd, shrunk=shrunk))
         if verbose and loop.rank == 0:
             print(f"[dist] iter {it}: overused={overused} "
-                  f"active={n_active} cpd={cpd*1e9:.2f}ns shrunk={shrunk}")
+                  f"active={n_active} cpd={cpd*1e9:.2f}ns shrunk={shrunk} "
+                  f"t_route={_t1-_t0:.2f}s t_sta={_t2-_t1:.2f}s "
+                  f"t_mask={_t3-_t2:.2f}s")
         if done:
             break
         pres = pres_fac_init if pres == 0.0 else pres * pres_fac_mult
     return dict(success=done, overused=int(overused), cpd=cpd,
-                iters=it, shrunk=shrunk, history=history)
+                iters=it, shrunk=shrunk, history=history, prof=prof)

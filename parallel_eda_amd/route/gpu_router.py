@@ -283,14 +283,29 @@ class GpuRouter:
     def _compute_bbs(self):
         xlow = np.asarray(self.g.xlow); ylow = np.asarray(self.g.ylow)
         nx, ny = self.arch.nx, self.arch.ny
-        bb = np.zeros((self.num_nets, 4), dtype=np.int16)
-        for n in range(self.num_nets):
-            terms = np.r_[self.src_rr[n],
-                          self.sink_rr[self.sink_ptr[n]:self.sink_ptr[n + 1]]]
-            xs = xlow[terms]; ys = ylow[terms]
-            m = self.bb_margin_per_net[n]
-            bb[n] = (max(0, xs.min() - m), max(0, ys.min() - m),
-                     min(nx + 1, xs.max() + m), min(ny + 1, ys.max() + m))
+        sx = xlow[self.src_rr].astype(np.int32)
+        sy = ylow[self.src_rr].astype(np.int32)
+        kx = xlow[self.sink_rr].astype(np.int32)
+        ky = ylow[self.sink_rr].astype(np.int32)
+        seg = self.sink_ptr[:-1]
+        has = self.sink_ptr[1:] > seg
+        # reduceat over each net's sink slice (empty slices produce the
+        # element AT the start index; masked back to the source below)
+        safe = np.minimum(seg, max(len(kx) - 1, 0))
+        xmin = np.minimum.reduceat(kx, safe) if len(kx) else sx.copy()
+        xmax = np.maximum.reduceat(kx, safe) if len(kx) else sx.copy()
+        ymin = np.minimum.reduceat(ky, safe) if len(ky) else sy.copy()
+        ymax = np.maximum.reduceat(ky, safe) if len(ky) else sy.copy()
+        xmin = np.where(has, np.minimum(xmin, sx), sx)
+        xmax = np.where(has, np.maximum(xmax, sx), sx)
+        ymin = np.where(has, np.minimum(ymin, sy), sy)
+        ymax = np.where(has, np.maximum(ymax, sy), sy)
+        m = self.bb_margin_per_net
+        bb = np.empty((self.num_nets, 4), dtype=np.int16)
+        bb[:, 0] = np.maximum(0, xmin - m)
+        bb[:, 1] = np.maximum(0, ymin - m)
+        bb[:, 2] = np.minimum(nx + 1, xmax + m)
+        bb[:, 3] = np.minimum(ny + 1, ymax + m)
         return bb
 
     def _bb_areas(self, bb):
@@ -344,6 +359,11 @@ class GpuRouter:
         pending round-2 GPU validation; both kernels honor it)."""
         self._partial = partial
         torch = self.torch
+        import time as _time
+        if not hasattr(self, "prof"):
+            self.prof = dict(prep=0.0, sched=0.0, kernel=0.0, bbgrow=0.0,
+                             launches=0)
+        _tp = _time.perf_counter()
         # order sinks by decreasing criticality within each net
         net_of_sink = np.repeat(np.arange(self.num_nets), np.diff(self.sink_ptr))
         perm = np.lexsort((-crit, net_of_sink))
@@ -378,6 +398,7 @@ class GpuRouter:
                 self._cong_mult = 1.0
         else:
             self._cong_mult = 1.0
+        self.prof["prep"] += _time.perf_counter() - _tp
         while True:
             self.t_fail.zero_()
             # Large reroute sets run as ONE concurrent launch (net-level
@@ -398,7 +419,10 @@ class GpuRouter:
                     order = xcd_interleaved_order(self.bb, order)
                 waves = [order]
             else:
+                _ts = _time.perf_counter()
                 waves = self._schedule_waves(todo)
+                self.prof["sched"] += _time.perf_counter() - _ts
+            _tk = _time.perf_counter()
             areas = self._bb_areas(self.bb)
             import os
             dbg = os.environ.get("PNR_ROUTE_DEBUG")
@@ -422,6 +446,8 @@ class GpuRouter:
                     print(f"    [done]", flush=True)
                 # no host sync between waves: stream order serializes them
             torch.cuda.synchronize(self.device)
+            self.prof["kernel"] += _time.perf_counter() - _tk
+            self.prof["launches"] += len(waves)
             fail = self.t_fail.cpu().numpy()
             failed = np.nonzero(fail)[0]
             if len(failed) == 0:
@@ -443,6 +469,7 @@ class GpuRouter:
                     f"router: {len(failed)} nets failed after retries "
                     f"(codes {np.unique(fail[failed])})")
             # grow bb of failed nets and retry just those
+            _tb = _time.perf_counter()
             self.bb_margin_per_net[failed] = np.minimum(
                 self.bb_margin_per_net[failed] * 2 + 4,
                 max(self.arch.nx, self.arch.ny) + 2)
@@ -451,6 +478,7 @@ class GpuRouter:
             self._bb_version += 1
             # state may be dirty for failed slots; refill (cheap)
             self._fill_state()
+            self.prof["bbgrow"] += _time.perf_counter() - _tb
             todo = failed
 
         overused = int((self.t_occ > self.t_cap.to(torch.int32)).sum().item())
