@@ -582,44 +582,6 @@ using namespace pnrh;
 
 extern "C" {
 
-struct RouteLaunchArgs {
-  // RRDev
-  const int8_t* type; const int16_t* xlow; const int16_t* ylow;
-  const int16_t* xhigh; const int16_t* yhigh; const int16_t* capacity;
-  const float* R; const float* C;
-  const int32_t* row_ptr; const int32_t* edge_dst; const int8_t* edge_sw;
-  const float* sw_R; const float* sw_Tdel; const float* base_cost;
-  const int32_t* idx_in_tile;
-  int32_t num_nodes, nx, ny, L, npt;
-  // NetsDev
-  const int32_t* net_src; const int32_t* sink_ptr; const int32_t* sink_rr;
-  const float* crit; const int32_t* sink_orig; const int16_t* bb;
-  int32_t num_nets;
-  // TreesDev
-  const int64_t* tree_off; int32_t* tree_node; int32_t* tree_parent;
-  int8_t* tree_sw; float* tree_delay; int32_t* tree_len; float* sink_delay;
-  // params
-  float astar_fac, pres_fac, seg_delay, ipin_delay, seg_base, ipin_base;
-  float delta_fac;
-  float cong_mult;
-  int32_t max_rounds;
-  int32_t strict_term;
-  // queues
-  const int32_t* queue_small; int32_t n_queue_small;
-  const int32_t* queue_large; int32_t n_queue_large;
-  int32_t* q_cursors;
-  int32_t* occ; const float* acc;
-  uint64_t* state_base; int64_t small_cap; int64_t large_cap;
-  int32_t n_small_slots; int32_t n_large_slots;
-  float4* frontier_base; int64_t f_cap_small; int64_t f_cap_large;
-  int32_t* touched_base; int64_t t_cap_small; int64_t t_cap_large;
-  int32_t* fail_flags;
-  unsigned long long* stats;   // [8] search counters or null
-  unsigned long long* net_scans;  // per-net scan counters or null
-  int32_t use_calendar;        // EXPERIMENTAL: calendar-queue frontier
-  int32_t partial;             // partial rip-up (see RouteParams)
-};
-
 int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
   RRDev g{a->type, a->xlow, a->ylow, a->xhigh, a->yhigh, a->capacity,
           a->R, a->C, a->row_ptr, a->edge_dst, a->edge_sw,

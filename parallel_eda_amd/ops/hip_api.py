@@ -77,6 +77,11 @@ def lib():
         _lib.pnr_flag_congested_nets.argtypes = [ct.c_void_p] * 5 + [ct.c_int32, ct.c_void_p, ct.c_void_p]
         _lib.pnr_fill_u64_launch.restype = ct.c_int
         _lib.pnr_fill_u64_launch.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_int64, ct.c_void_p]
+        _lib.pnr_mwg_route_net.restype = ct.c_int
+        _lib.pnr_mwg_route_net.argtypes = [
+            ct.POINTER(RouteLaunchArgs), ct.c_int32, ct.c_int32, ct.c_int32,
+            ct.c_void_p, ct.c_void_p, ct.c_void_p, ct.c_int64, ct.c_void_p,
+            ct.c_int32, ct.c_void_p]
         _lib.pnr_route_args_sizeof.restype = ct.c_int64
         # ABI guard: the ctypes mirror must match the C struct exactly —
         # a silent mismatch turns into near-null GPU pointer faults.

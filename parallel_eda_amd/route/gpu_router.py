@@ -279,6 +279,33 @@ class GpuRouter:
         self._fill_state()
         self.lib = hip_api.lib()
 
+        # multi-workgroup straggler engine (csrc/hip/router_mwg.hip):
+        # whole-device search for nets whose bb covers a large fraction of
+        # the chip — one workgroup per net serializes the endgame
+        # (measured 106 s for 123 nets at bitcoin scale). Buffers lazy.
+        self.mwg_area_threshold = max(
+            ((arch.nx + 2) * (arch.ny + 2)) // 8, self.bb_max_small_area * 2)
+        self.mwg_max_per_launch = 64
+        self._mwg_bufs = None
+
+    def _mwg_buffers(self):
+        if self._mwg_bufs is None:
+            torch = self.torch
+            f_cap = int(min(1 << 23, max(1 << 20, self.g.num_nodes)))
+            state = torch.empty(self.g.num_nodes, dtype=torch.int64,
+                                device=self.device)
+            rc = self.lib.pnr_fill_u64_launch(
+                ct_ptr(state), 0xFFFFFFFFFFFFFFFF, state.numel(),
+                self._stream())
+            hip_api.check(rc, "mwg_fill_state")
+            frA = torch.empty(f_cap * 4, dtype=torch.float32,
+                              device=self.device)
+            frB = torch.empty(f_cap * 4, dtype=torch.float32,
+                              device=self.device)
+            ctrl = torch.zeros(16, dtype=torch.int32, device=self.device)
+            self._mwg_bufs = (state, frA, frB, ctrl, f_cap)
+        return self._mwg_bufs
+
     # ---- helpers ----
     def _compute_bbs(self):
         xlow = np.asarray(self.g.xlow); ylow = np.asarray(self.g.ylow)
@@ -401,6 +428,23 @@ class GpuRouter:
         self.prof["prep"] += _time.perf_counter() - _tp
         while True:
             self.t_fail.zero_()
+            # straggler split: huge-bb nets go to the multi-workgroup
+            # engine (whole-device per-net search) instead of a wave /
+            # concurrent slot. Applied in the endgame and on late retries;
+            # capped so a large qualifying set can't serialize the launch.
+            mwg_nets = None
+            if (not self.deterministic and
+                    (len(todo) <= self.concurrent_threshold or attempts >= 2)):
+                ar = self._bb_areas(self.bb)[todo]
+                big = todo[ar >= self.mwg_area_threshold]
+                if len(big):
+                    if len(big) > self.mwg_max_per_launch:
+                        order_big = big[np.argsort(
+                            -self._bb_areas(self.bb)[big], kind="stable")]
+                        big = order_big[:self.mwg_max_per_launch]
+                    sel = np.isin(todo, big)
+                    mwg_nets = todo[sel]
+                    todo = todo[~sel]
             # Large reroute sets run as ONE concurrent launch (net-level
             # parallelism with atomic congestion, reference locking_route
             # family); small/endgame sets get the deterministic bb-disjoint
@@ -423,6 +467,20 @@ class GpuRouter:
                 waves = self._schedule_waves(todo)
                 self.prof["sched"] += _time.perf_counter() - _ts
             _tk = _time.perf_counter()
+            if mwg_nets is not None:
+                state, frA, frB, ctrl, f_cap = self._mwg_buffers()
+                empty_q = torch.zeros(0, dtype=torch.int32, device=self.device)
+                margs = self._make_args(t_sink_rr, t_crit, t_sink_orig,
+                                        empty_q, empty_q, pres_fac)
+                for inet in mwg_nets:
+                    s0 = int(self.sink_ptr[inet])
+                    s1 = int(self.sink_ptr[inet + 1])
+                    rc = self.lib.pnr_mwg_route_net(
+                        hip_api.ct.byref(margs), int(inet), s0, s1,
+                        ct_ptr(state), ct_ptr(frA), ct_ptr(frB), f_cap,
+                        ct_ptr(ctrl), 48, self._stream())
+                    hip_api.check(rc, "mwg_route_net")
+                self.prof["launches"] += len(mwg_nets)
             areas = self._bb_areas(self.bb)
             import os
             dbg = os.environ.get("PNR_ROUTE_DEBUG")
