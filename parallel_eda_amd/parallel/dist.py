@@ -413,7 +413,9 @@ def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
         # incremental cadence therefore alternates PARTIAL rip-up of the
         # active set with a FULL rip of the same set (timing refresh),
         # never a reroute of frozen feasible nets.
-        resync = incremental and it > 2 and (it - 2) % 2 == 0
+        resync_every = int(os.environ.get("PNR_RESYNC_EVERY", "2"))
+        resync = (incremental and resync_every > 0 and it > 2 and
+                  (it - 2) % resync_every == 0)
         _t0 = _time.perf_counter()
         overused, sd = loop.iteration(crit, pres, acc_fac,
                                       active_mask=active,

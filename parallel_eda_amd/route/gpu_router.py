@@ -504,8 +504,15 @@ class GpuRouter:
                     print(f"    [done]", flush=True)
                 # no host sync between waves: stream order serializes them
             torch.cuda.synchronize(self.device)
-            self.prof["kernel"] += _time.perf_counter() - _tk
+            _dtk = _time.perf_counter() - _tk
+            self.prof["kernel"] += _dtk
             self.prof["launches"] += len(waves)
+            if _os.environ.get("PNR_ATTEMPT_LOG"):
+                import sys as _sys
+                print(f"    [attempt {attempts}] nets={len(todo)} "
+                      f"mwg={0 if mwg_nets is None else len(mwg_nets)} "
+                      f"waves={len(waves)} t={_dtk:.2f}s",
+                      file=_sys.stderr, flush=True)
             fail = self.t_fail.cpu().numpy()
             failed = np.nonzero(fail)[0]
             if len(failed) == 0:
