@@ -403,6 +403,11 @@ class GpuRouter:
         attempted = todo
         attempts = 0
         self.last_retries = []
+        # nets forced to the MWG engine on retry: frontier/touched/path-cap
+        # overflows are SCRATCH limits of the per-WG slot, not bb problems —
+        # growing the bb makes the overflowing search even bigger. The MWG
+        # engine has a whole-device frontier and no touched list.
+        force_mwg = np.zeros(self.num_nets, dtype=bool)
         # congestion-aware lookahead: scale the heuristic's wire cost by
         # the mean effective cost of used wires so A* stays focused when
         # pres/acc inflate edge costs far beyond base (keeps the estimate
@@ -433,16 +438,18 @@ class GpuRouter:
             # concurrent slot. Applied in the endgame and on late retries;
             # capped so a large qualifying set can't serialize the launch.
             mwg_nets = None
-            if (not self.deterministic and
-                    (len(todo) <= self.concurrent_threshold or attempts >= 2)):
-                ar = self._bb_areas(self.bb)[todo]
-                big = todo[ar >= self.mwg_area_threshold]
-                if len(big):
+            if not self.deterministic:
+                big = np.zeros(0, dtype=todo.dtype)
+                if len(todo) <= self.concurrent_threshold or attempts >= 2:
+                    ar = self._bb_areas(self.bb)[todo]
+                    big = todo[ar >= self.mwg_area_threshold]
                     if len(big) > self.mwg_max_per_launch:
-                        order_big = big[np.argsort(
-                            -self._bb_areas(self.bb)[big], kind="stable")]
-                        big = order_big[:self.mwg_max_per_launch]
-                    sel = np.isin(todo, big)
+                        big = big[np.argsort(
+                            -self._bb_areas(self.bb)[big],
+                            kind="stable")][:self.mwg_max_per_launch]
+                forced = todo[force_mwg[todo]]
+                if len(big) or len(forced):
+                    sel = np.isin(todo, np.union1d(big, forced))
                     mwg_nets = todo[sel]
                     todo = todo[~sel]
             # Large reroute sets run as ONE concurrent launch (net-level
@@ -533,14 +540,25 @@ class GpuRouter:
                 raise RuntimeError(
                     f"router: {len(failed)} nets failed after retries "
                     f"(codes {np.unique(fail[failed])})")
-            # grow bb of failed nets and retry just those
+            # Scratch-limit failures (frontier/touched/path-cap overflow,
+            # codes 1/6/5) retry on the MWG engine at the SAME bb; genuine
+            # reachability failures grow the bb and retry.
             _tb = _time.perf_counter()
-            self.bb_margin_per_net[failed] = np.minimum(
-                self.bb_margin_per_net[failed] * 2 + 4,
-                max(self.arch.nx, self.arch.ny) + 2)
-            self.bb = self._compute_bbs()
-            self.t_bb.copy_(torch.from_numpy(self.bb).to(self.device))
-            self._bb_version += 1
+            codes = fail[failed]
+            # deterministic mode has no MWG path: it must keep growing
+            # (the doubled bb moves the net to the large whole-chip class,
+            # whose scratch is bigger — the round-1 behavior)
+            scratch = (np.isin(codes, (1, 5, 6)) if not self.deterministic
+                       else np.zeros(len(codes), dtype=bool))
+            force_mwg[failed[scratch]] = True
+            grow = failed[~scratch]
+            if len(grow):
+                self.bb_margin_per_net[grow] = np.minimum(
+                    self.bb_margin_per_net[grow] * 2 + 4,
+                    max(self.arch.nx, self.arch.ny) + 2)
+                self.bb = self._compute_bbs()
+                self.t_bb.copy_(torch.from_numpy(self.bb).to(self.device))
+                self._bb_version += 1
             # state may be dirty for failed slots; refill (cheap)
             self._fill_state()
             self.prof["bbgrow"] += _time.perf_counter() - _tb
