@@ -201,4 +201,6 @@ struct RouteLaunchArgs {
   unsigned long long* net_scans;  // per-net scan counters or null
   int32_t use_calendar;        // EXPERIMENTAL: calendar-queue frontier
   int32_t partial;             // partial rip-up (see RouteParams)
+  uint8_t* inq_base;           // per-state-entry in-queue flag (same slot
+                               // layout as state_base; frontier dedup)
 };

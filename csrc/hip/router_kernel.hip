@@ -76,7 +76,8 @@ __global__ void route_nets_kernel(
     float4* frontier_base, int64_t f_cap_small, int64_t f_cap_large,
     int32_t* touched_base, int64_t t_cap_small, int64_t t_cap_large,
     int32_t* fail_flags, unsigned long long* stats /*[8] or null*/,
-    unsigned long long* net_scans /* per-net scan counts or null */) {
+    unsigned long long* net_scans /* per-net scan counts or null */,
+    uint8_t* inq_base /* per-state-entry in-queue flag (frontier dedup) */) {
   const int tid = threadIdx.x;
   const bool is_small = (int)blockIdx.x < n_small_slots;
   const int slot = blockIdx.x;
@@ -84,11 +85,13 @@ __global__ void route_nets_kernel(
   __shared__ WgShared sh;
 
   uint64_t* state;
+  uint8_t* inq;
   float4* fr[2];
   int32_t* touched;
   int64_t f_cap, t_cap;
   if (is_small) {
     state = state_base + (int64_t)slot * small_cap;
+    inq = inq_base + (int64_t)slot * small_cap;
     float4* fb = frontier_base + (int64_t)slot * 2 * f_cap_small;
     fr[0] = fb; fr[1] = fb + f_cap_small;
     touched = touched_base + (int64_t)slot * t_cap_small;
@@ -97,6 +100,8 @@ __global__ void route_nets_kernel(
     int ls = slot - n_small_slots;
     uint64_t* lbase = state_base + (int64_t)n_small_slots * small_cap;
     state = lbase + (int64_t)ls * large_cap;
+    inq = inq_base + (int64_t)n_small_slots * small_cap
+        + (int64_t)ls * large_cap;
     float4* fb = frontier_base + (int64_t)n_small_slots * 2 * f_cap_small
                + (int64_t)ls * 2 * f_cap_large;
     fr[0] = fb; fr[1] = fb + f_cap_large;
@@ -278,6 +283,7 @@ __global__ void route_nets_kernel(
         // leave phantom occupancy). The frontier entry carries the REAL
         // back cost for expansion.
         state[li] = pack_state(0.0f, v);
+        inq[li] = 1;
         int ti = atomicAdd(&sh.touched_cnt, 1);
         if (ti < t_cap) touched[ti] = (int32_t)li;
         int fi = atomicAdd(&sh.fcnt[0], 1);
@@ -323,14 +329,30 @@ __global__ void route_nets_kernel(
           // their entry carries the real back cost
           const uint64_t expect = (prev == v) ? pack_state(0.0f, v)
                                               : pack_state(back, prev);
-          if (load_state(&state[li]) != expect) continue;  // stale entry
+          const uint64_t st = load_state(&state[li]);
+          if (st != expect) {
+            // stale: the in-queue dedup keeps exactly ONE live entry per
+            // node, so REPAIR it from the current state instead of
+            // dropping it (a drop would lose the node's only entry)
+            int32_t prev2 = (int32_t)(st & 0xffffffffu);
+            if (st == INF_STATE || prev2 == v) continue;
+            back = bits_f32((uint32_t)(st >> 32));
+            prev = prev2;
+            tot = back + S.astar_fac * expected_cost(g, P, v, S);
+            e = make_float4(tot, back, __int_as_float(v),
+                            __int_as_float(prev));
+          }
           if (tot > thr) {
-            // keep for a later bucket
+            // keep for a later bucket (stays in-queue)
             int fi = atomicAdd(&sh.fcnt[nxt], 1);
             if (fi < f_cap) fr[nxt][fi] = e;
             atomicMin(&sh.fmin_next, f32_bits(tot));
             continue;
           }
+          // leaving the queue: clear the membership flag with an
+          // L1-bypassing store so a racing relaxer's atomicExch sees it
+          __hip_atomic_store(&inq[li], (uint8_t)0, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
           if (v == S.sink_node) continue;  // settled sink; no expansion
           // expand
           int32_t e0 = g.row_ptr[v], e1 = g.row_ptr[v + 1];
@@ -362,12 +384,19 @@ __global__ void route_nets_kernel(
                 int ti = atomicAdd(&sh.touched_cnt, 1);
                 if (ti < t_cap) touched[ti] = (int32_t)lw;
               }
-              int fi = atomicAdd(&sh.fcnt[nxt], 1);
-              if (fi < f_cap)
-                fr[nxt][fi] = make_float4(tot_new, back_new,
-                                          __int_as_float(w), __int_as_float(v));
+              // fmin/best always track the improvement; the entry is
+              // pushed only if the node is not already queued (dedup —
+              // the queued entry is repaired from state at pop)
               atomicMin(&sh.fmin_next, f32_bits(tot_new));
               if (w == S.sink_node) atomicMin(&sh.best_sink_back, f32_bits(back_new));
+              if (__hip_atomic_exchange(&inq[lw], (uint8_t)1,
+                                        __ATOMIC_RELAXED,
+                                        __HIP_MEMORY_SCOPE_AGENT) == 0) {
+                int fi = atomicAdd(&sh.fcnt[nxt], 1);
+                if (fi < f_cap)
+                  fr[nxt][fi] = make_float4(tot_new, back_new,
+                                            __int_as_float(w), __int_as_float(v));
+              }
             }
           }
         }
@@ -469,8 +498,11 @@ __global__ void route_nets_kernel(
 
       // ---- sparse state reset (touched list) ----
       int nt = min((int64_t)sh.touched_cnt, t_cap);
-      for (int k = tid; k < nt; k += WG_THREADS)
-        state[(size_t)(uint32_t)touched[k]] = INF_STATE;
+      for (int k = tid; k < nt; k += WG_THREADS) {
+        size_t li = (size_t)(uint32_t)touched[k];
+        state[li] = INF_STATE;
+        inq[li] = 0;
+      }
       __syncthreads();
     }
 
@@ -483,12 +515,18 @@ __global__ void route_nets_kernel(
     // on success; do a full reset of touched here for safety
     if (sh.fail) {
       int nt = min((int64_t)sh.touched_cnt, t_cap);
-      for (int k = tid; k < nt; k += WG_THREADS)
-        state[(size_t)(uint32_t)touched[k]] = INF_STATE;
+      for (int k = tid; k < nt; k += WG_THREADS) {
+        size_t li = (size_t)(uint32_t)touched[k];
+        state[li] = INF_STATE;
+        inq[li] = 0;
+      }
       // if the touched list overflowed, fall back to a full clear
       if (sh.touched_cnt > t_cap) {
         int64_t cap = is_small ? small_cap : large_cap;
-        for (int64_t k = tid; k < cap; k += WG_THREADS) state[k] = INF_STATE;
+        for (int64_t k = tid; k < cap; k += WG_THREADS) {
+          state[k] = INF_STATE;
+          inq[k] = 0;
+        }
       }
       __syncthreads();
     }
@@ -625,7 +663,7 @@ int pnr_route_nets(const RouteLaunchArgs* a, void* stream) {
                      a->n_small_slots,
                      a->frontier_base, a->f_cap_small, a->f_cap_large,
                      a->touched_base, a->t_cap_small, a->t_cap_large,
-                     a->fail_flags, a->stats, a->net_scans);
+                     a->fail_flags, a->stats, a->net_scans, a->inq_base);
   return (int)hipGetLastError();
 }
 

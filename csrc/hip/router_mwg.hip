@@ -68,7 +68,8 @@ __global__ void mwg_init_ctrl_kernel(uint32_t* ctrl) {
 // workgroup kernel's per-sink seeding; state uses GLOBAL node indexing)
 __global__ void mwg_seed_kernel(RRDev g, NetsDev nets, TreesDev trees,
                                 RouteParams P, int32_t inet, int32_t si,
-                                uint64_t* state, float4* frA, int64_t f_cap,
+                                uint64_t* state, uint8_t* inq,
+                                float4* frA, int64_t f_cap,
                                 uint32_t* ctrl) {
   SinkCtx S;
   S.sink_node = nets.sink_rr[si];
@@ -93,6 +94,7 @@ __global__ void mwg_seed_kernel(RRDev g, NetsDev nets, TreesDev trees,
     float back = S.crit * trees.delay[off + i];
     float tot = back + S.astar_fac * expected_cost(g, P, v, S);
     state[v] = pack_state(0.0f, v);   // seed: unbeatable, prev==self
+    inq[v] = 1;
     uint32_t fi = atomicAdd(&ctrl[0], 1u);
     if (fi < (uint32_t)f_cap)
       frA[fi] = make_float4(tot, back, __int_as_float(v), __int_as_float(v));
@@ -103,6 +105,7 @@ __global__ void mwg_seed_kernel(RRDev g, NetsDev nets, TreesDev trees,
 __global__ __launch_bounds__(MWG_THREADS, 2)
 void mwg_round_kernel(RRDev g, NetsDev nets, RouteParams P,
                       int32_t inet, int32_t si, uint64_t* state,
+                      uint8_t* inq,
                       float4* frA, float4* frB, int64_t f_cap,
                       uint32_t* ctrl, const int32_t* __restrict__ occ,
                       const float* __restrict__ acc) {
@@ -137,13 +140,24 @@ void mwg_round_kernel(RRDev g, NetsDev nets, RouteParams P,
     int32_t prev = __float_as_int(e.w);
     const uint64_t expect = (prev == v) ? pack_state(0.0f, v)
                                         : pack_state(back, prev);
-    if (load_state(&state[v]) != expect) continue;   // stale
-    if (tot > thr) {   // keep for a later bucket
+    const uint64_t st = load_state(&state[v]);
+    if (st != expect) {
+      // stale: repair from current state (dedup keeps one entry/node)
+      int32_t prev2 = (int32_t)(st & 0xffffffffu);
+      if (st == MWG_INF_STATE || prev2 == v) continue;
+      back = bits_f32((uint32_t)(st >> 32));
+      prev = prev2;
+      tot = back + S.astar_fac * expected_cost(g, P, v, S);
+      e = make_float4(tot, back, __int_as_float(v), __int_as_float(prev));
+    }
+    if (tot > thr) {   // keep for a later bucket (stays in-queue)
       uint32_t fi = atomicAdd(&ctrl[1], 1u);
       if (fi < (uint32_t)f_cap) fout[fi] = e;
       atomicMin(&ctrl[3], f32_bits(tot));
       continue;
     }
+    __hip_atomic_store(&inq[v], (uint8_t)0, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
     if (v == S.sink_node) continue;   // settled sink; no expansion
     int32_t e0 = g.row_ptr[v], e1 = g.row_ptr[v + 1];
     for (int32_t ei = e0; ei < e1; ++ei) {
@@ -161,12 +175,15 @@ void mwg_round_kernel(RRDev g, NetsDev nets, RouteParams P,
       uint64_t old = atomicMin((unsigned long long*)&state[w],
                                (unsigned long long)pk);
       if (pk < old) {
-        uint32_t fi = atomicAdd(&ctrl[1], 1u);
-        if (fi < (uint32_t)f_cap)
-          fout[fi] = make_float4(tot_new, back_new,
-                                 __int_as_float(w), __int_as_float(v));
         atomicMin(&ctrl[3], f32_bits(tot_new));
         if (w == S.sink_node) atomicMin(&ctrl[4], f32_bits(back_new));
+        if (__hip_atomic_exchange(&inq[w], (uint8_t)1, __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT) == 0) {
+          uint32_t fi = atomicAdd(&ctrl[1], 1u);
+          if (fi < (uint32_t)f_cap)
+            fout[fi] = make_float4(tot_new, back_new,
+                                   __int_as_float(w), __int_as_float(v));
+        }
       }
     }
   }
@@ -292,7 +309,8 @@ extern "C" {
 // host syncs (stream-ordered; over-enqueued rounds exit on the done flag).
 int pnr_mwg_route_net(const RouteLaunchArgs* a, int32_t inet,
                       int32_t s_begin, int32_t s_end,
-                      uint64_t* state, float4* frA, float4* frB,
+                      uint64_t* state, uint8_t* inq,
+                      float4* frA, float4* frB,
                       int64_t f_cap, uint32_t* ctrl,
                       int32_t rounds_per_batch, void* stream_v) {
   hipStream_t stream = (hipStream_t)stream_v;
@@ -333,14 +351,14 @@ int pnr_mwg_route_net(const RouteLaunchArgs* a, int32_t inet,
                        ctrl);
     hipLaunchKernelGGL(mwg_seed_kernel, dim3(64), dim3(MWG_THREADS), 0,
                        stream, g, nets, trees, P, inet, si,
-                       state, frA, f_cap, ctrl);
+                       state, inq, frA, f_cap, ctrl);
     // bucket rounds in batches; device computes done
     for (;;) {
       for (int k = 0; k < rounds_per_batch; ++k) {
         hipLaunchKernelGGL(mwg_round_kernel, dim3(MWG_GRID),
                            dim3(MWG_THREADS), 0, stream,
-                           g, nets, P, inet, si, state, frA, frB, f_cap,
-                           ctrl, a->occ, a->acc);
+                           g, nets, P, inet, si, state, inq, frA, frB,
+                           f_cap, ctrl, a->occ, a->acc);
         hipLaunchKernelGGL(mwg_advance_kernel, dim3(1), dim3(1), 0, stream,
                            ctrl, f_cap, P.max_rounds, a->net_scans, inet);
       }
@@ -354,13 +372,16 @@ int pnr_mwg_route_net(const RouteLaunchArgs* a, int32_t inet,
     hipLaunchKernelGGL(mwg_backtrack_kernel, dim3(1), dim3(MWG_THREADS), 0,
                        stream, g, nets, trees, P, inet, si, state, ctrl,
                        a->occ, a->fail_flags);
-    // restore INF state for the next sink (full refill at HBM bandwidth)
+    // restore INF state + clear in-queue flags for the next sink (full
+    // refill at HBM bandwidth)
     {
       int64_t n = a->num_nodes;
       int64_t g64 = (n + 255) / 256;
       int grid = (int)(g64 < 4096 ? g64 : 4096);
       hipLaunchKernelGGL(fill_u64_kernel, dim3(grid), dim3(256), 0, stream,
                          state, MWG_INF_STATE, n);
+      hipError_t mrc = hipMemsetAsync(inq, 0, (size_t)n, stream);
+      if (mrc != hipSuccess) return (int)mrc;
     }
     // sticky net failure? (backtrack can fail) — stop routing its sinks
     hipError_t rc = hipMemcpyAsync(h_ctrl, ctrl, sizeof(h_ctrl),

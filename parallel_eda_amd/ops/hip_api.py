@@ -52,6 +52,7 @@ class RouteLaunchArgs(ct.Structure):
         ("net_scans", ct.c_void_p),
         ("use_calendar", ct.c_int32),
         ("partial", ct.c_int32),
+        ("inq_base", ct.c_void_p),
     ]
 
 
@@ -80,8 +81,8 @@ def lib():
         _lib.pnr_mwg_route_net.restype = ct.c_int
         _lib.pnr_mwg_route_net.argtypes = [
             ct.POINTER(RouteLaunchArgs), ct.c_int32, ct.c_int32, ct.c_int32,
-            ct.c_void_p, ct.c_void_p, ct.c_void_p, ct.c_int64, ct.c_void_p,
-            ct.c_int32, ct.c_void_p]
+            ct.c_void_p, ct.c_void_p, ct.c_void_p, ct.c_void_p, ct.c_int64,
+            ct.c_void_p, ct.c_int32, ct.c_void_p]
         _lib.pnr_route_args_sizeof.restype = ct.c_int64
         # ABI guard: the ctypes mirror must match the C struct exactly —
         # a silent mismatch turns into near-null GPU pointer faults.

@@ -234,6 +234,10 @@ class GpuRouter:
         self.t_touched = torch.empty(
             n_small_slots * self.t_cap_small + n_large_slots * self.t_cap_large,
             dtype=torch.int32, device=device)
+        # per-state-entry in-queue flag (frontier dedup; starts clear and
+        # is restored clear by the touched-list reset)
+        self.t_inq = torch.zeros(self.t_state.numel(), dtype=torch.uint8,
+                                 device=device)
 
         # trees: per-net capacity
         nsinks = np.diff(self.sink_ptr)
@@ -302,8 +306,10 @@ class GpuRouter:
                               device=self.device)
             frB = torch.empty(f_cap * 4, dtype=torch.float32,
                               device=self.device)
+            inq = torch.zeros(self.g.num_nodes, dtype=torch.uint8,
+                              device=self.device)
             ctrl = torch.zeros(16, dtype=torch.int32, device=self.device)
-            self._mwg_bufs = (state, frA, frB, ctrl, f_cap)
+            self._mwg_bufs = (state, inq, frA, frB, ctrl, f_cap)
         return self._mwg_bufs
 
     # ---- helpers ----
@@ -344,6 +350,8 @@ class GpuRouter:
             ct_ptr(self.t_state), 0xFFFFFFFFFFFFFFFF, self.t_state.numel(),
             self._stream())
         hip_api.check(rc, "fill_state")
+        if hasattr(self, "t_inq"):
+            self.t_inq.zero_()
 
     def _stream(self):
         return self.torch.cuda.current_stream().cuda_stream
@@ -475,7 +483,7 @@ class GpuRouter:
                 self.prof["sched"] += _time.perf_counter() - _ts
             _tk = _time.perf_counter()
             if mwg_nets is not None:
-                state, frA, frB, ctrl, f_cap = self._mwg_buffers()
+                state, inq, frA, frB, ctrl, f_cap = self._mwg_buffers()
                 empty_q = torch.zeros(0, dtype=torch.int32, device=self.device)
                 margs = self._make_args(t_sink_rr, t_crit, t_sink_orig,
                                         empty_q, empty_q, pres_fac)
@@ -484,8 +492,9 @@ class GpuRouter:
                     s1 = int(self.sink_ptr[inet + 1])
                     rc = self.lib.pnr_mwg_route_net(
                         hip_api.ct.byref(margs), int(inet), s0, s1,
-                        ct_ptr(state), ct_ptr(frA), ct_ptr(frB), f_cap,
-                        ct_ptr(ctrl), 48, self._stream())
+                        ct_ptr(state), ct_ptr(inq), ct_ptr(frA),
+                        ct_ptr(frB), f_cap, ct_ptr(ctrl), 48,
+                        self._stream())
                     hip_api.check(rc, "mwg_route_net")
                 self.prof["launches"] += len(mwg_nets)
             areas = self._bb_areas(self.bb)
@@ -700,6 +709,7 @@ class GpuRouter:
                                (getattr(self, "use_calendar", False) or
                                 _os.environ.get("PNR_CALENDAR"))) else 0
         a.partial = 1 if getattr(self, "_partial", False) else 0
+        a.inq_base = ct_ptr(self.t_inq)
         self._args_keepalive = (t_sink_rr, t_crit, t_sink_orig, q_small,
                                 q_large)
         return a
