@@ -130,7 +130,7 @@ class GpuRouter:
                  device="cuda:0", astar_fac=1.2, n_small_slots=1024,
                  n_large_slots=64, bb_margin=4, max_rounds=200000,
                  delta_fac=3.0, deterministic=False,
-                 concurrent_threshold=768, occ=None, dev_graph=None):
+                 concurrent_threshold=48, occ=None, dev_graph=None):
         torch = _torch()
         self.torch = torch
         self.device = device
@@ -523,14 +523,16 @@ class GpuRouter:
             _dtk = _time.perf_counter() - _tk
             self.prof["kernel"] += _dtk
             self.prof["launches"] += len(waves)
-            if _os.environ.get("PNR_ATTEMPT_LOG"):
-                import sys as _sys
-                print(f"    [attempt {attempts}] nets={len(todo)} "
-                      f"mwg={0 if mwg_nets is None else len(mwg_nets)} "
-                      f"waves={len(waves)} t={_dtk:.2f}s",
-                      file=_sys.stderr, flush=True)
             fail = self.t_fail.cpu().numpy()
             failed = np.nonzero(fail)[0]
+            if _os.environ.get("PNR_ATTEMPT_LOG"):
+                import sys as _sys
+                cd, cn = np.unique(fail[failed], return_counts=True)
+                print(f"    [attempt {attempts}] nets={len(todo)} "
+                      f"mwg={0 if mwg_nets is None else len(mwg_nets)} "
+                      f"waves={len(waves)} t={_dtk:.2f}s "
+                      f"fails={dict(zip(cd.tolist(), cn.tolist()))}",
+                      file=_sys.stderr, flush=True)
             if len(failed) == 0:
                 self.incomplete[attempted] = False
                 break
