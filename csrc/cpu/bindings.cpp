@@ -78,6 +78,10 @@ PYBIND11_MODULE(_pnr_cpu, m) {
     auto opt_int = [&](const char* k) {
       return a.contains(k) ? a[k].cast<int>() : 0;
     };
+    ap.sb_turn_fanin = a.contains("sb_turn_fanin")
+        ? a["sb_turn_fanin"].cast<int>() : 2;
+    if (ap.sb_turn_fanin < 1) ap.sb_turn_fanin = 1;
+    ap.w_l1 = a.contains("w_l1") ? a["w_l1"].cast<int>() : -1;
     ap.ram_col_every = opt_int("ram_col_every");
     ap.dsp_col_every = opt_int("dsp_col_every");
     ap.ram_in = opt_int("ram_in"); ap.ram_out = opt_int("ram_out");

@@ -38,6 +38,13 @@ constexpr int NUM_SWITCHES = 4;
 struct ArchParams {
   int nx, ny, W, L;
   int fc_in, fc_out;
+  int sb_turn_fanin = 1;  // in-wires per turn side at each wire's driver mux
+  int w_l1 = -1;          // tracks (of W) that are LENGTH-1 wires; -1 = auto
+                          // (W/8 rounded to a pair). A pure single-length
+                          // unidir fabric confines routing to a
+                          // (mod L x mod L) switch-block sublattice — the
+                          // length mix is what makes bb-local routing
+                          // possible (real fabrics mix L1/L4/L16).
   int clb_in, clb_out;
   int io_cap;
   // heterogeneous column tiles (0 = none); column phases mirror

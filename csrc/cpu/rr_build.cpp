@@ -14,6 +14,8 @@
 //    deterministic rotation) or at any tile along its span (IPIN edges).
 //  * OPINs drive wires that START at their tile; IPINs tap any wire
 //    covering their tile.  fc_in/fc_out are absolute track counts.
+#include <numeric>
+
 #include "pnr.h"
 
 namespace pnr {
@@ -56,6 +58,20 @@ RRGraph build_rr_graph(const ArchParams& ap) {
   g.nx = ap.nx; g.ny = ap.ny; g.W = ap.W; g.L = ap.L;
   const int nx = ap.nx, ny = ap.ny, W = ap.W, L = ap.L;
   if (W % 2) throw std::runtime_error("W must be even");
+  // Per-track wire length: the first w_l1 tracks are LENGTH-1 segments,
+  // the rest length L. See ArchParams::w_l1 for why a single-length
+  // fabric cannot route locally.
+  int w_l1 = ap.w_l1;
+  if (w_l1 < 0) {
+    // auto: W/8 length-1 tracks, but only where phase confinement can
+    // occur at all — on grids within ~2L of the edge every wire is
+    // clipped and already phase-mixes, so tiny fabrics keep the exact
+    // single-length channel the arch asked for
+    bool interior = L > 1 && nx > 4 * L && ny > 4 * L;
+    w_l1 = interior ? (((W / 8) & ~1) < 2 ? 2 : ((W / 8) & ~1)) : 0;
+  }
+  if (w_l1 > W) w_l1 = W;
+  auto track_L = [w_l1, L](int t) { return t < w_l1 ? 1 : L; };
 
   for (int i = 0; i < 6; ++i) g.base_cost[i] = ap.base_cost[i];
   // Switch table
@@ -134,7 +150,7 @@ RRGraph build_rr_graph(const ArchParams& ap) {
     for (int t = 0; t < W; ++t) {
       int p = 1;
       while (p <= nx) {
-        int b = span_high(p, t, L, nx);
+        int b = span_high(p, t, track_L(t), nx);
         float len = (float)(b - p + 1);
         int32_t id = add_node(CHANX, p, y, b, y, t, 1,
                               ap.R_wire * len, ap.C_wire * len);
@@ -148,7 +164,7 @@ RRGraph build_rr_graph(const ArchParams& ap) {
     for (int t = 0; t < W; ++t) {
       int p = 1;
       while (p <= ny) {
-        int b = span_high(p, t, L, ny);
+        int b = span_high(p, t, track_L(t), ny);
         float len = (float)(b - p + 1);
         int32_t id = add_node(CHANY, x, p, x, b, t, 1,
                               ap.R_wire * len, ap.C_wire * len);
@@ -169,25 +185,27 @@ RRGraph build_rr_graph(const ArchParams& ap) {
   // dec_lowstart[pos]: DEC tracks whose span_low == pos
   struct TrackLists {
     std::vector<std::vector<int16_t>> starts_at, inc_end, dec_lowstart;
-    void build(int N, int W, int L) {
+    void build(int N, int W, int w_l1, int L) {
+      auto track_L = [w_l1, L](int t) { return t < w_l1 ? 1 : L; };
       starts_at.assign(N + 1, {});
       inc_end.assign(N + 1, {});
       dec_lowstart.assign(N + 1, {});
       for (int pos = 1; pos <= N; ++pos)
         for (int t = 0; t < W; ++t) {
+          int Lt = track_L(t);
           if ((t & 1) == 0) {
-            if (span_low(pos, t, L) == pos) starts_at[pos].push_back(t);
-            if (span_high(pos, t, L, N) == pos) inc_end[pos].push_back(t);
+            if (span_low(pos, t, Lt) == pos) starts_at[pos].push_back(t);
+            if (span_high(pos, t, Lt, N) == pos) inc_end[pos].push_back(t);
           } else {
-            if (span_high(pos, t, L, N) == pos) starts_at[pos].push_back(t);
-            if (span_low(pos, t, L) == pos) dec_lowstart[pos].push_back(t);
+            if (span_high(pos, t, Lt, N) == pos) starts_at[pos].push_back(t);
+            if (span_low(pos, t, Lt) == pos) dec_lowstart[pos].push_back(t);
           }
         }
     }
   };
   TrackLists tlx, tly;
-  tlx.build(nx, W, L);
-  tly.build(ny, W, L);
+  tlx.build(nx, W, w_l1, L);
+  tly.build(ny, W, w_l1, L);
   // channel access for a tile side: returns (is_x, chan_index, pos)
   struct SideRef { bool is_x; int chan; int pos; bool valid; };
   auto side_ref = [&](int x, int y, int8_t side) -> SideRef {
@@ -275,15 +293,37 @@ RRGraph build_rr_graph(const ArchParams& ap) {
         int s = ins[out_dir][k];
         sb_in_list(sb_i, sb_j, s, cand);
         if (cand.empty()) continue;
-        int idx = -1;
+        int n = (int)cand.size();
         if (k == 0) {
-          // straight side: prefer the same track's incoming wire
-          for (size_t c = 0; c < cand.size(); ++c)
-            if (g.ptc[cand[c]] == t_out) { idx = (int)c; break; }
+          // straight side: the same track's incoming wire (standard
+          // unidir straight-through; guarantees long straight chains)
+          int idx = -1;
+          for (int c = 0; c < n; ++c)
+            if (g.ptc[cand[c]] == t_out) { idx = c; break; }
+          if (idx < 0) idx = (out_pos + sb_i + 2 * sb_j) % n;
+          emit(cand[idx], wnode, SW_SB);
+          continue;
         }
-        if (idx < 0)
-          idx = (out_pos + k + sb_i + 2 * sb_j) % (int)cand.size();
-        emit(cand[idx], wnode, SW_SB);
+        // TURN sides take sb_turn_fanin in-wires each (default 1: the
+        // bijective rotation). NOTE the permutation structure here is NOT
+        // what limits local routability — the wire-length mix is (see
+        // w_l1 above): a single-length fabric moves in strides of exactly
+        // L, so forward-reachable and can-reach-sink wire sets inside a
+        // bb live on disjoint (mod L, mod L) SB sublattices (measured at
+        // bitcoin scale: zero intersection for 11% of nets, which then
+        // needed bb margin ~124 — the chip edge — to route). Extra picks
+        // (fan-in > 1) spread by an affine map with a step coprime to the
+        // list size.
+        int picks = ap.sb_turn_fanin < n ? ap.sb_turn_fanin : n;
+        int base = (out_pos + k + sb_i + 2 * sb_j) % n;
+        if (picks == 1) {
+          emit(cand[base], wnode, SW_SB);
+        } else {
+          int b = 1 + ((sb_i + sb_j * 11 + k * 3) % 8);
+          while (std::gcd(b, n) != 1) ++b;
+          for (int r = 0; r < picks; ++r)
+            emit(cand[(base + r * b) % n], wnode, SW_SB);
+        }
       }
     };
     for (int y = 0; y <= ny; ++y) {
@@ -291,7 +331,7 @@ RRGraph build_rr_graph(const ArchParams& ap) {
       for (int t = 0; t < W; ++t) {
         int p = 1;
         while (p <= nx) {
-          int b = span_high(p, t, L, nx);
+          int b = span_high(p, t, track_L(t), nx);
           int32_t id = chanx[y].at(t, p);
           if ((t & 1) == 0) connect_out_wire(id, p - 1, y, 0, pos_inc[p]++);
           else connect_out_wire(id, b, y, 1, pos_dec[b]++);
@@ -304,7 +344,7 @@ RRGraph build_rr_graph(const ArchParams& ap) {
       for (int t = 0; t < W; ++t) {
         int p = 1;
         while (p <= ny) {
-          int b = span_high(p, t, L, ny);
+          int b = span_high(p, t, track_L(t), ny);
           int32_t id = chany[x].at(t, p);
           if ((t & 1) == 0) connect_out_wire(id, x, p - 1, 2, pos_inc[p]++);
           else connect_out_wire(id, x, b, 3, pos_dec[b]++);

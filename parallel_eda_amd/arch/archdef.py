@@ -48,6 +48,15 @@ class ArchDef:
     L: int = 4            # wire segment length in tiles
     fc_in: int = 8        # tracks each IPIN connects to (absolute count)
     fc_out: int = 8       # wire-starts each OPIN can drive (absolute count)
+    sb_turn_fanin: int = 1  # in-wires per TURN side at each wire's driver
+                            # mux (1 reproduces a pure permutation network,
+                            # whose orbit structure leaves ~11% of nets
+                            # locally unreachable — see rr_build.cpp)
+    w_l1: int = -1        # tracks that are LENGTH-1 wires (-1 = auto W/8).
+                          # A single-length unidir fabric moves in strides
+                          # of exactly L, confining every route to one
+                          # (mod L, mod L) SB sublattice; the length mix
+                          # restores bb-local reachability (rr_build.cpp)
     clb_in: int = 40      # CLB input pins (≡ SINK capacity)
     clb_out: int = 10     # CLB output pins (≡ SOURCE capacity)
     clb_n_ble: int = 10   # BLEs per CLB (packer cluster size, arch <pb_type num_pb>)
