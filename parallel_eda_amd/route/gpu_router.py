@@ -193,8 +193,11 @@ class GpuRouter:
         if self.large_cap <= self.small_cap:
             self.large_cap = self.small_cap
 
-        self.f_cap_small = 1 << 17
-        self.f_cap_large = 1 << 21
+        # with frontier dedup an overflow means > f_cap LIVE nodes; the
+        # pres-0 first iteration's tie plateaus get close at Titan scale,
+        # so size generously (memory is 16 B/entry: 2*2^18 / slot = 8 MiB)
+        self.f_cap_small = 1 << 18
+        self.f_cap_large = 1 << 22
         self.t_cap_small = self.small_cap
         self.t_cap_large = self.large_cap
 
