@@ -414,10 +414,11 @@ class GpuRouter:
         attempted = todo
         attempts = 0
         self.last_retries = []
-        # nets forced to the MWG engine on retry: frontier/touched/path-cap
-        # overflows are SCRATCH limits of the per-WG slot, not bb problems —
-        # growing the bb makes the overflowing search even bigger. The MWG
-        # engine has a whole-device frontier and no touched list.
+        # Scratch-overflow retry ladder (frontier/touched/path-cap, codes
+        # 1/5/6, are SLOT limits, not bb problems — growing the bb makes
+        # the overflowing search bigger): small slot -> large slot
+        # (concurrent, 32x frontier) -> MWG (whole-device, serial).
+        force_large = np.zeros(self.num_nets, dtype=bool)
         force_mwg = np.zeros(self.num_nets, dtype=bool)
         # congestion-aware lookahead: scale the heuristic's wire cost by
         # the mean effective cost of used wires so A* stays focused when
@@ -504,8 +505,10 @@ class GpuRouter:
             import os
             dbg = os.environ.get("PNR_ROUTE_DEBUG")
             for wi, wave in enumerate(waves):
-                small = wave[areas[wave] <= self.bb_max_small_area].astype(np.int32)
-                large = wave[areas[wave] > self.bb_max_small_area].astype(np.int32)
+                is_small = (areas[wave] <= self.bb_max_small_area) & \
+                    ~force_large[wave]
+                small = wave[is_small].astype(np.int32)
+                large = wave[~is_small].astype(np.int32)
                 q_small = torch.from_numpy(small).to(self.device)
                 q_large = torch.from_numpy(large).to(self.device)
                 self.t_cursors.zero_()
@@ -564,7 +567,13 @@ class GpuRouter:
             # whose scratch is bigger — the round-1 behavior)
             scratch = (np.isin(codes, (1, 5, 6)) if not self.deterministic
                        else np.zeros(len(codes), dtype=bool))
-            force_mwg[failed[scratch]] = True
+            sc = failed[scratch]
+            # scratch failure in the small class: retry in the large class
+            # (concurrent); failure in the large class: MWG engine
+            was_large = (self._bb_areas(self.bb)[sc] >
+                         self.bb_max_small_area) | force_large[sc]
+            force_mwg[sc[was_large]] = True
+            force_large[sc] = True
             grow = failed[~scratch]
             if len(grow):
                 self.bb_margin_per_net[grow] = np.minimum(
