@@ -68,6 +68,32 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
                 pass  # clock nets are global, don't count against pins
         return ins
 
+    model_outs = set(model.outputs)
+
+    def cluster_outputs(members, cand=None):
+        """Member outputs that leave the cluster (consumed by an outside
+        primitive or a model output) — bounded by clb_out (reference:
+        cluster_legality.c output-pin feasibility)."""
+        mem = set(members)
+        if cand is not None:
+            mem.add(cand)
+        outs = set()
+        for i in mem:
+            for o in prim_outputs(i):
+                if o in model_outs or any(j not in mem
+                                          for j in sig_users.get(o, ())):
+                    outs.add(o)
+        return outs
+
+    def cluster_clock(members):
+        """The single clock shared by the cluster's latches (None if
+        purely combinational). VPR legality: one clock per cluster
+        (cluster_legality.c clock feasibility)."""
+        for i in members:
+            if prims[i].clock:
+                return prims[i].clock
+        return None
+
     unclustered = set(range(np_prims)) - hard_set
     while unclustered:
         # seed: primitive with most inputs (hardest to place later)
@@ -90,12 +116,20 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
                             gain[j] += 1
             if not gain:
                 break
-            # best gain, feasibility-filtered on input pins
+            # best gain, legality-filtered: input pins, output pins, and
+            # single-clock (reference: cluster_legality.c feasibility)
             best = None
+            cur_clk = cluster_clock(members)
             for cand, gn in sorted(gain.items(), key=lambda kv: (-kv[1], kv[0])):
-                if len(cluster_inputs(members, cand)) <= arch.clb_in:
-                    best = cand
-                    break
+                cclk = prims[cand].clock
+                if cclk and cur_clk and cclk != cur_clk:
+                    continue
+                if len(cluster_inputs(members, cand)) > arch.clb_in:
+                    continue
+                if len(cluster_outputs(members, cand)) > arch.clb_out:
+                    continue
+                best = cand
+                break
             if best is None:
                 break
             members.append(best)
