@@ -83,6 +83,11 @@ def lib():
             ct.POINTER(RouteLaunchArgs), ct.c_int32, ct.c_int32, ct.c_int32,
             ct.c_void_p, ct.c_void_p, ct.c_void_p, ct.c_void_p, ct.c_int64,
             ct.c_void_p, ct.c_int32, ct.c_void_p]
+        _lib.pnr_mfma_gemm_f32.restype = ct.c_int
+        _lib.pnr_mfma_gemm_f32.argtypes = [ct.c_void_p, ct.c_void_p,
+                                           ct.c_void_p, ct.c_int32,
+                                           ct.c_int32, ct.c_int32,
+                                           ct.c_void_p]
         _lib.pnr_route_args_sizeof.restype = ct.c_int64
         # ABI guard: the ctypes mirror must match the C struct exactly —
         # a silent mismatch turns into near-null GPU pointer faults.

@@ -13,8 +13,55 @@ from ..arch.archdef import ArchDef
 from .. import ops, rrgraph
 
 
-def routed_delay_matrix(arch: ArchDef, g=None, engine="cpu", device="cuda:0"):
-    """Returns delay_mat shaped (nx+2, ny+2), indexed [|dx|, |dy|]."""
+def smoothing_band(n: int, radius: int) -> np.ndarray:
+    """Row-normalized band-averaging matrix (distance-weighted)."""
+    idx = np.arange(n)
+    w = np.maximum(0.0, radius + 1 - np.abs(idx[:, None] - idx[None, :]))
+    return (w / w.sum(axis=1, keepdims=True)).astype(np.float32)
+
+
+def mfma_gemm(a, b):
+    """C = A @ B on the MFMA pipes (f32 16x16x4 tiles;
+    csrc/hip/mfma_kernels.hip). a, b: torch CUDA f32 2-D tensors."""
+    import torch
+    from ..ops import hip_api
+    from ..route.gpu_router import ct_ptr
+    assert a.dtype == torch.float32 and b.dtype == torch.float32
+    a = a.contiguous(); b = b.contiguous()
+    M, K = a.shape
+    K2, N = b.shape
+    assert K == K2
+    c = torch.empty((M, N), dtype=torch.float32, device=a.device)
+    lib = hip_api.lib()
+    rc = lib.pnr_mfma_gemm_f32(
+        ct_ptr(a), ct_ptr(b), ct_ptr(c), M, N, K,
+        torch.cuda.current_stream(a.device).cuda_stream)
+    hip_api.check(rc, "mfma_gemm_f32")
+    return c
+
+
+def smooth_delay_matrix_gpu(dm: np.ndarray, radius: int = 1,
+                            device="cuda:0") -> np.ndarray:
+    """R = S_r @ D @ S_c^T band smoothing of the routed delay matrix on
+    the MFMA matrix cores (reference post-pass on the
+    timing_place_lookup.c tables; the raw router-measured matrix has
+    single-sample noise along the edge rows)."""
+    import torch
+    nr, nc = dm.shape
+    sr = torch.from_numpy(smoothing_band(nr, radius)).to(device)
+    sct = torch.from_numpy(smoothing_band(nc, radius).T.copy()).to(device)
+    d = torch.from_numpy(np.ascontiguousarray(dm, dtype=np.float32)).to(device)
+    t = mfma_gemm(sr, d)
+    r = mfma_gemm(t, sct)
+    return r.cpu().numpy()
+
+
+def routed_delay_matrix(arch: ArchDef, g=None, engine="cpu", device="cuda:0",
+                        smooth_radius=0):
+    """Returns delay_mat shaped (nx+2, ny+2), indexed [|dx|, |dy|].
+
+    smooth_radius > 0 (GPU engine): MFMA band smoothing of the measured
+    table (smooth_delay_matrix_gpu)."""
     if g is None:
         g = rrgraph.build_rr_graph(arch)
     nx, ny = arch.nx, arch.ny
@@ -64,4 +111,8 @@ def routed_delay_matrix(arch: ArchDef, g=None, engine="cpu", device="cuda:0"):
                 extra = (max(0, dx - (nx - 1)) + max(0, dy - (ny - 1)))
                 step = dm[1, 0] - dm[0, 0] if dm[1, 0] > 0 else 0.0
                 dm[dx, dy] = base + extra * max(step, 0.0)
-    return dm.astype(np.float32)
+    dm = dm.astype(np.float32)
+    if smooth_radius > 0 and engine == "gpu":
+        dm = smooth_delay_matrix_gpu(dm, radius=smooth_radius,
+                                     device=device)
+    return dm
