@@ -40,6 +40,12 @@ struct PlaceDev {
   int32_t rx0, rx1;             // move region (column strip) for the
                                 // distributed strip-sharded anneal
                                 // (parallel/dist_place.py); rx0 < 0 = off
+  // carry-chain macros (reference: place_macro.c): rigid block groups
+  // moved as one. macro_of: [num_blocks] macro id or -1 (nullptr = no
+  // macros); macro_ptr/macro_blk: CSR of member blocks.
+  const int32_t* macro_of;
+  const int32_t* macro_ptr;
+  const int32_t* macro_blk;
 };
 
 struct MovesDev {
@@ -92,6 +98,40 @@ __device__ float net_bb_cost(const PlaceDev& p, int n, int32_t b1, int x1,
     ymin = min(ymin, y); ymax = max(ymax, y);
   }
   return p.net_q[n] * ((xmax - xmin + 1) + (ymax - ymin + 1));
+}
+
+// bb cost of net n with every member of macro `mid` displaced (ddx,ddy)
+__device__ float net_bb_cost_macro(const PlaceDev& p, int n, int mid,
+                                   int ddx, int ddy) {
+  int xmin = 1 << 28, xmax = -1, ymin = 1 << 28, ymax = -1;
+  int32_t e0 = p.net_blk_ptr[n], e1 = p.net_blk_ptr[n + 1];
+  for (int32_t e = e0; e < e1; ++e) {
+    int32_t b = p.net_blks[e];
+    int x = p.bx[b], y = p.by[b];
+    if (p.macro_of[b] == mid) { x += ddx; y += ddy; }
+    xmin = min(xmin, x); xmax = max(xmax, x);
+    ymin = min(ymin, y); ymax = max(ymax, y);
+  }
+  return p.net_q[n] * ((xmax - xmin + 1) + (ymax - ymin + 1));
+}
+
+__device__ float net_td_cost_macro(const PlaceDev& p, int n, int mid,
+                                   int ddx, int ddy) {
+  if (p.delay_mat == nullptr) return 0.0f;
+  int32_t e0 = p.net_blk_ptr[n], e1 = p.net_blk_ptr[n + 1];
+  int32_t drv = p.net_blks[e0];
+  int dx0 = p.bx[drv], dy0 = p.by[drv];
+  if (p.macro_of[drv] == mid) { dx0 += ddx; dy0 += ddy; }
+  float t = 0.0f;
+  int32_t c0 = p.net_sink_ptr[n];
+  for (int32_t e = e0 + 1; e < e1; ++e) {
+    int32_t b = p.net_blks[e];
+    int x = p.bx[b], y = p.by[b];
+    if (p.macro_of[b] == mid) { x += ddx; y += ddy; }
+    int dx = abs(x - dx0), dy = abs(y - dy0);
+    t += p.conn_crit[c0 + (e - e0 - 1)] * p.delay_mat[dx * p.gy + dy];
+  }
+  return t;
 }
 
 // timing cost of net n (sum over conns of crit * delay) with overrides
@@ -182,9 +222,83 @@ __global__ void place_propose_kernel(PlaceDev p, MovesDev m, float T,
     }
   }
   if (x1 < 0) return;
+  const int mid = p.macro_of ? p.macro_of[blk] : -1;
+  if (mid >= 0) {
+    // -------- macro move: displace the whole chain rigidly --------
+    // (reference: place_macro.c + try_swap's macro branch; GPU variant
+    // requires FREE target tiles — no swap chains)
+    const int ddx = x1 - x0, ddy = y1 - y0;
+    const int32_t mm0 = p.macro_ptr[mid], mm1 = p.macro_ptr[mid + 1];
+    if (mm1 - mm0 > 16) return;
+    for (int32_t j = mm0; j < mm1; ++j) {
+      int32_t b = p.macro_blk[j];
+      if (p.fixed && p.fixed[b]) return;
+      int sx = p.bx[b], sy = p.by[b];
+      int tx = sx + ddx, ty = sy + ddy;
+      if (tx < 0 || tx >= p.gx || ty < 0 || ty >= p.gy) return;
+      if (p.rx0 >= 0 && (tx < p.rx0 || tx > p.rx1 ||
+                         sx < p.rx0 || sx > p.rx1)) return;
+      if (p.tile_btype) {
+        if (p.tile_btype[tx * p.gy + ty] != p.blk_type[b]) return;
+      } else if (is_io_loc(p, tx, ty) != (p.blk_type[b] == 0)) return;
+      if (cap_at(p, tx, ty) != 1) return;   // macros on cap-1 tiles only
+      int32_t occ = p.grid[((int64_t)tx * p.gy + ty) * p.cap];
+      if (occ >= 0 && p.macro_of[occ] != mid) return;   // target not free
+    }
+    atomicAdd(&m.counters[0], 1);
+    // union of member nets
+    int32_t nets[MAX_MOVE_NETS];
+    int nn = 0;
+    for (int32_t j = mm0; j < mm1; ++j) {
+      int32_t b = p.macro_blk[j];
+      for (int32_t k = p.blk_net_ptr[b]; k < p.blk_net_ptr[b + 1]; ++k) {
+        int32_t n = p.blk_nets[k];
+        bool dup = false;
+        for (int q = 0; q < nn; ++q) if (nets[q] == n) { dup = true; break; }
+        if (!dup) {
+          if (nn >= MAX_MOVE_NETS) return;
+          nets[nn++] = n;
+        }
+      }
+    }
+    float dbb = 0.0f, dtd = 0.0f;
+    for (int q = 0; q < nn; ++q) {
+      int32_t n = nets[q];
+      dbb += net_bb_cost_macro(p, n, mid, ddx, ddy) - p.net_cost[n];
+      if (timing_tradeoff > 0.0f)
+        dtd += net_td_cost_macro(p, n, mid, ddx, ddy) - p.net_tcost[n];
+    }
+    float delta = (1.0f - timing_tradeoff) * dbb * inv_bb_norm +
+                  timing_tradeoff * dtd * inv_td_norm;
+    bool accept;
+    if (delta <= 0.0f) accept = true;
+    else if (T <= 0.0f) accept = false;
+    else {
+      float u = (rng_hash(seed, batch, i * 131 + 5) & 0xFFFFFF) *
+                (1.0f / 16777216.0f);
+      accept = u < __expf(-delta / T);
+    }
+    if (!accept) return;
+    atomicAdd(&m.counters[1], 1);
+    m.mv_blk[i] = blk;
+    m.mv_to[i] = ((ddx + 4096) << 13) | (ddy + 4096);
+    m.mv_other[i] = -2 - mid;   // macro-move marker
+    m.mv_dbb[i] = dbb;
+    m.mv_dtd[i] = dtd;
+    m.mv_flags[i] = 1;
+    for (int q = 0; q < nn; ++q) atomicMin(&m.net_claim[nets[q]], i);
+    for (int32_t j = mm0; j < mm1; ++j) {
+      int32_t b = p.macro_blk[j];
+      atomicMin(&m.loc_claim[p.bx[b] * p.gy + p.by[b]], i);
+      atomicMin(&m.loc_claim[(p.bx[b] + ddx) * p.gy + p.by[b] + ddy], i);
+    }
+    return;
+  }
   int32_t other = p.grid[((int64_t)x1 * p.gy + y1) * p.cap + slot1];
   if (other == blk) return;
   if (other >= 0 && p.fixed && p.fixed[other]) return;
+  if (other >= 0 && p.macro_of && p.macro_of[other] >= 0)
+    return;   // never swap a chain member out from under its macro
   atomicAdd(&m.counters[0], 1);  // valid proposals only
 
   // collect affected nets (dedup)
@@ -242,6 +356,25 @@ __global__ void place_resolve_kernel(PlaceDev p, MovesDev m) {
   int32_t blk = m.mv_blk[i];
   int32_t other = m.mv_other[i];
   int32_t to = m.mv_to[i];
+  if (other <= -2) {
+    // macro move: every member's src+dst tile and every member net
+    const int mid = -2 - other;
+    const int ddx = (to >> 13) - 4096, ddy = (to & 0x1FFF) - 4096;
+    bool win = true;
+    const int32_t mm0 = p.macro_ptr[mid], mm1 = p.macro_ptr[mid + 1];
+    for (int32_t j = mm0; j < mm1 && win; ++j) {
+      int32_t b = p.macro_blk[j];
+      int sx = p.bx[b], sy = p.by[b];
+      if (m.loc_claim[sx * p.gy + sy] != i ||
+          m.loc_claim[(sx + ddx) * p.gy + sy + ddy] != i) win = false;
+      for (int32_t k = p.blk_net_ptr[b]; win && k < p.blk_net_ptr[b + 1]; ++k)
+        if (m.net_claim[p.blk_nets[k]] != i) win = false;
+    }
+    if (!win) { m.mv_flags[i] = 0; atomicAdd(&m.counters[3], 1); return; }
+    m.mv_flags[i] = 2;
+    atomicAdd(&m.counters[2], 1);
+    return;
+  }
   int x1 = to / p.cap / p.gy, y1 = (to / p.cap) % p.gy;
   int x0 = p.bx[blk], y0 = p.by[blk];
   bool win = m.loc_claim[x0 * p.gy + y0] == i &&
@@ -266,6 +399,35 @@ __global__ void place_apply_kernel(PlaceDev p, MovesDev m, float timing_tradeoff
   int32_t blk = m.mv_blk[i];
   int32_t other = m.mv_other[i];
   int32_t to = m.mv_to[i];
+  if (other <= -2) {
+    // macro move: clear every src slot, then write every dst (winner
+    // owns all tiles and nets exclusively; single thread per move)
+    const int mid = -2 - other;
+    const int ddx = (to >> 13) - 4096, ddy = (to & 0x1FFF) - 4096;
+    const int32_t mm0 = p.macro_ptr[mid], mm1 = p.macro_ptr[mid + 1];
+    for (int32_t j = mm0; j < mm1; ++j) {
+      int32_t b = p.macro_blk[j];
+      p.grid[((int64_t)p.bx[b] * p.gy + p.by[b]) * p.cap + p.bslot[b]] = -1;
+    }
+    for (int32_t j = mm0; j < mm1; ++j) {
+      int32_t b = p.macro_blk[j];
+      int tx = p.bx[b] + ddx, ty = p.by[b] + ddy;
+      p.grid[((int64_t)tx * p.gy + ty) * p.cap] = b;
+      p.bx[b] = tx; p.by[b] = ty; p.bslot[b] = 0;
+    }
+    for (int32_t j = mm0; j < mm1; ++j) {
+      int32_t b = p.macro_blk[j];
+      for (int32_t k = p.blk_net_ptr[b]; k < p.blk_net_ptr[b + 1]; ++k) {
+        int32_t n = p.blk_nets[k];
+        p.net_cost[n] = net_bb_cost(p, n, -1, 0, 0, -1, 0, 0);
+        if (timing_tradeoff > 0.0f)
+          p.net_tcost[n] = net_td_cost(p, n, -1, 0, 0, -1, 0, 0);
+      }
+    }
+    unsafeAtomicAdd(&cost_acc[0], (double)m.mv_dbb[i]);
+    unsafeAtomicAdd(&cost_acc[1], (double)m.mv_dtd[i]);
+    return;
+  }
   int slot1 = to % p.cap;
   int x1 = to / p.cap / p.gy, y1 = (to / p.cap) % p.gy;
   int x0 = p.bx[blk], y0 = p.by[blk], s0 = p.bslot[blk];
@@ -344,6 +506,8 @@ struct PlaceLaunchArgs {
   float* net_cost; float* net_tcost;
   int32_t num_blocks, num_nets, gx, gy, cap, nx, ny, io_cap;
   int32_t rx0, rx1;
+  const int32_t* macro_of; const int32_t* macro_ptr;
+  const int32_t* macro_blk;
   // moves
   int32_t* mv_blk; int32_t* mv_to; int32_t* mv_other;
   float* mv_dbb; float* mv_dtd; uint8_t* mv_flags;
@@ -371,6 +535,8 @@ static void unpack(const PlaceLaunchArgs* a, PlaceDev& p, MovesDev& m) {
   p.gx = a->gx; p.gy = a->gy; p.cap = a->cap;
   p.nx = a->nx; p.ny = a->ny; p.io_cap = a->io_cap;
   p.rx0 = a->rx0; p.rx1 = a->rx1;
+  p.macro_of = a->macro_of; p.macro_ptr = a->macro_ptr;
+  p.macro_blk = a->macro_blk;
   m.mv_blk = a->mv_blk; m.mv_to = a->mv_to; m.mv_other = a->mv_other;
   m.mv_dbb = a->mv_dbb; m.mv_dtd = a->mv_dtd; m.mv_flags = a->mv_flags;
   m.net_claim = a->net_claim; m.loc_claim = a->loc_claim;

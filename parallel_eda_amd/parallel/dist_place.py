@@ -41,10 +41,10 @@ class _GpuPlacerAdapter:
     with the CDNA4 batched-move kernels (BASELINE config 4: grid-
     partitioned SA on 8 GPUs over RCCL)."""
 
-    def __init__(self, netlist, arch, seed, timing, device):
+    def __init__(self, netlist, arch, seed, timing, device, macros=None):
         from ..place.gpu_placer import GpuPlacer
         self.p = GpuPlacer(netlist, arch, seed=seed, timing=timing,
-                           device=device)
+                           device=device, macros=macros)
         self.nl = netlist
         self.arch = arch
         self._delta_std = 0.0
@@ -144,9 +144,8 @@ def anneal_place_dist(netlist, arch, rank=0, world_size=1, seed=7,
 
     tt = timing_tradeoff if sta is not None else 0.0
     if engine == "gpu":
-        if macros:
-            raise NotImplementedError("macro moves are CPU-engine-only")
-        placer = _GpuPlacerAdapter(netlist, arch, seed, tt > 0, device)
+        placer = _GpuPlacerAdapter(netlist, arch, seed, tt > 0, device,
+                                   macros=macros)
     else:
         cpu = ops.cpu()
         if sta is not None and tt > 0:

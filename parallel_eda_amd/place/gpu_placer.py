@@ -32,6 +32,8 @@ class PlaceLaunchArgs(ct.Structure):
         ("gx", ct.c_int32), ("gy", ct.c_int32), ("cap", ct.c_int32),
         ("nx", ct.c_int32), ("ny", ct.c_int32), ("io_cap", ct.c_int32),
         ("rx0", ct.c_int32), ("rx1", ct.c_int32),
+        ("macro_of", ct.c_void_p), ("macro_ptr", ct.c_void_p),
+        ("macro_blk", ct.c_void_p),
         ("mv_blk", ct.c_void_p), ("mv_to", ct.c_void_p), ("mv_other", ct.c_void_p),
         ("mv_dbb", ct.c_void_p), ("mv_dtd", ct.c_void_p), ("mv_flags", ct.c_void_p),
         ("net_claim", ct.c_void_p), ("loc_claim", ct.c_void_p),
@@ -64,7 +66,7 @@ def ptr(t):
 class GpuPlacer:
     def __init__(self, netlist, arch: ArchDef, seed=7, timing=False,
                  device="cuda:0", n_moves=None, fixed=None,
-                 delay_matrix="analytic"):
+                 delay_matrix="analytic", macros=None):
         import torch
         self.torch = torch
         self.device = device
@@ -152,6 +154,26 @@ class GpuPlacer:
 
         # initial placement (host, deterministic)
         bx, by, bslot, grid = self._initial_placement(seed)
+        # carry-chain macros (reference: place_macro.c): legalize the
+        # initial placement so each chain sits at its rigid offsets, and
+        # upload the membership tables the kernels use to move chains
+        # as one (propose/resolve/apply macro branches)
+        self.t_macro_of = self.t_macro_ptr = self.t_macro_blk = None
+        self.macros = macros or []
+        if self.macros:
+            bx, by, bslot, grid = self._legalize_macros(
+                bx, by, bslot, grid, self.macros)
+            macro_of = np.full(nb, -1, dtype=np.int32)
+            mptr = [0]
+            mblk = []
+            for mi, grp in enumerate(self.macros):
+                for (b, dx, dy) in grp:
+                    macro_of[b] = mi
+                    mblk.append(b)
+                mptr.append(len(mblk))
+            self.t_macro_of = up(macro_of)
+            self.t_macro_ptr = up(np.asarray(mptr, dtype=np.int32))
+            self.t_macro_blk = up(np.asarray(mblk, dtype=np.int32))
         self.t_bx = up(bx); self.t_by = up(by); self.t_bslot = up(bslot)
         self.t_grid = up(grid)
         self.t_net_cost = torch.zeros(nn, dtype=torch.float32, device=device)
@@ -198,6 +220,50 @@ class GpuPlacer:
                 grid[x, y, sl] = b
             return bx2, by2, bs2, grid.reshape(-1)
         return self._initial_placement_free(rng, flat=True)
+
+    def _legalize_macros(self, bx, by, bslot, grid, macros):
+        """Re-place each macro's members at their rigid offsets on free
+        tiles (deterministic first-fit anchor scan). Members vacate their
+        initial tiles first; non-members are never displaced."""
+        arch = self.arch
+        g = grid.reshape(self.gx, self.gy, self.cap)
+        members = [b for grp in macros for (b, _, _) in grp]
+        for b in members:
+            g[bx[b], by[b], bslot[b]] = -1
+        tb = (arch.tile_btype_grid().reshape(self.gx, self.gy)
+              if arch.is_heterogeneous() else None)
+
+        def tile_ok(b, x, y):
+            if not (0 <= x < self.gx and 0 <= y < self.gy):
+                return False
+            bt = int(self.nl.block_type[b])
+            if tb is not None:
+                if tb[x, y] != bt:
+                    return False
+            else:
+                io = (x == 0 or x == self.gx - 1 or
+                      y == 0 or y == self.gy - 1)
+                inner = 1 <= x <= arch.nx and 1 <= y <= arch.ny
+                if bt == 0 or io or not inner:
+                    return False    # macros on logic tiles only
+            return g[x, y, 0] < 0
+
+        for grp in macros:
+            placed = False
+            for ax in range(1, arch.nx + 1):
+                for ay in range(1, arch.ny + 1):
+                    if all(tile_ok(b, ax + dx, ay + dy)
+                           for (b, dx, dy) in grp):
+                        for (b, dx, dy) in grp:
+                            bx[b], by[b], bslot[b] = ax + dx, ay + dy, 0
+                            g[ax + dx, ay + dy, 0] = b
+                        placed = True
+                        break
+                if placed:
+                    break
+            if not placed:
+                raise ValueError("cannot legalize macro (grid too full)")
+        return bx, by, bslot, g.reshape(-1)
 
     def _initial_placement_free(self, rng, flat=False):
         arch = self.arch
@@ -268,6 +334,12 @@ class GpuPlacer:
         a.gx = self.gx; a.gy = self.gy; a.cap = self.cap
         a.nx = self.arch.nx; a.ny = self.arch.ny; a.io_cap = self.arch.io_cap
         a.rx0, a.rx1 = getattr(self, "move_region", (-1, -1))
+        if getattr(self, "t_macro_of", None) is not None:
+            a.macro_of = ptr(self.t_macro_of)
+            a.macro_ptr = ptr(self.t_macro_ptr)
+            a.macro_blk = ptr(self.t_macro_blk)
+        else:
+            a.macro_of = a.macro_ptr = a.macro_blk = None
         a.mv_blk = ptr(self.t_mv_blk); a.mv_to = ptr(self.t_mv_to)
         a.mv_other = ptr(self.t_mv_other); a.mv_dbb = ptr(self.t_mv_dbb)
         a.mv_dtd = ptr(self.t_mv_dtd); a.mv_flags = ptr(self.t_mv_flags)
@@ -360,6 +432,13 @@ class GpuPlacer:
         occ = (grid >= 0).sum()
         if occ != nl.num_blocks:
             return False, f"grid count {occ} != blocks {nl.num_blocks}"
+        # carry-chain rigidity (reference: check_place macro member check)
+        for grp in getattr(self, "macros", []):
+            b0, dx0, dy0 = grp[0]
+            for (b, dx, dy) in grp[1:]:
+                if (bx[b] - bx[b0] != dx - dx0 or
+                        by[b] - by[b0] != dy - dy0):
+                    return False, f"macro offsets broken at block {b}"
         return True, ""
 
 
@@ -374,11 +453,13 @@ def _cross_count(n):
 
 def anneal_place_gpu(netlist, arch, seed=7, timing_tradeoff=0.5, inner_num=1.0,
                      sta=None, crit_exp=1.0, verbose=False, device="cuda:0",
-                     n_moves=None, fixed=None, delay_matrix="analytic"):
+                     n_moves=None, fixed=None, delay_matrix="analytic",
+                     macros=None):
     """GPU anneal with the same adaptive schedule as the CPU oracle."""
     timing = sta is not None and timing_tradeoff > 0
     placer = GpuPlacer(netlist, arch, seed=seed, timing=timing, device=device,
-                       n_moves=n_moves, fixed=fixed, delay_matrix=delay_matrix)
+                       n_moves=n_moves, fixed=fixed, delay_matrix=delay_matrix,
+                       macros=macros)
     nb = netlist.num_blocks
     move_lim = max(256, int(inner_num * (nb ** 1.3333)))
     rlim = float(max(arch.nx, arch.ny))
