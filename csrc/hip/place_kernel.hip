@@ -157,20 +157,205 @@ __device__ float net_td_cost(const PlaceDev& p, int n, int32_t b1, int x1,
 }
 
 #define MAX_MOVE_NETS 160
+#define PROP_WAVES 4
 
+__device__ __forceinline__ float wave_sum_f32(float v) {
+  for (int off = 32; off; off >>= 1) v += __shfl_xor(v, off);
+  return v;
+}
+
+// binary search in block b's sorted net list
+__device__ __forceinline__ bool blk_has_net(const PlaceDev& p, int32_t b,
+                                            int32_t n) {
+  int32_t lo = p.blk_net_ptr[b], hi = p.blk_net_ptr[b + 1];
+  while (lo < hi) {
+    int32_t mid2 = (lo + hi) >> 1;
+    if (p.blk_nets[mid2] < n) lo = mid2 + 1; else hi = mid2;
+  }
+  return lo < p.blk_net_ptr[b + 1] && p.blk_nets[lo] == n;
+}
+
+__device__ int propose_select_impl(const PlaceDev& p, int rlim,
+                                   uint32_t seed, uint32_t batch, int i,
+                                   int& x1o, int& y1o, int& slot1o);
+
+// One WAVEFRONT per proposal (the per-thread version left ~1 live
+// thread per CU and propose was 52% of LU32 kernel time at 158 us per
+// 554-move batch): lane 0 selects the candidate, the wave evaluates the
+// affected nets' exact deltas in parallel and reduces them, and claims
+// go out lane-strided (atomicMin is idempotent, so the two block lists
+// need no dedup — only the delta sum skips shared nets).
+__launch_bounds__(64 * PROP_WAVES, 2)
 __global__ void place_propose_kernel(PlaceDev p, MovesDev m, float T,
                                      int rlim, float timing_tradeoff,
                                      float inv_bb_norm, float inv_td_norm,
                                      uint32_t seed, uint32_t batch) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int lane = threadIdx.x & 63;
+  const int i = blockIdx.x * PROP_WAVES + (threadIdx.x >> 6);
   if (i >= m.n_moves) return;
-  m.mv_flags[i] = 0;
-  // pick block + destination
+  int sel = -1, sx1 = -1, sy1 = -1, sslot = 0;
+  if (lane == 0) {
+    m.mv_flags[i] = 0;
+    sel = propose_select_impl(p, rlim, seed, batch, i, sx1, sy1, sslot);
+  }
+  sel = __shfl(sel, 0);
+  if (sel < 0) return;
+  const int x1 = __shfl(sx1, 0);
+  const int y1 = __shfl(sy1, 0);
+  const int slot1 = __shfl(sslot, 0);
+  const int32_t blk = sel;
+  const int x0 = p.bx[blk], y0 = p.by[blk];
+  const int mid = p.macro_of ? p.macro_of[blk] : -1;
+  if (mid >= 0) {
+    // -------- macro move (reference: place_macro.c; free-target) -----
+    const int ddx = x1 - x0, ddy = y1 - y0;
+    const int32_t mm0 = p.macro_ptr[mid], mm1 = p.macro_ptr[mid + 1];
+    if (mm1 - mm0 > 16) return;
+    int ok = 0;
+    if (lane == 0) {
+      ok = 1;
+      for (int32_t j = mm0; j < mm1 && ok; ++j) {
+        int32_t b = p.macro_blk[j];
+        int sx = p.bx[b], sy = p.by[b];
+        int tx = sx + ddx, ty = sy + ddy;
+        if (p.fixed && p.fixed[b]) ok = 0;
+        else if (tx < 0 || tx >= p.gx || ty < 0 || ty >= p.gy) ok = 0;
+        else if (p.rx0 >= 0 && (tx < p.rx0 || tx > p.rx1 ||
+                                sx < p.rx0 || sx > p.rx1)) ok = 0;
+        else if (p.tile_btype
+                     ? (p.tile_btype[tx * p.gy + ty] != p.blk_type[b])
+                     : (is_io_loc(p, tx, ty) != (p.blk_type[b] == 0))) ok = 0;
+        else if (cap_at(p, tx, ty) != 1) ok = 0;
+        else {
+          int32_t occ = p.grid[((int64_t)tx * p.gy + ty) * p.cap];
+          if (occ >= 0 && p.macro_of[occ] != mid) ok = 0;
+        }
+      }
+      if (ok) atomicAdd(&m.counters[0], 1);
+    }
+    ok = __shfl(ok, 0);
+    if (!ok) return;
+    float dbb = 0.0f, dtd = 0.0f;
+    for (int32_t j = mm0; j < mm1; ++j) {
+      int32_t b = p.macro_blk[j];
+      for (int32_t k = p.blk_net_ptr[b] + lane; k < p.blk_net_ptr[b + 1];
+           k += 64) {
+        int32_t n = p.blk_nets[k];
+        bool dup = false;
+        for (int32_t j2 = mm0; j2 < j && !dup; ++j2)
+          dup = blk_has_net(p, p.macro_blk[j2], n);
+        if (dup) continue;
+        dbb += net_bb_cost_macro(p, n, mid, ddx, ddy) - p.net_cost[n];
+        if (timing_tradeoff > 0.0f)
+          dtd += net_td_cost_macro(p, n, mid, ddx, ddy) - p.net_tcost[n];
+      }
+    }
+    dbb = wave_sum_f32(dbb);
+    dtd = wave_sum_f32(dtd);
+    float delta = (1.0f - timing_tradeoff) * dbb * inv_bb_norm +
+                  timing_tradeoff * dtd * inv_td_norm;
+    bool accept;
+    if (delta <= 0.0f) accept = true;
+    else if (T <= 0.0f) accept = false;
+    else {
+      float u = (rng_hash(seed, batch, i * 131 + 5) & 0xFFFFFF) *
+                (1.0f / 16777216.0f);
+      accept = u < __expf(-delta / T);
+    }
+    if (!accept) return;
+    if (lane == 0) {
+      atomicAdd(&m.counters[1], 1);
+      m.mv_blk[i] = blk;
+      m.mv_to[i] = ((ddx + 4096) << 13) | (ddy + 4096);
+      m.mv_other[i] = -2 - mid;
+      m.mv_dbb[i] = dbb;
+      m.mv_dtd[i] = dtd;
+      m.mv_flags[i] = 1;
+    }
+    for (int32_t j = mm0; j < mm1; ++j) {
+      int32_t b = p.macro_blk[j];
+      for (int32_t k = p.blk_net_ptr[b] + lane; k < p.blk_net_ptr[b + 1];
+           k += 64)
+        atomicMin(&m.net_claim[p.blk_nets[k]], i);
+    }
+    for (int32_t j = mm0 + lane; j < mm1; j += 64) {
+      int32_t b = p.macro_blk[j];
+      atomicMin(&m.loc_claim[p.bx[b] * p.gy + p.by[b]], i);
+      atomicMin(&m.loc_claim[(p.bx[b] + ddx) * p.gy + p.by[b] + ddy], i);
+    }
+    return;
+  }
+  // -------- single-block move / swap (wave-uniform checks) --------
+  int32_t other = p.grid[((int64_t)x1 * p.gy + y1) * p.cap + slot1];
+  if (other == blk) return;
+  if (other >= 0 && p.fixed && p.fixed[other]) return;
+  if (other >= 0 && p.macro_of && p.macro_of[other] >= 0)
+    return;   // never swap a chain member out from under its macro
+  if (lane == 0) atomicAdd(&m.counters[0], 1);  // valid proposals only
+  // exact deltas, lane-strided over the two blocks' net lists; nets
+  // shared by both blocks are counted once (skip in other's pass)
+  float dbb = 0.0f, dtd = 0.0f;
+  const int ox = other >= 0 ? x0 : -1000, oy = other >= 0 ? y0 : -1000;
+  for (int32_t k = p.blk_net_ptr[blk] + lane; k < p.blk_net_ptr[blk + 1];
+       k += 64) {
+    int32_t n = p.blk_nets[k];
+    dbb += net_bb_cost(p, n, blk, x1, y1, other, ox, oy) - p.net_cost[n];
+    if (timing_tradeoff > 0.0f)
+      dtd += net_td_cost(p, n, blk, x1, y1, other, ox, oy) - p.net_tcost[n];
+  }
+  if (other >= 0) {
+    for (int32_t k = p.blk_net_ptr[other] + lane;
+         k < p.blk_net_ptr[other + 1]; k += 64) {
+      int32_t n = p.blk_nets[k];
+      if (blk_has_net(p, blk, n)) continue;
+      dbb += net_bb_cost(p, n, blk, x1, y1, other, ox, oy) - p.net_cost[n];
+      if (timing_tradeoff > 0.0f)
+        dtd += net_td_cost(p, n, blk, x1, y1, other, ox, oy) - p.net_tcost[n];
+    }
+  }
+  dbb = wave_sum_f32(dbb);
+  dtd = wave_sum_f32(dtd);
+  float delta = (1.0f - timing_tradeoff) * dbb * inv_bb_norm +
+                timing_tradeoff * dtd * inv_td_norm;
+  bool accept;
+  if (delta <= 0.0f) accept = true;
+  else if (T <= 0.0f) accept = false;
+  else {
+    float u = (rng_hash(seed, batch, i * 131 + 5) & 0xFFFFFF) *
+              (1.0f / 16777216.0f);
+    accept = u < __expf(-delta / T);
+  }
+  if (!accept) return;
+  if (lane == 0) {
+    atomicAdd(&m.counters[1], 1);
+    m.mv_blk[i] = blk;
+    m.mv_to[i] = ((x1 * p.gy + y1) * p.cap + slot1);
+    m.mv_other[i] = other;
+    m.mv_dbb[i] = dbb;
+    m.mv_dtd[i] = dtd;
+    m.mv_flags[i] = 1;
+    atomicMin(&m.loc_claim[x0 * p.gy + y0], i);
+    atomicMin(&m.loc_claim[x1 * p.gy + y1], i);
+  }
+  for (int32_t k = p.blk_net_ptr[blk] + lane; k < p.blk_net_ptr[blk + 1];
+       k += 64)
+    atomicMin(&m.net_claim[p.blk_nets[k]], i);
+  if (other >= 0)
+    for (int32_t k = p.blk_net_ptr[other] + lane;
+         k < p.blk_net_ptr[other + 1]; k += 64)
+      atomicMin(&m.net_claim[p.blk_nets[k]], i);
+}
+
+// lane-0 candidate selection (same RNG stream as the round-1 per-thread
+// kernel); returns the block or -1, target via out-params
+__device__ int propose_select_impl(const PlaceDev& p, int rlim,
+                                   uint32_t seed, uint32_t batch, int i,
+                                   int& x1o, int& y1o, int& slot1o) {
   uint32_t r0 = rng_hash(seed ^ 0x5BD1E995u, batch, i);
   int32_t blk = r0 % p.num_blocks;
-  if (p.fixed && p.fixed[blk]) return;   // pinned (pad_loc_file)
+  if (p.fixed && p.fixed[blk]) return -1;
   if (p.rx0 >= 0 && (p.bx[blk] < p.rx0 || p.bx[blk] > p.rx1))
-    return;                              // not this rank's strip
+    return -1;
   bool io = p.blk_type[blk] == 0;
   int x0 = p.bx[blk], y0 = p.by[blk];
   int x1 = -1, y1 = -1, slot1 = 0;
@@ -221,241 +406,127 @@ __global__ void place_propose_kernel(PlaceDev p, MovesDev m, float T,
       break;
     }
   }
-  if (x1 < 0) return;
-  const int mid = p.macro_of ? p.macro_of[blk] : -1;
-  if (mid >= 0) {
-    // -------- macro move: displace the whole chain rigidly --------
-    // (reference: place_macro.c + try_swap's macro branch; GPU variant
-    // requires FREE target tiles — no swap chains)
-    const int ddx = x1 - x0, ddy = y1 - y0;
-    const int32_t mm0 = p.macro_ptr[mid], mm1 = p.macro_ptr[mid + 1];
-    if (mm1 - mm0 > 16) return;
-    for (int32_t j = mm0; j < mm1; ++j) {
-      int32_t b = p.macro_blk[j];
-      if (p.fixed && p.fixed[b]) return;
-      int sx = p.bx[b], sy = p.by[b];
-      int tx = sx + ddx, ty = sy + ddy;
-      if (tx < 0 || tx >= p.gx || ty < 0 || ty >= p.gy) return;
-      if (p.rx0 >= 0 && (tx < p.rx0 || tx > p.rx1 ||
-                         sx < p.rx0 || sx > p.rx1)) return;
-      if (p.tile_btype) {
-        if (p.tile_btype[tx * p.gy + ty] != p.blk_type[b]) return;
-      } else if (is_io_loc(p, tx, ty) != (p.blk_type[b] == 0)) return;
-      if (cap_at(p, tx, ty) != 1) return;   // macros on cap-1 tiles only
-      int32_t occ = p.grid[((int64_t)tx * p.gy + ty) * p.cap];
-      if (occ >= 0 && p.macro_of[occ] != mid) return;   // target not free
-    }
-    atomicAdd(&m.counters[0], 1);
-    // union of member nets
-    int32_t nets[MAX_MOVE_NETS];
-    int nn = 0;
-    for (int32_t j = mm0; j < mm1; ++j) {
-      int32_t b = p.macro_blk[j];
-      for (int32_t k = p.blk_net_ptr[b]; k < p.blk_net_ptr[b + 1]; ++k) {
-        int32_t n = p.blk_nets[k];
-        bool dup = false;
-        for (int q = 0; q < nn; ++q) if (nets[q] == n) { dup = true; break; }
-        if (!dup) {
-          if (nn >= MAX_MOVE_NETS) return;
-          nets[nn++] = n;
-        }
-      }
-    }
-    float dbb = 0.0f, dtd = 0.0f;
-    for (int q = 0; q < nn; ++q) {
-      int32_t n = nets[q];
-      dbb += net_bb_cost_macro(p, n, mid, ddx, ddy) - p.net_cost[n];
-      if (timing_tradeoff > 0.0f)
-        dtd += net_td_cost_macro(p, n, mid, ddx, ddy) - p.net_tcost[n];
-    }
-    float delta = (1.0f - timing_tradeoff) * dbb * inv_bb_norm +
-                  timing_tradeoff * dtd * inv_td_norm;
-    bool accept;
-    if (delta <= 0.0f) accept = true;
-    else if (T <= 0.0f) accept = false;
-    else {
-      float u = (rng_hash(seed, batch, i * 131 + 5) & 0xFFFFFF) *
-                (1.0f / 16777216.0f);
-      accept = u < __expf(-delta / T);
-    }
-    if (!accept) return;
-    atomicAdd(&m.counters[1], 1);
-    m.mv_blk[i] = blk;
-    m.mv_to[i] = ((ddx + 4096) << 13) | (ddy + 4096);
-    m.mv_other[i] = -2 - mid;   // macro-move marker
-    m.mv_dbb[i] = dbb;
-    m.mv_dtd[i] = dtd;
-    m.mv_flags[i] = 1;
-    for (int q = 0; q < nn; ++q) atomicMin(&m.net_claim[nets[q]], i);
-    for (int32_t j = mm0; j < mm1; ++j) {
-      int32_t b = p.macro_blk[j];
-      atomicMin(&m.loc_claim[p.bx[b] * p.gy + p.by[b]], i);
-      atomicMin(&m.loc_claim[(p.bx[b] + ddx) * p.gy + p.by[b] + ddy], i);
-    }
-    return;
-  }
-  int32_t other = p.grid[((int64_t)x1 * p.gy + y1) * p.cap + slot1];
-  if (other == blk) return;
-  if (other >= 0 && p.fixed && p.fixed[other]) return;
-  if (other >= 0 && p.macro_of && p.macro_of[other] >= 0)
-    return;   // never swap a chain member out from under its macro
-  atomicAdd(&m.counters[0], 1);  // valid proposals only
-
-  // collect affected nets (dedup)
-  int32_t nets[MAX_MOVE_NETS];
-  int nn = 0;
-  for (int pass = 0; pass < 2; ++pass) {
-    int32_t b = pass == 0 ? blk : other;
-    if (b < 0) continue;
-    for (int32_t k = p.blk_net_ptr[b]; k < p.blk_net_ptr[b + 1]; ++k) {
-      int32_t n = p.blk_nets[k];
-      bool dup = false;
-      for (int j = 0; j < nn; ++j) if (nets[j] == n) { dup = true; break; }
-      if (!dup) {
-        if (nn >= MAX_MOVE_NETS) return;  // very-high-fanin block: skip move
-        nets[nn++] = n;
-      }
-    }
-  }
-  // exact deltas with overrides: blk -> (x1,y1); other -> (x0,y0)
-  float dbb = 0.0f, dtd = 0.0f;
-  int ox = other >= 0 ? x0 : -1000, oy = other >= 0 ? y0 : -1000;
-  for (int j = 0; j < nn; ++j) {
-    int32_t n = nets[j];
-    dbb += net_bb_cost(p, n, blk, x1, y1, other, ox, oy) - p.net_cost[n];
-    if (timing_tradeoff > 0.0f)
-      dtd += net_td_cost(p, n, blk, x1, y1, other, ox, oy) - p.net_tcost[n];
-  }
-  float delta = (1.0f - timing_tradeoff) * dbb * inv_bb_norm +
-                timing_tradeoff * dtd * inv_td_norm;
-  bool accept;
-  if (delta <= 0.0f) accept = true;
-  else if (T <= 0.0f) accept = false;
-  else {
-    float u = (rng_hash(seed, batch, i * 131 + 5) & 0xFFFFFF) * (1.0f / 16777216.0f);
-    accept = u < __expf(-delta / T);
-  }
-  if (!accept) return;
-  atomicAdd(&m.counters[1], 1);
-  m.mv_blk[i] = blk;
-  m.mv_to[i] = ((x1 * p.gy + y1) * p.cap + slot1);
-  m.mv_other[i] = other;
-  m.mv_dbb[i] = dbb;
-  m.mv_dtd[i] = dtd;
-  m.mv_flags[i] = 1;
-  // claim nets + the two tile locations (tile granularity serializes
-  // slot-level races within a tile)
-  for (int j = 0; j < nn; ++j) atomicMin(&m.net_claim[nets[j]], i);
-  atomicMin(&m.loc_claim[x0 * p.gy + y0], i);
-  atomicMin(&m.loc_claim[x1 * p.gy + y1], i);
+  if (x1 < 0) return -1;
+  x1o = x1; y1o = y1; slot1o = slot1;
+  return blk;
 }
 
+__launch_bounds__(64 * PROP_WAVES, 2)
 __global__ void place_resolve_kernel(PlaceDev p, MovesDev m) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int lane = threadIdx.x & 63;
+  const int i = blockIdx.x * PROP_WAVES + (threadIdx.x >> 6);
   if (i >= m.n_moves || m.mv_flags[i] != 1) return;
   int32_t blk = m.mv_blk[i];
   int32_t other = m.mv_other[i];
   int32_t to = m.mv_to[i];
+  bool bad = false;
   if (other <= -2) {
     // macro move: every member's src+dst tile and every member net
     const int mid = -2 - other;
     const int ddx = (to >> 13) - 4096, ddy = (to & 0x1FFF) - 4096;
-    bool win = true;
     const int32_t mm0 = p.macro_ptr[mid], mm1 = p.macro_ptr[mid + 1];
-    for (int32_t j = mm0; j < mm1 && win; ++j) {
+    for (int32_t j = mm0 + lane; j < mm1; j += 64) {
       int32_t b = p.macro_blk[j];
       int sx = p.bx[b], sy = p.by[b];
       if (m.loc_claim[sx * p.gy + sy] != i ||
-          m.loc_claim[(sx + ddx) * p.gy + sy + ddy] != i) win = false;
-      for (int32_t k = p.blk_net_ptr[b]; win && k < p.blk_net_ptr[b + 1]; ++k)
-        if (m.net_claim[p.blk_nets[k]] != i) win = false;
+          m.loc_claim[(sx + ddx) * p.gy + sy + ddy] != i) bad = true;
     }
-    if (!win) { m.mv_flags[i] = 0; atomicAdd(&m.counters[3], 1); return; }
-    m.mv_flags[i] = 2;
-    atomicAdd(&m.counters[2], 1);
-    return;
-  }
-  int x1 = to / p.cap / p.gy, y1 = (to / p.cap) % p.gy;
-  int x0 = p.bx[blk], y0 = p.by[blk];
-  bool win = m.loc_claim[x0 * p.gy + y0] == i &&
-             m.loc_claim[x1 * p.gy + y1] == i;
-  if (win) {
-    for (int pass = 0; pass < 2 && win; ++pass) {
-      int32_t b = pass == 0 ? blk : other;
-      if (b < 0) continue;
-      for (int32_t k = p.blk_net_ptr[b]; k < p.blk_net_ptr[b + 1]; ++k)
-        if (m.net_claim[p.blk_nets[k]] != i) { win = false; break; }
+    for (int32_t j = mm0; j < mm1; ++j) {
+      int32_t b = p.macro_blk[j];
+      for (int32_t k = p.blk_net_ptr[b] + lane; k < p.blk_net_ptr[b + 1];
+           k += 64)
+        if (m.net_claim[p.blk_nets[k]] != i) bad = true;
     }
+  } else {
+    int x1 = to / p.cap / p.gy, y1 = (to / p.cap) % p.gy;
+    int x0 = p.bx[blk], y0 = p.by[blk];
+    if (lane == 0 && (m.loc_claim[x0 * p.gy + y0] != i ||
+                      m.loc_claim[x1 * p.gy + y1] != i)) bad = true;
+    for (int32_t k = p.blk_net_ptr[blk] + lane; k < p.blk_net_ptr[blk + 1];
+         k += 64)
+      if (m.net_claim[p.blk_nets[k]] != i) bad = true;
+    if (other >= 0)
+      for (int32_t k = p.blk_net_ptr[other] + lane;
+           k < p.blk_net_ptr[other + 1]; k += 64)
+        if (m.net_claim[p.blk_nets[k]] != i) bad = true;
   }
-  if (!win) { m.mv_flags[i] = 0; atomicAdd(&m.counters[3], 1); return; }
-  m.mv_flags[i] = 2;
-  atomicAdd(&m.counters[2], 1);
+  const bool win = !__any(bad);
+  if (lane == 0) {
+    if (!win) { m.mv_flags[i] = 0; atomicAdd(&m.counters[3], 1); }
+    else { m.mv_flags[i] = 2; atomicAdd(&m.counters[2], 1); }
+  }
 }
 
+__launch_bounds__(64 * PROP_WAVES, 2)
 __global__ void place_apply_kernel(PlaceDev p, MovesDev m, float timing_tradeoff,
                                    double* cost_acc /*[2]: dbb, dtd*/) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int lane = threadIdx.x & 63;
+  const int i = blockIdx.x * PROP_WAVES + (threadIdx.x >> 6);
   if (i >= m.n_moves || m.mv_flags[i] != 2) return;
   int32_t blk = m.mv_blk[i];
   int32_t other = m.mv_other[i];
   int32_t to = m.mv_to[i];
+  // Net-cost refresh runs FIRST, lane-strided, using the override cost
+  // functions on the PRE-move positions (so lanes never read positions
+  // another lane just wrote); duplicate refreshes of nets shared by both
+  // blocks are idempotent. Then lane 0 commits the grid/position update.
   if (other <= -2) {
-    // macro move: clear every src slot, then write every dst (winner
-    // owns all tiles and nets exclusively; single thread per move)
     const int mid = -2 - other;
     const int ddx = (to >> 13) - 4096, ddy = (to & 0x1FFF) - 4096;
     const int32_t mm0 = p.macro_ptr[mid], mm1 = p.macro_ptr[mid + 1];
     for (int32_t j = mm0; j < mm1; ++j) {
       int32_t b = p.macro_blk[j];
-      p.grid[((int64_t)p.bx[b] * p.gy + p.by[b]) * p.cap + p.bslot[b]] = -1;
-    }
-    for (int32_t j = mm0; j < mm1; ++j) {
-      int32_t b = p.macro_blk[j];
-      int tx = p.bx[b] + ddx, ty = p.by[b] + ddy;
-      p.grid[((int64_t)tx * p.gy + ty) * p.cap] = b;
-      p.bx[b] = tx; p.by[b] = ty; p.bslot[b] = 0;
-    }
-    for (int32_t j = mm0; j < mm1; ++j) {
-      int32_t b = p.macro_blk[j];
-      for (int32_t k = p.blk_net_ptr[b]; k < p.blk_net_ptr[b + 1]; ++k) {
+      for (int32_t k = p.blk_net_ptr[b] + lane; k < p.blk_net_ptr[b + 1];
+           k += 64) {
         int32_t n = p.blk_nets[k];
-        p.net_cost[n] = net_bb_cost(p, n, -1, 0, 0, -1, 0, 0);
+        p.net_cost[n] = net_bb_cost_macro(p, n, mid, ddx, ddy);
         if (timing_tradeoff > 0.0f)
-          p.net_tcost[n] = net_td_cost(p, n, -1, 0, 0, -1, 0, 0);
+          p.net_tcost[n] = net_td_cost_macro(p, n, mid, ddx, ddy);
       }
     }
-    unsafeAtomicAdd(&cost_acc[0], (double)m.mv_dbb[i]);
-    unsafeAtomicAdd(&cost_acc[1], (double)m.mv_dtd[i]);
+    if (lane == 0) {
+      for (int32_t j = mm0; j < mm1; ++j) {
+        int32_t b = p.macro_blk[j];
+        p.grid[((int64_t)p.bx[b] * p.gy + p.by[b]) * p.cap + p.bslot[b]] = -1;
+      }
+      for (int32_t j = mm0; j < mm1; ++j) {
+        int32_t b = p.macro_blk[j];
+        int tx = p.bx[b] + ddx, ty = p.by[b] + ddy;
+        p.grid[((int64_t)tx * p.gy + ty) * p.cap] = b;
+        p.bx[b] = tx; p.by[b] = ty; p.bslot[b] = 0;
+      }
+      unsafeAtomicAdd(&cost_acc[0], (double)m.mv_dbb[i]);
+      unsafeAtomicAdd(&cost_acc[1], (double)m.mv_dtd[i]);
+    }
     return;
   }
   int slot1 = to % p.cap;
   int x1 = to / p.cap / p.gy, y1 = (to / p.cap) % p.gy;
   int x0 = p.bx[blk], y0 = p.by[blk], s0 = p.bslot[blk];
-  // grid + placement updates (winner owns both tiles exclusively)
-  p.grid[((int64_t)x0 * p.gy + y0) * p.cap + s0] = other >= 0 ? other : -1;
-  p.grid[((int64_t)x1 * p.gy + y1) * p.cap + slot1] = blk;
-  p.bx[blk] = x1; p.by[blk] = y1; p.bslot[blk] = slot1;
-  if (other >= 0) { p.bx[other] = x0; p.by[other] = y0; p.bslot[other] = s0; }
-  // refresh cached per-net costs for affected nets (exclusive ownership)
-  for (int pass = 0; pass < 2; ++pass) {
-    int32_t b = pass == 0 ? blk : other;
-    if (b < 0) continue;
-    for (int32_t k = p.blk_net_ptr[b]; k < p.blk_net_ptr[b + 1]; ++k) {
-      int32_t n = p.blk_nets[k];
-      if (pass == 1) {
-        // skip nets already refreshed via blk
-        bool shared = false;
-        for (int32_t k2 = p.blk_net_ptr[blk]; k2 < p.blk_net_ptr[blk + 1]; ++k2)
-          if (p.blk_nets[k2] == n) { shared = true; break; }
-        if (shared) continue;
-      }
-      p.net_cost[n] = net_bb_cost(p, n, -1, 0, 0, -1, 0, 0);
-      if (timing_tradeoff > 0.0f)
-        p.net_tcost[n] = net_td_cost(p, n, -1, 0, 0, -1, 0, 0);
-    }
+  const int ox = other >= 0 ? x0 : -1000, oy = other >= 0 ? y0 : -1000;
+  for (int32_t k = p.blk_net_ptr[blk] + lane; k < p.blk_net_ptr[blk + 1];
+       k += 64) {
+    int32_t n = p.blk_nets[k];
+    p.net_cost[n] = net_bb_cost(p, n, blk, x1, y1, other, ox, oy);
+    if (timing_tradeoff > 0.0f)
+      p.net_tcost[n] = net_td_cost(p, n, blk, x1, y1, other, ox, oy);
   }
-  unsafeAtomicAdd(&cost_acc[0], (double)m.mv_dbb[i]);
-  unsafeAtomicAdd(&cost_acc[1], (double)m.mv_dtd[i]);
+  if (other >= 0)
+    for (int32_t k = p.blk_net_ptr[other] + lane;
+         k < p.blk_net_ptr[other + 1]; k += 64) {
+      int32_t n = p.blk_nets[k];
+      p.net_cost[n] = net_bb_cost(p, n, blk, x1, y1, other, ox, oy);
+      if (timing_tradeoff > 0.0f)
+        p.net_tcost[n] = net_td_cost(p, n, blk, x1, y1, other, ox, oy);
+    }
+  if (lane == 0) {
+    p.grid[((int64_t)x0 * p.gy + y0) * p.cap + s0] = other >= 0 ? other : -1;
+    p.grid[((int64_t)x1 * p.gy + y1) * p.cap + slot1] = blk;
+    p.bx[blk] = x1; p.by[blk] = y1; p.bslot[blk] = slot1;
+    if (other >= 0) { p.bx[other] = x0; p.by[other] = y0; p.bslot[other] = s0; }
+    unsafeAtomicAdd(&cost_acc[0], (double)m.mv_dbb[i]);
+    unsafeAtomicAdd(&cost_acc[1], (double)m.mv_dtd[i]);
+  }
 }
 
 // claims reset per batch; counters accumulate across batches (host zeroes
@@ -552,13 +623,15 @@ int pnr_place_batch(const PlaceLaunchArgs* a, void* stream) {
   hipLaunchKernelGGL(place_reset_claims_kernel, dim3(rg), dim3(256), 0, s,
                      m.net_claim, p.num_nets, m.loc_claim, p.gx * p.gy,
                      m.counters);
-  int mg = (m.n_moves + 255) / 256;
-  hipLaunchKernelGGL(place_propose_kernel, dim3(mg), dim3(256), 0, s,
-                     p, m, a->T, a->rlim, a->timing_tradeoff,
+  // one wave64 per proposal, PROP_WAVES waves per workgroup
+  int mg = (m.n_moves + PROP_WAVES - 1) / PROP_WAVES;
+  hipLaunchKernelGGL(place_propose_kernel, dim3(mg), dim3(64 * PROP_WAVES),
+                     0, s, p, m, a->T, a->rlim, a->timing_tradeoff,
                      a->inv_bb_norm, a->inv_td_norm, a->seed, a->batch);
-  hipLaunchKernelGGL(place_resolve_kernel, dim3(mg), dim3(256), 0, s, p, m);
-  hipLaunchKernelGGL(place_apply_kernel, dim3(mg), dim3(256), 0, s,
-                     p, m, a->timing_tradeoff, a->cost_acc);
+  hipLaunchKernelGGL(place_resolve_kernel, dim3(mg), dim3(64 * PROP_WAVES),
+                     0, s, p, m);
+  hipLaunchKernelGGL(place_apply_kernel, dim3(mg), dim3(64 * PROP_WAVES),
+                     0, s, p, m, a->timing_tradeoff, a->cost_acc);
   return (int)hipGetLastError();
 }
 
