@@ -137,6 +137,9 @@ class GpuRouter:
         self.arch = arch
         self.g = g
         self.num_nets = len(src_rr)
+        import os as _os
+        astar_fac = float(_os.environ.get("PNR_ASTAR", astar_fac))
+        delta_fac = float(_os.environ.get("PNR_DELTA", delta_fac))
         # deterministic mode needs an ADMISSIBLE lookahead (astar_fac<=1):
         # with an inflated heuristic the termination point depends on
         # which paths the racing relaxations discovered first
