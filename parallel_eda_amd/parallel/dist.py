@@ -102,7 +102,8 @@ class CpuEngine:
         self.r = router
         self.num_nodes = num_nodes
 
-    def route_subset(self, crit, pres_fac, net_ids, partial=False):
+    def route_subset(self, crit, pres_fac, net_ids, partial=False,
+                     force_waves=False):
         import numpy as _np
         self.r.set_pres_fac(pres_fac)
         # refresh pres array from occ under the new pres_fac
@@ -152,9 +153,11 @@ class GpuEngine:
         self.g = router
         self._last_sd = np.zeros(router.n_sinks, dtype=np.float32)
 
-    def route_subset(self, crit, pres_fac, net_ids, partial=False):
+    def route_subset(self, crit, pres_fac, net_ids, partial=False,
+                     force_waves=False):
         _, sd = self.g.route_iteration(crit, pres_fac, net_subset=net_ids,
-                                       fail_ok=True, partial=partial)
+                                       fail_ok=True, partial=partial,
+                                       force_waves=force_waves)
         self._last_sd = sd
 
     def occ_tensor(self):
@@ -334,7 +337,7 @@ class DistRouteLoop:
         return self.set_partition(new_rank)
 
     def iteration(self, crit, pres_fac, acc_fac, active_mask=None,
-                  partial=False):
+                  partial=False, force_waves=False):
         """One distributed PathFinder iteration. active_mask: optional
         bool mask over nets (selective reroute); only owned ACTIVE nets
         are routed, but every rank joins the collectives. partial:
@@ -349,7 +352,8 @@ class DistRouteLoop:
         if self.ws > 1:
             occ_before = eng.occ_tensor().clone()
         if len(nets):
-            eng.route_subset(crit, pres_fac, nets, partial=partial)
+            eng.route_subset(crit, pres_fac, nets, partial=partial,
+                             force_waves=force_waves)
         sd = eng.sink_delays_local(self.my_nets)
         if self.ws > 1:
             occ = eng.occ_tensor()
@@ -402,6 +406,7 @@ def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
     cpd = 0.0
     history = []
     overused = -1
+    prev_overused = 1 << 30
     it = 0
     for it in range(1, max_iters + 1):
         # After iteration 1 ONLY the congested ∪ incomplete set is ever
@@ -416,11 +421,19 @@ def pathfinder_route_dist(loop, cmap, sta, max_iters=60, pres_fac_init=0.5,
         resync_every = int(os.environ.get("PNR_RESYNC_EVERY", "2"))
         resync = (incremental and resync_every > 0 and it > 2 and
                   (it - 2) % resync_every == 0)
+        # stall breaker (see route.gpu_router.pathfinder_route_gpu): a
+        # stagnant concurrent endgame switches to the wave schedule
+        n_act = int(active.sum()) if active is not None else 0
+        force_waves = (it >= 4 and 0 < overused and
+                       overused > 0.85 * prev_overused and
+                       0 < n_act <= 2048)
+        prev_overused = overused if overused > 0 else prev_overused
         _t0 = _time.perf_counter()
         overused, sd = loop.iteration(crit, pres, acc_fac,
                                       active_mask=active,
                                       partial=incremental and
-                                      active is not None and not resync)
+                                      active is not None and not resync,
+                                      force_waves=force_waves)
         _t1 = _time.perf_counter()
         if sta is not None:
             cmap.conn_delays(sd, out=conn_delay, fill=intra_delay)

@@ -387,7 +387,7 @@ class GpuRouter:
 
     # ---- one PathFinder iteration ----
     def route_iteration(self, crit, pres_fac, net_subset=None, fail_ok=False,
-                        partial=False):
+                        partial=False, force_waves=False):
         """crit: per-sink criticality aligned with sink_rr (original order).
         Returns (overused_count, sink_delays aligned with original order).
         net_subset: optional array of net ids to (re)route; others keep
@@ -476,7 +476,8 @@ class GpuRouter:
             if attempts > 0 and _os.environ.get("PNR_RETRY_SERIAL"):
                 waves = [np.asarray([n]) for n in todo]   # bisection mode
             elif not self.deterministic and (
-                    len(todo) > self.concurrent_threshold or attempts > 0):
+                    (len(todo) > self.concurrent_threshold
+                     and not force_waves) or attempts > 0):
                 # one concurrent launch; biggest work first for load balance
                 areas_t = self._bb_areas(self.bb)[todo]
                 nsk = (self.sink_ptr[todo + 1] - self.sink_ptr[todo]).astype(np.int64)
@@ -762,6 +763,7 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
     history = []
     it = 0
     overused = -1
+    prev_overused = 1 << 30
     for it in range(1, max_iters + 1):
         # selective reroute (reference: build_phase_two congested-only):
         # iteration 1 routes everything; later iterations re-route only
@@ -781,9 +783,20 @@ def pathfinder_route_gpu(netlist, placement, g, arch, sta=None, max_iters=60,
             if len(subset) == 0:
                 subset = None
         router.reset_search_stats()
+        # Stall breaker: concurrent selective reroute can limit-cycle on
+        # the last contested nodes (snapshot collisions — nets re-divert
+        # onto each other forever; seen at LU32: 22 overused for 60
+        # iterations). When overused stops improving and the active set
+        # is small, force the bb-disjoint wave schedule, whose in-order
+        # congestion views resolve the tail.
+        force_waves = (it >= 4 and 0 < overused and
+                       overused > 0.85 * prev_overused and
+                       subset is not None and len(subset) <= 2048)
+        prev_overused = overused if overused > 0 else prev_overused
         overused, sink_delays = router.route_iteration(
             crit, pres_fac, net_subset=subset,
-            partial=incremental and subset is not None and not resync)
+            partial=incremental and subset is not None and not resync,
+            force_waves=force_waves)
         st = router.search_stats()
         history.append(dict(iter=it, overused=int(overused), cpd=cpd,
                             rounds=st["rounds"], scanned=st["scanned"],
