@@ -317,14 +317,34 @@ PYBIND11_MODULE(_pnr_cpu, m) {
       .def("analyze_domains", [](TimingGraph& t,
                                  py::array_t<float, py::array::c_style | py::array::forcecast> conn_delay,
                                  py::array_t<int32_t, py::array::c_style | py::array::forcecast> block_clock,
-                                 py::array_t<float, py::array::c_style | py::array::forcecast> periods) {
+                                 py::array_t<float, py::array::c_style | py::array::forcecast> periods,
+                                 py::object pair_skip, py::object pair_mult) {
         py::ssize_t n = conn_delay.size();
         py::array_t<float> slack(n), crit(n);
+        const uint8_t* ps = nullptr;
+        const float* pm = nullptr;
+        py::array_t<uint8_t, py::array::c_style | py::array::forcecast> psa;
+        py::array_t<float, py::array::c_style | py::array::forcecast> pma;
+        int K = (int)periods.size();
+        if (!pair_skip.is_none()) {
+          psa = pair_skip.cast<decltype(psa)>();
+          if ((int)psa.size() != K * K)
+            throw std::runtime_error("pair_skip must be KxK");
+          ps = psa.data();
+        }
+        if (!pair_mult.is_none()) {
+          pma = pair_mult.cast<decltype(pma)>();
+          if ((int)pma.size() != K * K)
+            throw std::runtime_error("pair_mult must be KxK");
+          pm = pma.data();
+        }
         float wp = t.analyze_domains(conn_delay.data(), block_clock.data(),
-                                     periods.data(), (int)periods.size(),
-                                     slack.mutable_data(), crit.mutable_data());
+                                     periods.data(), K,
+                                     slack.mutable_data(),
+                                     crit.mutable_data(), ps, pm);
         return py::make_tuple(wp, slack, crit);
-      })
+      }, py::arg("conn_delay"), py::arg("block_clock"), py::arg("periods"),
+         py::arg("pair_skip") = py::none(), py::arg("pair_mult") = py::none())
       .def("level_arrays", [](TimingGraph& t) {
         std::vector<int32_t> blocks, start;
         t.level_arrays(blocks, start);

@@ -12,6 +12,7 @@
 // (csrc/hip/sta_kernel.hip), which run the same sweeps level-synchronously.
 #include "pnr.h"
 #include <algorithm>
+#include <map>
 
 namespace pnr {
 
@@ -91,9 +92,17 @@ class TimingGraph {
   // Outputs: per-conn worst slack and criticality (max over pairs, each
   // normalized by its pair's constraint). Returns the worst cpd/period
   // ratio times its period (the binding domain's achieved period).
+  // pair_skip / pair_mult: optional KxK (src,sink)-domain-pair
+  // constraints (reference read_sdc.c: set_false_path and
+  // set_multicycle_path between clock domains). A skipped pair is not
+  // analyzed; a multiplied pair's setup constraint is periods[cj]*mult
+  // (the backward pass is re-run per distinct (cj, mult) so comb
+  // fan-in cones see the adjusted requirement exactly).
   float analyze_domains(const float* conn_delay, const int32_t* block_clock,
                         const float* periods, int K,
-                        float* slack, float* crit) {
+                        float* slack, float* crit,
+                        const uint8_t* pair_skip = nullptr,
+                        const float* pair_mult = nullptr) {
     int nb = nl_->num_blocks;
     int64_t nconn = (int64_t)nl_->net_sinks.size();
     const float NEG = -3.0e38f, POS = 3.0e38f;
@@ -117,11 +126,10 @@ class TimingGraph {
         a[b] = (m > NEG) ? m + blk_delay_[b] : NEG;
       }
     }
-    // backward per sink domain (required time at block outputs)
-    for (int cj = 0; cj < K; ++cj) {
-      auto& r = req[cj];
+    // backward per (sink domain, constraint period)
+    auto backward = [&](int cj, float period, std::vector<float>& r) {
       r.assign(nb, POS);
-      float req_ep = periods[cj] - T_seq_in_;
+      float req_ep = period - T_seq_in_;
       for (auto it = topo_.rbegin(); it != topo_.rend(); ++it) {
         int b = *it;
         float m = POS;
@@ -138,13 +146,27 @@ class TimingGraph {
         }
         r[b] = m;
       }
-    }
+    };
+    for (int cj = 0; cj < K; ++cj) backward(cj, periods[cj], req[cj]);
+    // multicycle pairs need the backward pass at the adjusted period
+    std::map<std::pair<int, float>, std::vector<float>> req_mult;
+    if (pair_mult)
+      for (int ci = 0; ci < K; ++ci)
+        for (int cj = 0; cj < K; ++cj) {
+          float mu = pair_mult[ci * K + cj];
+          if (mu != 1.0f && !req_mult.count({cj, mu}))
+            backward(cj, periods[cj] * mu, req_mult[{cj, mu}]);
+        }
     for (int64_t c = 0; c < nconn; ++c) { slack[c] = POS; crit[c] = 0.0f; }
     float worst_ratio = 0.0f;  // achieved/required; >1 means violated
     float worst_period = periods[0];
     for (int ci = 0; ci < K; ++ci)
       for (int cj = 0; cj < K; ++cj) {
-        float constraint = periods[cj];
+        if (pair_skip && pair_skip[ci * K + cj]) continue;  // false path
+        float mu = pair_mult ? pair_mult[ci * K + cj] : 1.0f;
+        float constraint = periods[cj] * mu;
+        const std::vector<float>& R =
+            (mu == 1.0f) ? req[cj] : req_mult[{cj, mu}];
         for (int64_t c = 0; c < nconn; ++c) {
           int drv = conn_driver_[c];
           int snk = nl_->net_sinks[c];
@@ -153,7 +175,7 @@ class TimingGraph {
           if (nl_->block_is_seq[snk])
             ri = (block_clock[snk] == cj) ? constraint - T_seq_in_ : POS;
           else
-            ri = (req[cj][snk] < POS) ? req[cj][snk] - blk_delay_[snk] : POS;
+            ri = (R[snk] < POS) ? R[snk] - blk_delay_[snk] : POS;
           if (ri >= POS) continue;
           float s = ri - (arr[ci][drv] + conn_delay[c]);
           if (s < slack[c]) slack[c] = s;

@@ -80,6 +80,74 @@ def parse_sdc(text):
     return list(clocks.values())[-1] if clocks else None
 
 
+def parse_sdc_constraints(text):
+    """Full supported SDC subset (reference read_sdc.c):
+    create_clock, set_false_path -from/-to [get_clocks ...],
+    set_multicycle_path N -from/-to, set_input_delay / set_output_delay
+    -clock C V. Returns a dict:
+      clocks: {name: period_s}
+      false_paths: [(from_clk_or_None, to_clk_or_None)]
+      multicycle: [(from_clk_or_None, to_clk_or_None, N)]
+      input_delay / output_delay: {clock: seconds}
+    None in a from/to slot means 'all clocks' (SDC wildcard)."""
+    import re
+
+    def clk_arg(rest, flag):
+        m = re.search(flag + r"\s+\[\s*get_clocks\s+\{?\s*(\S+?)\s*\}?\s*\]",
+                      rest)
+        if m:
+            return m.group(1)
+        m = re.search(flag + r"\s+(?!\[)(\S+)", rest)
+        return m.group(1) if m else None
+
+    out = dict(clocks=parse_sdc_clocks(text), false_paths=[],
+               multicycle=[], input_delay={}, output_delay={})
+    for line in text.splitlines():
+        line = line.split("#", 1)[0].strip()
+        if line.startswith("set_false_path"):
+            out["false_paths"].append((clk_arg(line, "-from"),
+                                       clk_arg(line, "-to")))
+        elif line.startswith("set_multicycle_path"):
+            m = re.search(r"set_multicycle_path\s+(?:-setup\s+)?(\d+)", line)
+            n = int(m.group(1)) if m else 1
+            out["multicycle"].append((clk_arg(line, "-from"),
+                                      clk_arg(line, "-to"), n))
+        elif line.startswith("set_input_delay") or \
+                line.startswith("set_output_delay"):
+            c = clk_arg(line, "-clock")
+            v = re.search(r"(?:-clock\s+\S+\s+|\]\s+)([0-9.eE+-]+)", line)
+            key = ("input_delay" if line.startswith("set_input") else
+                   "output_delay")
+            if c and v:
+                out[key][c] = float(v.group(1)) * 1e-9
+    return out
+
+
+def pair_constraints(sdc, clock_names):
+    """Build the (pair_skip, pair_mult) KxK arrays for
+    STA.analyze_domains from parse_sdc_constraints output. clock_names:
+    ordered domain names (index = domain id)."""
+    import numpy as np
+    K = len(clock_names)
+    idx = {c: i for i, c in enumerate(clock_names)}
+    skip = np.zeros((K, K), dtype=np.uint8)
+    mult = np.ones((K, K), dtype=np.float32)
+
+    def rows(frm):
+        return [idx[frm]] if frm in idx else list(range(K)) if frm is None \
+            else []
+
+    for (f, t) in sdc.get("false_paths", ()):
+        for i in rows(f):
+            for j in rows(t):
+                skip[i, j] = 1
+    for (f, t, n) in sdc.get("multicycle", ()):
+        for i in rows(f):
+            for j in rows(t):
+                mult[i, j] = float(n)
+    return skip, mult
+
+
 def parse_sdc_clocks(text):
     """All create_clock constraints as {name: period_seconds}; unnamed
     clocks get the port expression or 'clk<i>'."""

@@ -45,14 +45,22 @@ class STA:
         cpd, slack, crit = self.tg.analyze(conn_delay)
         return float(cpd), np.asarray(slack), np.asarray(crit)
 
-    def analyze_domains(self, conn_delay, block_clock, periods):
+    def analyze_domains(self, conn_delay, block_clock, periods,
+                        pair_skip=None, pair_mult=None):
         """Multi-clock analysis (reference: do_timing_analysis_new per
         (src,sink)-domain pairs). block_clock: per-block domain id (-1
-        comb); periods: seconds per domain. Returns
-        (worst_achieved_period, slack[], crit[]) — slack/crit are the
-        worst/max over domain pairs."""
+        comb); periods: seconds per domain. pair_skip / pair_mult:
+        optional KxK (src,sink)-pair constraint arrays from SDC
+        set_false_path / set_multicycle_path (reference read_sdc.c).
+        Returns (worst_achieved_period, slack[], crit[]) — slack/crit
+        are the worst/max over analyzed domain pairs."""
         conn_delay = np.ascontiguousarray(conn_delay, dtype=np.float32)
         bc = np.ascontiguousarray(block_clock, dtype=np.int32)
         pr = np.ascontiguousarray(periods, dtype=np.float32)
-        wp, slack, crit = self.tg.analyze_domains(conn_delay, bc, pr)
+        if pair_skip is not None:
+            pair_skip = np.ascontiguousarray(pair_skip, dtype=np.uint8)
+        if pair_mult is not None:
+            pair_mult = np.ascontiguousarray(pair_mult, dtype=np.float32)
+        wp, slack, crit = self.tg.analyze_domains(conn_delay, bc, pr,
+                                                  pair_skip, pair_mult)
         return float(wp), np.asarray(slack), np.asarray(crit)

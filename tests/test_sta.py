@@ -164,3 +164,55 @@ def test_sta_random_dag_property():
         # max criticality must be ~1 and its conn slack ~0
         assert crit.max() == pytest.approx(1.0, abs=1e-4)
         assert abs(slack[np.argmax(crit)]) <= 1e-12 + 1e-4 * cpd
+
+
+def test_sdc_pair_constraints():
+    """set_false_path and set_multicycle_path between clock domains
+    (reference read_sdc.c): a false-path pair is not analyzed; a
+    multicycle pair's setup constraint is N periods."""
+    import numpy as np
+    from parallel_eda_amd.arch.archdef import get_arch
+    from parallel_eda_amd.io.synth import synth_netlist, spec_for_arch
+    from parallel_eda_amd.timing.sta import STA
+    from parallel_eda_amd.timing.report import (parse_sdc_constraints,
+                                                pair_constraints)
+    sdc = parse_sdc_constraints("""
+create_clock -period 5.0 -name clkA
+create_clock -period 8.0 -name clkB
+set_false_path -from [get_clocks clkA] -to [get_clocks clkB]
+set_multicycle_path 2 -from [get_clocks clkB] -to [get_clocks clkA]
+set_input_delay -clock clkA 1.0
+""")
+    assert sdc["clocks"] == {"clkA": 5e-9, "clkB": 8e-9}
+    assert sdc["false_paths"] == [("clkA", "clkB")]
+    assert sdc["multicycle"] == [("clkB", "clkA", 2)]
+    assert abs(sdc["input_delay"]["clkA"] - 1e-9) < 1e-15
+    skip, mult = pair_constraints(sdc, ["clkA", "clkB"])
+    assert skip[0, 1] == 1 and skip.sum() == 1
+    assert mult[1, 0] == 2.0 and mult.sum() == 5.0
+
+    arch = get_arch("tseng")
+    nl = synth_netlist(spec_for_arch(arch, fill=0.5, seed=3))
+    rng = np.random.default_rng(3)
+    bc = np.where(np.asarray(nl.block_is_seq) > 0,
+                  rng.integers(0, 2, nl.num_blocks), -1).astype(np.int32)
+    periods = np.asarray([5e-9, 8e-9], dtype=np.float32)
+    dly = rng.uniform(0.1e-9, 2e-9, nl.num_conns).astype(np.float32)
+    sta = STA(nl, arch)
+    wp0, sl0, cr0 = sta.analyze_domains(dly, bc, periods)
+    # all pairs false -> nothing analyzed: zero slack everywhere
+    all_skip = np.ones((2, 2), dtype=np.uint8)
+    wpf, slf, crf = sta.analyze_domains(dly, bc, periods,
+                                        pair_skip=all_skip)
+    assert np.all(crf == 0.0)
+    # multicycle 2 on every pair: slacks can only grow
+    m2 = np.full((2, 2), 2.0, dtype=np.float32)
+    wpm, slm, crm = sta.analyze_domains(dly, bc, periods, pair_mult=m2)
+    assert np.all(slm >= sl0 - 1e-12)
+    assert wpm <= wp0 + 1e-12 or crm.max() <= cr0.max() + 1e-6
+    # skipping one pair never worsens slack on still-constrained conns
+    # (conns whose only pair was skipped become unconstrained -> slack 0)
+    wps, sls, crs = sta.analyze_domains(dly, bc, periods, pair_skip=skip)
+    constrained = sls != 0.0
+    assert np.all(sls[constrained] >= sl0[constrained] - 1e-12)
+    assert crs.max() <= cr0.max() + 1e-6
