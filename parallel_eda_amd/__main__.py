@@ -161,12 +161,18 @@ def main(argv=None):
     timing = args.router_algorithm == "timing_driven"
     sta = STA(netlist, arch) if timing else None
     sdc_clocks = {}
+    sdc_all = None
     if args.sdc:
-        from .timing.report import parse_sdc_clocks
+        from .timing.report import parse_sdc_constraints
         with open(args.sdc) as f:
-            sdc_clocks = parse_sdc_clocks(f.read())
+            sdc_all = parse_sdc_constraints(f.read())
+        sdc_clocks = sdc_all["clocks"]
         for nm, p_ in sdc_clocks.items():
             print(f"SDC: clock '{nm}' period {p_*1e9:.3f} ns")
+        for (f_, t_) in sdc_all["false_paths"]:
+            print(f"SDC: false path {f_ or '*'} -> {t_ or '*'}")
+        for (f_, t_, n_) in sdc_all["multicycle"]:
+            print(f"SDC: multicycle {n_} {f_ or '*'} -> {t_ or '*'}")
 
     # ---- placement ----
     t0 = time.perf_counter()
@@ -278,7 +284,12 @@ def main(argv=None):
             periods = [sdc_clocks.get(c, list(sdc_clocks.values())[0])
                        for c in cnames] or list(sdc_clocks.values())[:1]
             bc = getattr(netlist, "block_clock")
-            wp, sl, cr = sta.analyze_domains(cd3, bc, periods)
+            ps = pm = None
+            if sdc_all and (sdc_all["false_paths"] or sdc_all["multicycle"]):
+                from .timing.report import pair_constraints
+                ps, pm = pair_constraints(sdc_all, cnames)
+            wp, sl, cr = sta.analyze_domains(cd3, bc, periods,
+                                             pair_skip=ps, pair_mult=pm)
             ok_sdc = all(wp <= p_ * (1 + 1e-6) for p_ in periods[:1])
             print(f"SDC analysis: worst achieved period {wp*1e9:.3f} ns "
                   f"across {max(1, len(periods))} clock domain(s)")
