@@ -20,7 +20,10 @@ def test_mfma_gemm_matches_numpy():
         c = mfma_gemm(torch.from_numpy(a).cuda(),
                       torch.from_numpy(b).cuda()).cpu().numpy()
         ref = a.astype(np.float64) @ b.astype(np.float64)
-        assert np.allclose(c, ref, rtol=1e-5, atol=1e-5), (m, n, k)
+        # MFMA f32 is an exact fmaf chain; vs the float64 reference the
+        # error grows with the K-long accumulation — scale the bound
+        tol = 1e-6 + 4e-7 * k
+        assert np.allclose(c, ref, rtol=tol, atol=tol), (m, n, k)
 
 
 def test_mfma_delay_matrix_smoothing():
