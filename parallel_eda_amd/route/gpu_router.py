@@ -129,7 +129,7 @@ class GpuRouter:
     def __init__(self, g, arch: ArchDef, src_rr, sink_ptr, sink_rr,
                  device="cuda:0", astar_fac=1.8, n_small_slots=1024,
                  n_large_slots=128, bb_margin=4, max_rounds=200000,
-                 delta_fac=3.0, deterministic=False,
+                 delta_fac=1.5, deterministic=False,
                  concurrent_threshold=48, occ=None, dev_graph=None):
         torch = _torch()
         self.torch = torch
