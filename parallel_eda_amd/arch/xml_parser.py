@@ -58,19 +58,36 @@ def parse_arch_xml(path_or_text, nx=None, ny=None, W=64, name=None) -> ArchDef:
     a.W = W if W % 2 == 0 else W + 1
 
     # ---- segments ----
-    seg = root.find(".//segmentlist/segment")
-    if seg is not None:
-        length = seg.get("length", "4")
-        a.L = 1 if length in ("longline",) else max(1, int(length))
+    # Multiple <segment> entries (VTR arches declare a length
+    # distribution, e.g. L1 @ freq 0.3 + L4 @ 0.7): the LONGEST length
+    # becomes L and the length-1 share of the channel becomes w_l1
+    # (rr_build's two-length fabric; see archdef.w_l1 for why a
+    # single-length unidirectional channel cannot route locally).
+    segs = root.findall(".//segmentlist/segment")
+    if segs:
+        def seg_len(el):
+            s = el.get("length", "4")
+            return 1 if s in ("longline",) else max(1, int(s))
+
+        def seg_freq(el):
+            try:
+                return float(el.get("freq", el.get("frequency", "1")))
+            except ValueError:
+                return 1.0
+
+        lengths = [(seg_len(el), seg_freq(el)) for el in segs]
+        a.L = max(l for l, _ in lengths)
+        f1 = sum(f for l, f in lengths if l == 1)
+        ftot = sum(f for _, f in lengths) or 1.0
+        if a.L > 1 and f1 > 0:
+            a.w_l1 = max(2, int(a.W * f1 / ftot) & ~1)
+        seg = max(segs, key=seg_len)   # timing from the workhorse segment
         rm = _first_float(seg, "Rmetal", default=None)
         cm = _first_float(seg, "Cmetal", default=None)
         if rm is not None:
             a.R_wire = rm
         if cm is not None:
             a.C_wire = cm
-        # Fc override on the segment
-        for fc_el in seg.findall("fc") + ([] if seg.find("fc") is None else []):
-            pass
 
     # ---- switches ----
     sw = None

@@ -337,3 +337,36 @@ def test_big_blif_end_to_end(tmp_path):
     rc = main([str(bf), str(xf), "--route_chan_width", "80",
                "--max_router_iterations", "80"])
     assert rc == 0
+
+
+def test_arch_xml_segment_distribution():
+    """Multiple <segment> lengths map onto the two-length fabric: the
+    longest becomes L, the length-1 frequency share becomes w_l1
+    (reference: <segmentlist> distributions in VTR arches)."""
+    from parallel_eda_amd.arch.xml_parser import parse_arch_xml
+    xml = """<architecture>
+ <complexblocklist>
+  <pb_type name="io" capacity="2">
+   <input name="outpad" num_pins="1"/><output name="inpad" num_pins="1"/>
+  </pb_type>
+  <pb_type name="clb">
+   <input name="I" num_pins="16"/><output name="O" num_pins="4"/>
+  </pb_type>
+ </complexblocklist>
+ <segmentlist>
+  <segment length="1" freq="0.25" Rmetal="50" Cmetal="1e-14"/>
+  <segment length="4" freq="0.75" Rmetal="101" Cmetal="2.25e-14"/>
+ </segmentlist>
+ <switchlist>
+  <switch type="mux" name="0" R="551" Cin="7.7e-16" Tdel="5.8e-11"/>
+ </switchlist>
+</architecture>"""
+    a = parse_arch_xml(xml, nx=20, ny=20, W=40)
+    assert a.L == 4
+    assert a.w_l1 == 10   # 25% of W=40, even
+    assert a.R_wire == 101.0   # timing from the workhorse (longest) segment
+    # single-segment arch keeps auto behavior
+    a2 = parse_arch_xml(xml.replace(
+        '<segment length="1" freq="0.25" Rmetal="50" Cmetal="1e-14"/>', ""),
+        nx=20, ny=20, W=40)
+    assert a2.L == 4 and a2.w_l1 == -1
