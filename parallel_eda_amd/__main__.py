@@ -75,6 +75,10 @@ def build_parser():
     p.add_argument("--activity_file", type=str, default=None,
                    help="per-net switching activities (.act lines: "
                         "'netname activity')")
+    p.add_argument("--route_file", type=str, default=None,
+                   help="read this .route and ANALYZE it (occupancy + "
+                        "connectivity validation, Elmore delays, STA) "
+                        "instead of routing — VPR's ROUTE_NEVER flow")
     p.add_argument("--timing_report", type=str, default=None,
                    help="write a critical-path report here after routing")
     p.add_argument("--place_only", action="store_true",
@@ -213,6 +217,55 @@ def main(argv=None):
         write_placement_svg(args.draw_place, placement, netlist, arch)
         print(f"wrote {args.draw_place}")
     if args.place_only:
+        print(f"entire flow took {time.perf_counter()-t_start:.2f}s")
+        return 0
+
+    # ---- analysis-only: re-read an existing routing (reference: VPR's
+    # ROUTE_NEVER / --route_file analysis flow) ----
+    if args.route_file:
+        from .io.route_file import read_route, tree_elmore_delays
+        from .route.router import ConnMap
+        g = rrgraph.build_rr_graph(arch)
+        names_r, trees_r = read_route(args.route_file, g, arch)
+        net_ids, src_rr, sink_ptr, sink_rr, conn_index = net_rr_terminals(
+            netlist, placement, g, arch)
+        if len(trees_r) != len(net_ids):
+            print(f"route file has {len(trees_r)} nets, netlist expects "
+                  f"{len(net_ids)}")
+            return 1
+        # occupancy + connectivity validation (check_route.c semantics)
+        occ = np.zeros(g.num_nodes, dtype=np.int32)
+        cap = np.asarray(g.capacity)
+        sink_delays = np.zeros(len(sink_rr), dtype=np.float32)
+        for k, (nodes_k, parents_k) in enumerate(trees_r):
+            np.add.at(occ, nodes_k, 1)
+            if len(nodes_k) == 0 or nodes_k[0] != src_rr[k]:
+                print(f"net {k}: traceback does not start at its SOURCE")
+                return 1
+            d = tree_elmore_delays(g, nodes_k, parents_k)
+            pos = {int(v): i for i, v in enumerate(nodes_k)}
+            for s in range(sink_ptr[k], sink_ptr[k + 1]):
+                v = int(sink_rr[s])
+                if v not in pos:
+                    print(f"net {k}: sink rr node {v} not in traceback")
+                    return 1
+                sink_delays[s] = d[pos[v]]
+        over = int((occ > cap).sum())
+        print(f"read {args.route_file}: {len(trees_r)} nets, "
+              f"overused nodes: {over}")
+        if over:
+            return 1
+        if sta is not None:
+            cmap = ConnMap(conn_index, sink_ptr, netlist.num_conns,
+                           len(sink_rr))
+            cd = cmap.conn_delays(sink_delays,
+                                  fill=float(arch.T_opin + arch.T_ipin))
+            cpd, slack, crit = sta.analyze(cd)
+            print(f"analysis: crit_path={cpd*1e9:.3f}ns")
+            if args.timing_report:
+                from .timing.report import write_timing_report
+                write_timing_report(args.timing_report, netlist, sta, cd)
+                print(f"wrote {args.timing_report}")
         print(f"entire flow took {time.perf_counter()-t_start:.2f}s")
         return 0
 

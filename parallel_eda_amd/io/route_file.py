@@ -53,3 +53,83 @@ def write_route(path, g, arch, net_ids, trees, netlist=None):
                 f.write(_node_line(g, nodes[i], ty_arr, xl, yl, xh, yh, ptc)
                         + "\n")
                 prev = i
+
+
+def read_route(path, g, arch):
+    """Read a .route traceback back into per-net trees (reference:
+    route/read_route-style analysis flow — VPR re-reads its own routing
+    for timing analysis without re-routing).
+
+    Returns (net_names, trees) where trees[k] = (nodes, parents) numpy
+    arrays in tree order (parents index into the net's own arrays).
+    Node identity is recovered from (type, xlow, ylow, ptc), which is
+    unique in this fabric."""
+    ty_arr = np.asarray(g.type)
+    xl = np.asarray(g.xlow); yl = np.asarray(g.ylow)
+    ptc = np.asarray(g.ptc)
+    lookup = {}
+    for v in range(g.num_nodes):
+        lookup[(int(ty_arr[v]), int(xl[v]), int(yl[v]), int(ptc[v]))] = v
+    tname_id = {n: i for i, n in enumerate(RR_TYPE_NAMES)}
+
+    import re
+    line_re = re.compile(
+        r"^(\w+) \((\d+),(\d+)\)(?: to \(\d+,\d+\))?\s+"
+        r"(?:Class|Pin|Track): (\d+)")
+    net_names = []
+    trees = []
+    nodes = []
+    parents = []
+    index_of = {}
+    prev_idx = -1
+
+    def flush():
+        if net_names and len(net_names) > len(trees):
+            trees.append((np.asarray(nodes, dtype=np.int32),
+                          np.asarray(parents, dtype=np.int32)))
+
+    with open(path) as f:
+        for line in f:
+            line = line.rstrip("\n")
+            m = re.match(r"^Net (\d+) \((.*)\)$", line)
+            if m:
+                flush()
+                net_names.append(m.group(2))
+                nodes = []; parents = []; index_of = {}; prev_idx = -1
+                continue
+            m = line_re.match(line)
+            if not m:
+                continue
+            key = (tname_id[m.group(1)], int(m.group(2)), int(m.group(3)),
+                   int(m.group(4)))
+            if key not in lookup:
+                raise ValueError(f"route references unknown rr node {key}")
+            v = lookup[key]
+            if v in index_of:
+                prev_idx = index_of[v]     # branch attach re-print
+                continue
+            nodes.append(v)
+            parents.append(prev_idx)
+            index_of[v] = len(nodes) - 1
+            prev_idx = len(nodes) - 1
+    flush()
+    return net_names, trees
+
+
+def tree_elmore_delays(g, nodes, parents):
+    """Per-tree-node source->node delay via the same per-hop Elmore the
+    routers use: Tdel(sw) + C(v) * (R(sw) + R(v)/2)."""
+    row_ptr = np.asarray(g.row_ptr); edge_dst = np.asarray(g.edge_dst)
+    edge_sw = np.asarray(g.edge_sw)
+    sw_R = np.asarray(g.sw_R); sw_T = np.asarray(g.sw_Tdel)
+    R = np.asarray(g.node_R); C = np.asarray(g.node_C)
+    d = np.zeros(len(nodes), dtype=np.float64)
+    for i in range(1, len(nodes)):
+        u = nodes[i]; pu = nodes[parents[i]]
+        sw = 0
+        for e in range(row_ptr[pu], row_ptr[pu + 1]):
+            if edge_dst[e] == u:
+                sw = int(edge_sw[e])
+                break
+        d[i] = d[parents[i]] + sw_T[sw] + C[u] * (sw_R[sw] + 0.5 * R[u])
+    return d
