@@ -396,8 +396,11 @@ class GpuRouter:
         instead of raising (multi-rank benches must not kill a rank
         mid-collective).
         partial: partial rip-up (keep clean subtrees, skip connected
-        sinks; reference route_tree_mark_congested_...; EXPERIMENTAL,
-        pending round-2 GPU validation; both kernels honor it)."""
+        sinks; reference route_tree_mark_congested_...; production in
+        the incremental flows, GPU-validated round 2).
+        force_waves: route even medium sets on the bb-disjoint wave
+        schedule (the drivers' stall breaker for stagnant concurrent
+        endgames)."""
         self._partial = partial
         torch = self.torch
         import time as _time
