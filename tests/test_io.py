@@ -405,3 +405,38 @@ def test_route_file_roundtrip_and_analysis(tmp_path):
     rc3 = main([str(blif), str(xml), "--route_chan_width", "20",
                 "--place_file", str(pfile), "--route_file", str(bad)])
     assert rc3 == 1
+
+
+def test_route_file_delay_fidelity():
+    """read_route + tree_elmore_delays reproduce the router's own
+    per-sink delays exactly (same per-hop Elmore model)."""
+    import numpy as np
+    from parallel_eda_amd.arch.archdef import get_arch
+    from parallel_eda_amd.io.synth import synth_placed_netlist
+    from parallel_eda_amd import rrgraph
+    from parallel_eda_amd.route.router import (pathfinder_route,
+                                               net_rr_terminals)
+    from parallel_eda_amd.timing.sta import STA
+    from parallel_eda_amd.io.route_file import (write_route, read_route,
+                                                tree_elmore_delays)
+    arch = get_arch("tseng")
+    nl, pl = synth_placed_netlist(arch, fill=0.4, seed=21)
+    g = rrgraph.build_rr_graph(arch)
+    res = pathfinder_route(nl, pl, g, arch, sta=STA(nl, arch), max_iters=40)
+    assert res.success
+    net_ids, src_rr, sink_ptr, sink_rr, _ = net_rr_terminals(nl, pl, g, arch)
+    import tempfile, os
+    path = os.path.join(tempfile.mkdtemp(), "t.route")
+    write_route(path, g, arch, net_ids, lambda k: res.router.tree(k),
+                netlist=nl)
+    names, trees = read_route(path, g, arch)
+    assert len(trees) == len(net_ids)
+    sd_router = np.asarray(res.router.sink_delays())
+    for k, (nodes, parents) in enumerate(trees):
+        rn, rp, rs, rd = res.router.tree(k)
+        assert set(nodes.tolist()) == set(np.asarray(rn).tolist()), k
+        d = tree_elmore_delays(g, nodes, parents)
+        pos = {int(v): i for i, v in enumerate(nodes)}
+        for s in range(sink_ptr[k], sink_ptr[k + 1]):
+            v = int(sink_rr[s])
+            assert abs(d[pos[v]] - sd_router[s]) < 1e-12, (k, s)
