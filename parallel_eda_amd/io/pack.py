@@ -114,22 +114,34 @@ def pack_blif(model: BlifModel, arch: ArchDef, n_ble: int = 10):
                     for j in sig_users.get(o, ()):
                         if j in unclustered:
                             gain[j] += 1
-            if not gain:
-                break
             # best gain, legality-filtered: input pins, output pins, and
             # single-clock (reference: cluster_legality.c feasibility)
+            def legal(cand):
+                cclk = prims[cand].clock
+                if cclk and cur_clk and cclk != cur_clk:
+                    return False
+                if len(cluster_inputs(members, cand)) > arch.clb_in:
+                    return False
+                return len(cluster_outputs(members, cand)) <= arch.clb_out
+
             best = None
             cur_clk = cluster_clock(members)
             for cand, gn in sorted(gain.items(), key=lambda kv: (-kv[1], kv[0])):
-                cclk = prims[cand].clock
-                if cclk and cur_clk and cclk != cur_clk:
-                    continue
-                if len(cluster_inputs(members, cand)) > arch.clb_in:
-                    continue
-                if len(cluster_outputs(members, cand)) > arch.clb_out:
-                    continue
-                best = cand
-                break
+                if legal(cand):
+                    best = cand
+                    break
+            if best is None:
+                # unrelated fill (reference: do_clustering's
+                # allow_unrelated_clustering — when no connected
+                # candidate fits, pack the hardest-to-place remaining
+                # primitive that is still legal, keeping cluster count
+                # near capacity instead of leaving fragments)
+                for cand in sorted(unclustered,
+                                   key=lambda i: (-len(prims[i].inputs),
+                                                  i))[:64]:
+                    if legal(cand):
+                        best = cand
+                        break
             if best is None:
                 break
             members.append(best)
