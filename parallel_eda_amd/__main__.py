@@ -148,6 +148,18 @@ def main(argv=None):
                                               seed=args.seed))
         print(f"synthetic netlist '{args.synth}': {netlist.num_blocks} blocks,"
               f" {netlist.num_nets} nets")
+    elif args.circuit and args.circuit.endswith(".net"):
+        # packed netlist input: skip the packer (reference: VPR reads
+        # circuit.net directly when packing already happened)
+        from .io.net_file import read_net
+        if not args.arch:
+            print("error: need arch.xml with a .net input", file=sys.stderr)
+            return 2
+        arch = parse_arch_xml(args.arch, W=args.route_chan_width or 64)
+        netlist = read_net(args.circuit)
+        size_grid_for_netlist(netlist, arch)
+        print(f"read {args.circuit}: {netlist.num_blocks} blocks, "
+              f"{netlist.num_nets} nets on {arch.nx}x{arch.ny} grid")
     else:
         if not args.circuit or not args.arch:
             print("error: need circuit.blif + arch.xml (or --synth NAME)",

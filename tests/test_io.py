@@ -440,3 +440,19 @@ def test_route_file_delay_fidelity():
         for s in range(sink_ptr[k], sink_ptr[k + 1]):
             v = int(sink_rr[s])
             assert abs(d[pos[v]] - sd_router[s]) < 1e-12, (k, s)
+
+
+def test_cli_net_input(tmp_path):
+    """A packed .net is accepted directly as the circuit (reference: VPR
+    reads circuit.net and skips packing)."""
+    from parallel_eda_amd.__main__ import main
+    blif = tmp_path / "c.blif"
+    blif.write_text(BLIF)
+    xml = tmp_path / "a.xml"
+    xml.write_text(ARCH_XML)
+    netf = tmp_path / "c.net"
+    rc = main([str(blif), str(xml), "--route_chan_width", "20",
+               "--out_net", str(netf), "--place_only"])
+    assert rc == 0 and netf.exists()
+    rc2 = main([str(netf), str(xml), "--route_chan_width", "20"])
+    assert rc2 == 0
