@@ -328,13 +328,17 @@ class GpuRouter:
         ky = ylow[self.sink_rr].astype(np.int32)
         seg = self.sink_ptr[:-1]
         has = self.sink_ptr[1:] > seg
-        # reduceat over each net's sink slice (empty slices produce the
-        # element AT the start index; masked back to the source below)
-        safe = np.minimum(seg, max(len(kx) - 1, 0))
-        xmin = np.minimum.reduceat(kx, safe) if len(kx) else sx.copy()
-        xmax = np.maximum.reduceat(kx, safe) if len(kx) else sx.copy()
-        ymin = np.minimum.reduceat(ky, safe) if len(ky) else sy.copy()
-        ymax = np.maximum.reduceat(ky, safe) if len(ky) else sy.copy()
+        # reduceat over each net's sink slice. A SENTINEL element keeps
+        # every start index valid (trailing zero-sink nets have
+        # seg == len(kx)); clamping indices instead would silently
+        # truncate the PREVIOUS net's segment (caught by
+        # tests/test_property.py hypothesis fuzzing). Empty slices
+        # produce the sentinel and are masked back to the source below.
+        BIG = np.int32(1 << 30)
+        xmin = np.minimum.reduceat(np.r_[kx, BIG], seg)
+        xmax = np.maximum.reduceat(np.r_[kx, -BIG], seg)
+        ymin = np.minimum.reduceat(np.r_[ky, BIG], seg)
+        ymax = np.maximum.reduceat(np.r_[ky, -BIG], seg)
         xmin = np.where(has, np.minimum(xmin, sx), sx)
         xmax = np.where(has, np.maximum(xmax, sx), sx)
         ymin = np.where(has, np.minimum(ymin, sy), sy)
