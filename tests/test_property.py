@@ -75,3 +75,63 @@ def test_smoothing_band_row_stochastic(n, radius):
     assert s.shape == (n, n)
     assert np.allclose(s.sum(axis=1), 1.0, atol=1e-5)
     assert (s >= 0).all()
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.integers(0, 10**6), st.integers(1, 60), st.integers(8, 40))
+def test_wave_schedule_invariants(seed, n_nets, grid):
+    """schedule_bb_waves: every net appears exactly once; within a wave
+    the coarse-cell footprints are disjoint."""
+    from parallel_eda_amd.route.gpu_router import schedule_bb_waves
+    rng = np.random.default_rng(seed)
+    x0 = rng.integers(0, grid, n_nets)
+    y0 = rng.integers(0, grid, n_nets)
+    bb = np.stack([x0, y0,
+                   np.minimum(grid + 1, x0 + rng.integers(0, grid, n_nets)),
+                   np.minimum(grid + 1, y0 + rng.integers(0, grid, n_nets))],
+                  axis=1).astype(np.int16)
+    areas = ((bb[:, 2] - bb[:, 0] + 1).astype(np.int64) *
+             (bb[:, 3] - bb[:, 1] + 1))
+    ids = np.arange(n_nets)
+    waves = schedule_bb_waves(bb, ids, areas, grid, grid, cell=8)
+    seen = np.concatenate(waves) if waves else np.zeros(0, dtype=int)
+    assert sorted(seen.tolist()) == ids.tolist()
+    for w in waves:
+        cells = set()
+        for n in w:
+            cs = {(cx, cy)
+                  for cx in range(bb[n, 0] // 8, bb[n, 2] // 8 + 1)
+                  for cy in range(bb[n, 1] // 8, bb[n, 3] // 8 + 1)}
+            assert not (cells & cs), "overlapping nets in one wave"
+            cells |= cs
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.integers(0, 10**6), st.integers(1, 200), st.integers(1, 8))
+def test_spatial_partition_invariants(seed, n_nets, world):
+    """spatial_partition: every net assigned to a valid rank; weighted
+    split is contiguous in bb-center order."""
+    from parallel_eda_amd.parallel.dist import spatial_partition
+    rng = np.random.default_rng(seed)
+    x0 = rng.integers(0, 60, n_nets)
+    bb = np.stack([x0, np.zeros(n_nets, dtype=np.int64),
+                   x0 + rng.integers(0, 10, n_nets),
+                   np.full(n_nets, 5)], axis=1).astype(np.int16)
+    w = rng.uniform(0.1, 10.0, n_nets)
+    r = spatial_partition(bb, world, weight=w)
+    assert r.min() >= 0 and r.max() < world
+    # rank is monotone in bb-center order
+    cx = bb[:, 0].astype(int) + bb[:, 2]
+    order = np.argsort(cx, kind="stable")
+    assert (np.diff(r[order]) >= 0).all()
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.integers(0, 10**6), st.integers(1, 500))
+def test_xcd_order_is_permutation(seed, n):
+    from parallel_eda_amd.route.gpu_router import xcd_interleaved_order
+    rng = np.random.default_rng(seed)
+    bb = rng.integers(0, 100, (n, 4)).astype(np.int16)
+    order = rng.permutation(n)
+    out = xcd_interleaved_order(bb, order)
+    assert sorted(out.tolist()) == sorted(order.tolist())
