@@ -69,29 +69,44 @@ def allreduce_(t, op=None):
 
 
 def spatial_partition(bb, world_size, weight=None):
-    """Assign nets to ranks by bb-center strips along x (region-tree top
-    cuts; reference analogue: fpga_bipartition / build_net_tree
-    partitioning_multi_sink...cxx:3064,3295), balanced by estimated route
-    cost (reference's measured-time load balancing, mpi_route...cxx:249,
-    approximated by nsinks x bb semiperimeter until measured times land).
-    Returns rank_of_net (int array)."""
+    """Assign nets to ranks by RECURSIVE WEIGHTED BISECTION of bb
+    centers (the reference's region tree: fpga_bipartition /
+    build_net_tree, partitioning_multi_sink...cxx:3064,3295) — each
+    level splits the heavier child count along its WIDER axis, so 8
+    ranks get a 2-D tiling instead of 8 thin strips (half the cut
+    boundary, fewer cross-rank congestion conflicts). Balanced by
+    estimated route cost (reference's measured-time load balancing,
+    mpi_route...cxx:249, approximated by nsinks x bb semiperimeter until
+    measured times land). Deterministic pure-numpy: every rank computes
+    the identical partition. Returns rank_of_net (int array)."""
     n = len(bb)
-    cx = (bb[:, 0].astype(np.int64) + bb[:, 2].astype(np.int64))
-    order = np.argsort(cx, kind="stable")
     rank_of = np.zeros(n, dtype=np.int32)
-    if weight is None:
-        chunks = np.array_split(order, world_size)
-        for r, ch in enumerate(chunks):
-            rank_of[ch] = r
+    if world_size <= 1 or n == 0:
         return rank_of
-    w = np.asarray(weight, dtype=np.float64)[order]
-    cum = np.cumsum(w)
-    total = cum[-1] if len(cum) else 0.0
-    bounds = np.searchsorted(cum, total * np.arange(1, world_size) / world_size)
-    prev = 0
-    for r, b in enumerate(list(bounds) + [n]):
-        rank_of[order[prev:b]] = r
-        prev = b
+    cx = (bb[:, 0].astype(np.int64) + bb[:, 2].astype(np.int64))
+    cy = (bb[:, 1].astype(np.int64) + bb[:, 3].astype(np.int64))
+    w = (np.ones(n, dtype=np.float64) if weight is None
+         else np.asarray(weight, dtype=np.float64))
+
+    def split(ids, k, r0):
+        if k == 1 or len(ids) == 0:
+            rank_of[ids] = r0
+            return
+        k_lo = k // 2
+        # split along the wider axis of this region's bb-center extent
+        ex = int(cx[ids].max() - cx[ids].min()) if len(ids) else 0
+        ey = int(cy[ids].max() - cy[ids].min()) if len(ids) else 0
+        key = cx[ids] if ex >= ey else cy[ids]
+        order = ids[np.argsort(key, kind="stable")]
+        cum = np.cumsum(w[order])
+        total = cum[-1]
+        # cut minimizing |left weight - fair share|
+        cut = int(np.argmin(np.abs(cum - total * k_lo / k))) + 1
+        cut = min(max(cut, 0), len(order))
+        split(order[:cut], k_lo, r0)
+        split(order[cut:], k - k_lo, r0 + k_lo)
+
+    split(np.arange(n, dtype=np.int64), world_size, 0)
     return rank_of
 
 
