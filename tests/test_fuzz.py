@@ -189,3 +189,48 @@ def test_fuzz_blif_front_end(seed):
     if res.success:
         ok, err = res.router.check_routed()
         assert ok, err
+
+
+def test_fuzz_route_file_roundtrip():
+    """read_route reproduces the router's trees exactly across random
+    fabrics (incl. heterogeneous columns) and placements."""
+    import numpy as np
+    from parallel_eda_amd.arch.archdef import ArchDef
+    from parallel_eda_amd.io.synth import synth_placed_netlist
+    from parallel_eda_amd import rrgraph
+    from parallel_eda_amd.route.router import (pathfinder_route,
+                                               net_rr_terminals)
+    from parallel_eda_amd.io.route_file import (write_route, read_route,
+                                                tree_elmore_delays)
+    import tempfile, os
+    rng = np.random.default_rng(777)
+    for trial in range(6):
+        het = trial % 2 == 1
+        arch = ArchDef(name=f"rt{trial}", nx=int(rng.integers(6, 14)),
+                       ny=int(rng.integers(6, 14)),
+                       W=int(rng.integers(10, 20)) * 2,
+                       L=int(rng.integers(1, 4)),
+                       clb_in=10, clb_out=4, io_cap=3,
+                       ram_col_every=4 if het else 0,
+                       ram_in=6, ram_out=3)
+        nl, pl = synth_placed_netlist(arch, fill=0.35,
+                                      seed=int(rng.integers(1 << 30)))
+        g = rrgraph.build_rr_graph(arch)
+        res = pathfinder_route(nl, pl, g, arch, sta=None, max_iters=40)
+        if not res.success:
+            continue
+        net_ids, src_rr, sink_ptr, sink_rr, _ = net_rr_terminals(
+            nl, pl, g, arch)
+        path = os.path.join(tempfile.mkdtemp(), "f.route")
+        write_route(path, g, arch, net_ids,
+                    lambda k: res.router.tree(k), netlist=nl)
+        names, trees = read_route(path, g, arch)
+        assert len(trees) == len(net_ids), trial
+        sd = np.asarray(res.router.sink_delays())
+        for k, (nodes, parents) in enumerate(trees):
+            rn, *_ = res.router.tree(k)
+            assert set(nodes.tolist()) == set(np.asarray(rn).tolist())
+            d = tree_elmore_delays(g, nodes, parents)
+            pos = {int(v): i for i, v in enumerate(nodes)}
+            for s in range(sink_ptr[k], sink_ptr[k + 1]):
+                assert abs(d[pos[int(sink_rr[s])]] - sd[s]) < 1e-12
